@@ -1,0 +1,131 @@
+"""Distributed tests without a cluster: gloo backend,
+world_size=2, single node — DP=2 must produce the same parameters as DP=1
+on the equal global batch (the reference's MirroredStrategy semantics,
+SURVEY.md §2.4 X1/X2 + §8 Q4 loss scaling)."""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from transformer_amd.models import Transformer
+from transformer_amd.runtime import NoamAdam
+from transformer_amd.parallel import BucketedDataParallel
+from transformer_amd.ops import reference as R
+
+
+def _model():
+    torch.manual_seed(0)
+    return Transformer(num_layers=1, d_model=16, num_heads=2, dff=32,
+                       input_vocab_size=50, target_vocab_size=50, rate=0.0,
+                       max_position=32)
+
+
+def _batches(n_steps, global_batch=4, seq=6):
+    torch.manual_seed(99)
+    out = []
+    for _ in range(n_steps):
+        out.append((torch.randint(1, 50, (global_batch, seq)),
+                    torch.randint(1, 50, (global_batch, seq))))
+    return out
+
+
+def _train_steps(model, opt, batches, ddp=None, rank=0, world=1):
+    losses = []
+    for inp, tar in batches:
+        b = inp.shape[0]
+        lo, hi = rank * b // world, (rank + 1) * b // world
+        logits, _ = model((inp[lo:hi], tar[lo:hi]), training=True)
+        loss = R.masked_cross_entropy(logits, tar[lo:hi], batch_size=b)
+        opt.zero_grad()
+        loss.backward()
+        if ddp is not None:
+            ddp.finalize()
+        opt.step()
+        losses.append(loss.item())
+    return losses
+
+
+def _worker(rank, world, tmpdir, q):
+    dist.init_process_group("gloo", init_method=f"file://{tmpdir}/pg",
+                            rank=rank, world_size=world)
+    try:
+        model = _model()
+        opt = NoamAdam(model, 16, warmup_steps=10, use_flat=True)
+        ddp = BucketedDataParallel(opt.flat, bucket_mb=0.01)  # many buckets
+        ddp.broadcast_parameters()
+        # one backward + all-reduce, capture the synchronized gradient
+        (inp, tar) = _batches(1)[0]
+        b = inp.shape[0]
+        lo, hi = rank * b // world, (rank + 1) * b // world
+        logits, _ = model((inp[lo:hi], tar[lo:hi]), training=True)
+        loss = R.masked_cross_entropy(logits, tar[lo:hi], batch_size=b)
+        opt.zero_grad()
+        loss.backward()
+        ddp.finalize()
+        if rank == 0:
+            q.put((opt.flat.flat_g.detach().clone(), loss.item()))
+        # then 2 full optimizer steps must run without deadlock
+        _train_steps(model, opt, _batches(2), ddp=ddp, rank=rank, world=world)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_dp2_gradients_match_dp1(tmp_path):
+    """After bucketed SUM all-reduce, the DP=2 gradient equals the full
+    global-batch gradient (X2 semantics; Q4 loss pre-scaling makes SUM the
+    correct mean — no extra division)."""
+    model = _model()
+    opt = NoamAdam(model, 16, warmup_steps=10, use_flat=True)
+    (inp, tar) = _batches(1)[0]
+    logits, _ = model((inp, tar), training=True)
+    loss = R.masked_cross_entropy(logits, tar, batch_size=inp.shape[0])
+    opt.zero_grad()
+    loss.backward()
+    base_g = opt.flat.flat_g.detach().clone()
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    ps = [ctx.Process(target=_worker, args=(r, 2, str(tmp_path), q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    dp_g, dp_loss = q.get()
+    for p in ps:
+        p.join(120)
+        assert p.exitcode == 0
+    # sum-of-halves == full-batch gradient up to fp32 summation roundoff
+    scale = base_g.abs().max().item()
+    assert torch.allclose(base_g, dp_g, atol=1e-5 * max(scale, 1.0)), \
+        (base_g - dp_g).abs().max()
+
+
+def test_bucket_partition_covers_all_params():
+    model = _model()
+    opt = NoamAdam(model, 16, use_flat=True)
+    ddp = BucketedDataParallel(opt.flat, bucket_mb=0.01)
+    covered = set()
+    for b in ddp.buckets:
+        for p in b["params"]:
+            covered.add(id(p))
+        assert b["hi"] > b["lo"]
+    assert covered == {id(p) for p in opt.flat.params}
+    # buckets assemble in reverse registration order
+    first_bucket_params = ddp.buckets[0]["params"]
+    assert opt.flat.params[-1] in first_bucket_params
+
+
+def test_flat_grad_aliasing_survives_backward():
+    model = _model()
+    opt = NoamAdam(model, 16, use_flat=True)
+    inp = torch.randint(1, 50, (2, 5))
+    tar = torch.randint(1, 50, (2, 5))
+    for _ in range(2):  # twice: first sets, second accumulates
+        logits, _ = model((inp, tar), training=True)
+        loss = R.masked_cross_entropy(logits, tar, 2)
+        loss.backward()
+    opt.flat.check()
+    assert opt.flat.flat_g.abs().sum() > 0

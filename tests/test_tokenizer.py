@@ -1,0 +1,42 @@
+import os
+
+from transformer_amd.data import SubwordTokenizer
+
+
+CORPUS = [
+    "he goes to school",
+    "the cat sees a dog",
+    "one two three four five",
+    "the dog is big and the cat is small",
+    "she sees the red house",
+]
+
+
+def test_roundtrip():
+    tok = SubwordTokenizer.build_from_corpus(CORPUS, target_vocab_size=500)
+    for line in CORPUS:
+        ids = tok.encode(line)
+        assert all(0 < i < tok.vocab_size for i in ids)
+        assert tok.decode(ids) == line
+
+
+def test_oov_bytes_roundtrip():
+    tok = SubwordTokenizer.build_from_corpus(CORPUS, target_vocab_size=300)
+    s = "zürich blitz 😀"
+    assert tok.decode(tok.encode(s)) == s
+
+
+def test_pad_id_reserved():
+    tok = SubwordTokenizer.build_from_corpus(CORPUS, target_vocab_size=300)
+    assert 0 not in tok.encode("the cat sees the dog")
+
+
+def test_save_load(tmp_path):
+    tok = SubwordTokenizer.build_from_corpus(CORPUS, target_vocab_size=300)
+    prefix = str(tmp_path / "vocab")
+    tok.save_to_file(prefix)
+    assert os.path.exists(prefix + ".subwords")
+    tok2 = SubwordTokenizer.load_from_file(prefix)
+    assert tok2.vocab_size == tok.vocab_size
+    for line in CORPUS:
+        assert tok2.encode(line) == tok.encode(line)

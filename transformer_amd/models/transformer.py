@@ -1,0 +1,261 @@
+"""Encoder-decoder Transformer, MI355X-first.
+
+Capability parity with the reference model (reference Transformer.py:5-32,
+Encoder.py, Decoder.py, Attention.py, point_ffn.py — SURVEY.md §2.1 C1-C10):
+post-LN residuals, three attention variants (encoder self bidirectional+pad,
+decoder causal+pad, cross q-len≠kv-len+pad), sinusoidal PE in the reference's
+CONCAT layout, embedding ×√d, final linear logits head.
+
+MI355X-first differences from the reference's structure (not behaviour):
+  * QKV projections are packed into one fused MFMA GEMM;
+    cross-attention packs KV.
+  * Attention runs as one fused flash-style HIP kernel in (B,S,H,dh) layout —
+    no head split/merge transposes, no S×S materialization.
+  * Residual-add + LayerNorm are fused (K9); embedding+scale+PE fused (K10).
+  * Attention weights are only materialized in inspection mode
+    (`return_attention=True`, eager path) — SURVEY.md §8 Q13.
+
+On CPU the same modules run the fp32 reference math (ops/reference.py), which
+is the numerics oracle for every kernel.
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+from ..ops import reference as R
+
+
+def _glorot(out_f: int, in_f: int) -> torch.Tensor:
+    """Keras Dense default initializer (glorot_uniform), matching the
+    reference's tf.keras.layers.Dense (reference Attention.py:46-50)."""
+    limit = math.sqrt(6.0 / (in_f + out_f))
+    return torch.empty(out_f, in_f).uniform_(-limit, limit)
+
+
+class SelfAttention(nn.Module):
+    """Multi-head self-attention with packed QKV projection (C2 / K1)."""
+
+    def __init__(self, d_model: int, num_heads: int):
+        super().__init__()
+        assert d_model % num_heads == 0
+        self.d_model, self.num_heads = d_model, num_heads
+        self.depth = d_model // num_heads
+        w = torch.cat([_glorot(d_model, d_model) for _ in range(3)], dim=0)
+        self.w_qkv = nn.Parameter(w)              # (3d, d)
+        self.b_qkv = nn.Parameter(torch.zeros(3 * d_model))
+        self.w_o = nn.Parameter(_glorot(d_model, d_model))
+        self.b_o = nn.Parameter(torch.zeros(d_model))
+
+    def forward(self, x, kv_pad=None, causal=False, return_weights=False):
+        B, S, d = x.shape
+        qkv = ops.linear(x, self.w_qkv, self.b_qkv)          # (B,S,3d)
+        qkv = qkv.view(B, S, 3, self.num_heads, self.depth)
+        q, k, v = qkv.unbind(dim=2)                          # (B,S,H,dh) each
+        out = ops.fused_attention(q, k, v, kv_pad=kv_pad, causal=causal,
+                                  return_weights=return_weights)
+        if return_weights:
+            out, w = out
+        out = out.reshape(B, S, d)
+        out = ops.linear(out, self.w_o, self.b_o)
+        return (out, w) if return_weights else (out, None)
+
+
+class CrossAttention(nn.Module):
+    """Decoder-encoder cross attention with packed KV projection (C2)."""
+
+    def __init__(self, d_model: int, num_heads: int):
+        super().__init__()
+        assert d_model % num_heads == 0
+        self.d_model, self.num_heads = d_model, num_heads
+        self.depth = d_model // num_heads
+        self.w_q = nn.Parameter(_glorot(d_model, d_model))
+        self.b_q = nn.Parameter(torch.zeros(d_model))
+        w = torch.cat([_glorot(d_model, d_model) for _ in range(2)], dim=0)
+        self.w_kv = nn.Parameter(w)               # (2d, d)
+        self.b_kv = nn.Parameter(torch.zeros(2 * d_model))
+        self.w_o = nn.Parameter(_glorot(d_model, d_model))
+        self.b_o = nn.Parameter(torch.zeros(d_model))
+
+    def forward(self, x, enc_output, kv_pad=None, return_weights=False):
+        B, Sq, d = x.shape
+        Sk = enc_output.shape[1]
+        q = ops.linear(x, self.w_q, self.b_q).view(B, Sq, self.num_heads, self.depth)
+        kv = ops.linear(enc_output, self.w_kv, self.b_kv)
+        kv = kv.view(B, Sk, 2, self.num_heads, self.depth)
+        k, v = kv.unbind(dim=2)
+        out = ops.fused_attention(q, k, v, kv_pad=kv_pad, causal=False,
+                                  return_weights=return_weights)
+        if return_weights:
+            out, w = out
+        out = out.reshape(B, Sq, d)
+        out = ops.linear(out, self.w_o, self.b_o)
+        return (out, w) if return_weights else (out, None)
+
+
+class FeedForward(nn.Module):
+    """Position-wise FFN: Dense(dff, relu) -> Dense(d) (C3 / K8), ReLU fused
+    into the first GEMM's epilogue."""
+
+    def __init__(self, d_model: int, dff: int):
+        super().__init__()
+        self.w1 = nn.Parameter(_glorot(dff, d_model))
+        self.b1 = nn.Parameter(torch.zeros(dff))
+        self.w2 = nn.Parameter(_glorot(d_model, dff))
+        self.b2 = nn.Parameter(torch.zeros(d_model))
+
+    def forward(self, x):
+        h = ops.linear(x, self.w1, self.b1, activation="relu")
+        return ops.linear(h, self.w2, self.b2)
+
+
+class _LN(nn.Module):
+    def __init__(self, d_model: int, eps: float = 1e-6):
+        super().__init__()
+        self.gamma = nn.Parameter(torch.ones(d_model))
+        self.beta = nn.Parameter(torch.zeros(d_model))
+        self.eps = eps
+
+
+class EncoderLayer(nn.Module):
+    """MHA -> dropout -> LN(x+.) -> FFN -> dropout -> LN(.+.) (C6,
+    reference Encoder.py:6-29, post-LN)."""
+
+    def __init__(self, d_model, num_heads, dff, rate=0.1):
+        super().__init__()
+        self.mha = SelfAttention(d_model, num_heads)
+        self.ffn = FeedForward(d_model, dff)
+        self.ln1 = _LN(d_model)
+        self.ln2 = _LN(d_model)
+        self.rate = rate
+
+    def forward(self, x, kv_pad, training):
+        attn, _ = self.mha(x, kv_pad=kv_pad, causal=False)
+        attn = ops.dropout(attn, self.rate, training)
+        out1 = ops.residual_layernorm(attn, x, self.ln1.gamma, self.ln1.beta, self.ln1.eps)
+        ffn = self.ffn(out1)
+        ffn = ops.dropout(ffn, self.rate, training)
+        return ops.residual_layernorm(ffn, out1, self.ln2.gamma, self.ln2.beta, self.ln2.eps)
+
+
+class DecoderLayer(nn.Module):
+    """masked self-MHA -> cross-MHA -> FFN, each with dropout + post-LN
+    residual (C8, reference Decoder.py:7-42)."""
+
+    def __init__(self, d_model, num_heads, dff, rate=0.1):
+        super().__init__()
+        self.mha1 = SelfAttention(d_model, num_heads)
+        self.mha2 = CrossAttention(d_model, num_heads)
+        self.ffn = FeedForward(d_model, dff)
+        self.ln1 = _LN(d_model)
+        self.ln2 = _LN(d_model)
+        self.ln3 = _LN(d_model)
+        self.rate = rate
+
+    def forward(self, x, enc_output, tgt_pad, src_pad, training,
+                return_weights=False):
+        attn1, w1 = self.mha1(x, kv_pad=tgt_pad, causal=True,
+                              return_weights=return_weights)
+        attn1 = ops.dropout(attn1, self.rate, training)
+        out1 = ops.residual_layernorm(attn1, x, self.ln1.gamma, self.ln1.beta, self.ln1.eps)
+        attn2, w2 = self.mha2(out1, enc_output, kv_pad=src_pad,
+                              return_weights=return_weights)
+        attn2 = ops.dropout(attn2, self.rate, training)
+        out2 = ops.residual_layernorm(attn2, out1, self.ln2.gamma, self.ln2.beta, self.ln2.eps)
+        ffn = self.ffn(out2)
+        ffn = ops.dropout(ffn, self.rate, training)
+        out3 = ops.residual_layernorm(ffn, out2, self.ln3.gamma, self.ln3.beta, self.ln3.eps)
+        return out3, w1, w2
+
+
+class Encoder(nn.Module):
+    """Embedding ×√d + PE + dropout -> L encoder layers (C7,
+    reference Encoder.py:31-60)."""
+
+    def __init__(self, num_layers, d_model, num_heads, dff, input_vocab_size,
+                 rate=0.1, max_position=4096):
+        super().__init__()
+        self.d_model, self.num_layers = d_model, num_layers
+        self.embedding = nn.Parameter(torch.empty(input_vocab_size, d_model)
+                                      .uniform_(-0.05, 0.05))
+        # PE table sized by max seq len, not vocab; concat
+        # layout (Q2). Stored fp32, registered as a buffer (not a parameter).
+        pe = R.positional_encoding(max_position, d_model).squeeze(0)
+        self.register_buffer("pe", pe, persistent=False)
+        self.layers = nn.ModuleList(
+            EncoderLayer(d_model, num_heads, dff, rate) for _ in range(num_layers))
+        self.rate = rate
+
+    def forward(self, tokens, src_pad, training):
+        x = ops.embedding_scale_pe(tokens, self.embedding, self.pe)
+        x = ops.dropout(x, self.rate, training)
+        for layer in self.layers:
+            x = layer(x, src_pad, training)
+        return x
+
+
+class Decoder(nn.Module):
+    """Embedding + PE -> L decoder layers; collects per-layer attention
+    weights (keys 'decoder_layer{i}_block{1,2}', reference Decoder.py:75-76)
+    only in inspection mode."""
+
+    def __init__(self, num_layers, d_model, num_heads, dff, target_vocab_size,
+                 rate=0.1, max_position=4096):
+        super().__init__()
+        self.d_model, self.num_layers = d_model, num_layers
+        self.embedding = nn.Parameter(torch.empty(target_vocab_size, d_model)
+                                      .uniform_(-0.05, 0.05))
+        pe = R.positional_encoding(max_position, d_model).squeeze(0)
+        self.register_buffer("pe", pe, persistent=False)
+        self.layers = nn.ModuleList(
+            DecoderLayer(d_model, num_heads, dff, rate) for _ in range(num_layers))
+        self.rate = rate
+
+    def forward(self, tokens, enc_output, tgt_pad, src_pad, training,
+                return_weights=False):
+        x = ops.embedding_scale_pe(tokens, self.embedding, self.pe)
+        x = ops.dropout(x, self.rate, training)
+        attention_weights = {}
+        for i, layer in enumerate(self.layers):
+            x, w1, w2 = layer(x, enc_output, tgt_pad, src_pad, training,
+                              return_weights=return_weights)
+            if return_weights:
+                attention_weights[f"decoder_layer{i + 1}_block1"] = w1
+                attention_weights[f"decoder_layer{i + 1}_block2"] = w2
+        return x, attention_weights
+
+
+class Transformer(nn.Module):
+    """Top-level model (C10, reference Transformer.py:5-32).
+
+    call((inp, tar), training) -> (logits, attention_weights) — the reference
+    signature; attention_weights is {} unless return_weights=True (Q13)."""
+
+    def __init__(self, num_layers, d_model, num_heads, dff, input_vocab_size,
+                 target_vocab_size, rate=0.1, max_position=4096):
+        super().__init__()
+        self.encoder = Encoder(num_layers, d_model, num_heads, dff,
+                               input_vocab_size, rate, max_position)
+        self.decoder = Decoder(num_layers, d_model, num_heads, dff,
+                               target_vocab_size, rate, max_position)
+        self.w_final = nn.Parameter(_glorot(target_vocab_size, d_model))
+        self.b_final = nn.Parameter(torch.zeros(target_vocab_size))
+
+    def forward(self, inputs, training=False, return_weights=False):
+        inp, tar = inputs
+        # mask construction (C5) folds into the attention kernels' predicate
+        # logic: only per-token pad flags cross the boundary.
+        src_pad = (inp == 0)
+        tgt_pad = (tar == 0)
+        enc_output = self.encoder(inp, src_pad, training)
+        dec_output, attention_weights = self.decoder(
+            tar, enc_output, tgt_pad, src_pad, training, return_weights)
+        logits = ops.linear(dec_output, self.w_final, self.b_final)
+        return logits, attention_weights
+
+    # convenience alias matching the reference's .call
+    call = forward

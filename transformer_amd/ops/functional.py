@@ -1,0 +1,272 @@
+"""Autograd-integrated op layer with device dispatch.
+
+GPU (CUDA/ROCm) tensors run the hand-written CDNA4 HIP kernels via
+`torch.autograd.Function` wrappers with hand-written backward kernels
+(SURVEY.md §2.3); CPU tensors run the fp32 reference math in reference.py
+through plain differentiable torch ops.  The GPU path never falls back:
+if the extension is missing it raises (see ops/__init__.py).
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+
+from . import reference as R
+from . import ext
+
+
+# ---------------------------------------------------------------------------
+# Linear: y = x @ W^T + b, optional fused ReLU epilogue (K1/K7/K8/K12).
+# ---------------------------------------------------------------------------
+
+class _LinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b, activation):
+        # x: (M, K) bf16; w: (N, K) bf16; b: (N,) bf16 or None
+        E = ext()
+        y = E.gemm_nt(x, w, b if b is not None else torch.Tensor(),
+                      1 if activation == "relu" else 0)
+        ctx.activation = activation
+        ctx.has_bias = b is not None
+        ctx.save_for_backward(x, w, y if activation == "relu" else torch.Tensor())
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        E = ext()
+        x, w, y = ctx.saved_tensors
+        dy = dy.contiguous()
+        if ctx.activation == "relu":
+            dy = E.relu_bwd(dy, y)  # dz = dy * (y > 0)
+        # dx[M,K] = dy[M,N] @ W[N,K]  -> NT gemm with B = W^T
+        wt = E.transpose2d(w)
+        dx = E.gemm_nt(dy, wt, torch.Tensor(), 0)
+        # dW[N,K] = dy^T[N,M] @ x[M,K] -> NT gemm with A = dy^T, B = x^T
+        dyt = E.transpose2d(dy)
+        xt = E.transpose2d(x)
+        dw = E.gemm_nt(dyt, xt, torch.Tensor(), 0)
+        db = E.colsum(dy) if ctx.has_bias else None
+        return dx, dw, db, None
+
+
+def linear(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor | None = None,
+           activation: str | None = None) -> torch.Tensor:
+    """y = x @ W^T + b (+ReLU). x may have leading batch dims."""
+    if x.is_cuda:
+        shp = x.shape
+        y = _LinearFn.apply(x.reshape(-1, shp[-1]).contiguous(), w, b, activation)
+        return y.view(*shp[:-1], w.shape[0])
+    y = torch.nn.functional.linear(x, w, b)
+    if activation == "relu":
+        y = torch.relu(y)
+    return y
+
+
+# ---------------------------------------------------------------------------
+# Fused flash-style attention (K2-K5, K14). Layout (B, S, H, dh) throughout —
+# the head split/merge transposes of reference Attention.py:52-57,74-76 are
+# eliminated.
+# ---------------------------------------------------------------------------
+
+class _AttentionFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, kv_pad, causal, scale):
+        E = ext()
+        o, lse = E.attn_fwd(q, k, v,
+                            kv_pad if kv_pad is not None else torch.Tensor(),
+                            causal, scale)
+        ctx.causal = causal
+        ctx.scale = scale
+        ctx.save_for_backward(q, k, v, o, lse,
+                              kv_pad if kv_pad is not None else torch.Tensor())
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        E = ext()
+        q, k, v, o, lse, kv_pad = ctx.saved_tensors
+        dq, dk, dv = E.attn_bwd(q, k, v, o, do.contiguous(), lse, kv_pad,
+                                ctx.causal, ctx.scale)
+        return dq, dk, dv, None, None, None
+
+
+def fused_attention(q, k, v, kv_pad=None, causal=False, return_weights=False):
+    """q: (B,Sq,H,dh), k/v: (B,Sk,H,dh); kv_pad: (B,Sk) bool/uint8, True=pad.
+
+    Masking semantics match reference Attention.py:25-26: masked logits get
+    -1e9 added before softmax. Returns (B,Sq,H,dh); attention weights only on
+    the eager/inspection path."""
+    scale = 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda and not return_weights:
+        kp = kv_pad.to(torch.uint8).contiguous() if kv_pad is not None else None
+        return _AttentionFn.apply(q.contiguous(), k.contiguous(), v.contiguous(),
+                                  kp, causal, scale)
+    # eager path (CPU, or weight inspection): (B,S,H,dh) -> (B,H,S,dh)
+    qt, kt, vt = (t.permute(0, 2, 1, 3) for t in (q, k, v))
+    mask = None
+    B, Sq, H, _ = q.shape
+    Sk = k.shape[1]
+    if kv_pad is not None:
+        mask = kv_pad.to(torch.float32)[:, None, None, :]
+    if causal:
+        la = R.create_look_ahead_mask(Sq, device=q.device)
+        assert Sq == Sk, "causal attention requires Sq == Sk"
+        mask = la[None, None] if mask is None else torch.maximum(mask, la[None, None])
+    if q.dtype in (torch.bfloat16, torch.float16):
+        qt, kt, vt = qt.float(), kt.float(), vt.float()
+    out, w = R.scaled_dot_product_attention(qt, kt, vt, mask,
+                                            return_weights=True)
+    out = out.permute(0, 2, 1, 3).to(q.dtype)
+    if return_weights:
+        return out, w
+    return out
+
+
+# ---------------------------------------------------------------------------
+# Fused residual-add + LayerNorm, eps=1e-6 (K9).
+# ---------------------------------------------------------------------------
+
+class _ResidualLNFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, res, gamma, beta, eps):
+        E = ext()
+        y, s, mean, rstd = E.ln_fwd(x, res, gamma, beta, eps)
+        ctx.save_for_backward(s, gamma, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        E = ext()
+        s, gamma, mean, rstd = ctx.saved_tensors
+        dx, dgamma, dbeta = E.ln_bwd(dy.contiguous(), s, gamma, mean, rstd)
+        # d/dx and d/dres are identical (y = LN(x + res))
+        return dx, dx, dgamma, dbeta, None
+
+
+def residual_layernorm(x, res, gamma, beta, eps: float = 1e-6):
+    if x.is_cuda:
+        shp = x.shape
+        d = shp[-1]
+        y = _ResidualLNFn.apply(x.reshape(-1, d).contiguous(),
+                                res.reshape(-1, d).contiguous(), gamma, beta, eps)
+        return y.view(shp)
+    if x.dtype in (torch.bfloat16, torch.float16):
+        return R.residual_layernorm(x.float(), res.float(), gamma.float(),
+                                    beta.float(), eps).to(x.dtype)
+    return R.residual_layernorm(x, res, gamma, beta, eps)
+
+
+# ---------------------------------------------------------------------------
+# Fused embedding * sqrt(d) + positional encoding (K10).
+# ---------------------------------------------------------------------------
+
+class _EmbedPEFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, tokens, weight, pe):
+        E = ext()
+        y = E.embed_pe_fwd(tokens, weight, pe)
+        ctx.vocab = weight.shape[0]
+        ctx.save_for_backward(tokens)
+        ctx.wdtype = weight.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        E = ext()
+        (tokens,) = ctx.saved_tensors
+        dw = E.embed_pe_bwd(dy.contiguous(), tokens, ctx.vocab)
+        return None, dw, None
+
+
+def embedding_scale_pe(tokens, weight, pe):
+    """tokens (B,S) int; weight (V,d); pe (P,d) fp32 table, P >= S."""
+    if weight.is_cuda:
+        return _EmbedPEFn.apply(tokens.contiguous(), weight, pe)
+    if weight.dtype in (torch.bfloat16, torch.float16):
+        return R.embedding_scale_pe(tokens, weight.float(), pe[None].float()).to(weight.dtype)
+    return R.embedding_scale_pe(tokens, weight, pe[None].to(weight.dtype))
+
+
+# ---------------------------------------------------------------------------
+# Dropout (K11): mask saved for exact backward; seeds from torch RNG.
+# ---------------------------------------------------------------------------
+
+class _DropoutFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, p):
+        E = ext()
+        seed = int(torch.randint(0, 2**31 - 1, (1,)).item())
+        y, mask = E.dropout_fwd(x, p, seed)
+        ctx.p = p
+        ctx.save_for_backward(mask)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        E = ext()
+        (mask,) = ctx.saved_tensors
+        return E.dropout_bwd(dy.contiguous(), mask, ctx.p), None
+
+
+def dropout(x, p: float, training: bool):
+    if not training or p == 0.0:
+        return x
+    if x.is_cuda:
+        shp = x.shape
+        return _DropoutFn.apply(x.reshape(-1).contiguous(), p).view(shp)
+    return torch.nn.functional.dropout(x, p=p, training=True)
+
+
+# ---------------------------------------------------------------------------
+# Fused padding-masked (label-smoothed) cross entropy (K13).
+# ---------------------------------------------------------------------------
+
+class _CrossEntropyFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, targets, batch_size, label_smoothing):
+        E = ext()
+        loss, lse = E.ce_fwd(logits, targets, float(batch_size), label_smoothing)
+        ctx.batch_size = float(batch_size)
+        ctx.ls = label_smoothing
+        ctx.save_for_backward(logits, targets, lse)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        E = ext()
+        logits, targets, lse = ctx.saved_tensors
+        dlogits = E.ce_bwd(logits, targets, lse, float(dloss.item()),
+                           ctx.batch_size, ctx.ls)
+        return dlogits, None, None, None
+
+
+def masked_cross_entropy(logits, targets, batch_size, label_smoothing=0.0):
+    """sum(CE * pad_mask)/batch_size (reference train.py:83-88)."""
+    if logits.is_cuda:
+        B, T, V = logits.shape
+        return _CrossEntropyFn.apply(logits.reshape(-1, V).contiguous(),
+                                     targets.reshape(-1).contiguous(),
+                                     batch_size, label_smoothing)
+    return R.masked_cross_entropy(logits, targets, batch_size, label_smoothing)
+
+
+# ---------------------------------------------------------------------------
+# Metrics / decode kernels (K16, K17).
+# ---------------------------------------------------------------------------
+
+def argmax_lastdim(logits):
+    if logits.is_cuda:
+        shp = logits.shape
+        return ext().argmax_lastdim(logits.reshape(-1, shp[-1]).contiguous()).view(shp[:-1])
+    return logits.argmax(dim=-1)
+
+
+def masked_accuracy(logits, targets):
+    if logits.is_cuda:
+        B, T, V = logits.shape
+        correct, total = ext().accuracy(logits.reshape(-1, V).contiguous(),
+                                        targets.reshape(-1).contiguous())
+        return correct / max(total, 1)
+    return float(R.masked_accuracy(logits, targets))

@@ -1,0 +1,158 @@
+"""Noam-scheduled Adam (C11+C12, K15) with flat-buffer fused GPU path.
+
+MI355X design: all model parameters live as views into ONE contiguous bf16
+buffer, gradients accumulate into one contiguous bf16 buffer, and the Adam
+update (β1=0.9, β2=0.98, ε=1e-9 — reference train.py:65-66) runs as a single
+fused multi-element HIP kernel over fp32 master weights + m/v state
+(SURVEY.md §2.4 design (b): bf16 wire gradients, fp32 Adam master).  The
+flat gradient buffer is also what the DP runtime all-reduces in bucket
+slices (parallel/ddp.py) — fewer, larger collectives for xGMI.
+
+CPU path: per-parameter fp32 Adam via ops/reference.py (the kernel's
+numerics oracle).
+"""
+
+from __future__ import annotations
+
+import torch
+
+from ..ops import ext, has_ext, reference as R
+from .schedule import NoamSchedule
+
+_ALIGN = 64  # elements; keeps every param slice 128-byte aligned for kernels
+
+
+class FlatParams:
+    """Rebind every parameter of `model` as a view into one flat buffer and
+    pre-assign .grad views into a flat gradient buffer.
+
+    Must be called AFTER model.to(device, dtype).  Autograd accumulates
+    in-place into existing .grad tensors, so gradients land directly in the
+    flat buffer; `check()` asserts that aliasing survived (fails loudly if a
+    torch version rebinds .grad)."""
+
+    def __init__(self, model: torch.nn.Module):
+        params = [p for p in model.parameters() if p.requires_grad]
+        if not params:
+            raise ValueError("model has no trainable parameters")
+        dev, dt = params[0].device, params[0].dtype
+        offs, total = [], 0
+        for p in params:
+            offs.append(total)
+            total += (p.numel() + _ALIGN - 1) // _ALIGN * _ALIGN
+        self.flat_w = torch.zeros(total, device=dev, dtype=dt)
+        self.flat_g = torch.zeros(total, device=dev, dtype=dt)
+        self.params = params
+        self.offsets = offs
+        self.numel = total
+        for p, off in zip(params, offs):
+            n = p.numel()
+            self.flat_w[off:off + n].copy_(p.data.reshape(-1))
+            p.data = self.flat_w[off:off + n].view(p.shape)
+            p.grad = self.flat_g[off:off + n].view(p.shape)
+
+    def check(self):
+        base = self.flat_g.data_ptr()
+        end = base + self.flat_g.numel() * self.flat_g.element_size()
+        for p in self.params:
+            if p.grad is None or not (base <= p.grad.data_ptr() < end):
+                raise RuntimeError(
+                    "flat-gradient aliasing broken: autograd rebound .grad "
+                    "(check torch accumulate_grad in-place semantics)")
+
+    def zero_grad(self):
+        self.flat_g.zero_()
+
+
+class NoamAdam:
+    """Adam with the Noam schedule computed host-side per step (K15).
+
+    GPU: one fused HIP kernel over the whole flat parameter set.
+    CPU: reference per-tensor fp32 Adam."""
+
+    def __init__(self, model: torch.nn.Module, d_model: int,
+                 warmup_steps: int = 60000, betas=(0.9, 0.98), eps: float = 1e-9,
+                 use_flat: bool | None = None):
+        self.schedule = NoamSchedule(d_model, warmup_steps)
+        self.betas, self.eps = betas, eps
+        self.step_count = 0
+        p0 = next(model.parameters())
+        self.is_cuda = p0.is_cuda
+        if use_flat is None:
+            use_flat = self.is_cuda
+        self.flat: FlatParams | None = None
+        if use_flat:
+            self.flat = FlatParams(model)
+            self.master = self.flat.flat_w.float()
+            self.m = torch.zeros_like(self.master)
+            self.v = torch.zeros_like(self.master)
+        else:
+            self.params = [p for p in model.parameters() if p.requires_grad]
+            self.state = [
+                {"m": torch.zeros_like(p, dtype=torch.float32),
+                 "v": torch.zeros_like(p, dtype=torch.float32),
+                 "master": p.detach().float().clone()}
+                for p in self.params
+            ]
+
+    @property
+    def lr(self) -> float:
+        return self.schedule(max(self.step_count, 1))
+
+    def zero_grad(self, set_to_none: bool = False):
+        if self.flat is not None:
+            self.flat.zero_grad()
+        else:
+            for p in self.params:
+                if p.grad is not None:
+                    p.grad.zero_()
+
+    @torch.no_grad()
+    def step(self):
+        self.step_count += 1
+        lr = self.schedule(self.step_count)
+        b1, b2 = self.betas
+        if self.flat is not None:
+            self.flat.check()
+            if self.is_cuda:
+                ext().adam_fused(self.master, self.m, self.v,
+                                 self.flat.flat_g, self.flat.flat_w,
+                                 lr, b1, b2, self.eps, self.step_count)
+            else:
+                g = self.flat.flat_g.float()
+                R.adam_step_reference(self.master, g, self.m, self.v,
+                                      self.step_count, lr, b1, b2, self.eps)
+                self.flat.flat_w.copy_(self.master.to(self.flat.flat_w.dtype))
+        else:
+            for p, st in zip(self.params, self.state):
+                if p.grad is None:
+                    continue
+                R.adam_step_reference(st["master"], p.grad.float(), st["m"],
+                                      st["v"], self.step_count, lr, b1, b2,
+                                      self.eps)
+                p.data.copy_(st["master"].to(p.dtype))
+
+    # -- checkpointing (C16) ------------------------------------------------
+    def state_dict(self):
+        d = {"step": self.step_count,
+             "d_model": self.schedule.d_model,
+             "warmup_steps": self.schedule.warmup_steps}
+        if self.flat is not None:
+            d.update(master=self.master, m=self.m, v=self.v)
+        else:
+            d.update(state=[{k: v for k, v in st.items()} for st in self.state])
+        return d
+
+    def load_state_dict(self, d):
+        self.step_count = d["step"]
+        if self.flat is not None:
+            self.master.copy_(d["master"].to(self.master.device))
+            self.m.copy_(d["m"].to(self.m.device))
+            self.v.copy_(d["v"].to(self.v.device))
+            self.flat.flat_w.copy_(self.master.to(self.flat.flat_w.dtype))
+        else:
+            for st, sd in zip(self.state, d["state"]):
+                for k in ("m", "v", "master"):
+                    st[k].copy_(sd[k].to(st[k].device))
+            for p, st in zip(self.params, self.state):
+                p.data.copy_(st["master"].to(p.dtype))

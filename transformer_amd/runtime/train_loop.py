@@ -1,0 +1,205 @@
+"""Training runtime (C17/C18/C19 — reference train.py:37-213) and the DP
+subclass (C20 — reference distributed_train.py:25-121).
+
+Train owns the optimizer (Noam-Adam), loss (padding-masked CE, sum/global
+batch — SURVEY.md §8 Q4), metrics, summary writers and checkpoint manager,
+exactly the reference's responsibilities (train.py:64-80).  Deliberate
+behaviour fixes over the reference, per SURVEY.md §8: masked accuracy (Q5),
+tgt-tokenizer ids + EOS stop in predict (Q6), checkpoint cadence intent
+"every 5 epochs or at end" (Q7), fixed eval-batch count (Q8).
+"""
+
+from __future__ import annotations
+
+import time
+
+import torch
+
+from .. import ops
+from ..runtime.optimizer import NoamAdam
+from ..runtime.metrics import Mean
+from ..runtime.summary import SummaryWriter
+from ..runtime.checkpoint import CheckpointManager
+
+
+class Train:
+    def __init__(self, epochs, enable_function, transformer, src_tokenizer,
+                 tgt_tokenizer, batch_size, train_log_dir, test_log_dir,
+                 max_ckpt_keep, ckpt_path, d_model,
+                 warmup_steps: int = 60000, label_smoothing: float = 0.0,
+                 device=None, log_interval: int = 100, eval_steps: int = 50,
+                 max_decode_len: int = 10, is_rank0: bool = True,
+                 use_flat: bool | None = None):
+        self.epochs = epochs
+        self.enable_function = enable_function
+        self.transformer = transformer
+        self.src_tokenizer = src_tokenizer
+        self.tgt_tokenizer = tgt_tokenizer
+        self.batch_size = batch_size  # GLOBAL batch (loss scale, Q4)
+        self.label_smoothing = label_smoothing
+        self.log_interval = log_interval
+        self.eval_steps = eval_steps
+        self.max_decode_len = max_decode_len
+        self.is_rank0 = is_rank0
+        p0 = next(transformer.parameters())
+        self.device = device if device is not None else p0.device
+
+        self.optimizer = NoamAdam(transformer, d_model, warmup_steps,
+                                  use_flat=use_flat)
+        self.train_loss = Mean("train_loss")
+        self.test_loss = Mean("test_loss")
+        self.train_accuracy = Mean("train_accuracy")
+        self.test_accuracy = Mean("test_accuracy")
+        self.train_summary_writer = (SummaryWriter(train_log_dir)
+                                     if (is_rank0 and train_log_dir) else None)
+        self.test_summary_writer = (SummaryWriter(test_log_dir)
+                                    if (is_rank0 and test_log_dir) else None)
+        self.ckpt_manager = CheckpointManager(transformer, self.optimizer,
+                                              ckpt_path, max_ckpt_keep)
+
+    # -- loss (C13, reference train.py:83-88) -------------------------------
+    def loss_function(self, real, pred):
+        return ops.masked_cross_entropy(pred, real, self.batch_size,
+                                        self.label_smoothing)
+
+    # -- steps (C18, reference train.py:124-156) ----------------------------
+    def _grad_sync(self):
+        """Hook point for the DP subclass (bucket finalize)."""
+
+    def train_step(self, inputs):
+        src, tar = inputs
+        src = src.to(self.device, non_blocking=True)
+        tar = tar.to(self.device, non_blocking=True)
+        tar_inp, tar_real = tar[:, :-1].contiguous(), tar[:, 1:].contiguous()
+
+        predictions, _ = self.transformer((src, tar_inp), training=True)
+        loss = self.loss_function(tar_real, predictions)
+        self.optimizer.zero_grad()
+        loss.backward()
+        self._grad_sync()
+        self.optimizer.step()
+
+        self.train_loss.update(loss.detach().item())
+        self.train_accuracy.update(ops.masked_accuracy(predictions.detach(),
+                                                       tar_real))
+        return loss
+
+    @torch.no_grad()
+    def test_step(self, inputs):
+        src, tar = inputs
+        src = src.to(self.device, non_blocking=True)
+        tar = tar.to(self.device, non_blocking=True)
+        tar_inp, tar_real = tar[:, :-1].contiguous(), tar[:, 1:].contiguous()
+        predictions, _ = self.transformer((src, tar_inp), training=False)
+        t_loss = self.loss_function(tar_real, predictions)
+        self.test_loss.update(t_loss.item())
+        self.test_accuracy.update(ops.masked_accuracy(predictions, tar_real))
+
+    # -- greedy inference (C19, reference train.py:91-121; Q6 fixes) --------
+    @torch.no_grad()
+    def predict(self, input_sentence):
+        if isinstance(input_sentence, (list, tuple)):
+            input_sentence = input_sentence[0]
+        src_start = self.src_tokenizer.vocab_size
+        src_end = src_start + 1
+        tgt_start = self.tgt_tokenizer.vocab_size
+        tgt_end = tgt_start + 1
+        tokens = [src_start] + self.src_tokenizer.encode(input_sentence) + [src_end]
+        encoder_input = torch.tensor([tokens], dtype=torch.int64,
+                                     device=self.device)
+        output = torch.tensor([[tgt_start]], dtype=torch.int64,
+                              device=self.device)
+        for _ in range(self.max_decode_len):
+            predictions, _ = self.transformer((encoder_input, output),
+                                              training=False)
+            predicted_id = ops.argmax_lastdim(predictions[:, -1:, :].float())
+            output = torch.cat([output, predicted_id.view(1, 1)], dim=-1)
+            if int(predicted_id) == tgt_end:
+                break
+        return output.squeeze(0).cpu()
+
+    def load_ckpt(self):
+        meta = self.ckpt_manager.restore()
+        if meta is not None:
+            print("Latest checkpoint restored!!")
+        return meta
+
+    # -- epoch loop (C17, reference train.py:167-213) -----------------------
+    def _run_eval(self, test_dataset):
+        self.test_loss.reset()
+        self.test_accuracy.reset()
+        n = 0
+        for batch in test_dataset:
+            self.test_step(batch)
+            n += 1
+            if n >= self.eval_steps:
+                break
+
+    def _save_if_due(self, epoch: int, step: int):
+        # Q7 intent: save every 5 epochs or at the end.
+        if self.is_rank0 and ((epoch + 1) % 5 == 0 or (epoch + 1) == self.epochs):
+            self.ckpt_manager.save(step, epoch)
+
+    def training_loop(self, train_dataset, test_dataset):
+        template = ("Epoch {}  Loss {:.4f} Accuracy {:.4f}, "
+                    "Test Loss {:.4f}, Test Accuracy {:.4f}")
+        step = self.optimizer.step_count
+        for epoch in range(self.epochs):
+            start = time.time()
+            for m in (self.train_loss, self.train_accuracy,
+                      self.test_loss, self.test_accuracy):
+                m.reset()
+            if hasattr(train_dataset, "set_epoch"):
+                train_dataset.set_epoch(epoch)
+
+            for inputs in train_dataset:
+                self.train_step(inputs)
+                step += 1
+                if step % self.log_interval == 0:
+                    self._run_eval(test_dataset)
+                    if self.is_rank0:
+                        print(template.format(
+                            epoch + 1, self.train_loss.result(),
+                            self.train_accuracy.result(),
+                            self.test_loss.result(),
+                            self.test_accuracy.result()))
+
+            for m in (self.train_loss, self.train_accuracy,
+                      self.test_loss, self.test_accuracy):
+                m.sync()
+            if self.train_summary_writer:
+                self.train_summary_writer.add_scalar("loss", self.train_loss.result(), epoch)
+                self.train_summary_writer.add_scalar("accuracy", self.train_accuracy.result(), epoch)
+            if self.test_summary_writer:
+                self.test_summary_writer.add_scalar("loss", self.test_loss.result(), epoch)
+                self.test_summary_writer.add_scalar("accuracy", self.test_accuracy.result(), epoch)
+            self._save_if_due(epoch, step)
+            if self.is_rank0:
+                print(template.format(
+                    epoch + 1, self.train_loss.result(),
+                    self.train_accuracy.result(),
+                    self.test_loss.result(), self.test_accuracy.result()))
+                print(f"Time taken for 1 epoch: {time.time() - start} secs\n")
+
+
+class DistributedTrain(Train):
+    """DP training over N GPUs (C20).  Differences from Train: bucketed
+    RCCL all-reduce finalization before each optimizer step, X1 parameter
+    broadcast at start, rank-0-only logging/checkpointing."""
+
+    def __init__(self, *args, ddp=None, **kwargs):
+        kwargs.setdefault("use_flat", True)
+        super().__init__(*args, **kwargs)
+        if ddp is None:
+            import torch.distributed as dist
+            if dist.is_available() and dist.is_initialized() \
+                    and dist.get_world_size() > 1:
+                from ..parallel import BucketedDataParallel
+                ddp = BucketedDataParallel(self.optimizer.flat)
+        self.ddp = ddp
+        if ddp is not None:
+            ddp.broadcast_parameters()
+
+    def _grad_sync(self):
+        if self.ddp is not None:
+            self.ddp.finalize()
