@@ -1,0 +1,304 @@
+"""Kernel numerics tests: every HIP kernel against the
+plain PyTorch fp32 reference in ops/reference.py, on the same bf16-rounded
+inputs.  All @pytest.mark.gpu — run on an MI355X via gpurun."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from transformer_amd.ops import ext
+    E = None
+
+
+def _ext():
+    global E
+    if E is None:
+        from transformer_amd.ops import ext as _e
+        E = _e()
+    return E
+
+
+def assert_close(got, ref, tol=0.03, name=""):
+    got = got.float().cpu()
+    ref = ref.float().cpu()
+    assert got.shape == ref.shape, (name, got.shape, ref.shape)
+    scale = ref.abs().max().clamp(min=1.0)
+    err = (got - ref).abs().max() / scale
+    assert err < tol, f"{name}: rel-max err {err:.4f} (tol {tol})"
+
+
+# ---------------------------------------------------------------------------
+# GEMM family
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("m,n,k", [(128, 128, 64), (256, 512, 512),
+                                   (100, 130, 72), (64, 32770 // 10, 512),
+                                   (33, 17, 24)])
+def test_gemm_nt(m, n, k):
+    torch.manual_seed(0)
+    a = torch.randn(m, k, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(n, k, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    c = _ext().gemm_nt(a, w, b, 0)
+    ref = a.float() @ w.float().T + b.float()
+    assert_close(c, ref, 0.03, "gemm_nt")
+
+
+def test_gemm_nt_relu():
+    torch.manual_seed(1)
+    a = torch.randn(200, 512, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(300, 512, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(300, device="cuda", dtype=torch.bfloat16)
+    c = _ext().gemm_nt(a, w, b, 1)
+    ref = torch.relu(a.float() @ w.float().T + b.float())
+    assert_close(c, ref, 0.03, "gemm_relu")
+    assert (c.float() >= 0).all()
+
+
+def test_gemm_no_bias():
+    a = torch.randn(64, 128, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(96, 128, device="cuda", dtype=torch.bfloat16)
+    c = _ext().gemm_nt(a, w, torch.Tensor(), 0)
+    assert_close(c, a.float() @ w.float().T, 0.03, "gemm_nobias")
+
+
+def test_transpose2d():
+    a = torch.randn(130, 70, device="cuda", dtype=torch.bfloat16)
+    t = _ext().transpose2d(a)
+    assert torch.equal(t.cpu(), a.cpu().T.contiguous())
+
+
+def test_colsum():
+    a = torch.randn(500, 300, device="cuda", dtype=torch.bfloat16)
+    s = _ext().colsum(a)
+    assert_close(s, a.float().sum(0), 0.03, "colsum")
+
+
+def test_relu_bwd():
+    dy = torch.randn(1000, device="cuda", dtype=torch.bfloat16)
+    y = torch.randn(1000, device="cuda", dtype=torch.bfloat16)
+    dz = _ext().relu_bwd(dy, y)
+    ref = dy.float() * (y.float() > 0)
+    assert_close(dz, ref, 1e-3, "relu_bwd")
+
+
+# ---------------------------------------------------------------------------
+# LayerNorm
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("r,d", [(256, 512), (100, 1024), (64, 128)])
+def test_ln_fwd_bwd(r, d):
+    from transformer_amd.ops import reference as R
+    torch.manual_seed(0)
+    x = torch.randn(r, d, device="cuda", dtype=torch.bfloat16)
+    res = torch.randn(r, d, device="cuda", dtype=torch.bfloat16)
+    gamma = (torch.randn(d, device="cuda") * 0.1 + 1).bfloat16()
+    beta = (torch.randn(d, device="cuda") * 0.1).bfloat16()
+    y, s, mean, rstd = _ext().ln_fwd(x, res, gamma, beta, 1e-6)
+    xf = x.float().requires_grad_(False)
+    ref_in = (x.float() + res.float()).requires_grad_(True)
+    gf = gamma.float().requires_grad_(True)
+    bf = beta.float().requires_grad_(True)
+    ref = torch.nn.functional.layer_norm(ref_in, (d,), gf, bf, 1e-6)
+    assert_close(y, ref, 0.03, "ln_fwd")
+    assert_close(s, x.float() + res.float(), 0.02, "ln_s")
+
+    dy = torch.randn(r, d, device="cuda", dtype=torch.bfloat16)
+    ref.backward(dy.float())
+    dx, dgamma, dbeta = _ext().ln_bwd(dy, s, gamma, mean, rstd)
+    assert_close(dx, ref_in.grad, 0.05, "ln_dx")
+    assert_close(dgamma, gf.grad, 0.05, "ln_dgamma")
+    assert_close(dbeta, bf.grad, 0.05, "ln_dbeta")
+
+
+# ---------------------------------------------------------------------------
+# Embedding + PE
+# ---------------------------------------------------------------------------
+
+def test_embed_pe_fwd_bwd():
+    from transformer_amd.ops import reference as R
+    torch.manual_seed(0)
+    V, d, B, S = 1000, 512, 4, 37
+    w = torch.randn(V, d, device="cuda", dtype=torch.bfloat16)
+    pe = R.positional_encoding(64, d).squeeze(0).to("cuda", torch.bfloat16)
+    toks = torch.randint(0, V, (B, S), device="cuda")
+    y = _ext().embed_pe_fwd(toks, w, pe)
+    ref = R.embedding_scale_pe(toks, w.float(), pe.float()[None])
+    assert_close(y, ref, 0.03, "embed_fwd")
+
+    dy = torch.randn(B, S, d, device="cuda", dtype=torch.bfloat16)
+    dw = _ext().embed_pe_bwd(dy, toks, V)
+    ref_w = w.float().requires_grad_(True)
+    out = R.embedding_scale_pe(toks, ref_w, pe.float()[None])
+    out.backward(dy.float())
+    assert_close(dw, ref_w.grad, 0.05, "embed_bwd")
+
+
+# ---------------------------------------------------------------------------
+# Dropout
+# ---------------------------------------------------------------------------
+
+def test_dropout_stats_and_bwd():
+    x = torch.ones(100000, device="cuda", dtype=torch.bfloat16)
+    y, mask = _ext().dropout_fwd(x, 0.1, 1234)
+    keep = mask.float().mean().item()
+    assert abs(keep - 0.9) < 0.01
+    # kept elements scaled by 1/(1-p)
+    yv = y.float()
+    assert ((yv - mask.float() * (1 / 0.9)).abs() < 0.01).all()
+    dy = torch.randn(100000, device="cuda", dtype=torch.bfloat16)
+    dx = _ext().dropout_bwd(dy, mask, 0.1)
+    ref = dy.float() * mask.float() / 0.9
+    assert_close(dx, ref, 0.01, "dropout_bwd")
+    # deterministic for same seed
+    y2, mask2 = _ext().dropout_fwd(x, 0.1, 1234)
+    assert torch.equal(mask, mask2)
+
+
+# ---------------------------------------------------------------------------
+# Cross entropy
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("ls", [0.0, 0.1])
+def test_ce_fwd_bwd(ls):
+    from transformer_amd.ops import reference as R
+    torch.manual_seed(0)
+    Rn, V = 64, 32770
+    logits = (torch.randn(Rn, V, device="cuda") * 2).bfloat16()
+    targets = torch.randint(0, V, (Rn,), device="cuda")
+    targets[::7] = 0  # some padding
+    loss, lse = _ext().ce_fwd(logits, targets, 8.0, ls)
+    lref = R.masked_cross_entropy(logits.float().view(1, Rn, V),
+                                  targets.view(1, Rn), 8, ls)
+    assert abs(loss.item() - lref.item()) / max(abs(lref.item()), 1) < 0.01
+
+    lg = logits.float().requires_grad_(True)
+    lr2 = R.masked_cross_entropy(lg.view(1, Rn, V), targets.view(1, Rn), 8, ls)
+    lr2.backward()
+    dl = _ext().ce_bwd(logits, targets, lse, 1.0, 8.0, ls)
+    assert_close(dl, lg.grad, 0.02, "ce_bwd")
+
+
+# ---------------------------------------------------------------------------
+# Adam
+# ---------------------------------------------------------------------------
+
+def test_adam_fused():
+    from transformer_amd.ops import reference as R
+    torch.manual_seed(0)
+    n = 10000
+    master = torch.randn(n, device="cuda")
+    m = torch.randn(n, device="cuda").abs() * 0.01
+    v = torch.randn(n, device="cuda").abs() * 0.01
+    grad = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    param = master.bfloat16()
+    mm, vv, mast = m.clone(), v.clone(), master.clone()
+    _ext().adam_fused(master, m, v, grad, param, 1e-3, 0.9, 0.98, 1e-9, 3)
+    R.adam_step_reference(mast, grad.float(), mm, vv, 3, 1e-3, 0.9, 0.98, 1e-9)
+    assert_close(master, mast, 1e-4, "adam_master")
+    assert_close(m, mm, 1e-4, "adam_m")
+    assert_close(v, vv, 1e-4, "adam_v")
+    assert_close(param, mast.bfloat16(), 0.01, "adam_param")
+
+
+# ---------------------------------------------------------------------------
+# argmax / accuracy
+# ---------------------------------------------------------------------------
+
+def test_argmax_accuracy():
+    torch.manual_seed(0)
+    logits = torch.randn(257, 1003, device="cuda", dtype=torch.bfloat16)
+    am = _ext().argmax_lastdim(logits)
+    ref = logits.float().argmax(-1)
+    assert torch.equal(am.cpu(), ref.cpu())
+    targets = torch.randint(0, 1003, (257,), device="cuda")
+    targets[:40] = 0
+    correct, total = _ext().accuracy(logits, targets)
+    mask = targets != 0
+    assert total == int(mask.sum())
+    assert correct == int(((ref == targets) & mask).sum())
+
+
+# ---------------------------------------------------------------------------
+# Attention (3 mask variants), fwd + bwd vs fp32 reference
+# ---------------------------------------------------------------------------
+
+def _attn_ref(q, k, v, kv_pad, causal):
+    """fp32 reference in (B,S,H,dh); mirrors ops.fused_attention eager path."""
+    import transformer_amd.ops.reference as R
+    qt, kt, vt = (t.float().permute(0, 2, 1, 3) for t in (q, k, v))
+    mask = None
+    if kv_pad is not None:
+        mask = kv_pad.float()[:, None, None, :]
+    if causal:
+        la = R.create_look_ahead_mask(q.shape[1], device=q.device)
+        mask = la[None, None] if mask is None else torch.maximum(mask, la[None, None])
+    out = R.scaled_dot_product_attention(qt, kt, vt, mask)
+    return out.permute(0, 2, 1, 3)
+
+
+@pytest.mark.parametrize("dh", [32, 64, 128])
+@pytest.mark.parametrize("causal", [False, True])
+def test_attn_fwd(dh, causal):
+    torch.manual_seed(0)
+    B, S, H = 2, 100, 3
+    q = torch.randn(B, S, H, dh, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, S, H, dh, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, S, H, dh, device="cuda", dtype=torch.bfloat16)
+    pad = torch.zeros(B, S, device="cuda", dtype=torch.uint8)
+    pad[0, 80:] = 1
+    o, lse = _ext().attn_fwd(q, k, v, pad, causal, 1.0 / math.sqrt(dh))
+    ref = _attn_ref(q, k, v, pad, causal)
+    assert_close(o, ref, 0.04, f"attn_fwd dh{dh} causal{causal}")
+
+
+def test_attn_fwd_cross_shapes():
+    torch.manual_seed(1)
+    B, Sq, Sk, H, dh = 2, 37, 75, 4, 64
+    q = torch.randn(B, Sq, H, dh, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, Sk, H, dh, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, Sk, H, dh, device="cuda", dtype=torch.bfloat16)
+    pad = torch.zeros(B, Sk, device="cuda", dtype=torch.uint8)
+    pad[1, 60:] = 1
+    o, _ = _ext().attn_fwd(q, k, v, pad, False, 1.0 / math.sqrt(dh))
+    ref = _attn_ref(q, k, v, pad, False)
+    assert_close(o, ref, 0.04, "attn_cross")
+
+
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize("dh", [64, 128])
+def test_attn_bwd(causal, dh):
+    torch.manual_seed(0)
+    B, S, H = 2, 64, 2
+    mk = lambda: torch.randn(B, S, H, dh, device="cuda", dtype=torch.bfloat16)
+    q, k, v = mk(), mk(), mk()
+    pad = torch.zeros(B, S, device="cuda", dtype=torch.uint8)
+    pad[0, 50:] = 1
+    scale = 1.0 / math.sqrt(dh)
+    o, lse = _ext().attn_fwd(q, k, v, pad, causal, scale)
+    do = mk()
+    dq, dk, dv = _ext().attn_bwd(q, k, v, o, do, lse, pad, causal, scale)
+
+    qf = q.float().requires_grad_(True)
+    kf = k.float().requires_grad_(True)
+    vf = v.float().requires_grad_(True)
+    ref = _attn_ref(qf, kf, vf, pad, causal)
+    ref.backward(do.float())
+    assert_close(dq, qf.grad, 0.06, "attn_dq")
+    assert_close(dk, kf.grad, 0.06, "attn_dk")
+    assert_close(dv, vf.grad, 0.06, "attn_dv")
+
+
+def test_attn_fwd_no_pad_mask():
+    torch.manual_seed(2)
+    B, S, H, dh = 1, 256, 2, 64
+    q = torch.randn(B, S, H, dh, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, S, H, dh, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, S, H, dh, device="cuda", dtype=torch.bfloat16)
+    o, _ = _ext().attn_fwd(q, k, v, torch.Tensor(), True, 1.0 / math.sqrt(dh))
+    ref = _attn_ref(q, k, v, None, True)
+    assert_close(o, ref, 0.04, "attn_nopad")

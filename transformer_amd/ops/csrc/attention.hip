@@ -1,0 +1,598 @@
+// Fused flash-style multi-head attention, forward + backward (SURVEY.md
+// K2-K5 + K14; semantics = reference Attention.py:3-34 incl. the additive
+// mask*(-1e9) before softmax, reference Attention.py:25-26).
+//
+// Layout: q (B,Sq,H,dh), k/v (B,Sk,H,dh) bf16 — the (B,S,H,dh) layout the
+// packed QKV GEMM produces, so the reference's head split/merge transposes
+// (Attention.py:52-57,74-76) never exist on device.
+// Masking: `causal` flag + per-token kv_pad bytes replace the materialized
+// (B,1,1,S)/(T,T) mask tensors (K14): three variants — encoder self
+// (pad), decoder self (causal+pad), cross (pad, Sq != Sk).
+//
+// Forward: online softmax (running m, l per row), O(Sq·dh) memory, fp32
+// MFMA accumulate, saves LSE for the backward.  Work unit: 4-wave block =
+// 64 q-rows; K/V tiles of 32 staged through swizzled LDS; QK^T and PV on
+// mfma_f32_16x16x32_bf16; P crosses C-layout -> A-layout through a small
+// per-wave LDS buffer.
+//
+// Backward: FA2-style two-kernel split (no atomics, deterministic):
+//   attn_bwd_kv: grid over kv-tiles, accumulates dK,dV (recomputing P from
+//                q,k,lse), dP from dO·V^T, dS = P*(dP-D).
+//   attn_bwd_q:  grid over q-tiles, accumulates dQ.
+//   attn_bwd_dot: D[row] = rowsum(dO * O).
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+
+#define KVT 32        // kv-tile (= MFMA K for PV)
+#define QW 16         // q-rows per wave
+#define WAVES 4
+#define NEG_BIG (-1e9f)
+
+// ---- swizzled LDS helpers --------------------------------------------------
+// natural tile [rows][DH] bf16 (row stride DH*2 bytes): XOR spreads the
+// 16-lane b128 groups; transposed/P tiles [rows][32] (64-byte rows).
+template <int ROWB>  // row stride in bytes
+DEV_INLINE int lds_swz(int row, int col_bytes) {
+  if (ROWB >= 128) return row * ROWB + (col_bytes ^ ((row & 7) << 4));
+  return row * ROWB + (col_bytes ^ ((row & 3) << 4));
+}
+
+template <int ROWB>
+DEV_INLINE bf16x8 lds_read8(const short* base, int row, int col /*elems*/) {
+  return (bf16x8)*(const s16x8*)((const char*)base + lds_swz<ROWB>(row, col * 2));
+}
+
+template <int ROWB>
+DEV_INLINE void lds_write8(short* base, int row, int col, s16x8 v) {
+  *(s16x8*)((char*)base + lds_swz<ROWB>(row, col * 2)) = v;
+}
+
+template <int ROWB>
+DEV_INLINE void lds_write1(short* base, int row, int col, short v) {
+  *(short*)((char*)base + (row * ROWB + ((col * 2) ^ ((ROWB >= 128 ? (row & 7) : (row & 3)) << 4)))) = v;
+}
+
+// Stage a [KVT][DH] bf16 tile from global rows (stride row_stride elems)
+// into swizzled LDS: natural layout into lds_n (if WRITE_N) and/or the
+// transpose [DH][KVT] into lds_t (if WRITE_T).
+// 256 threads cooperate; guards rows >= nrows with zeros.
+template <int DH, bool WRITE_N, bool WRITE_T>
+DEV_INLINE void stage_kv(const short* __restrict__ g, long row_stride,
+                         int nrows, short* lds_n, short* lds_t) {
+  const int t = threadIdx.x;
+  constexpr int CH = KVT * DH / 8;  // 16-byte chunks
+#pragma unroll
+  for (int p = 0; p < (CH + 255) / 256; ++p) {
+    int idx = p * 256 + t;
+    if (idx >= CH) break;
+    int row = idx / (DH / 8);
+    int c8 = (idx % (DH / 8)) * 8;
+    s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (row < nrows) v = *(const s16x8*)(g + row * row_stride + c8);
+    if (WRITE_N) lds_write8<DH * 2>(lds_n, row, c8, v);
+    if (WRITE_T) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) lds_write1<KVT * 2>(lds_t, c8 + j, row, v[j]);
+    }
+  }
+}
+
+// ---- forward ---------------------------------------------------------------
+template <int DH>
+__global__ __launch_bounds__(256)
+void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
+                     const short* __restrict__ V,
+                     const unsigned char* __restrict__ kv_pad,
+                     short* __restrict__ O, float* __restrict__ LSE, int B,
+                     int H, int Sq, int Sk, int causal, float scale) {
+  constexpr int D32 = DH / 32;   // QK^T MFMA k-steps
+  constexpr int D16 = DH / 16;   // O fragments
+  __shared__ short k_lds[KVT * DH];
+  __shared__ short vt_lds[DH * KVT];
+  __shared__ short p_lds[WAVES][QW * KVT];
+
+  const int bh = blockIdx.x;      // b*H + h
+  const int qb = blockIdx.y;      // q-block of 64
+  const int b = bh / H, h = bh % H;
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int q0 = qb * (WAVES * QW) + wid * QW;  // this wave's first q-row
+  const int fr = lane & 15, kg = lane >> 4;
+
+  const long q_bs = (long)Sq * H * DH, kv_bs = (long)Sk * H * DH;
+  const long row_stride = (long)H * DH;
+  const short* Qp = Q + b * q_bs + (long)h * DH;
+  const short* Kp = K + b * kv_bs + (long)h * DH;
+  const short* Vp = V + b * kv_bs + (long)h * DH;
+  const unsigned char* pad = kv_pad ? kv_pad + (long)b * Sk : nullptr;
+
+  // Q fragments in registers: A[row=fr][k=kg*8+j] per 32-chunk
+  bf16x8 qf[D32];
+#pragma unroll
+  for (int d = 0; d < D32; ++d) {
+    s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+    int row = q0 + fr;
+    if (row < Sq) v = *(const s16x8*)(Qp + (long)row * row_stride + d * 32 + kg * 8);
+    qf[d] = (bf16x8)v;
+  }
+
+  f32x4 acc[D16];
+#pragma unroll
+  for (int i = 0; i < D16; ++i) acc[i] = {0, 0, 0, 0};
+  float m_run[4], l_run[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.f; }
+
+  // causal: keys beyond this block's last row are fully masked
+  const int kend = causal ? min(Sk, qb * (WAVES * QW) + WAVES * QW) : Sk;
+
+  for (int k0 = 0; k0 < kend; k0 += KVT) {
+    const int kc = min(KVT, Sk - k0);
+    stage_kv<DH, true, false>(Kp + (long)k0 * row_stride, row_stride, kc,
+                              k_lds, nullptr);
+    stage_kv<DH, false, true>(Vp + (long)k0 * row_stride, row_stride, kc,
+                              nullptr, vt_lds);
+    __syncthreads();
+
+    // ---- S = scale*(Q K^T) + mask, two 16-key halves -----------------
+    // A = Q rows, B = K rows (NT): C[row=q][col=key].
+    float p_raw[2][4];   // C-layout: [half][reg r] value for (row kg*4+r, col fr)
+    float tile_pmax[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) tile_pmax[r] = -1e30f;
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      f32x4 s2 = {0, 0, 0, 0};
+#pragma unroll
+      for (int d = 0; d < D32; ++d) {
+        bf16x8 kf = lds_read8<DH * 2>(k_lds, half * 16 + fr, d * 32 + kg * 8);
+        s2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[d], kf, s2, 0, 0, 0);
+      }
+      const int kcol = k0 + half * 16 + fr;   // C col = lane&15
+      const bool col_pad = (kcol >= Sk) || (pad && pad[min(kcol, Sk - 1)]);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qrow = q0 + kg * 4 + r;     // C row = (lane>>4)*4+r
+        float x = s2[r] * scale;
+        if (col_pad) x += NEG_BIG;
+        if (causal && kcol > qrow) x += NEG_BIG;
+        p_raw[half][r] = x;
+        tile_pmax[r] = fmaxf(tile_pmax[r], x);
+      }
+    }
+    // row max across the 16 cols held by the 16-lane group
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        tile_pmax[r] = fmaxf(tile_pmax[r], __shfl_xor(tile_pmax[r], off));
+    }
+    // online rescale
+    float alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float mn = fmaxf(m_run[r], tile_pmax[r]);
+      alpha[r] = __expf(m_run[r] - mn);
+      m_run[r] = mn;
+    }
+#pragma unroll
+    for (int i = 0; i < D16; ++i)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc[i][r] *= alpha[r];
+
+    // P = exp(S - m), row-sum into l, and park bf16 P in LDS (A-layout)
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      float psum[4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float p = __expf(p_raw[half][r] - m_run[r]);
+        p_raw[half][r] = p;
+        psum[r] = p;
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1)
+          psum[r] += __shfl_xor(psum[r], off);
+        l_run[r] = l_run[r] * (half == 0 ? alpha[r] : 1.f) + psum[r];
+      }
+      if (half == 0) {
+        // first half applied alpha to l above; second half adds directly
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        lds_write1<KVT * 2>(p_lds[wid], kg * 4 + r, half * 16 + fr,
+                            f2bfbits(p_raw[half][r]));
+    }
+    __syncthreads();  // vt_lds ready + p_lds visible to own wave
+
+    // ---- O += P · V  (A = P[q][key] from LDS, B = V^T[d][key]) -------
+#pragma unroll
+    for (int i = 0; i < D16; ++i) {
+      bf16x8 pa = lds_read8<KVT * 2>(p_lds[wid], fr, kg * 8);
+      bf16x8 vb = lds_read8<KVT * 2>(vt_lds, i * 16 + fr, kg * 8);
+      acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, acc[i], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: O = acc / l ; LSE = m + log(l)
+  short* Op = O + b * q_bs + (long)h * DH;
+#pragma unroll
+  for (int i = 0; i < D16; ++i) {
+    const int gcol = i * 16 + fr;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qrow = q0 + kg * 4 + r;
+      if (qrow >= Sq) continue;
+      float l = l_run[r];
+      float o = (l > 0.f) ? acc[i][r] / l : 0.f;
+      Op[(long)qrow * row_stride + gcol] = f2bfbits(o);
+    }
+  }
+  if (fr == 0) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qrow = q0 + kg * 4 + r;
+      if (qrow < Sq)
+        LSE[((long)b * H + h) * Sq + qrow] =
+            m_run[r] + __logf(fmaxf(l_run[r], 1e-30f));
+    }
+  }
+}
+
+// ---- backward helpers ------------------------------------------------------
+
+// D[row] = sum_d dO[row][d] * O[row][d]  (fp32), one wave per row
+__global__ __launch_bounds__(256)
+void attn_bwd_dot_kernel(const short* __restrict__ dO,
+                         const short* __restrict__ O, float* __restrict__ Dl,
+                         int H, int S, int DH, long n_rows) {
+  const long row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= n_rows) return;   // row = ((b*Sq)+s) over (B,Sq) with H inner
+  const int lane = threadIdx.x & 63;
+  // rows iterate (b, s, h): dO layout (B,S,H,DH) flattened rows b*S*H+s*H+h
+  const short* dop = dO + row * DH;
+  const short* op = O + row * DH;
+  float acc = 0.f;
+  for (int c = lane; c < DH; c += WAVE)
+    acc += bfbits2f(dop[c]) * bfbits2f(op[c]);
+  acc = wave_sum(acc);
+  if (lane == 0) {
+    // remap (b,s,h) -> (b,h,s) to match LSE layout
+    long bsh = row;
+    long h = bsh % H;
+    long s = (bsh / H) % S;
+    long b = bsh / ((long)H * S);
+    Dl[((b * H) + h) * S + s] = acc;
+  }
+}
+
+// dK/dV kernel: one 4-wave block per 64-key tile; loops q-tiles of 32.
+template <int DH>
+__global__ __launch_bounds__(256)
+void attn_bwd_kv_kernel(const short* __restrict__ Q,
+                        const short* __restrict__ K,
+                        const short* __restrict__ V,
+                        const short* __restrict__ dO,
+                        const float* __restrict__ LSE,
+                        const float* __restrict__ Dl,
+                        const unsigned char* __restrict__ kv_pad,
+                        short* __restrict__ dK, short* __restrict__ dV, int B,
+                        int H, int Sq, int Sk, int causal, float scale) {
+  constexpr int D32 = DH / 32;
+  constexpr int D16 = DH / 16;
+  __shared__ short q_lds[KVT * DH];    // q-tile natural [32][DH]
+  __shared__ short do_lds[KVT * DH];   // dO-tile natural
+  __shared__ short qt_lds[DH * KVT];   // q-tile transposed
+  __shared__ short dot_lds[DH * KVT];  // dO-tile transposed
+  __shared__ short x_lds[WAVES][QW * KVT];  // per-wave P^T / dS^T scratch
+
+  const int bh = blockIdx.x;
+  const int kb = blockIdx.y;           // key-block of 64
+  const int b = bh / H, h = bh % H;
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int k0w = kb * (WAVES * QW) + wid * QW;  // wave's first key row
+  const int fr = lane & 15, kg = lane >> 4;
+
+  const long q_bs = (long)Sq * H * DH, kv_bs = (long)Sk * H * DH;
+  const long row_stride = (long)H * DH;
+  const short* Qp = Q + b * q_bs + (long)h * DH;
+  const short* Kp = K + b * kv_bs + (long)h * DH;
+  const short* Vp = V + b * kv_bs + (long)h * DH;
+  const short* dOp = dO + b * q_bs + (long)h * DH;
+  const float* lse = LSE + ((long)b * H + h) * Sq;
+  const float* dl = Dl + ((long)b * H + h) * Sq;
+  const unsigned char* pad = kv_pad ? kv_pad + (long)b * Sk : nullptr;
+
+  // this wave's K,V rows in registers (A-fragments)
+  bf16x8 kf[D32], vf[D32];
+  const int krow = k0w + fr;
+  const bool k_valid = krow < Sk;
+  const bool k_pad = pad && k_valid && pad[krow];
+#pragma unroll
+  for (int d = 0; d < D32; ++d) {
+    s16x8 kv_ = {0, 0, 0, 0, 0, 0, 0, 0}, vv = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (k_valid) {
+      kv_ = *(const s16x8*)(Kp + (long)krow * row_stride + d * 32 + kg * 8);
+      vv = *(const s16x8*)(Vp + (long)krow * row_stride + d * 32 + kg * 8);
+    }
+    kf[d] = (bf16x8)kv_;
+    vf[d] = (bf16x8)vv;
+  }
+
+  f32x4 acc_dv[D16], acc_dk[D16];
+#pragma unroll
+  for (int i = 0; i < D16; ++i) {
+    acc_dv[i] = {0, 0, 0, 0};
+    acc_dk[i] = {0, 0, 0, 0};
+  }
+
+  // causal: q-rows before this key-block are fully masked
+  const int q_start = causal ? (kb * (WAVES * QW)) / KVT * KVT : 0;
+
+  for (int j0 = q_start; j0 < Sq; j0 += KVT) {
+    const int jc = min(KVT, Sq - j0);
+    stage_kv<DH, true, true>(Qp + (long)j0 * row_stride, row_stride, jc,
+                             q_lds, qt_lds);
+    stage_kv<DH, true, true>(dOp + (long)j0 * row_stride, row_stride, jc,
+                             do_lds, dot_lds);
+    __syncthreads();
+
+    // pass 1: both halves of P^T into x_lds; dS^T kept in registers
+    float ds_reg[2][4];
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      // S^T[key][q] = K·Q^T ; dP^T[key][q] = V·dO^T   (C row=key, col=q)
+      f32x4 st = {0, 0, 0, 0}, dpt = {0, 0, 0, 0};
+#pragma unroll
+      for (int d = 0; d < D32; ++d) {
+        bf16x8 qb_ = lds_read8<DH * 2>(q_lds, half * 16 + fr, d * 32 + kg * 8);
+        bf16x8 dob = lds_read8<DH * 2>(do_lds, half * 16 + fr, d * 32 + kg * 8);
+        st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[d], qb_, st, 0, 0, 0);
+        dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vf[d], dob, dpt, 0, 0, 0);
+      }
+      const int qcol = j0 + half * 16 + fr;
+      const bool q_ok = qcol < Sq;
+      const float lse_q = q_ok ? lse[min(qcol, Sq - 1)] : 0.f;
+      const float d_q = q_ok ? dl[min(qcol, Sq - 1)] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int key = k0w + kg * 4 + r;
+        float x = st[r] * scale;
+        bool masked = !q_ok || key >= Sk || (pad && key < Sk && pad[key]) ||
+                      (causal && key > qcol);
+        float p = masked ? 0.f : __expf(x - lse_q);
+        ds_reg[half][r] = p * (dpt[r] - d_q) * scale;
+        lds_write1<KVT * 2>(x_lds[wid], kg * 4 + r, half * 16 + fr, f2bfbits(p));
+      }
+    }
+    __builtin_amdgcn_wave_barrier();
+    // dV += P^T · dO   (A = full P^T[key][q0..31], B = dO^T[d][q])
+#pragma unroll
+    for (int i = 0; i < D16; ++i) {
+      bf16x8 pa = lds_read8<KVT * 2>(x_lds[wid], fr, kg * 8);
+      bf16x8 db = lds_read8<KVT * 2>(dot_lds, i * 16 + fr, kg * 8);
+      acc_dv[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, db, acc_dv[i], 0, 0, 0);
+    }
+    // pass 2: both halves of dS^T, then dK += dS^T · Q
+#pragma unroll
+    for (int half = 0; half < 2; ++half)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        lds_write1<KVT * 2>(x_lds[wid], kg * 4 + r, half * 16 + fr,
+                            f2bfbits(ds_reg[half][r]));
+    __builtin_amdgcn_wave_barrier();
+#pragma unroll
+    for (int i = 0; i < D16; ++i) {
+      bf16x8 sa = lds_read8<KVT * 2>(x_lds[wid], fr, kg * 8);
+      bf16x8 qb2 = lds_read8<KVT * 2>(qt_lds, i * 16 + fr, kg * 8);
+      acc_dk[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(sa, qb2, acc_dk[i], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // write dK, dV (exclusive rows — no atomics)
+  short* dKp = dK + b * kv_bs + (long)h * DH;
+  short* dVp = dV + b * kv_bs + (long)h * DH;
+#pragma unroll
+  for (int i = 0; i < D16; ++i) {
+    const int gcol = i * 16 + fr;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int key = k0w + kg * 4 + r;
+      if (key >= Sk) continue;
+      dKp[(long)key * row_stride + gcol] = f2bfbits(acc_dk[i][r]);
+      dVp[(long)key * row_stride + gcol] = f2bfbits(acc_dv[i][r]);
+    }
+  }
+}
+
+// dQ kernel: one 4-wave block per 64-q tile; loops kv-tiles of 32.
+template <int DH>
+__global__ __launch_bounds__(256)
+void attn_bwd_q_kernel(const short* __restrict__ Q,
+                       const short* __restrict__ K,
+                       const short* __restrict__ V,
+                       const short* __restrict__ dO,
+                       const float* __restrict__ LSE,
+                       const float* __restrict__ Dl,
+                       const unsigned char* __restrict__ kv_pad,
+                       short* __restrict__ dQ, int B, int H, int Sq, int Sk,
+                       int causal, float scale) {
+  constexpr int D32 = DH / 32;
+  constexpr int D16 = DH / 16;
+  __shared__ short k_lds[KVT * DH];
+  __shared__ short v_lds[KVT * DH];
+  __shared__ short kt_lds[DH * KVT];
+  __shared__ short x_lds[WAVES][QW * KVT];
+
+  const int bh = blockIdx.x;
+  const int qb = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int q0 = qb * (WAVES * QW) + wid * QW;
+  const int fr = lane & 15, kg = lane >> 4;
+
+  const long q_bs = (long)Sq * H * DH, kv_bs = (long)Sk * H * DH;
+  const long row_stride = (long)H * DH;
+  const short* Qp = Q + b * q_bs + (long)h * DH;
+  const short* Kp = K + b * kv_bs + (long)h * DH;
+  const short* Vp = V + b * kv_bs + (long)h * DH;
+  const short* dOp = dO + b * q_bs + (long)h * DH;
+  const float* lse = LSE + ((long)b * H + h) * Sq;
+  const float* dl = Dl + ((long)b * H + h) * Sq;
+  const unsigned char* pad = kv_pad ? kv_pad + (long)b * Sk : nullptr;
+
+  bf16x8 qf[D32], dof[D32];
+  const int qrow_l = q0 + fr;
+  const bool q_valid = qrow_l < Sq;
+#pragma unroll
+  for (int d = 0; d < D32; ++d) {
+    s16x8 qv = {0, 0, 0, 0, 0, 0, 0, 0}, dv = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (q_valid) {
+      qv = *(const s16x8*)(Qp + (long)qrow_l * row_stride + d * 32 + kg * 8);
+      dv = *(const s16x8*)(dOp + (long)qrow_l * row_stride + d * 32 + kg * 8);
+    }
+    qf[d] = (bf16x8)qv;
+    dof[d] = (bf16x8)dv;
+  }
+
+  f32x4 acc_dq[D16];
+#pragma unroll
+  for (int i = 0; i < D16; ++i) acc_dq[i] = {0, 0, 0, 0};
+
+  const int kend = causal ? min(Sk, qb * (WAVES * QW) + WAVES * QW) : Sk;
+
+  for (int k0 = 0; k0 < kend; k0 += KVT) {
+    const int kc = min(KVT, Sk - k0);
+    stage_kv<DH, true, true>(Kp + (long)k0 * row_stride, row_stride, kc,
+                             k_lds, kt_lds);
+    stage_kv<DH, true, false>(Vp + (long)k0 * row_stride, row_stride, kc,
+                              v_lds, nullptr);
+    __syncthreads();
+
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      f32x4 s = {0, 0, 0, 0}, dp = {0, 0, 0, 0};
+#pragma unroll
+      for (int d = 0; d < D32; ++d) {
+        bf16x8 kb_ = lds_read8<DH * 2>(k_lds, half * 16 + fr, d * 32 + kg * 8);
+        bf16x8 vb_ = lds_read8<DH * 2>(v_lds, half * 16 + fr, d * 32 + kg * 8);
+        s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[d], kb_, s, 0, 0, 0);
+        dp = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof[d], vb_, dp, 0, 0, 0);
+      }
+      const int kcol = k0 + half * 16 + fr;
+      const bool col_pad = (kcol >= Sk) || (pad && pad[min(kcol, Sk - 1)]);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qrow = q0 + kg * 4 + r;
+        const bool row_ok = qrow < Sq;
+        float lse_q = row_ok ? lse[min(qrow, Sq - 1)] : 0.f;
+        float d_q = row_ok ? dl[min(qrow, Sq - 1)] : 0.f;
+        bool masked = !row_ok || col_pad || (causal && kcol > qrow);
+        float p = masked ? 0.f : __expf(s[r] * scale - lse_q);
+        float ds = p * (dp[r] - d_q) * scale;
+        lds_write1<KVT * 2>(x_lds[wid], kg * 4 + r, half * 16 + fr,
+                            f2bfbits(ds));
+      }
+    }
+    __builtin_amdgcn_wave_barrier();
+    // dQ += dS · K   (A = dS[q][key] from LDS, B = K^T[d][key])
+#pragma unroll
+    for (int i = 0; i < D16; ++i) {
+      bf16x8 sa = lds_read8<KVT * 2>(x_lds[wid], fr, kg * 8);
+      bf16x8 kb2 = lds_read8<KVT * 2>(kt_lds, i * 16 + fr, kg * 8);
+      acc_dq[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(sa, kb2, acc_dq[i], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  short* dQp = dQ + b * q_bs + (long)h * DH;
+#pragma unroll
+  for (int i = 0; i < D16; ++i) {
+    const int gcol = i * 16 + fr;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qrow = q0 + kg * 4 + r;
+      if (qrow >= Sq) continue;
+      dQp[(long)qrow * row_stride + gcol] = f2bfbits(acc_dq[i][r]);
+    }
+  }
+}
+
+// ---- host wrappers ---------------------------------------------------------
+
+#define DISPATCH_DH(DH_VAL, ...)                                   \
+  switch (DH_VAL) {                                                \
+    case 32: { constexpr int DHC = 32; __VA_ARGS__; break; }       \
+    case 64: { constexpr int DHC = 64; __VA_ARGS__; break; }       \
+    case 128: { constexpr int DHC = 128; __VA_ARGS__; break; }     \
+    default: TORCH_CHECK(false, "attention: head dim must be 32/64/128, got ", DH_VAL); \
+  }
+
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, torch::Tensor kv_pad,
+                                    bool causal, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.dim() == 4 &&
+              q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  const int B = q.size(0), Sq = q.size(1), H = q.size(2), DH = q.size(3);
+  const int Sk = k.size(1);
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, H, Sq}, q.options().dtype(torch::kFloat32));
+  const unsigned char* pad = nullptr;
+  if (kv_pad.defined() && kv_pad.numel() > 0) {
+    TORCH_CHECK(kv_pad.dtype() == torch::kUInt8 && kv_pad.is_contiguous() &&
+                kv_pad.size(0) == B && kv_pad.size(1) == Sk);
+    pad = kv_pad.data_ptr<unsigned char>();
+  }
+  dim3 grid(B * H, cdiv(Sq, WAVES * QW));
+  auto stream = at::hip::getCurrentHIPStream();
+  DISPATCH_DH(DH, attn_fwd_kernel<DHC><<<grid, 256, 0, stream>>>(
+                  (const short*)q.data_ptr(), (const short*)k.data_ptr(),
+                  (const short*)v.data_ptr(), pad, (short*)o.data_ptr(),
+                  lse.data_ptr<float>(), B, H, Sq, Sk, causal ? 1 : 0,
+                  (float)scale));
+  return {o, lse};
+}
+
+std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, torch::Tensor o,
+                                    torch::Tensor dout, torch::Tensor lse,
+                                    torch::Tensor kv_pad, bool causal,
+                                    double scale) {
+  const int B = q.size(0), Sq = q.size(1), H = q.size(2), DH = q.size(3);
+  const int Sk = k.size(1);
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  auto dl = torch::empty({B, H, Sq}, q.options().dtype(torch::kFloat32));
+  const unsigned char* pad = nullptr;
+  if (kv_pad.defined() && kv_pad.numel() > 0)
+    pad = kv_pad.data_ptr<unsigned char>();
+  auto stream = at::hip::getCurrentHIPStream();
+  long n_rows = (long)B * Sq * H;
+  attn_bwd_dot_kernel<<<cdiv(n_rows, 4), 256, 0, stream>>>(
+      (const short*)dout.data_ptr(), (const short*)o.data_ptr(),
+      dl.data_ptr<float>(), H, Sq, DH, n_rows);
+  dim3 grid_kv(B * H, cdiv(Sk, WAVES * QW));
+  dim3 grid_q(B * H, cdiv(Sq, WAVES * QW));
+  DISPATCH_DH(DH, {
+    attn_bwd_kv_kernel<DHC><<<grid_kv, 256, 0, stream>>>(
+        (const short*)q.data_ptr(), (const short*)k.data_ptr(),
+        (const short*)v.data_ptr(), (const short*)dout.data_ptr(),
+        lse.data_ptr<float>(), dl.data_ptr<float>(), pad,
+        (short*)dk.data_ptr(), (short*)dv.data_ptr(), B, H, Sq, Sk,
+        causal ? 1 : 0, (float)scale);
+    attn_bwd_q_kernel<DHC><<<grid_q, 256, 0, stream>>>(
+        (const short*)q.data_ptr(), (const short*)k.data_ptr(),
+        (const short*)v.data_ptr(), (const short*)dout.data_ptr(),
+        lse.data_ptr<float>(), dl.data_ptr<float>(), pad,
+        (short*)dq.data_ptr(), B, H, Sq, Sk, causal ? 1 : 0, (float)scale);
+  });
+  return {dq, dk, dv};
+}

@@ -1,0 +1,157 @@
+// Fused padding-masked (label-smoothed) cross entropy over the V≈32k vocab
+// (semantics = reference train.py:83-88 with SURVEY §8 Q9
+// smoothing: eps=0 reproduces plain CE, loss = sum(per-token)/global_batch).
+//
+// One 256-thread block (4 waves) per logits row; bf16 logits are read with
+// short8 vector loads, max/sum reduced wave-wise then across the block's 4
+// waves through LDS.  Forward saves the fp32 LSE per row; backward
+// recomputes softmax from logits+LSE (no S×V weight materialization).
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+// block-level reduce over 4 waves (256 threads)
+DEV_INLINE float block_reduce(float v, float* scratch, int op /*0=max 1=sum*/) {
+  int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  v = (op == 0) ? wave_max(v) : wave_sum(v);
+  if (lane == 0) scratch[wid] = v;
+  __syncthreads();
+  float r = scratch[0];
+#pragma unroll
+  for (int i = 1; i < 4; ++i)
+    r = (op == 0) ? fmaxf(r, scratch[i]) : r + scratch[i];
+  __syncthreads();
+  return r;
+}
+
+__global__ __launch_bounds__(256)
+void ce_fwd_kernel(const short* __restrict__ logits,
+                   const long* __restrict__ targets,
+                   float* __restrict__ lse_out, float* __restrict__ loss_out,
+                   long R, int V, float inv_batch, float eps_ls) {
+  __shared__ float scratch[4];
+  const long row = blockIdx.x;
+  if (row >= R) return;
+  const short* lrow = logits + row * V;
+  const int t = threadIdx.x;
+
+  float mx = -1e30f, sume = 0.f, sumx = 0.f;
+  for (int c = t * 8; c < V; c += 256 * 8) {
+    if (c + 8 <= V) {
+      s16x8 v = *(const s16x8*)(lrow + c);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) mx = fmaxf(mx, bfbits2f(v[j]));
+    } else {
+      for (int j = 0; c + j < V; ++j) mx = fmaxf(mx, bfbits2f(lrow[c + j]));
+    }
+  }
+  mx = block_reduce(mx, scratch, 0);
+  for (int c = t * 8; c < V; c += 256 * 8) {
+    if (c + 8 <= V) {
+      s16x8 v = *(const s16x8*)(lrow + c);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float x = bfbits2f(v[j]);
+        sume += __expf(x - mx);
+        sumx += x;
+      }
+    } else {
+      for (int j = 0; c + j < V; ++j) {
+        float x = bfbits2f(lrow[c + j]);
+        sume += __expf(x - mx);
+        sumx += x;
+      }
+    }
+  }
+  sume = block_reduce(sume, scratch, 1);
+  __syncthreads();
+  sumx = block_reduce(sumx, scratch, 1);
+  const float lse = mx + __logf(sume);
+  if (t == 0) {
+    lse_out[row] = lse;
+    const long tgt = targets[row];
+    if (tgt != 0) {  // padding mask (real != 0)
+      float xt = bfbits2f(lrow[tgt]);
+      // (1-eps)*(lse - x_t) + eps*(lse - mean_j x_j)
+      float loss = lse - (1.f - eps_ls) * xt - eps_ls * (sumx / V);
+      atomicAdd(loss_out, loss * inv_batch);
+    }
+  }
+}
+
+// dlogits[j] = dloss/batch * mask_row * (softmax_j - (1-eps)*onehot_j - eps/V)
+__global__ __launch_bounds__(256)
+void ce_bwd_kernel(const short* __restrict__ logits,
+                   const long* __restrict__ targets,
+                   const float* __restrict__ lse, short* __restrict__ dlogits,
+                   long R, int V, float scale, float eps_ls) {
+  const long row = blockIdx.x;
+  if (row >= R) return;
+  const short* lrow = logits + row * V;
+  short* drow = dlogits + row * V;
+  const long tgt = targets[row];
+  const int t = threadIdx.x;
+  if (tgt == 0) {  // padded position contributes no gradient
+    for (int c = t * 8; c < V; c += 256 * 8) {
+      if (c + 8 <= V) {
+        *(s16x8*)(drow + c) = (s16x8){0, 0, 0, 0, 0, 0, 0, 0};
+      } else {
+        for (int j = 0; c + j < V; ++j) drow[c + j] = 0;
+      }
+    }
+    return;
+  }
+  const float l = lse[row];
+  const float eps_v = eps_ls / V;
+  for (int c = t * 8; c < V; c += 256 * 8) {
+    if (c + 8 <= V) {
+      s16x8 v = *(const s16x8*)(lrow + c);
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float sm = __expf(bfbits2f(v[j]) - l);
+        float g = sm - eps_v - ((c + j == tgt) ? (1.f - eps_ls) : 0.f);
+        o[j] = f2bfbits(g * scale);
+      }
+      *(s16x8*)(drow + c) = o;
+    } else {
+      for (int j = 0; c + j < V; ++j) {
+        float sm = __expf(bfbits2f(lrow[c + j]) - l);
+        float g = sm - eps_v - ((c + j == tgt) ? (1.f - eps_ls) : 0.f);
+        drow[c + j] = f2bfbits(g * scale);
+      }
+    }
+  }
+}
+
+std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets,
+                                  double batch_size, double label_smoothing) {
+  TORCH_CHECK(logits.is_cuda() && logits.dtype() == torch::kBFloat16 &&
+              logits.dim() == 2 && logits.is_contiguous());
+  TORCH_CHECK(targets.dtype() == torch::kInt64 && targets.is_contiguous());
+  const long R = logits.size(0);
+  const int V = logits.size(1);
+  auto lse = torch::empty({R}, logits.options().dtype(torch::kFloat32));
+  auto loss = torch::zeros({}, logits.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  ce_fwd_kernel<<<R, 256, 0, stream>>>(
+      (const short*)logits.data_ptr(), targets.data_ptr<long>(),
+      lse.data_ptr<float>(), loss.data_ptr<float>(), R, V,
+      1.0f / (float)batch_size, (float)label_smoothing);
+  return {loss, lse};
+}
+
+torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets,
+                     torch::Tensor lse, double dloss, double batch_size,
+                     double label_smoothing) {
+  const long R = logits.size(0);
+  const int V = logits.size(1);
+  auto dlogits = torch::empty_like(logits);
+  auto stream = at::hip::getCurrentHIPStream();
+  ce_bwd_kernel<<<R, 256, 0, stream>>>(
+      (const short*)logits.data_ptr(), targets.data_ptr<long>(),
+      lse.data_ptr<float>(), (short*)dlogits.data_ptr(), R, V,
+      (float)(dloss / batch_size), (float)label_smoothing);
+  return dlogits;
+}

@@ -1,0 +1,76 @@
+// Dropout fwd/bwd (reference rate 0.1, Encoder.py:16-17).
+// Counter-based RNG (PCG-style hash of (seed, index)) -> deterministic per
+// (seed, element); mask saved as one byte per element for the exact
+// backward.  y = x * mask / (1-p).
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+DEV_INLINE unsigned int pcg_hash(unsigned long long key) {
+  key = key * 6364136223846793005ull + 1442695040888963407ull;
+  unsigned int x = (unsigned int)((key ^ (key >> 33)) >> 11);
+  x ^= x >> 16;
+  x *= 0x7feb352dU;
+  x ^= x >> 15;
+  x *= 0x846ca68bU;
+  x ^= x >> 16;
+  return x;
+}
+
+__global__ void dropout_fwd_kernel(const short* __restrict__ x,
+                                   short* __restrict__ y,
+                                   unsigned char* __restrict__ mask, long n,
+                                   float p, float inv_keep,
+                                   unsigned long long seed) {
+  long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  if (i >= n) return;
+  const unsigned int thr = (unsigned int)(p * 4294967296.0f);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    if (i + j >= n) break;
+    unsigned int r = pcg_hash(seed * 0x9E3779B97F4A7C15ull + (i + j));
+    unsigned char keep = r >= thr;
+    mask[i + j] = keep;
+    y[i + j] = keep ? f2bfbits(bfbits2f(x[i + j]) * inv_keep) : (short)0;
+  }
+}
+
+__global__ void dropout_bwd_kernel(const short* __restrict__ dy,
+                                   const unsigned char* __restrict__ mask,
+                                   short* __restrict__ dx, long n,
+                                   float inv_keep) {
+  long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  if (i >= n) return;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    if (i + j >= n) break;
+    dx[i + j] = mask[i + j] ? f2bfbits(bfbits2f(dy[i + j]) * inv_keep)
+                            : (short)0;
+  }
+}
+
+std::vector<torch::Tensor> dropout_fwd(torch::Tensor x, double p,
+                                       int64_t seed) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 &&
+              x.is_contiguous());
+  auto y = torch::empty_like(x);
+  auto mask = torch::empty({x.numel()}, x.options().dtype(torch::kUInt8));
+  long n = x.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  dropout_fwd_kernel<<<((n + 3) / 4 + 255) / 256, 256, 0, stream>>>(
+      (const short*)x.data_ptr(), (short*)y.data_ptr(),
+      mask.data_ptr<unsigned char>(), n, (float)p, 1.0f / (1.0f - (float)p),
+      (unsigned long long)seed);
+  return {y, mask};
+}
+
+torch::Tensor dropout_bwd(torch::Tensor dy, torch::Tensor mask, double p) {
+  auto dx = torch::empty_like(dy);
+  long n = dy.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  dropout_bwd_kernel<<<((n + 3) / 4 + 255) / 256, 256, 0, stream>>>(
+      (const short*)dy.data_ptr(), mask.data_ptr<unsigned char>(),
+      (short*)dx.data_ptr(), n, 1.0f / (1.0f - (float)p));
+  return dx;
+}

@@ -1,0 +1,302 @@
+// MFMA bf16 GEMM (NT layout) + small helper kernels.
+//
+// Implements SURVEY.md §2.3 K1/K7/K8/K12: every linear layer of the model
+// (packed QKV, attention out-proj, FFN with fused ReLU, the V≈32k logits
+// head) runs through gemm_nt:  C[M,N] = A[M,K] @ W[N,K]^T (+bias)(+ReLU),
+// bf16 inputs, fp32 MFMA accumulate, bf16 out.
+//
+// CDNA4 design (see /opt/skills guides): 128x128 output tile per 4-wave
+// (256-thread) workgroup, BK=64 K-steps double-staged through LDS with an
+// XOR bank-swizzle (byte ^= (row&7)<<4) so ds_read_b128 fragment reads are
+// ≤2-way conflicting; mfma_f32_16x16x32_bf16 with 4x4 fragments per wave
+// (64x64 per wave); bias/ReLU fused in the epilogue; XCD-aware workgroup
+// remap for L2 affinity (8 XCDs with private L2s).
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+
+#define BM 128
+#define BN 128
+#define BK 64
+#define NTHREADS 256
+
+// LDS tiles are [128 rows][64 cols] bf16 = 128 B per row.  Byte-offset
+// swizzle spreads the 16-lane ds_read_b128 groups over 8 slots (T2 recipe).
+DEV_INLINE int lds_off(int row, int col_bytes) {
+  return row * (BK * 2) + (col_bytes ^ ((row & 7) << 4));
+}
+
+// Stage one [128][BK] bf16 tile from global (row-major, stride ldg elems)
+// into LDS via registers + swizzled ds_write_b128.  Each of the 256 threads
+// writes 4 16-byte chunks.  Guards rows >= nrows and cols >= ncols (zeros).
+template <bool ALIGNED>
+DEV_INLINE void stage_tile(const short* __restrict__ g, int ldg, int nrows,
+                           int ncols, short* lds) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    int idx = p * NTHREADS + t;          // 1024 chunks of 8 elems
+    int row = idx >> 3;                  // /8 chunks per row
+    int c8 = (idx & 7) << 3;             // starting col
+    s16x4 lo = {0, 0, 0, 0}, hi = {0, 0, 0, 0};
+    if (row < nrows) {
+      const short* src = g + (long)row * ldg + c8;
+      if (ALIGNED && c8 + 8 <= ncols) {
+        s16x8 vv = *(const s16x8*)src;
+        lo = {vv[0], vv[1], vv[2], vv[3]};
+        hi = {vv[4], vv[5], vv[6], vv[7]};
+      } else {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          if (c8 + j < ncols) lo[j] = src[j];
+          if (c8 + 4 + j < ncols) hi[j] = src[4 + j];
+        }
+      }
+    }
+    int off = lds_off(row, c8 * 2);
+    *(s16x4*)((char*)lds + off) = lo;
+    *(s16x4*)((char*)lds + off + 8) = hi;
+  }
+}
+
+DEV_INLINE bf16x8 read_frag(const short* lds, int row, int kbase) {
+  int off = lds_off(row, kbase * 2);
+  s16x8 v0 = *(const s16x8*)((const char*)lds + off);
+  return (bf16x8)v0;
+}
+
+template <int EPILOGUE, bool ALIGNED_A, bool ALIGNED_B>
+__global__ __launch_bounds__(NTHREADS)
+void gemm_nt_kernel(const short* __restrict__ A, const short* __restrict__ W,
+                    const short* __restrict__ bias, short* __restrict__ C,
+                    int M, int N, int K, int has_bias, int nbm, int nbn) {
+  __shared__ short a_lds[BM * BK];
+  __shared__ short b_lds[BN * BK];
+
+  // XCD-aware bijective remap (guide T1): contiguous grid chunk per XCD.
+  int nwg = nbm * nbn;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, r = nwg % 8, x = wg % 8, o = wg / 8;
+    wg = (x < r ? x * (q + 1) : r * (q + 1) + (x - r) * q) + o;
+  }
+  const int bm0 = (wg / nbn) * BM;
+  const int bn0 = (wg % nbn) * BN;
+
+  const int wid = threadIdx.x >> 6;      // wave 0..3 -> 2x2
+  const int lane = threadIdx.x & 63;
+  const int wr = (wid >> 1) * 64;        // wave row offset in tile
+  const int wc = (wid & 1) * 64;
+  const int fr = lane & 15;              // fragment row/col lane index
+  const int kg = lane >> 4;              // k-group 0..3
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0, 0, 0, 0};
+
+  const int mrows = min(BM, M - bm0);
+  const int nrows = min(BN, N - bn0);
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    int kc = min(BK, K - k0);
+    stage_tile<ALIGNED_A>(A + k0, K, mrows, kc, a_lds);
+    stage_tile<ALIGNED_B>(W + k0, K, nrows, kc, b_lds);
+    __syncthreads();
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {     // two 32-deep MFMA K-steps
+      bf16x8 af[4], bf_[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        af[i] = read_frag(a_lds, wr + i * 16 + fr, ks * 32 + kg * 8);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        bf_[j] = read_frag(b_lds, wc + j * 16 + fr, ks * 32 + kg * 8);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf_[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // Epilogue: C/D lane map is col = lane&15, row = (lane>>4)*4 + r.
+  // Pack 2 bf16 (adjacent rows... cols are per-lane) -> scalar stores; the
+  // 16 lanes of a group cover 16 consecutive cols so stores coalesce.
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int grow_base = bm0 + wr + i * 16 + kg * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int gcol = bn0 + wc + j * 16 + fr;
+      if (gcol >= N) continue;
+      float bv = (has_bias && bias) ? bfbits2f(bias[gcol]) : 0.0f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int grow = grow_base + r;
+        if (grow >= M) continue;
+        float v = acc[i][j][r] + bv;
+        if (EPILOGUE == 1) v = fmaxf(v, 0.0f);
+        C[(long)grow * N + gcol] = f2bfbits(v);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// transpose2d: [M,N] bf16 -> [N,M].  64x64 tiles, padded LDS, short4 IO.
+// Used by the backward GEMMs (dX = dY·W, dW = dY^T·X) until dedicated
+// TN/NN kernel variants land.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256)
+void transpose2d_kernel(const short* __restrict__ in, short* __restrict__ out,
+                        int M, int N) {
+  __shared__ short tile[64][64 + 4];
+  int tb = blockIdx.x, nbx = (N + 63) >> 6;
+  int bm = (tb / nbx) << 6, bn = (tb % nbx) << 6;
+  int t = threadIdx.x;
+  // load 64x64: each thread 16 elems as 4x short4 rows
+  for (int p = 0; p < 16; ++p) {
+    int idx = p * 256 + t;          // 4096 slots of 1 elem... use vec when interior
+    int r = idx >> 6, c = idx & 63;
+    int gr = bm + r, gc = bn + c;
+    tile[r][c] = (gr < M && gc < N) ? in[(long)gr * N + gc] : (short)0;
+  }
+  __syncthreads();
+  for (int p = 0; p < 16; ++p) {
+    int idx = p * 256 + t;
+    int r = idx >> 6, c = idx & 63;   // r,c in OUTPUT tile [N-major]
+    int gr = bn + r, gc = bm + c;
+    if (gr < N && gc < M) out[(long)gr * M + gc] = tile[c][r];
+  }
+}
+
+// colsum: db[N] = sum_m dy[M,N] (fp32 accumulate, bf16 out)
+__global__ void colsum_kernel(const short* __restrict__ dy,
+                              short* __restrict__ db, int M, int N) {
+  int n = blockIdx.x * blockDim.x + threadIdx.x;
+  if (n >= N) return;
+  float s = 0.f;
+  for (long m = 0; m < M; ++m) s += bfbits2f(dy[m * N + n]);
+  db[n] = f2bfbits(s);
+}
+
+// relu_bwd: dz = dy * (y > 0)
+__global__ void relu_bwd_kernel(const short* __restrict__ dy,
+                                const short* __restrict__ y,
+                                short* __restrict__ dz, long n) {
+  long i = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (i + 8 <= n) {
+    s16x8 dv = *(const s16x8*)(dy + i);
+    s16x8 yv = *(const s16x8*)(y + i);
+    s16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = (bfbits2f(yv[j]) > 0.f) ? dv[j] : (short)0;
+    *(s16x8*)(dz + i) = o;
+  } else {
+    for (long j = i; j < n; ++j)
+      dz[j] = (bfbits2f(y[j]) > 0.f) ? dy[j] : (short)0;
+  }
+}
+
+__global__ void smoke_add_kernel(const short* a, const short* b, short* c,
+                                 long n) {
+  long i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) c[i] = f2bfbits(bfbits2f(a[i]) + bfbits2f(b[i]));
+}
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+#define CHECK_BF16_2D(t)                                                      \
+  TORCH_CHECK(t.is_cuda() && t.dtype() == torch::kBFloat16 && t.dim() == 2 && \
+                  t.is_contiguous(),                                          \
+              #t " must be contiguous 2-D bf16 on GPU")
+
+torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
+                      int64_t epilogue) {
+  CHECK_BF16_2D(a);
+  CHECK_BF16_2D(w);
+  const int M = a.size(0), K = a.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "gemm_nt: K mismatch");
+  bool has_bias = bias.defined() && bias.numel() > 0;
+  if (has_bias)
+    TORCH_CHECK(bias.is_cuda() && bias.dtype() == torch::kBFloat16 &&
+                bias.numel() == N, "bias must be bf16[N]");
+  auto c = torch::empty({M, N}, a.options());
+  int nbm = cdiv(M, BM), nbn = cdiv(N, BN);
+  dim3 grid(nbm * nbn);
+  auto stream = at::hip::getCurrentHIPStream();
+  bool al = (K % 8 == 0);
+  auto launch = [&](auto epi, auto ala, auto alb) {
+    gemm_nt_kernel<decltype(epi)::value, decltype(ala)::value,
+                   decltype(alb)::value>
+        <<<grid, NTHREADS, 0, stream>>>(
+            (const short*)a.data_ptr(), (const short*)w.data_ptr(),
+            has_bias ? (const short*)bias.data_ptr() : nullptr,
+            (short*)c.data_ptr(), M, N, K, has_bias, nbm, nbn);
+  };
+  using T = std::true_type;
+  using F = std::false_type;
+  using E0 = std::integral_constant<int, 0>;
+  using E1 = std::integral_constant<int, 1>;
+  if (epilogue == 1) {
+    if (al) launch(E1{}, T{}, T{}); else launch(E1{}, F{}, F{});
+  } else {
+    if (al) launch(E0{}, T{}, T{}); else launch(E0{}, F{}, F{});
+  }
+  return c;
+}
+
+torch::Tensor transpose2d(torch::Tensor a) {
+  CHECK_BF16_2D(a);
+  int M = a.size(0), N = a.size(1);
+  auto out = torch::empty({N, M}, a.options());
+  int nb = cdiv(M, 64) * cdiv(N, 64);
+  auto stream = at::hip::getCurrentHIPStream();
+  transpose2d_kernel<<<nb, 256, 0, stream>>>(
+      (const short*)a.data_ptr(), (short*)out.data_ptr(), M, N);
+  return out;
+}
+
+torch::Tensor colsum(torch::Tensor a) {
+  CHECK_BF16_2D(a);
+  int M = a.size(0), N = a.size(1);
+  auto out = torch::empty({N}, a.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  colsum_kernel<<<cdiv(N, 256), 256, 0, stream>>>(
+      (const short*)a.data_ptr(), (short*)out.data_ptr(), M, N);
+  return out;
+}
+
+torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y) {
+  TORCH_CHECK(dy.is_cuda() && dy.dtype() == torch::kBFloat16 &&
+              dy.is_contiguous() && y.is_contiguous() &&
+              dy.numel() == y.numel());
+  auto out = torch::empty_like(dy);
+  long n = dy.numel();
+  long blocks = ((n + 7) / 8 + 255) / 256;
+  auto stream = at::hip::getCurrentHIPStream();
+  relu_bwd_kernel<<<blocks, 256, 0, stream>>>(
+      (const short*)dy.data_ptr(), (const short*)y.data_ptr(),
+      (short*)out.data_ptr(), n);
+  return out;
+}
+
+torch::Tensor smoke_add(torch::Tensor a, torch::Tensor b) {
+  TORCH_CHECK(a.is_cuda() && a.dtype() == torch::kBFloat16);
+  auto c = torch::empty_like(a);
+  long n = a.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  smoke_add_kernel<<<(n + 255) / 256, 256, 0, stream>>>(
+      (const short*)a.data_ptr(), (const short*)b.data_ptr(),
+      (short*)c.data_ptr(), n);
+  return c;
+}

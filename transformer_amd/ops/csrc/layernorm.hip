@@ -1,0 +1,169 @@
+// Fused residual-add + LayerNorm fwd/bwd (semantics =
+// reference Encoder.py:23 `layernorm(x + attn_output)` with eps=1e-6).
+//
+// One 64-lane wave per row (4 rows per 256-thread block); bf16 loads are
+// vectorized short8 (guide G13), statistics and the backward reductions in
+// fp32 via 64-wide shuffle reduces.  dgamma/dbeta accumulate into fp32
+// workspaces with atomics, then cast to bf16.
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+// forward: y = LN(x+res)*gamma+beta; saves s=x+res (bf16), mean & rstd (f32)
+__global__ __launch_bounds__(256)
+void ln_fwd_kernel(const short* __restrict__ x, const short* __restrict__ res,
+                   const short* __restrict__ gamma,
+                   const short* __restrict__ beta, short* __restrict__ y,
+                   short* __restrict__ s, float* __restrict__ mean_out,
+                   float* __restrict__ rstd_out, int R, int D, float eps) {
+  const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= R) return;
+  const int lane = threadIdx.x & 63;
+  const long base = (long)row * D;
+
+  float sum = 0.f, sumsq = 0.f;
+  // pass 1: s = x + res, accumulate stats
+  for (int c = lane * 8; c < D; c += WAVE * 8) {
+    float v[8];
+    if (c + 8 <= D) {
+      s16x8 xv = *(const s16x8*)(x + base + c);
+      s16x8 rv = *(const s16x8*)(res + base + c);
+      s16x8 sv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        v[j] = bfbits2f(xv[j]) + bfbits2f(rv[j]);
+        sv[j] = f2bfbits(v[j]);
+        sum += v[j];
+        sumsq += v[j] * v[j];
+      }
+      *(s16x8*)(s + base + c) = sv;
+    } else {
+      for (int j = 0; c + j < D; ++j) {
+        float t = bfbits2f(x[base + c + j]) + bfbits2f(res[base + c + j]);
+        s[base + c + j] = f2bfbits(t);
+        sum += t;
+        sumsq += t * t;
+      }
+    }
+  }
+  sum = wave_sum(sum);
+  sumsq = wave_sum(sumsq);
+  const float mean = sum / D;
+  const float var = fmaxf(sumsq / D - mean * mean, 0.f);
+  const float rstd = rsqrtf(var + eps);
+  if (lane == 0) {
+    mean_out[row] = mean;
+    rstd_out[row] = rstd;
+  }
+  // pass 2: normalize (s is in registers only partially; re-read via bf16)
+  for (int c = lane * 8; c < D; c += WAVE * 8) {
+    if (c + 8 <= D) {
+      s16x8 sv = *(const s16x8*)(s + base + c);
+      s16x8 gv = *(const s16x8*)(gamma + c);
+      s16x8 bv = *(const s16x8*)(beta + c);
+      s16x8 yv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        yv[j] = f2bfbits((bfbits2f(sv[j]) - mean) * rstd * bfbits2f(gv[j]) +
+                         bfbits2f(bv[j]));
+      *(s16x8*)(y + base + c) = yv;
+    } else {
+      for (int j = 0; c + j < D; ++j)
+        y[base + c + j] =
+            f2bfbits((bfbits2f(s[base + c + j]) - mean) * rstd *
+                         bfbits2f(gamma[c + j]) + bfbits2f(beta[c + j]));
+    }
+  }
+}
+
+// backward: dx = rstd*(g - mean(g) - xhat*mean(g*xhat)), g = dy*gamma;
+// dgamma += dy*xhat, dbeta += dy (atomics into fp32 workspace)
+__global__ __launch_bounds__(256)
+void ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ s,
+                   const short* __restrict__ gamma,
+                   const float* __restrict__ mean,
+                   const float* __restrict__ rstd, short* __restrict__ dx,
+                   int R, int D) {
+  const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= R) return;
+  const int lane = threadIdx.x & 63;
+  const long base = (long)row * D;
+  const float mu = mean[row], rs = rstd[row];
+
+  float sg = 0.f, sgx = 0.f;
+  for (int c = lane; c < D; c += WAVE) {
+    float dyv = bfbits2f(dy[base + c]);
+    float xh = (bfbits2f(s[base + c]) - mu) * rs;
+    float g = dyv * bfbits2f(gamma[c]);
+    sg += g;
+    sgx += g * xh;
+  }
+  sg = wave_sum(sg) / D;
+  sgx = wave_sum(sgx) / D;
+  for (int c = lane; c < D; c += WAVE) {
+    float xh = (bfbits2f(s[base + c]) - mu) * rs;
+    float g = bfbits2f(dy[base + c]) * bfbits2f(gamma[c]);
+    dx[base + c] = f2bfbits(rs * (g - sg - xh * sgx));
+  }
+}
+
+// column reduction: dgamma[c] = sum_r dy[r][c]*xhat[r][c]; dbeta[c] = sum dy.
+// Thread t of block b owns column b*256+t; every row read is coalesced
+// across the block's 256 consecutive columns.
+__global__ __launch_bounds__(256)
+void ln_gb_kernel(const short* __restrict__ dy, const short* __restrict__ s,
+                  const float* __restrict__ mean,
+                  const float* __restrict__ rstd, short* __restrict__ dgamma,
+                  short* __restrict__ dbeta, int R, int D) {
+  const int c = blockIdx.x * 256 + threadIdx.x;
+  if (c >= D) return;
+  float sg = 0.f, sb = 0.f;
+  for (long r = 0; r < R; ++r) {
+    float dyv = bfbits2f(dy[r * D + c]);
+    float xh = (bfbits2f(s[r * D + c]) - mean[r]) * rstd[r];
+    sg += dyv * xh;
+    sb += dyv;
+  }
+  dgamma[c] = f2bfbits(sg);
+  dbeta[c] = f2bfbits(sb);
+}
+
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor res,
+                                  torch::Tensor gamma, torch::Tensor beta,
+                                  double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.dim() == 2 &&
+              x.is_contiguous() && res.is_contiguous());
+  const int R = x.size(0), D = x.size(1);
+  auto y = torch::empty_like(x);
+  auto s = torch::empty_like(x);
+  auto mean = torch::empty({R}, x.options().dtype(torch::kFloat32));
+  auto rstd = torch::empty({R}, x.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  ln_fwd_kernel<<<cdiv(R, 4), 256, 0, stream>>>(
+      (const short*)x.data_ptr(), (const short*)res.data_ptr(),
+      (const short*)gamma.data_ptr(), (const short*)beta.data_ptr(),
+      (short*)y.data_ptr(), (short*)s.data_ptr(), mean.data_ptr<float>(),
+      rstd.data_ptr<float>(), R, D, (float)eps);
+  return {y, s, mean, rstd};
+}
+
+std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor s,
+                                  torch::Tensor gamma, torch::Tensor mean,
+                                  torch::Tensor rstd) {
+  const int R = dy.size(0), D = dy.size(1);
+  auto dx = torch::empty_like(dy);
+  auto dgamma = torch::empty({D}, dy.options());
+  auto dbeta = torch::empty({D}, dy.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  ln_bwd_kernel<<<cdiv(R, 4), 256, 0, stream>>>(
+      (const short*)dy.data_ptr(), (const short*)s.data_ptr(),
+      (const short*)gamma.data_ptr(), mean.data_ptr<float>(),
+      rstd.data_ptr<float>(), (short*)dx.data_ptr(), R, D);
+  ln_gb_kernel<<<cdiv(D, 256), 256, 0, stream>>>(
+      (const short*)dy.data_ptr(), (const short*)s.data_ptr(),
+      mean.data_ptr<float>(), rstd.data_ptr<float>(),
+      (short*)dgamma.data_ptr(), (short*)dbeta.data_ptr(), R, D);
+  return {dx, dgamma, dbeta};
+}
