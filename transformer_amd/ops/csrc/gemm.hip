@@ -104,8 +104,8 @@ void gemm_nt_kernel(const short* __restrict__ A, const short* __restrict__ W,
 
   for (int k0 = 0; k0 < K; k0 += BK) {
     int kc = min(BK, K - k0);
-    stage_tile<ALIGNED_A>(A + k0, K, mrows, kc, a_lds);
-    stage_tile<ALIGNED_B>(W + k0, K, nrows, kc, b_lds);
+    stage_tile<ALIGNED_A>(A + (long)bm0 * K + k0, K, mrows, kc, a_lds);
+    stage_tile<ALIGNED_B>(W + (long)bn0 * K + k0, K, nrows, kc, b_lds);
     __syncthreads();
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {     // two 32-deep MFMA K-steps
