@@ -48,6 +48,15 @@ DEV_INLINE void stage_tile(const short* __restrict__ g, int ldg, int nrows,
         s16x8 vv = *(const s16x8*)src;
         lo = {vv[0], vv[1], vv[2], vv[3]};
         hi = {vv[4], vv[5], vv[6], vv[7]};
+      } else if ((ldg & 1) == 0 && c8 + 8 <= ncols) {
+        // 4-byte-aligned rows (even leading dim): s16x2 loads, 4x fewer
+        // transactions than scalar (hits e.g. the V=32770 logits backward)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          s16x2 p = *(const s16x2*)(src + 2 * j);
+          if (j < 2) { lo[2 * j] = p[0]; lo[2 * j + 1] = p[1]; }
+          else { hi[2 * (j - 2)] = p[0]; hi[2 * (j - 2) + 1] = p[1]; }
+        }
       } else {
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
@@ -177,14 +186,26 @@ void transpose2d_kernel(const short* __restrict__ in, short* __restrict__ out,
   }
 }
 
-// colsum: db[N] = sum_m dy[M,N] (fp32 accumulate, bf16 out)
-__global__ void colsum_kernel(const short* __restrict__ dy,
-                              short* __restrict__ db, int M, int N) {
-  int n = blockIdx.x * blockDim.x + threadIdx.x;
+// colsum: db[N] = sum_m dy[M,N] (fp32 accumulate, bf16 out).
+// Two-stage: blocks cover (col-chunk, row-chunk) so the whole chip
+// participates; fp32 atomics into a zeroed workspace, then cast.
+#define COLSUM_ROWS 128
+__global__ __launch_bounds__(256)
+void colsum_part_kernel(const short* __restrict__ dy, float* __restrict__ acc,
+                        int M, int N) {
+  int n = blockIdx.x * 256 + threadIdx.x;
   if (n >= N) return;
+  long m0 = (long)blockIdx.y * COLSUM_ROWS;
+  long m1 = min((long)M, m0 + COLSUM_ROWS);
   float s = 0.f;
-  for (long m = 0; m < M; ++m) s += bfbits2f(dy[m * N + n]);
-  db[n] = f2bfbits(s);
+  for (long m = m0; m < m1; ++m) s += bfbits2f(dy[m * N + n]);
+  atomicAdd(&acc[n], s);
+}
+
+__global__ void cast_colsum_kernel(const float* __restrict__ in,
+                                   short* __restrict__ out, int n) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) out[i] = f2bfbits(in[i]);
 }
 
 // relu_bwd: dz = dy * (y > 0)
@@ -269,10 +290,14 @@ torch::Tensor transpose2d(torch::Tensor a) {
 torch::Tensor colsum(torch::Tensor a) {
   CHECK_BF16_2D(a);
   int M = a.size(0), N = a.size(1);
+  auto acc = torch::zeros({N}, a.options().dtype(torch::kFloat32));
   auto out = torch::empty({N}, a.options());
   auto stream = at::hip::getCurrentHIPStream();
-  colsum_kernel<<<cdiv(N, 256), 256, 0, stream>>>(
-      (const short*)a.data_ptr(), (short*)out.data_ptr(), M, N);
+  dim3 grid(cdiv(N, 256), cdiv(M, COLSUM_ROWS));
+  colsum_part_kernel<<<grid, 256, 0, stream>>>(
+      (const short*)a.data_ptr(), acc.data_ptr<float>(), M, N);
+  cast_colsum_kernel<<<cdiv(N, 256), 256, 0, stream>>>(
+      acc.data_ptr<float>(), (short*)out.data_ptr(), N);
   return out;
 }
 

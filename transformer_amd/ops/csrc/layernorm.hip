@@ -111,22 +111,36 @@ void ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ s,
 // column reduction: dgamma[c] = sum_r dy[r][c]*xhat[r][c]; dbeta[c] = sum dy.
 // Thread t of block b owns column b*256+t; every row read is coalesced
 // across the block's 256 consecutive columns.
+#define LNGB_ROWS 128
 __global__ __launch_bounds__(256)
 void ln_gb_kernel(const short* __restrict__ dy, const short* __restrict__ s,
                   const float* __restrict__ mean,
-                  const float* __restrict__ rstd, short* __restrict__ dgamma,
-                  short* __restrict__ dbeta, int R, int D) {
+                  const float* __restrict__ rstd, float* __restrict__ acc_g,
+                  float* __restrict__ acc_b, int R, int D) {
   const int c = blockIdx.x * 256 + threadIdx.x;
   if (c >= D) return;
+  const long r0 = (long)blockIdx.y * LNGB_ROWS;
+  const long r1 = min((long)R, r0 + LNGB_ROWS);
   float sg = 0.f, sb = 0.f;
-  for (long r = 0; r < R; ++r) {
+  for (long r = r0; r < r1; ++r) {
     float dyv = bfbits2f(dy[r * D + c]);
     float xh = (bfbits2f(s[r * D + c]) - mean[r]) * rstd[r];
     sg += dyv * xh;
     sb += dyv;
   }
-  dgamma[c] = f2bfbits(sg);
-  dbeta[c] = f2bfbits(sb);
+  atomicAdd(&acc_g[c], sg);
+  atomicAdd(&acc_b[c], sb);
+}
+
+__global__ void ln_gb_cast_kernel(const float* __restrict__ ag,
+                                  const float* __restrict__ ab,
+                                  short* __restrict__ dgamma,
+                                  short* __restrict__ dbeta, int D) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < D) {
+    dgamma[i] = f2bfbits(ag[i]);
+    dbeta[i] = f2bfbits(ab[i]);
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -161,9 +175,15 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor s,
       (const short*)dy.data_ptr(), (const short*)s.data_ptr(),
       (const short*)gamma.data_ptr(), mean.data_ptr<float>(),
       rstd.data_ptr<float>(), (short*)dx.data_ptr(), R, D);
-  ln_gb_kernel<<<cdiv(D, 256), 256, 0, stream>>>(
+  auto acc_g = torch::zeros({D}, dy.options().dtype(torch::kFloat32));
+  auto acc_b = torch::zeros({D}, dy.options().dtype(torch::kFloat32));
+  dim3 gbgrid(cdiv(D, 256), cdiv(R, LNGB_ROWS));
+  ln_gb_kernel<<<gbgrid, 256, 0, stream>>>(
       (const short*)dy.data_ptr(), (const short*)s.data_ptr(),
       mean.data_ptr<float>(), rstd.data_ptr<float>(),
-      (short*)dgamma.data_ptr(), (short*)dbeta.data_ptr(), R, D);
+      acc_g.data_ptr<float>(), acc_b.data_ptr<float>(), R, D);
+  ln_gb_cast_kernel<<<cdiv(D, 256), 256, 0, stream>>>(
+      acc_g.data_ptr<float>(), acc_b.data_ptr<float>(),
+      (short*)dgamma.data_ptr(), (short*)dbeta.data_ptr(), D);
   return {dx, dgamma, dbeta};
 }
