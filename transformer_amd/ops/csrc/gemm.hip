@@ -77,6 +77,27 @@ DEV_INLINE bf16x8 read_frag(const short* lds, int row, int kbase) {
   return (bf16x8)v0;
 }
 
+// global_load_lds staging (guide §5 step-3 / rule 21): LDS image stays the
+// SWIZZLED layout, built with a lane-linear LDS destination by inverse-
+// swizzling the per-lane SOURCE address.  16 wave-instructions stage a
+// [128][64] bf16 tile.  Requires a full tile (no row/col guards).
+DEV_INLINE void stage_tile_glds(const short* __restrict__ g, int ldg,
+                                short* lds) {
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    const int chunk = p * 4 + wid;            // 16 chunks of 1 KiB
+    const int idx = chunk * 64 + lane;        // 16-B unit index
+    const int row = idx >> 3;
+    const int colb = ((idx & 7) << 4) ^ ((row & 7) << 4);  // inverse swizzle
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)(g + (long)row * ldg +
+                                                        (colb >> 1)),
+        (__attribute__((address_space(3))) void*)((char*)lds + chunk * 1024),
+        16, 0, 0);
+  }
+}
+
 template <int EPILOGUE, bool ALIGNED_A, bool ALIGNED_B>
 __global__ __launch_bounds__(NTHREADS)
 void gemm_nt_kernel(const short* __restrict__ A, const short* __restrict__ W,
@@ -110,11 +131,20 @@ void gemm_nt_kernel(const short* __restrict__ A, const short* __restrict__ W,
 
   const int mrows = min(BM, M - bm0);
   const int nrows = min(BN, N - bn0);
+  // glds fast path: full output tile and K a multiple of BK (no guards);
+  // block-uniform condition.
+  const bool glds = ALIGNED_A && ALIGNED_B && mrows == BM && nrows == BN &&
+                    (K % BK) == 0;
 
   for (int k0 = 0; k0 < K; k0 += BK) {
     int kc = min(BK, K - k0);
-    stage_tile<ALIGNED_A>(A + (long)bm0 * K + k0, K, mrows, kc, a_lds);
-    stage_tile<ALIGNED_B>(W + (long)bn0 * K + k0, K, nrows, kc, b_lds);
+    if (glds) {
+      stage_tile_glds(A + (long)bm0 * K + k0, K, a_lds);
+      stage_tile_glds(W + (long)bn0 * K + k0, K, b_lds);
+    } else {
+      stage_tile<ALIGNED_A>(A + (long)bm0 * K + k0, K, mrows, kc, a_lds);
+      stage_tile<ALIGNED_B>(W + (long)bn0 * K + k0, K, nrows, kc, b_lds);
+    }
     __syncthreads();
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {     // two 32-deep MFMA K-steps
