@@ -98,11 +98,17 @@ DEV_INLINE void stage_tile_glds(const short* __restrict__ g, int ldg,
   }
 }
 
-template <int EPILOGUE, bool ALIGNED_A, bool ALIGNED_B>
+// SPLITK=false: writes bf16 C directly with fused bias/ReLU.
+// SPLITK=true (blockIdx.y = K-slice): fp32 atomicAdd partials into CW;
+// bias/epilogue applied by gemm_finalize_kernel after all slices land.
+// Split-K keeps the chip full on deep-K small-MN GEMMs (the dW shapes:
+// e.g. 512x512xK=16384 is only 16 workgroups unsplit).
+template <int EPILOGUE, bool ALIGNED_A, bool ALIGNED_B, bool SPLITK>
 __global__ __launch_bounds__(NTHREADS)
 void gemm_nt_kernel(const short* __restrict__ A, const short* __restrict__ W,
                     const short* __restrict__ bias, short* __restrict__ C,
-                    int M, int N, int K, int has_bias, int nbm, int nbn) {
+                    float* __restrict__ CW, int M, int N, int K, int has_bias,
+                    int nbm, int nbn, int k_per_slice) {
   __shared__ short a_lds[BM * BK];
   __shared__ short b_lds[BN * BK];
 
@@ -135,8 +141,13 @@ void gemm_nt_kernel(const short* __restrict__ A, const short* __restrict__ W,
   // block-uniform condition.
   const bool glds = ALIGNED_A && ALIGNED_B && mrows == BM && nrows == BN &&
                     (K % BK) == 0;
+  int k_lo = 0, k_hi = K;
+  if (SPLITK) {
+    k_lo = blockIdx.y * k_per_slice;
+    k_hi = min(K, k_lo + k_per_slice);
+  }
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
+  for (int k0 = k_lo; k0 < k_hi; k0 += BK) {
     int kc = min(BK, K - k0);
     if (glds) {
       stage_tile_glds(A + (long)bm0 * K + k0, K, a_lds);
@@ -175,17 +186,35 @@ void gemm_nt_kernel(const short* __restrict__ A, const short* __restrict__ W,
     for (int j = 0; j < 4; ++j) {
       int gcol = bn0 + wc + j * 16 + fr;
       if (gcol >= N) continue;
-      float bv = (has_bias && bias) ? bfbits2f(bias[gcol]) : 0.0f;
+      float bv = (!SPLITK && has_bias && bias) ? bfbits2f(bias[gcol]) : 0.0f;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int grow = grow_base + r;
         if (grow >= M) continue;
-        float v = acc[i][j][r] + bv;
-        if (EPILOGUE == 1) v = fmaxf(v, 0.0f);
-        C[(long)grow * N + gcol] = f2bfbits(v);
+        if (SPLITK) {
+          atomicAdd(&CW[(long)grow * N + gcol], acc[i][j][r]);
+        } else {
+          float v = acc[i][j][r] + bv;
+          if (EPILOGUE == 1) v = fmaxf(v, 0.0f);
+          C[(long)grow * N + gcol] = f2bfbits(v);
+        }
       }
     }
   }
+}
+
+// finalize for the split-K path: C = epi(CW + bias) in bf16
+template <int EPILOGUE>
+__global__ void gemm_finalize_kernel(const float* __restrict__ cw,
+                                     const short* __restrict__ bias,
+                                     short* __restrict__ c, long mn, int N,
+                                     int has_bias) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= mn) return;
+  float v = cw[i];
+  if (has_bias && bias) v += bfbits2f(bias[i % N]);
+  if (EPILOGUE == 1) v = fmaxf(v, 0.0f);
+  c[i] = f2bfbits(v);
 }
 
 // ---------------------------------------------------------------------------
@@ -283,25 +312,51 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
                 bias.numel() == N, "bias must be bf16[N]");
   auto c = torch::empty({M, N}, a.options());
   int nbm = cdiv(M, BM), nbn = cdiv(N, BN);
-  dim3 grid(nbm * nbn);
+  int nwg = nbm * nbn;
+  // split-K when the unsplit grid underfills the 256-CU chip and K is deep
+  int sk = 1;
+  if (nwg < 256 && K >= 4 * BK) {
+    sk = std::min({(512 + nwg - 1) / nwg, K / BK, 32});
+    sk = std::max(sk, 1);
+  }
+  int k_per_slice = (cdiv(K, BK) + sk - 1) / sk * BK;
+  sk = cdiv(K, k_per_slice);
+  dim3 grid(nwg, sk);
   auto stream = at::hip::getCurrentHIPStream();
   bool al = (K % 8 == 0);
-  auto launch = [&](auto epi, auto ala, auto alb) {
+  torch::Tensor cw;
+  float* cwp = nullptr;
+  if (sk > 1) {
+    cw = torch::zeros({M, N}, a.options().dtype(torch::kFloat32));
+    cwp = cw.data_ptr<float>();
+  }
+  auto launch = [&](auto epi, auto ala, auto splitk) {
     gemm_nt_kernel<decltype(epi)::value, decltype(ala)::value,
-                   decltype(alb)::value>
+                   decltype(ala)::value, decltype(splitk)::value>
         <<<grid, NTHREADS, 0, stream>>>(
             (const short*)a.data_ptr(), (const short*)w.data_ptr(),
             has_bias ? (const short*)bias.data_ptr() : nullptr,
-            (short*)c.data_ptr(), M, N, K, has_bias, nbm, nbn);
+            (short*)c.data_ptr(), cwp, M, N, K, has_bias, nbm, nbn,
+            k_per_slice);
   };
   using T = std::true_type;
   using F = std::false_type;
   using E0 = std::integral_constant<int, 0>;
   using E1 = std::integral_constant<int, 1>;
-  if (epilogue == 1) {
-    if (al) launch(E1{}, T{}, T{}); else launch(E1{}, F{}, F{});
+  if (sk > 1) {
+    if (epilogue == 1) { if (al) launch(E1{}, T{}, T{}); else launch(E1{}, F{}, T{}); }
+    else { if (al) launch(E0{}, T{}, T{}); else launch(E0{}, F{}, T{}); }
+    long mn = (long)M * N;
+    auto fin = [&](auto epi) {
+      gemm_finalize_kernel<decltype(epi)::value>
+          <<<(mn + 1023) / 1024, 1024, 0, stream>>>(
+              cwp, has_bias ? (const short*)bias.data_ptr() : nullptr,
+              (short*)c.data_ptr(), mn, N, has_bias);
+    };
+    if (epilogue == 1) fin(E1{}); else fin(E0{});
   } else {
-    if (al) launch(E0{}, T{}, T{}); else launch(E0{}, F{}, F{});
+    if (epilogue == 1) { if (al) launch(E1{}, T{}, F{}); else launch(E1{}, F{}, F{}); }
+    else { if (al) launch(E0{}, T{}, F{}); else launch(E0{}, F{}, F{}); }
   }
   return c;
 }
