@@ -109,8 +109,11 @@ void gemm_nt_kernel(const short* __restrict__ A, const short* __restrict__ W,
                     const short* __restrict__ bias, short* __restrict__ C,
                     float* __restrict__ CW, int M, int N, int K, int has_bias,
                     int nbm, int nbn, int k_per_slice) {
-  __shared__ short a_lds[BM * BK];
-  __shared__ short b_lds[BN * BK];
+  // Single-buffered 32 KiB LDS: measured FASTER than a 64 KiB double
+  // buffer here — at 3-4 blocks/CU, block-level overlap already hides the
+  // glds latency, and doubling LDS halves occupancy (guide §5 regime note).
+  __shared__ short a_lds2[1][BM * BK];
+  __shared__ short b_lds2[1][BN * BK];
 
   // XCD-aware bijective remap (guide T1): contiguous grid chunk per XCD.
   int nwg = nbm * nbn;
@@ -147,34 +150,42 @@ void gemm_nt_kernel(const short* __restrict__ A, const short* __restrict__ W,
     k_hi = min(K, k_lo + k_per_slice);
   }
 
-  for (int k0 = k_lo; k0 < k_hi; k0 += BK) {
-    int kc = min(BK, K - k0);
-    if (glds) {
-      stage_tile_glds(A + (long)bm0 * K + k0, K, a_lds);
-      stage_tile_glds(W + (long)bn0 * K + k0, K, b_lds);
-    } else {
-      stage_tile<ALIGNED_A>(A + (long)bm0 * K + k0, K, mrows, kc, a_lds);
-      stage_tile<ALIGNED_B>(W + (long)bn0 * K + k0, K, nrows, kc, b_lds);
-    }
-    __syncthreads();
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {     // two 32-deep MFMA K-steps
-      bf16x8 af[4], bf_[4];
-#pragma unroll
-      for (int i = 0; i < 4; ++i)
-        af[i] = read_frag(a_lds, wr + i * 16 + fr, ks * 32 + kg * 8);
-#pragma unroll
-      for (int j = 0; j < 4; ++j)
-        bf_[j] = read_frag(b_lds, wc + j * 16 + fr, ks * 32 + kg * 8);
-#pragma unroll
-      for (int i = 0; i < 4; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[i], bf_[j], acc[i][j], 0, 0, 0);
-    }
-    __syncthreads();
+  const long Aoff = (long)bm0 * K, Woff = (long)bn0 * K;
+
+#define MFMA_TILE(a_lds, b_lds)                                            \
+  _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) {                       \
+    bf16x8 af[4], bf_[4];                                                  \
+    _Pragma("unroll") for (int i = 0; i < 4; ++i)                          \
+        af[i] = read_frag(a_lds, wr + i * 16 + fr, ks * 32 + kg * 8);      \
+    _Pragma("unroll") for (int j = 0; j < 4; ++j)                          \
+        bf_[j] = read_frag(b_lds, wc + j * 16 + fr, ks * 32 + kg * 8);     \
+    _Pragma("unroll") for (int i = 0; i < 4; ++i)                          \
+        _Pragma("unroll") for (int j = 0; j < 4; ++j)                      \
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(           \
+                af[i], bf_[j], acc[i][j], 0, 0, 0);                        \
   }
+
+  if (glds) {
+    for (int k0 = k_lo; k0 < k_hi; k0 += BK) {
+      stage_tile_glds(A + Aoff + k0, K, a_lds2[0]);
+      stage_tile_glds(W + Woff + k0, K, b_lds2[0]);
+      __syncthreads();              // drains the LDS-DMA queue (vmcnt 0)
+      MFMA_TILE(a_lds2[0], b_lds2[0]);
+      __syncthreads();
+    }
+  } else {
+    short* a_lds = a_lds2[0];
+    short* b_lds = b_lds2[0];
+    for (int k0 = k_lo; k0 < k_hi; k0 += BK) {
+      int kc = min(BK, K - k0);
+      stage_tile<ALIGNED_A>(A + Aoff + k0, K, mrows, kc, a_lds);
+      stage_tile<ALIGNED_B>(W + Woff + k0, K, nrows, kc, b_lds);
+      __syncthreads();
+      MFMA_TILE(a_lds, b_lds);
+      __syncthreads();
+    }
+  }
+#undef MFMA_TILE
 
   // Epilogue: C/D lane map is col = lane&15, row = (lane>>4)*4 + r.
   // Pack 2 bf16 (adjacent rows... cols are per-lane) -> scalar stores; the
@@ -225,23 +236,43 @@ __global__ void gemm_finalize_kernel(const float* __restrict__ cw,
 __global__ __launch_bounds__(256)
 void transpose2d_kernel(const short* __restrict__ in, short* __restrict__ out,
                         int M, int N) {
-  __shared__ short tile[64][64 + 4];
+  __shared__ short tile[64][64 + 8];  // +8 elems (16 B) row pad vs conflicts
   int tb = blockIdx.x, nbx = (N + 63) >> 6;
   int bm = (tb / nbx) << 6, bn = (tb % nbx) << 6;
   int t = threadIdx.x;
-  // load 64x64: each thread 16 elems as 4x short4 rows
-  for (int p = 0; p < 16; ++p) {
-    int idx = p * 256 + t;          // 4096 slots of 1 elem... use vec when interior
-    int r = idx >> 6, c = idx & 63;
-    int gr = bm + r, gc = bn + c;
-    tile[r][c] = (gr < M && gc < N) ? in[(long)gr * N + gc] : (short)0;
-  }
-  __syncthreads();
-  for (int p = 0; p < 16; ++p) {
-    int idx = p * 256 + t;
-    int r = idx >> 6, c = idx & 63;   // r,c in OUTPUT tile [N-major]
-    int gr = bn + r, gc = bm + c;
-    if (gr < N && gc < M) out[(long)gr * M + gc] = tile[c][r];
+  const bool interior = (bm + 64 <= M) && (bn + 64 <= N);
+  if (interior) {
+    // vectorized: 512 short8 chunks, 2 per thread
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      int idx = p * 256 + t;
+      int r = idx >> 3, c8 = (idx & 7) << 3;
+      *(s16x8*)&tile[r][c8] = *(const s16x8*)(in + (long)(bm + r) * N + bn + c8);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      int idx = p * 256 + t;
+      int r = idx >> 3, c8 = (idx & 7) << 3;  // output row r (= col of in)
+      s16x8 v;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = tile[c8 + j][r];
+      *(s16x8*)(out + (long)(bn + r) * M + bm + c8) = v;
+    }
+  } else {
+    for (int p = 0; p < 16; ++p) {
+      int idx = p * 256 + t;
+      int r = idx >> 6, c = idx & 63;
+      int gr = bm + r, gc = bn + c;
+      tile[r][c] = (gr < M && gc < N) ? in[(long)gr * N + gc] : (short)0;
+    }
+    __syncthreads();
+    for (int p = 0; p < 16; ++p) {
+      int idx = p * 256 + t;
+      int r = idx >> 6, c = idx & 63;
+      int gr = bn + r, gc = bm + c;
+      if (gr < N && gc < M) out[(long)gr * M + gc] = tile[c][r];
+    }
   }
 }
 
@@ -256,9 +287,17 @@ void colsum_part_kernel(const short* __restrict__ dy, float* __restrict__ acc,
   if (n >= N) return;
   long m0 = (long)blockIdx.y * COLSUM_ROWS;
   long m1 = min((long)M, m0 + COLSUM_ROWS);
-  float s = 0.f;
-  for (long m = m0; m < m1; ++m) s += bfbits2f(dy[m * N + n]);
-  atomicAdd(&acc[n], s);
+  // 4 independent accumulators for memory-level parallelism
+  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+  long m = m0;
+  for (; m + 4 <= m1; m += 4) {
+    s0 += bfbits2f(dy[m * N + n]);
+    s1 += bfbits2f(dy[(m + 1) * N + n]);
+    s2 += bfbits2f(dy[(m + 2) * N + n]);
+    s3 += bfbits2f(dy[(m + 3) * N + n]);
+  }
+  for (; m < m1; ++m) s0 += bfbits2f(dy[m * N + n]);
+  atomicAdd(&acc[n], (s0 + s1) + (s2 + s3));
 }
 
 __global__ void cast_colsum_kernel(const float* __restrict__ in,
