@@ -66,6 +66,41 @@ def test_gemm_no_bias():
     assert_close(c, a.float() @ w.float().T, 0.03, "gemm_nobias")
 
 
+@pytest.mark.parametrize("m,n,k", [(256, 512, 512), (256, 512, 1536),
+                                   (100, 130, 72), (64, 512, 32770),
+                                   (33, 17, 24)])
+def test_gemm_nn(m, n, k):
+    # dX = dY @ W: contraction over W's row dim, no physical transpose
+    torch.manual_seed(2)
+    a = torch.randn(m, k, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(k, n, device="cuda", dtype=torch.bfloat16)
+    c = _ext().gemm_nn(a, b)
+    assert_close(c, a.float() @ b.float(), 0.03, "gemm_nn")
+
+
+@pytest.mark.parametrize("m,n,k", [(512, 512, 4096), (1536, 512, 1000),
+                                   (130, 70, 100), (2048, 512, 16384),
+                                   (17, 33, 50)])
+def test_gemm_tn(m, n, k):
+    # dW = dY^T @ X: both operands contraction(row)-major
+    torch.manual_seed(3)
+    a = torch.randn(k, m, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(k, n, device="cuda", dtype=torch.bfloat16)
+    c = _ext().gemm_tn(a, b)
+    assert_close(c, a.float().T @ b.float(), 0.03, "gemm_tn")
+
+
+def test_gemm_tn_out_destination():
+    # dW written straight into a provided (flat-grad view) buffer
+    torch.manual_seed(4)
+    a = torch.randn(777, 96, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(777, 64, device="cuda", dtype=torch.bfloat16)
+    out = torch.empty(96, 64, device="cuda", dtype=torch.bfloat16)
+    c = _ext().gemm_tn(a, b, out)
+    assert c.data_ptr() == out.data_ptr()
+    assert_close(out, a.float().T @ b.float(), 0.03, "gemm_tn_out")
+
+
 def test_transpose2d():
     a = torch.randn(130, 70, device="cuda", dtype=torch.bfloat16)
     t = _ext().transpose2d(a)
@@ -179,7 +214,8 @@ def test_ce_fwd_bwd(ls):
     lg = logits.float().requires_grad_(True)
     lr2 = R.masked_cross_entropy(lg.view(1, Rn, V), targets.view(1, Rn), 8, ls)
     lr2.backward()
-    dl = _ext().ce_bwd(logits, targets, lse, 1.0, 8.0, ls)
+    one = torch.ones(1, device="cuda", dtype=torch.float32)
+    dl = _ext().ce_bwd(logits, targets, lse, one, 8.0, ls)
     assert_close(dl, lg.grad, 0.02, "ce_bwd")
 
 
@@ -302,3 +338,59 @@ def test_attn_fwd_no_pad_mask():
     o, _ = _ext().attn_fwd(q, k, v, torch.Tensor(), True, 1.0 / math.sqrt(dh))
     ref = _attn_ref(q, k, v, None, True)
     assert_close(o, ref, 0.04, "attn_nopad")
+
+
+# ---------------------------------------------------------------------------
+# Packed-QKV attention (strided kernel I/O, packed dQKV backward)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("causal", [False, True])
+def test_attn_packed_self_matches_unpacked(causal):
+    from transformer_amd import ops
+    torch.manual_seed(11)
+    B, S, H, dh = 3, 80, 4, 64
+    qkv = torch.randn(B, S, 3, H, dh, device="cuda", dtype=torch.bfloat16,
+                      requires_grad=True)
+    pad = torch.zeros(B, S, dtype=torch.bool, device="cuda")
+    pad[:, -7:] = True
+    out = ops.self_attention(qkv, kv_pad=pad, causal=causal)
+    g = torch.randn_like(out)
+    out.backward(g)
+    dqkv_packed = qkv.grad.clone()
+
+    qkv2 = qkv.detach().clone().requires_grad_(True)
+    q, k, v = (t.contiguous() for t in qkv2.unbind(dim=2))
+    out2 = ops.fused_attention(q.detach().clone().requires_grad_(True),
+                               k.detach().clone().requires_grad_(True),
+                               v.detach().clone().requires_grad_(True),
+                               kv_pad=pad, causal=causal)
+    assert_close(out, out2, 0.02, "packed fwd vs unpacked")
+
+    # backward parity vs fp32 reference through the eager path
+    qkv3 = qkv.detach().float().clone().requires_grad_(True)
+    out3 = ops.self_attention(qkv3.cpu(), kv_pad=pad.cpu(), causal=causal)
+    out3.backward(g.float().cpu())
+    assert_close(dqkv_packed, qkv3.grad, 0.06, "packed dqkv vs fp32 ref")
+
+
+def test_attn_packed_cross_matches_reference():
+    from transformer_amd import ops
+    torch.manual_seed(12)
+    B, Sq, Sk, H, dh = 2, 40, 96, 4, 64
+    q = torch.randn(B, Sq, H, dh, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    kv = torch.randn(B, Sk, 2, H, dh, device="cuda", dtype=torch.bfloat16,
+                     requires_grad=True)
+    pad = torch.zeros(B, Sk, dtype=torch.bool, device="cuda")
+    pad[:, -5:] = True
+    out = ops.cross_attention(q, kv, kv_pad=pad)
+    g = torch.randn_like(out)
+    out.backward(g)
+
+    q3 = q.detach().float().cpu().clone().requires_grad_(True)
+    kv3 = kv.detach().float().cpu().clone().requires_grad_(True)
+    out3 = ops.cross_attention(q3, kv3, kv_pad=pad.cpu())
+    out3.backward(g.float().cpu())
+    assert_close(out, out3, 0.02, "cross fwd")
+    assert_close(q.grad, q3.grad, 0.06, "cross dq")
+    assert_close(kv.grad, kv3.grad, 0.06, "cross dkv")

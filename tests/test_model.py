@@ -112,3 +112,36 @@ def test_pe_buffer_not_in_state_dict_params():
     assert "encoder.pe" not in dict(m.named_parameters())
     # non-persistent buffer: excluded from state_dict (recomputed on build)
     assert "encoder.pe" not in m.state_dict()
+
+
+def test_kv_cached_decode_matches_full_rerun():
+    """greedy_decode (KV cache, encoder reuse) must produce exactly the same
+    tokens as the reference's naive loop (full model re-run per step,
+    reference train.py:109-118)."""
+    import torch
+    from transformer_amd.models import Transformer
+    from transformer_amd.models.transformer import greedy_decode
+    from transformer_amd import ops
+
+    torch.manual_seed(7)
+    model = Transformer(num_layers=2, d_model=64, num_heads=4, dff=128,
+                        input_vocab_size=50, target_vocab_size=60,
+                        rate=0.0, max_position=64)
+    model.eval()
+    inp = torch.randint(1, 48, (2, 9))
+    inp[:, 0] = 48
+    inp[0, -1] = 49
+    inp[0, -2:] = 0  # padded tail on one sequence
+
+    start, end, max_len = 58, 59, 7
+    out_cached = greedy_decode(model, inp, start, end, max_len=max_len)
+
+    # naive loop
+    output = torch.full((2, 1), start, dtype=torch.int64)
+    for _ in range(max_len):
+        logits, _ = model((inp, output), training=False)
+        nxt = logits[:, -1:, :].argmax(dim=-1)
+        output = torch.cat([output, nxt], dim=-1)
+    assert out_cached.shape[1] <= output.shape[1]
+    T = out_cached.shape[1]
+    assert torch.equal(out_cached, output[:, :T]), (out_cached, output)

@@ -55,9 +55,10 @@ class SelfAttention(nn.Module):
         B, S, d = x.shape
         qkv = ops.linear(x, self.w_qkv, self.b_qkv)          # (B,S,3d)
         qkv = qkv.view(B, S, 3, self.num_heads, self.depth)
-        q, k, v = qkv.unbind(dim=2)                          # (B,S,H,dh) each
-        out = ops.fused_attention(q, k, v, kv_pad=kv_pad, causal=causal,
-                                  return_weights=return_weights)
+        # packed path: the attention kernels read q/k/v slots through strides
+        # and backward emits one packed dQKV (no copies)
+        out = ops.self_attention(qkv, kv_pad=kv_pad, causal=causal,
+                                 return_weights=return_weights)
         if return_weights:
             out, w = out
         out = out.reshape(B, S, d)
@@ -87,8 +88,7 @@ class CrossAttention(nn.Module):
         q = ops.linear(x, self.w_q, self.b_q).view(B, Sq, self.num_heads, self.depth)
         kv = ops.linear(enc_output, self.w_kv, self.b_kv)
         kv = kv.view(B, Sk, 2, self.num_heads, self.depth)
-        k, v = kv.unbind(dim=2)
-        out = ops.fused_attention(q, k, v, kv_pad=kv_pad, causal=False,
+        out = ops.cross_attention(q, kv, kv_pad=kv_pad,
                                   return_weights=return_weights)
         if return_weights:
             out, w = out
@@ -259,3 +259,107 @@ class Transformer(nn.Module):
 
     # convenience alias matching the reference's .call
     call = forward
+
+
+# ---------------------------------------------------------------------------
+# KV-cached incremental decoding (inference).  The reference's predict loop
+# re-runs the FULL encoder + decoder prefix for every generated token
+# (reference train.py:109-118); SURVEY.md §3.3 marks that as the naive
+# contract, not a design to copy.  Here: the encoder runs once, each decoder
+# layer keeps a packed (B, S_max, 2, H, dh) self-attention KV cache appended
+# in place, and cross-attention K/V are projected from the encoder output
+# once.  Each decode step costs O(1) in sequence length for the projections
+# and O(t) for the cached attention reads.
+# ---------------------------------------------------------------------------
+
+class DecodeCache:
+    """Per-layer packed KV caches + precomputed cross K/V."""
+
+    def __init__(self, model: "Transformer", B: int, max_len: int,
+                 enc_output: torch.Tensor, src_pad: torch.Tensor):
+        dec = model.decoder
+        d = dec.d_model
+        layer0 = dec.layers[0]
+        H, dh = layer0.mha1.num_heads, layer0.mha1.depth
+        dev, dt = enc_output.device, enc_output.dtype
+        self.pos = 0
+        self.max_len = max_len
+        self.enc_output = enc_output
+        self.src_pad = src_pad
+        self.self_kv = [torch.empty(B, max_len, 2, H, dh, device=dev, dtype=dt)
+                        for _ in dec.layers]
+        # cross K/V once per sequence (B, Sk, 2, H, dh)
+        self.cross_kv = []
+        Sk = enc_output.shape[1]
+        for layer in dec.layers:
+            kv = ops.linear(enc_output, layer.mha2.w_kv, layer.mha2.b_kv)
+            self.cross_kv.append(kv.view(B, Sk, 2, H, dh))
+
+
+@torch.no_grad()
+def encode(model, inp, training=False):
+    """Run the encoder once; returns (enc_output, src_pad)."""
+    src_pad = (inp == 0)
+    return model.encoder(inp, src_pad, training), src_pad
+
+
+@torch.no_grad()
+def decode_step(model, tokens_new, cache: DecodeCache):
+    """One incremental decoder step.
+
+    tokens_new: (B, 1) int64 — the latest target token per sequence.
+    Returns logits (B, 1, V) for the next-token distribution."""
+    dec = model.decoder
+    B = tokens_new.shape[0]
+    t = cache.pos
+    # embedding * sqrt(d) + PE at position t
+    x = ops.embedding_scale_pe_at(tokens_new, dec.embedding, dec.pe, t)
+    for li, layer in enumerate(dec.layers):
+        H, dh = layer.mha1.num_heads, layer.mha1.depth
+        # masked self-attention against the cache: project packed qkv for the
+        # new row, append k/v in place, attend over positions [0, t].
+        qkv = ops.linear(x, layer.mha1.w_qkv, layer.mha1.b_qkv)
+        qkv = qkv.view(B, 1, 3, H, dh)
+        cache.self_kv[li][:, t:t + 1] = qkv[:, :, 1:3]
+        q = qkv[:, :, 0].contiguous()
+        kv_hist = cache.self_kv[li][:, :t + 1]
+        k, v = kv_hist.unbind(dim=2)
+        attn1 = ops.fused_attention(q, k, v)  # all cached keys visible
+        attn1 = ops.linear(attn1.reshape(B, 1, H * dh),
+                           layer.mha1.w_o, layer.mha1.b_o)
+        out1 = ops.residual_layernorm(attn1, x, layer.ln1.gamma,
+                                      layer.ln1.beta, layer.ln1.eps)
+        # cross-attention against precomputed encoder K/V
+        q2 = ops.linear(out1, layer.mha2.w_q, layer.mha2.b_q)
+        q2 = q2.view(B, 1, H, dh)
+        k2, v2 = cache.cross_kv[li].unbind(dim=2)
+        attn2 = ops.fused_attention(q2, k2, v2, kv_pad=cache.src_pad)
+        attn2 = ops.linear(attn2.reshape(B, 1, H * dh),
+                           layer.mha2.w_o, layer.mha2.b_o)
+        out2 = ops.residual_layernorm(attn2, out1, layer.ln2.gamma,
+                                      layer.ln2.beta, layer.ln2.eps)
+        ffn = layer.ffn(out2)
+        x = ops.residual_layernorm(ffn, out2, layer.ln3.gamma,
+                                   layer.ln3.beta, layer.ln3.eps)
+    cache.pos = t + 1
+    return ops.linear(x, model.w_final, model.b_final)
+
+
+@torch.no_grad()
+def greedy_decode(model, inp, start_id, end_id, max_len=10):
+    """KV-cached greedy decode.  inp (B, S) int64; returns (B, <=max_len+1)
+    token ids starting with start_id; stops early when every sequence has
+    emitted end_id."""
+    B = inp.shape[0]
+    enc_output, src_pad = encode(model, inp)
+    cache = DecodeCache(model, B, max_len + 1, enc_output, src_pad)
+    out = torch.full((B, 1), start_id, dtype=torch.int64, device=inp.device)
+    finished = torch.zeros(B, dtype=torch.bool, device=inp.device)
+    for _ in range(max_len):
+        logits = decode_step(model, out[:, -1:], cache)
+        nxt = ops.argmax_lastdim(logits.float()).view(B, 1)
+        out = torch.cat([out, nxt], dim=-1)
+        finished |= (nxt.squeeze(1) == end_id)
+        if bool(finished.all()):
+            break
+    return out

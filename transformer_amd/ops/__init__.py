@@ -50,8 +50,11 @@ from . import reference  # noqa: E402
 from .functional import (  # noqa: E402
     linear,
     fused_attention,
+    self_attention,
+    cross_attention,
     residual_layernorm,
     embedding_scale_pe,
+    embedding_scale_pe_at,
     dropout,
     masked_cross_entropy,
     masked_accuracy,

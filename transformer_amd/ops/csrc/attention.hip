@@ -88,7 +88,8 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
                      const short* __restrict__ V,
                      const unsigned char* __restrict__ kv_pad,
                      short* __restrict__ O, float* __restrict__ LSE, int B,
-                     int H, int Sq, int Sk, int causal, float scale) {
+                     int H, int Sq, int Sk, int causal, float scale,
+                     long q_rs, long kv_rs, long o_rs) {
   constexpr int D32 = DH / 32;   // QK^T MFMA k-steps
   constexpr int D16 = DH / 16;   // O fragments
   __shared__ short k_lds[KVT * DH];
@@ -102,11 +103,9 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
   const int q0 = qb * (WAVES * QW) + wid * QW;  // this wave's first q-row
   const int fr = lane & 15, kg = lane >> 4;
 
-  const long q_bs = (long)Sq * H * DH, kv_bs = (long)Sk * H * DH;
-  const long row_stride = (long)H * DH;
-  const short* Qp = Q + b * q_bs + (long)h * DH;
-  const short* Kp = K + b * kv_bs + (long)h * DH;
-  const short* Vp = V + b * kv_bs + (long)h * DH;
+  const short* Qp = Q + b * (Sq * q_rs) + (long)h * DH;
+  const short* Kp = K + b * (Sk * kv_rs) + (long)h * DH;
+  const short* Vp = V + b * (Sk * kv_rs) + (long)h * DH;
   const unsigned char* pad = kv_pad ? kv_pad + (long)b * Sk : nullptr;
 
   // Q fragments in registers: A[row=fr][k=kg*8+j] per 32-chunk
@@ -115,7 +114,7 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
   for (int d = 0; d < D32; ++d) {
     s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
     int row = q0 + fr;
-    if (row < Sq) v = *(const s16x8*)(Qp + (long)row * row_stride + d * 32 + kg * 8);
+    if (row < Sq) v = *(const s16x8*)(Qp + (long)row * q_rs + d * 32 + kg * 8);
     qf[d] = (bf16x8)v;
   }
 
@@ -131,9 +130,9 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
 
   for (int k0 = 0; k0 < kend; k0 += KVT) {
     const int kc = min(KVT, Sk - k0);
-    stage_kv<DH, true, false>(Kp + (long)k0 * row_stride, row_stride, kc,
+    stage_kv<DH, true, false>(Kp + (long)k0 * kv_rs, kv_rs, kc,
                               k_lds, nullptr);
-    stage_kv<DH, false, true>(Vp + (long)k0 * row_stride, row_stride, kc,
+    stage_kv<DH, false, true>(Vp + (long)k0 * kv_rs, kv_rs, kc,
                               nullptr, vt_lds);
     __syncthreads();
 
@@ -221,7 +220,7 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
   }
 
   // epilogue: O = acc / l ; LSE = m + log(l)
-  short* Op = O + b * q_bs + (long)h * DH;
+  short* Op = O + b * (Sq * o_rs) + (long)h * DH;
 #pragma unroll
   for (int i = 0; i < D16; ++i) {
     const int gcol = i * 16 + fr;
@@ -231,7 +230,7 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
       if (qrow >= Sq) continue;
       float l = l_run[r];
       float o = (l > 0.f) ? acc[i][r] / l : 0.f;
-      Op[(long)qrow * row_stride + gcol] = f2bfbits(o);
+      Op[(long)qrow * o_rs + gcol] = f2bfbits(o);
     }
   }
   if (fr == 0) {
@@ -283,7 +282,8 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
                         const float* __restrict__ Dl,
                         const unsigned char* __restrict__ kv_pad,
                         short* __restrict__ dK, short* __restrict__ dV, int B,
-                        int H, int Sq, int Sk, int causal, float scale) {
+                        int H, int Sq, int Sk, int causal, float scale,
+                        long q_rs, long kv_rs, long do_rs, long dkv_rs) {
   constexpr int D32 = DH / 32;
   constexpr int D16 = DH / 16;
   __shared__ short q_lds[KVT * DH];    // q-tile natural [32][DH]
@@ -299,12 +299,10 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
   const int k0w = kb * (WAVES * QW) + wid * QW;  // wave's first key row
   const int fr = lane & 15, kg = lane >> 4;
 
-  const long q_bs = (long)Sq * H * DH, kv_bs = (long)Sk * H * DH;
-  const long row_stride = (long)H * DH;
-  const short* Qp = Q + b * q_bs + (long)h * DH;
-  const short* Kp = K + b * kv_bs + (long)h * DH;
-  const short* Vp = V + b * kv_bs + (long)h * DH;
-  const short* dOp = dO + b * q_bs + (long)h * DH;
+  const short* Qp = Q + b * (Sq * q_rs) + (long)h * DH;
+  const short* Kp = K + b * (Sk * kv_rs) + (long)h * DH;
+  const short* Vp = V + b * (Sk * kv_rs) + (long)h * DH;
+  const short* dOp = dO + b * (Sq * do_rs) + (long)h * DH;
   const float* lse = LSE + ((long)b * H + h) * Sq;
   const float* dl = Dl + ((long)b * H + h) * Sq;
   const unsigned char* pad = kv_pad ? kv_pad + (long)b * Sk : nullptr;
@@ -318,8 +316,8 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
   for (int d = 0; d < D32; ++d) {
     s16x8 kv_ = {0, 0, 0, 0, 0, 0, 0, 0}, vv = {0, 0, 0, 0, 0, 0, 0, 0};
     if (k_valid) {
-      kv_ = *(const s16x8*)(Kp + (long)krow * row_stride + d * 32 + kg * 8);
-      vv = *(const s16x8*)(Vp + (long)krow * row_stride + d * 32 + kg * 8);
+      kv_ = *(const s16x8*)(Kp + (long)krow * kv_rs + d * 32 + kg * 8);
+      vv = *(const s16x8*)(Vp + (long)krow * kv_rs + d * 32 + kg * 8);
     }
     kf[d] = (bf16x8)kv_;
     vf[d] = (bf16x8)vv;
@@ -337,9 +335,9 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
 
   for (int j0 = q_start; j0 < Sq; j0 += KVT) {
     const int jc = min(KVT, Sq - j0);
-    stage_kv<DH, true, true>(Qp + (long)j0 * row_stride, row_stride, jc,
+    stage_kv<DH, true, true>(Qp + (long)j0 * q_rs, q_rs, jc,
                              q_lds, qt_lds);
-    stage_kv<DH, true, true>(dOp + (long)j0 * row_stride, row_stride, jc,
+    stage_kv<DH, true, true>(dOp + (long)j0 * do_rs, do_rs, jc,
                              do_lds, dot_lds);
     __syncthreads();
 
@@ -397,8 +395,8 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
   }
 
   // write dK, dV (exclusive rows — no atomics)
-  short* dKp = dK + b * kv_bs + (long)h * DH;
-  short* dVp = dV + b * kv_bs + (long)h * DH;
+  short* dKp = dK + b * (Sk * dkv_rs) + (long)h * DH;
+  short* dVp = dV + b * (Sk * dkv_rs) + (long)h * DH;
 #pragma unroll
   for (int i = 0; i < D16; ++i) {
     const int gcol = i * 16 + fr;
@@ -406,8 +404,8 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
     for (int r = 0; r < 4; ++r) {
       const int key = k0w + kg * 4 + r;
       if (key >= Sk) continue;
-      dKp[(long)key * row_stride + gcol] = f2bfbits(acc_dk[i][r]);
-      dVp[(long)key * row_stride + gcol] = f2bfbits(acc_dv[i][r]);
+      dKp[(long)key * dkv_rs + gcol] = f2bfbits(acc_dk[i][r]);
+      dVp[(long)key * dkv_rs + gcol] = f2bfbits(acc_dv[i][r]);
     }
   }
 }
@@ -423,7 +421,8 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
                        const float* __restrict__ Dl,
                        const unsigned char* __restrict__ kv_pad,
                        short* __restrict__ dQ, int B, int H, int Sq, int Sk,
-                       int causal, float scale) {
+                       int causal, float scale, long q_rs, long kv_rs,
+                       long do_rs, long dq_rs) {
   constexpr int D32 = DH / 32;
   constexpr int D16 = DH / 16;
   __shared__ short k_lds[KVT * DH];
@@ -438,12 +437,10 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
   const int q0 = qb * (WAVES * QW) + wid * QW;
   const int fr = lane & 15, kg = lane >> 4;
 
-  const long q_bs = (long)Sq * H * DH, kv_bs = (long)Sk * H * DH;
-  const long row_stride = (long)H * DH;
-  const short* Qp = Q + b * q_bs + (long)h * DH;
-  const short* Kp = K + b * kv_bs + (long)h * DH;
-  const short* Vp = V + b * kv_bs + (long)h * DH;
-  const short* dOp = dO + b * q_bs + (long)h * DH;
+  const short* Qp = Q + b * (Sq * q_rs) + (long)h * DH;
+  const short* Kp = K + b * (Sk * kv_rs) + (long)h * DH;
+  const short* Vp = V + b * (Sk * kv_rs) + (long)h * DH;
+  const short* dOp = dO + b * (Sq * do_rs) + (long)h * DH;
   const float* lse = LSE + ((long)b * H + h) * Sq;
   const float* dl = Dl + ((long)b * H + h) * Sq;
   const unsigned char* pad = kv_pad ? kv_pad + (long)b * Sk : nullptr;
@@ -455,8 +452,8 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
   for (int d = 0; d < D32; ++d) {
     s16x8 qv = {0, 0, 0, 0, 0, 0, 0, 0}, dv = {0, 0, 0, 0, 0, 0, 0, 0};
     if (q_valid) {
-      qv = *(const s16x8*)(Qp + (long)qrow_l * row_stride + d * 32 + kg * 8);
-      dv = *(const s16x8*)(dOp + (long)qrow_l * row_stride + d * 32 + kg * 8);
+      qv = *(const s16x8*)(Qp + (long)qrow_l * q_rs + d * 32 + kg * 8);
+      dv = *(const s16x8*)(dOp + (long)qrow_l * do_rs + d * 32 + kg * 8);
     }
     qf[d] = (bf16x8)qv;
     dof[d] = (bf16x8)dv;
@@ -470,9 +467,9 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
 
   for (int k0 = 0; k0 < kend; k0 += KVT) {
     const int kc = min(KVT, Sk - k0);
-    stage_kv<DH, true, true>(Kp + (long)k0 * row_stride, row_stride, kc,
+    stage_kv<DH, true, true>(Kp + (long)k0 * kv_rs, kv_rs, kc,
                              k_lds, kt_lds);
-    stage_kv<DH, true, false>(Vp + (long)k0 * row_stride, row_stride, kc,
+    stage_kv<DH, true, false>(Vp + (long)k0 * kv_rs, kv_rs, kc,
                               v_lds, nullptr);
     __syncthreads();
 
@@ -512,7 +509,7 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
     __syncthreads();
   }
 
-  short* dQp = dQ + b * q_bs + (long)h * DH;
+  short* dQp = dQ + b * (Sq * dq_rs) + (long)h * DH;
 #pragma unroll
   for (int i = 0; i < D16; ++i) {
     const int gcol = i * 16 + fr;
@@ -520,7 +517,7 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
     for (int r = 0; r < 4; ++r) {
       const int qrow = q0 + kg * 4 + r;
       if (qrow >= Sq) continue;
-      dQp[(long)qrow * row_stride + gcol] = f2bfbits(acc_dq[i][r]);
+      dQp[(long)qrow * dq_rs + gcol] = f2bfbits(acc_dq[i][r]);
     }
   }
 }
@@ -535,14 +532,27 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
     default: TORCH_CHECK(false, "attention: head dim must be 32/64/128, got ", DH_VAL); \
   }
 
+static void check_attn_view(const torch::Tensor& t, int DH, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.dtype() == torch::kBFloat16 && t.dim() == 4,
+              name, " must be 4-D bf16 on GPU");
+  TORCH_CHECK(t.stride(3) == 1 && t.stride(2) == DH,
+              name, " must have contiguous (head, dh) trailing layout");
+  TORCH_CHECK(t.stride(0) == t.size(1) * t.stride(1),
+              name, " batch stride must equal S * row stride");
+}
+
+// q (B,Sq,H,dh), k/v (B,Sk,H,dh) — possibly strided views (e.g. slots of a
+// packed (B,S,3,H,dh) QKV tensor); k and v must share their row stride.
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor kv_pad,
                                     bool causal, double scale) {
-  TORCH_CHECK(q.is_cuda() && q.dtype() == torch::kBFloat16 && q.dim() == 4 &&
-              q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
   const int B = q.size(0), Sq = q.size(1), H = q.size(2), DH = q.size(3);
   const int Sk = k.size(1);
-  auto o = torch::empty_like(q);
+  check_attn_view(q, DH, "q");
+  check_attn_view(k, DH, "k");
+  check_attn_view(v, DH, "v");
+  TORCH_CHECK(k.stride(1) == v.stride(1), "k/v row strides must match");
+  auto o = torch::empty({B, Sq, H, DH}, q.options());
   auto lse = torch::empty({B, H, Sq}, q.options().dtype(torch::kFloat32));
   const unsigned char* pad = nullptr;
   if (kv_pad.defined() && kv_pad.numel() > 0) {
@@ -556,20 +566,59 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                   (const short*)q.data_ptr(), (const short*)k.data_ptr(),
                   (const short*)v.data_ptr(), pad, (short*)o.data_ptr(),
                   lse.data_ptr<float>(), B, H, Sq, Sk, causal ? 1 : 0,
-                  (float)scale));
+                  (float)scale, q.stride(1), k.stride(1), (long)H * DH));
   return {o, lse};
 }
 
+// mode 0: dq, dk, dv separate contiguous tensors.
+// mode 1: one packed dqkv (B,Sq,3,H,dh) — the self-attention path, gradient
+//         flows straight into the packed-QKV linear with no cat/stack.
+// mode 2: dq (B,Sq,H,dh) + packed dkv (B,Sk,2,H,dh) — the cross-attn path.
 std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor o,
                                     torch::Tensor dout, torch::Tensor lse,
                                     torch::Tensor kv_pad, bool causal,
-                                    double scale) {
+                                    double scale, int64_t mode) {
   const int B = q.size(0), Sq = q.size(1), H = q.size(2), DH = q.size(3);
   const int Sk = k.size(1);
-  auto dq = torch::empty_like(q);
-  auto dk = torch::empty_like(k);
-  auto dv = torch::empty_like(v);
+  check_attn_view(q, DH, "q");
+  check_attn_view(k, DH, "k");
+  check_attn_view(v, DH, "v");
+  TORCH_CHECK(k.stride(1) == v.stride(1), "k/v row strides must match");
+  TORCH_CHECK(o.is_contiguous() && dout.is_contiguous(),
+              "o/dout must be contiguous");
+  std::vector<torch::Tensor> outs;
+  short *dq_p, *dk_p, *dv_p;
+  long dq_rs, dkv_rs;
+  if (mode == 1) {
+    TORCH_CHECK(Sq == Sk, "packed self mode requires Sq == Sk");
+    auto dqkv = torch::empty({B, Sq, 3, H, DH}, q.options());
+    short* base = (short*)dqkv.data_ptr();
+    dq_p = base;
+    dk_p = base + (long)H * DH;
+    dv_p = base + 2L * H * DH;
+    dq_rs = dkv_rs = 3L * H * DH;
+    outs = {dqkv};
+  } else if (mode == 2) {
+    auto dq = torch::empty({B, Sq, H, DH}, q.options());
+    auto dkv = torch::empty({B, Sk, 2, H, DH}, q.options());
+    dq_p = (short*)dq.data_ptr();
+    short* base = (short*)dkv.data_ptr();
+    dk_p = base;
+    dv_p = base + (long)H * DH;
+    dq_rs = (long)H * DH;
+    dkv_rs = 2L * H * DH;
+    outs = {dq, dkv};
+  } else {
+    auto dq = torch::empty({B, Sq, H, DH}, q.options());
+    auto dk = torch::empty({B, Sk, H, DH}, q.options());
+    auto dv = torch::empty({B, Sk, H, DH}, q.options());
+    dq_p = (short*)dq.data_ptr();
+    dk_p = (short*)dk.data_ptr();
+    dv_p = (short*)dv.data_ptr();
+    dq_rs = dkv_rs = (long)H * DH;
+    outs = {dq, dk, dv};
+  }
   auto dl = torch::empty({B, H, Sq}, q.options().dtype(torch::kFloat32));
   const unsigned char* pad = nullptr;
   if (kv_pad.defined() && kv_pad.numel() > 0)
@@ -581,18 +630,19 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
       dl.data_ptr<float>(), H, Sq, DH, n_rows);
   dim3 grid_kv(B * H, cdiv(Sk, WAVES * QW));
   dim3 grid_q(B * H, cdiv(Sq, WAVES * QW));
+  const long q_rs = q.stride(1), kv_rs = k.stride(1);
+  const long do_rs = (long)H * DH;
   DISPATCH_DH(DH, {
     attn_bwd_kv_kernel<DHC><<<grid_kv, 256, 0, stream>>>(
         (const short*)q.data_ptr(), (const short*)k.data_ptr(),
         (const short*)v.data_ptr(), (const short*)dout.data_ptr(),
-        lse.data_ptr<float>(), dl.data_ptr<float>(), pad,
-        (short*)dk.data_ptr(), (short*)dv.data_ptr(), B, H, Sq, Sk,
-        causal ? 1 : 0, (float)scale);
+        lse.data_ptr<float>(), dl.data_ptr<float>(), pad, dk_p, dv_p, B, H,
+        Sq, Sk, causal ? 1 : 0, (float)scale, q_rs, kv_rs, do_rs, dkv_rs);
     attn_bwd_q_kernel<DHC><<<grid_q, 256, 0, stream>>>(
         (const short*)q.data_ptr(), (const short*)k.data_ptr(),
         (const short*)v.data_ptr(), (const short*)dout.data_ptr(),
-        lse.data_ptr<float>(), dl.data_ptr<float>(), pad,
-        (short*)dq.data_ptr(), B, H, Sq, Sk, causal ? 1 : 0, (float)scale);
+        lse.data_ptr<float>(), dl.data_ptr<float>(), pad, dq_p, B, H, Sq, Sk,
+        causal ? 1 : 0, (float)scale, q_rs, kv_rs, do_rs, dq_rs);
   });
-  return {dq, dk, dv};
+  return outs;
 }

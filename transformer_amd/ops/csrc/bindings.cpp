@@ -8,8 +8,12 @@
 // ---- implemented in the .hip translation units ----------------------------
 torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
                       int64_t epilogue);
+torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b,
+                      c10::optional<torch::Tensor> out);
+torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b,
+                      c10::optional<torch::Tensor> out);
 torch::Tensor transpose2d(torch::Tensor a);
-torch::Tensor colsum(torch::Tensor a);
+torch::Tensor colsum(torch::Tensor a, c10::optional<torch::Tensor> out);
 torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y);
 torch::Tensor smoke_add(torch::Tensor a, torch::Tensor b);
 
@@ -18,7 +22,9 @@ std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor res,
                                   double eps);
 std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor s,
                                   torch::Tensor gamma, torch::Tensor mean,
-                                  torch::Tensor rstd);
+                                  torch::Tensor rstd,
+                                  c10::optional<torch::Tensor> dgamma_out,
+                                  c10::optional<torch::Tensor> dbeta_out);
 
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor kv_pad,
@@ -27,12 +33,13 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor o,
                                     torch::Tensor dout, torch::Tensor lse,
                                     torch::Tensor kv_pad, bool causal,
-                                    double scale);
+                                    double scale, int64_t mode);
 
 torch::Tensor embed_pe_fwd(torch::Tensor tokens, torch::Tensor weight,
                            torch::Tensor pe);
 torch::Tensor embed_pe_bwd(torch::Tensor dy, torch::Tensor tokens,
-                           int64_t vocab);
+                           int64_t vocab,
+                           c10::optional<torch::Tensor> out);
 
 std::vector<torch::Tensor> dropout_fwd(torch::Tensor x, double p,
                                        int64_t seed);
@@ -41,8 +48,8 @@ torch::Tensor dropout_bwd(torch::Tensor dy, torch::Tensor mask, double p);
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets,
                                   double batch_size, double label_smoothing);
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets,
-                     torch::Tensor lse, double dloss, double batch_size,
-                     double label_smoothing);
+                     torch::Tensor lse, torch::Tensor dloss,
+                     double batch_size, double label_smoothing);
 
 void adam_fused(torch::Tensor master, torch::Tensor m, torch::Tensor v,
                 torch::Tensor grad, torch::Tensor param, double lr,
@@ -53,16 +60,31 @@ std::vector<int64_t> accuracy(torch::Tensor logits, torch::Tensor targets);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_nt", &gemm_nt, "C[M,N] = A[M,K] @ W[N,K]^T + bias (epilogue: 0=none 1=relu)");
+  m.def("gemm_nn", &gemm_nn, "C[M,N] = A[M,K] @ B[K,N] (dX)",
+        pybind11::arg("a"), pybind11::arg("b"),
+        pybind11::arg("out") = pybind11::none());
+  m.def("gemm_tn", &gemm_tn, "C[M,N] = A[K,M]^T @ B[K,N] (dW)",
+        pybind11::arg("a"), pybind11::arg("b"),
+        pybind11::arg("out") = pybind11::none());
   m.def("transpose2d", &transpose2d);
-  m.def("colsum", &colsum);
+  m.def("colsum", &colsum, pybind11::arg("a"),
+        pybind11::arg("out") = pybind11::none());
   m.def("relu_bwd", &relu_bwd);
   m.def("smoke_add", &smoke_add);
   m.def("ln_fwd", &ln_fwd);
-  m.def("ln_bwd", &ln_bwd);
+  m.def("ln_bwd", &ln_bwd, pybind11::arg("dy"), pybind11::arg("s"),
+        pybind11::arg("gamma"), pybind11::arg("mean"), pybind11::arg("rstd"),
+        pybind11::arg("dgamma_out") = pybind11::none(),
+        pybind11::arg("dbeta_out") = pybind11::none());
   m.def("attn_fwd", &attn_fwd);
-  m.def("attn_bwd", &attn_bwd);
+  m.def("attn_bwd", &attn_bwd, pybind11::arg("q"), pybind11::arg("k"),
+        pybind11::arg("v"), pybind11::arg("o"), pybind11::arg("dout"),
+        pybind11::arg("lse"), pybind11::arg("kv_pad"), pybind11::arg("causal"),
+        pybind11::arg("scale"), pybind11::arg("mode") = 0);
   m.def("embed_pe_fwd", &embed_pe_fwd);
-  m.def("embed_pe_bwd", &embed_pe_bwd);
+  m.def("embed_pe_bwd", &embed_pe_bwd, pybind11::arg("dy"),
+        pybind11::arg("tokens"), pybind11::arg("vocab"),
+        pybind11::arg("out") = pybind11::none());
   m.def("dropout_fwd", &dropout_fwd);
   m.def("dropout_bwd", &dropout_bwd);
   m.def("ce_fwd", &ce_fwd);

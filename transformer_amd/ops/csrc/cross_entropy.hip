@@ -81,13 +81,17 @@ void ce_fwd_kernel(const short* __restrict__ logits,
 }
 
 // dlogits[j] = dloss/batch * mask_row * (softmax_j - (1-eps)*onehot_j - eps/V)
+// dloss arrives as a device scalar so backward never synchronizes the host.
 __global__ __launch_bounds__(256)
 void ce_bwd_kernel(const short* __restrict__ logits,
                    const long* __restrict__ targets,
-                   const float* __restrict__ lse, short* __restrict__ dlogits,
-                   long R, int V, float scale, float eps_ls) {
+                   const float* __restrict__ lse,
+                   const float* __restrict__ dloss_dev,
+                   short* __restrict__ dlogits,
+                   long R, int V, float inv_batch, float eps_ls) {
   const long row = blockIdx.x;
   if (row >= R) return;
+  const float scale = dloss_dev[0] * inv_batch;
   const short* lrow = logits + row * V;
   short* drow = dlogits + row * V;
   const long tgt = targets[row];
@@ -143,15 +147,18 @@ std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets,
 }
 
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets,
-                     torch::Tensor lse, double dloss, double batch_size,
-                     double label_smoothing) {
+                     torch::Tensor lse, torch::Tensor dloss,
+                     double batch_size, double label_smoothing) {
   const long R = logits.size(0);
   const int V = logits.size(1);
+  TORCH_CHECK(dloss.is_cuda() && dloss.dtype() == torch::kFloat32 &&
+              dloss.numel() == 1, "dloss must be a device fp32 scalar");
   auto dlogits = torch::empty_like(logits);
   auto stream = at::hip::getCurrentHIPStream();
   ce_bwd_kernel<<<R, 256, 0, stream>>>(
       (const short*)logits.data_ptr(), targets.data_ptr<long>(),
-      lse.data_ptr<float>(), (short*)dlogits.data_ptr(), R, V,
-      (float)(dloss / batch_size), (float)label_smoothing);
+      lse.data_ptr<float>(), dloss.data_ptr<float>(),
+      (short*)dlogits.data_ptr(), R, V, (float)(1.0 / batch_size),
+      (float)label_smoothing);
   return dlogits;
 }

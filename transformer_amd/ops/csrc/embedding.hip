@@ -77,7 +77,8 @@ torch::Tensor embed_pe_fwd(torch::Tensor tokens, torch::Tensor weight,
 }
 
 torch::Tensor embed_pe_bwd(torch::Tensor dy, torch::Tensor tokens,
-                           int64_t vocab) {
+                           int64_t vocab,
+                           c10::optional<torch::Tensor> out_opt) {
   const int D = dy.size(-1);
   long n_rows = dy.numel() / D;
   auto dw_f = torch::zeros({vocab, (long)D},
@@ -86,8 +87,15 @@ torch::Tensor embed_pe_bwd(torch::Tensor dy, torch::Tensor tokens,
   embed_pe_bwd_kernel<<<cdiv(n_rows, 4), 256, 0, stream>>>(
       (const short*)dy.data_ptr(), tokens.data_ptr<long>(),
       dw_f.data_ptr<float>(), D, n_rows, sqrtf((float)D));
-  auto dw = torch::empty({vocab, (long)D}, dy.options());
-  long n = dw.numel();
+  torch::Tensor dw;
+  if (out_opt.has_value()) {
+    dw = *out_opt;
+    TORCH_CHECK(dw.is_cuda() && dw.dtype() == torch::kBFloat16 &&
+                dw.is_contiguous() && dw.numel() == vocab * (long)D);
+  } else {
+    dw = torch::empty({vocab, (long)D}, dy.options());
+  }
+  long n = vocab * (long)D;
   cast_f32_bf16_kernel<<<(n + 1023) / 1024, 1024, 0, stream>>>(
       dw_f.data_ptr<float>(), (short*)dw.data_ptr(), n);
   return dw;

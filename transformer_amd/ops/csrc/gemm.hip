@@ -1,12 +1,18 @@
-// MFMA bf16 GEMM (NT layout) + small helper kernels.
+// MFMA bf16 GEMM (NT / NN / TN layouts) + small helper kernels.
 //
 // Implements SURVEY.md §2.3 K1/K7/K8/K12: every linear layer of the model
 // (packed QKV, attention out-proj, FFN with fused ReLU, the V≈32k logits
-// head) runs through gemm_nt:  C[M,N] = A[M,K] @ W[N,K]^T (+bias)(+ReLU),
-// bf16 inputs, fp32 MFMA accumulate, bf16 out.
+// head) runs through these kernels:
+//   gemm_nt: C[M,N] = A[M,K] @ B[N,K]^T (+bias)(+ReLU)   — forward
+//   gemm_nn: C[M,N] = A[M,K] @ B[K,N]                    — dX = dY @ W
+//   gemm_tn: C[M,N] = A[K,M]^T @ B[K,N]                  — dW = dY^T @ X
+// bf16 inputs, fp32 MFMA accumulate, bf16 out.  The NN/TN variants read the
+// "transposed" operand through an LDS-bounce staging (global->linear scratch
+// coalesced, scratch->swizzled tile conflict-free), so linear backward needs
+// NO physical transpose kernels or extra HBM round-trips.
 //
 // CDNA4 design (see /opt/skills guides): 128x128 output tile per 4-wave
-// (256-thread) workgroup, BK=64 K-steps double-staged through LDS with an
+// (256-thread) workgroup, BK=64 K-steps staged through LDS with an
 // XOR bank-swizzle (byte ^= (row&7)<<4) so ds_read_b128 fragment reads are
 // ≤2-way conflicting; mfma_f32_16x16x32_bf16 with 4x4 fragments per wave
 // (64x64 per wave); bias/ReLU fused in the epilogue; XCD-aware workgroup
@@ -22,6 +28,9 @@ typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 #define BN 128
 #define BK 64
 #define NTHREADS 256
+// scratch row stride (elems) for transposed staging: 68 dwords per row keeps
+// rows 16B-aligned and spreads the step-2 column reads over the banks.
+#define SCR_LD 136
 
 // LDS tiles are [128 rows][64 cols] bf16 = 128 B per row.  Byte-offset
 // swizzle spreads the 16-lane ds_read_b128 groups over 8 slots (T2 recipe).
@@ -71,6 +80,62 @@ DEV_INLINE void stage_tile(const short* __restrict__ g, int ldg, int nrows,
   }
 }
 
+// Transposed staging, step 1 (global -> linear scratch, coalesced):
+// scratch[c][r] = g[c*ldg + r] for c<64 contraction rows, r<128 tile rows.
+// g points at the (k0, tilerow0) corner of the region.
+template <bool ALIGNED>
+DEV_INLINE void stage_tr_scr(const short* __restrict__ g, int ldg, int nrows,
+                             int ncols, short* scr) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    int idx = p * NTHREADS + t;   // 1024 chunks of 8 elems
+    int c = idx >> 4;             // contraction row 0..63
+    int r8 = (idx & 15) << 3;     // tile-row 0..120
+    s16x4 lo = {0, 0, 0, 0}, hi = {0, 0, 0, 0};
+    if (c < ncols) {
+      const short* src = g + (long)c * ldg + r8;
+      if (ALIGNED && r8 + 8 <= nrows) {
+        s16x8 vv = *(const s16x8*)src;
+        lo = {vv[0], vv[1], vv[2], vv[3]};
+        hi = {vv[4], vv[5], vv[6], vv[7]};
+      } else if ((ldg & 1) == 0 && r8 + 8 <= nrows) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          s16x2 pr = *(const s16x2*)(src + 2 * j);
+          if (j < 2) { lo[2 * j] = pr[0]; lo[2 * j + 1] = pr[1]; }
+          else { hi[2 * (j - 2)] = pr[0]; hi[2 * (j - 2) + 1] = pr[1]; }
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          if (r8 + j < nrows) lo[j] = src[j];
+          if (r8 + 4 + j < nrows) hi[j] = src[4 + j];
+        }
+      }
+    }
+    *(s16x4*)&scr[c * SCR_LD + r8] = lo;
+    *(s16x4*)&scr[c * SCR_LD + r8 + 4] = hi;
+  }
+}
+
+// Transposed staging, step 2 (scratch -> swizzled tile): tile[r][c] =
+// scr[c][r].  Per wave-instruction: rows rbase..rbase+7 (j = lane&7) x
+// column-pairs cbase..cbase+7 (i = lane>>3) as b32 writes — conflict-free
+// writes, 2-way scratch reads (SCR_LD = 68 dwords spreads 8c+r/2).
+DEV_INLINE void stage_tr_tile(const short* __restrict__ scr, short* lds) {
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int j = lane & 7, i = lane >> 3;
+#pragma unroll
+  for (int p = 0; p < 16; ++p) {
+    int task = p * 4 + wid;            // 64 (rbase, cbase) wave-tasks
+    int r = ((task & 15) << 3) + j;    // tile row
+    int c2 = ((task >> 4) << 3) + i;   // column pair index (of 32)
+    s16x2 v = {scr[(2 * c2) * SCR_LD + r], scr[(2 * c2 + 1) * SCR_LD + r]};
+    *(s16x2*)((char*)lds + lds_off(r, 4 * c2)) = v;
+  }
+}
+
 DEV_INLINE bf16x8 read_frag(const short* lds, int row, int kbase) {
   int off = lds_off(row, kbase * 2);
   s16x8 v0 = *(const s16x8*)((const char*)lds + off);
@@ -103,17 +168,24 @@ DEV_INLINE void stage_tile_glds(const short* __restrict__ g, int ldg,
 // bias/epilogue applied by gemm_finalize_kernel after all slices land.
 // Split-K keeps the chip full on deep-K small-MN GEMMs (the dW shapes:
 // e.g. 512x512xK=16384 is only 16 workgroups unsplit).
-template <int EPILOGUE, bool ALIGNED_A, bool ALIGNED_B, bool SPLITK>
+//
+// TA/TB select transposed staging for A/B: the operand is stored
+// contraction-major (element (tile_row r, contraction c) at g[c*ld + r]).
+template <int EPILOGUE, bool TA, bool TB, bool ALIGNED_A, bool ALIGNED_B,
+          bool SPLITK>
 __global__ __launch_bounds__(NTHREADS)
-void gemm_nt_kernel(const short* __restrict__ A, const short* __restrict__ W,
-                    const short* __restrict__ bias, short* __restrict__ C,
-                    float* __restrict__ CW, int M, int N, int K, int has_bias,
-                    int nbm, int nbn, int k_per_slice) {
-  // Single-buffered 32 KiB LDS: measured FASTER than a 64 KiB double
-  // buffer here — at 3-4 blocks/CU, block-level overlap already hides the
-  // glds latency, and doubling LDS halves occupancy (guide §5 regime note).
-  __shared__ short a_lds2[1][BM * BK];
-  __shared__ short b_lds2[1][BN * BK];
+void gemm_kernel(const short* __restrict__ A, const short* __restrict__ B,
+                 const short* __restrict__ bias, short* __restrict__ C,
+                 float* __restrict__ CW, int M, int N, int K, int lda, int ldb,
+                 int has_bias, int nbm, int nbn, int k_per_slice) {
+  // Dynamic LDS: [a_tile BM*BK][b_tile BN*BK][scratch iff TA||TB].
+  // Single-buffered tiles: measured FASTER than a 64 KiB double buffer at
+  // this occupancy — block-level overlap already hides the glds latency
+  // (guide §5 regime note).
+  extern __shared__ short smem[];
+  short* a_lds = smem;
+  short* b_lds = smem + BM * BK;
+  short* scr = smem + BM * BK + BN * BK;  // only sized when TA||TB
 
   // XCD-aware bijective remap (guide T1): contiguous grid chunk per XCD.
   int nwg = nbm * nbn;
@@ -140,25 +212,22 @@ void gemm_nt_kernel(const short* __restrict__ A, const short* __restrict__ W,
 
   const int mrows = min(BM, M - bm0);
   const int nrows = min(BN, N - bn0);
-  // glds fast path: full output tile and K a multiple of BK (no guards);
-  // block-uniform condition.
-  const bool glds = ALIGNED_A && ALIGNED_B && mrows == BM && nrows == BN &&
-                    (K % BK) == 0;
+  // glds fast path (NT only): full output tile and K a multiple of BK.
+  const bool glds = !TA && !TB && ALIGNED_A && ALIGNED_B && mrows == BM &&
+                    nrows == BN && (K % BK) == 0;
   int k_lo = 0, k_hi = K;
   if (SPLITK) {
     k_lo = blockIdx.y * k_per_slice;
     k_hi = min(K, k_lo + k_per_slice);
   }
 
-  const long Aoff = (long)bm0 * K, Woff = (long)bn0 * K;
-
-#define MFMA_TILE(a_lds, b_lds)                                            \
+#define MFMA_TILE(a_lds_, b_lds_)                                          \
   _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) {                       \
     bf16x8 af[4], bf_[4];                                                  \
     _Pragma("unroll") for (int i = 0; i < 4; ++i)                          \
-        af[i] = read_frag(a_lds, wr + i * 16 + fr, ks * 32 + kg * 8);      \
+        af[i] = read_frag(a_lds_, wr + i * 16 + fr, ks * 32 + kg * 8);     \
     _Pragma("unroll") for (int j = 0; j < 4; ++j)                          \
-        bf_[j] = read_frag(b_lds, wc + j * 16 + fr, ks * 32 + kg * 8);     \
+        bf_[j] = read_frag(b_lds_, wc + j * 16 + fr, ks * 32 + kg * 8);    \
     _Pragma("unroll") for (int i = 0; i < 4; ++i)                          \
         _Pragma("unroll") for (int j = 0; j < 4; ++j)                      \
             acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(           \
@@ -167,19 +236,30 @@ void gemm_nt_kernel(const short* __restrict__ A, const short* __restrict__ W,
 
   if (glds) {
     for (int k0 = k_lo; k0 < k_hi; k0 += BK) {
-      stage_tile_glds(A + Aoff + k0, K, a_lds2[0]);
-      stage_tile_glds(W + Woff + k0, K, b_lds2[0]);
+      stage_tile_glds(A + (long)bm0 * lda + k0, lda, a_lds);
+      stage_tile_glds(B + (long)bn0 * ldb + k0, ldb, b_lds);
       __syncthreads();              // drains the LDS-DMA queue (vmcnt 0)
-      MFMA_TILE(a_lds2[0], b_lds2[0]);
+      MFMA_TILE(a_lds, b_lds);
       __syncthreads();
     }
   } else {
-    short* a_lds = a_lds2[0];
-    short* b_lds = b_lds2[0];
     for (int k0 = k_lo; k0 < k_hi; k0 += BK) {
       int kc = min(BK, K - k0);
-      stage_tile<ALIGNED_A>(A + Aoff + k0, K, mrows, kc, a_lds);
-      stage_tile<ALIGNED_B>(W + Woff + k0, K, nrows, kc, b_lds);
+      if (!TA)
+        stage_tile<ALIGNED_A>(A + (long)bm0 * lda + k0, lda, mrows, kc, a_lds);
+      if (!TB)
+        stage_tile<ALIGNED_B>(B + (long)bn0 * ldb + k0, ldb, nrows, kc, b_lds);
+      if (TA) {
+        stage_tr_scr<ALIGNED_A>(A + (long)k0 * lda + bm0, lda, mrows, kc, scr);
+        __syncthreads();
+        stage_tr_tile(scr, a_lds);
+      }
+      if (TB) {
+        if (TA) __syncthreads();  // a_lds staging read scr; wait before reuse
+        stage_tr_scr<ALIGNED_B>(B + (long)k0 * ldb + bn0, ldb, nrows, kc, scr);
+        __syncthreads();
+        stage_tr_tile(scr, b_lds);
+      }
       __syncthreads();
       MFMA_TILE(a_lds, b_lds);
       __syncthreads();
@@ -188,8 +268,6 @@ void gemm_nt_kernel(const short* __restrict__ A, const short* __restrict__ W,
 #undef MFMA_TILE
 
   // Epilogue: C/D lane map is col = lane&15, row = (lane>>4)*4 + r.
-  // Pack 2 bf16 (adjacent rows... cols are per-lane) -> scalar stores; the
-  // 16 lanes of a group cover 16 consecutive cols so stores coalesce.
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     int grow_base = bm0 + wr + i * 16 + kg * 4;
@@ -230,8 +308,7 @@ __global__ void gemm_finalize_kernel(const float* __restrict__ cw,
 
 // ---------------------------------------------------------------------------
 // transpose2d: [M,N] bf16 -> [N,M].  64x64 tiles, padded LDS, short4 IO.
-// Used by the backward GEMMs (dX = dY·W, dW = dY^T·X) until dedicated
-// TN/NN kernel variants land.
+// (No longer on the linear-backward hot path — kept as a utility op.)
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256)
 void transpose2d_kernel(const short* __restrict__ in, short* __restrict__ out,
@@ -339,17 +416,26 @@ __global__ void smoke_add_kernel(const short* a, const short* b, short* c,
                   t.is_contiguous(),                                          \
               #t " must be contiguous 2-D bf16 on GPU")
 
-torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
-                      int64_t epilogue) {
-  CHECK_BF16_2D(a);
-  CHECK_BF16_2D(w);
-  const int M = a.size(0), K = a.size(1), N = w.size(0);
-  TORCH_CHECK(w.size(1) == K, "gemm_nt: K mismatch");
+namespace {
+
+// Shared launcher: out dims MxN, contraction K; lda/ldb per TA/TB semantics.
+template <bool TA, bool TB>
+torch::Tensor gemm_launch(torch::Tensor a, torch::Tensor b, torch::Tensor bias,
+                          int64_t epilogue, int M, int N, int K, int lda,
+                          int ldb, c10::optional<torch::Tensor> out_opt) {
   bool has_bias = bias.defined() && bias.numel() > 0;
   if (has_bias)
     TORCH_CHECK(bias.is_cuda() && bias.dtype() == torch::kBFloat16 &&
                 bias.numel() == N, "bias must be bf16[N]");
-  auto c = torch::empty({M, N}, a.options());
+  torch::Tensor c;
+  if (out_opt.has_value()) {
+    c = *out_opt;
+    TORCH_CHECK(c.is_cuda() && c.dtype() == torch::kBFloat16 &&
+                c.is_contiguous() && c.numel() == (long)M * N,
+                "out must be contiguous bf16 with M*N elements");
+  } else {
+    c = torch::empty({M, N}, a.options());
+  }
   int nbm = cdiv(M, BM), nbn = cdiv(N, BN);
   int nwg = nbm * nbn;
   // split-K when the unsplit grid underfills the 256-CU chip and K is deep
@@ -362,29 +448,37 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
   sk = cdiv(K, k_per_slice);
   dim3 grid(nwg, sk);
   auto stream = at::hip::getCurrentHIPStream();
-  bool al = (K % 8 == 0);
+  bool ala = (lda % 8 == 0);
+  bool alb = (ldb % 8 == 0);
+  size_t smem = (BM * BK + BN * BK + ((TA || TB) ? 64 * SCR_LD : 0)) *
+                sizeof(short);
   torch::Tensor cw;
   float* cwp = nullptr;
   if (sk > 1) {
     cw = torch::zeros({M, N}, a.options().dtype(torch::kFloat32));
     cwp = cw.data_ptr<float>();
   }
-  auto launch = [&](auto epi, auto ala, auto splitk) {
-    gemm_nt_kernel<decltype(epi)::value, decltype(ala)::value,
-                   decltype(ala)::value, decltype(splitk)::value>
-        <<<grid, NTHREADS, 0, stream>>>(
-            (const short*)a.data_ptr(), (const short*)w.data_ptr(),
+  auto launch = [&](auto epi, auto aa, auto ab, auto splitk) {
+    gemm_kernel<decltype(epi)::value, TA, TB, decltype(aa)::value,
+                decltype(ab)::value, decltype(splitk)::value>
+        <<<grid, NTHREADS, smem, stream>>>(
+            (const short*)a.data_ptr(), (const short*)b.data_ptr(),
             has_bias ? (const short*)bias.data_ptr() : nullptr,
-            (short*)c.data_ptr(), cwp, M, N, K, has_bias, nbm, nbn,
+            (short*)c.data_ptr(), cwp, M, N, K, lda, ldb, has_bias, nbm, nbn,
             k_per_slice);
   };
   using T = std::true_type;
   using F = std::false_type;
   using E0 = std::integral_constant<int, 0>;
   using E1 = std::integral_constant<int, 1>;
+  auto dis_al = [&](auto epi, auto splitk) {
+    if (ala) { if (alb) launch(epi, T{}, T{}, splitk);
+               else launch(epi, T{}, F{}, splitk); }
+    else { if (alb) launch(epi, F{}, T{}, splitk);
+           else launch(epi, F{}, F{}, splitk); }
+  };
   if (sk > 1) {
-    if (epilogue == 1) { if (al) launch(E1{}, T{}, T{}); else launch(E1{}, F{}, T{}); }
-    else { if (al) launch(E0{}, T{}, T{}); else launch(E0{}, F{}, T{}); }
+    if (epilogue == 1) dis_al(E1{}, T{}); else dis_al(E0{}, T{});
     long mn = (long)M * N;
     auto fin = [&](auto epi) {
       gemm_finalize_kernel<decltype(epi)::value>
@@ -394,10 +488,44 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
     };
     if (epilogue == 1) fin(E1{}); else fin(E0{});
   } else {
-    if (epilogue == 1) { if (al) launch(E1{}, T{}, F{}); else launch(E1{}, F{}, F{}); }
-    else { if (al) launch(E0{}, T{}, F{}); else launch(E0{}, F{}, F{}); }
+    if (epilogue == 1) dis_al(E1{}, F{}); else dis_al(E0{}, F{});
   }
   return c;
+}
+
+}  // namespace
+
+// C[M,N] = A[M,K] @ W[N,K]^T (+bias)(+ReLU) — forward layout.
+torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
+                      int64_t epilogue) {
+  CHECK_BF16_2D(a);
+  CHECK_BF16_2D(w);
+  const int M = a.size(0), K = a.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "gemm_nt: K mismatch");
+  return gemm_launch<false, false>(a, w, bias, epilogue, M, N, K, K, K,
+                                   c10::nullopt);
+}
+
+// C[M,N] = A[M,K] @ B[K,N] — dX = dY @ W without transposing W.
+torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b,
+                      c10::optional<torch::Tensor> out) {
+  CHECK_BF16_2D(a);
+  CHECK_BF16_2D(b);
+  const int M = a.size(0), K = a.size(1), N = b.size(1);
+  TORCH_CHECK(b.size(0) == K, "gemm_nn: K mismatch");
+  return gemm_launch<false, true>(a, b, torch::Tensor(), 0, M, N, K, K, N,
+                                  out);
+}
+
+// C[M,N] = A[K,M]^T @ B[K,N] — dW = dY^T @ X without physical transposes.
+torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b,
+                      c10::optional<torch::Tensor> out) {
+  CHECK_BF16_2D(a);
+  CHECK_BF16_2D(b);
+  const int K = a.size(0), M = a.size(1), N = b.size(1);
+  TORCH_CHECK(b.size(0) == K, "gemm_tn: contraction mismatch");
+  return gemm_launch<true, true>(a, b, torch::Tensor(), 0, M, N, K, M, N,
+                                 out);
 }
 
 torch::Tensor transpose2d(torch::Tensor a) {
@@ -411,11 +539,18 @@ torch::Tensor transpose2d(torch::Tensor a) {
   return out;
 }
 
-torch::Tensor colsum(torch::Tensor a) {
+torch::Tensor colsum(torch::Tensor a, c10::optional<torch::Tensor> out_opt) {
   CHECK_BF16_2D(a);
   int M = a.size(0), N = a.size(1);
   auto acc = torch::zeros({N}, a.options().dtype(torch::kFloat32));
-  auto out = torch::empty({N}, a.options());
+  torch::Tensor out;
+  if (out_opt.has_value()) {
+    out = *out_opt;
+    TORCH_CHECK(out.is_cuda() && out.dtype() == torch::kBFloat16 &&
+                out.is_contiguous() && out.numel() == N);
+  } else {
+    out = torch::empty({N}, a.options());
+  }
   auto stream = at::hip::getCurrentHIPStream();
   dim3 grid(cdiv(N, 256), cdiv(M, COLSUM_ROWS));
   colsum_part_kernel<<<grid, 256, 0, stream>>>(

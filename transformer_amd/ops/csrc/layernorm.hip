@@ -165,11 +165,21 @@ std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor res,
 
 std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor s,
                                   torch::Tensor gamma, torch::Tensor mean,
-                                  torch::Tensor rstd) {
+                                  torch::Tensor rstd,
+                                  c10::optional<torch::Tensor> dgamma_out,
+                                  c10::optional<torch::Tensor> dbeta_out) {
   const int R = dy.size(0), D = dy.size(1);
   auto dx = torch::empty_like(dy);
-  auto dgamma = torch::empty({D}, dy.options());
-  auto dbeta = torch::empty({D}, dy.options());
+  auto dest = [&](c10::optional<torch::Tensor>& o) {
+    if (o.has_value()) {
+      TORCH_CHECK(o->is_cuda() && o->dtype() == torch::kBFloat16 &&
+                  o->is_contiguous() && o->numel() == D);
+      return *o;
+    }
+    return torch::empty({D}, dy.options());
+  };
+  auto dgamma = dest(dgamma_out);
+  auto dbeta = dest(dbeta_out);
   auto stream = at::hip::getCurrentHIPStream();
   ln_bwd_kernel<<<cdiv(R, 4), 256, 0, stream>>>(
       (const short*)dy.data_ptr(), (const short*)s.data_ptr(),

@@ -18,6 +18,38 @@ from . import ext
 
 
 # ---------------------------------------------------------------------------
+# Flat-gradient fast path.  When a parameter carries a `_flat_grad` view
+# (runtime/optimizer.py FlatParams), the op wrappers below hide it from
+# autograd entirely: the backward kernel writes the weight gradient straight
+# into the flat buffer slice (no per-parameter AccumulateGrad kernel, no
+# clone — autograd ALWAYS clones view gradients, so returning the view would
+# cost an extra copy per parameter per step).  The DP bucket manager
+# subscribes here for gradient-readiness instead of post-accumulate hooks.
+# ---------------------------------------------------------------------------
+
+_GRAD_READY_CB = None
+
+
+def set_grad_ready_callback(fn):
+    """fn(param) is invoked (backward order) right after a parameter's
+    gradient kernels are enqueued on the compute stream.  Used by
+    parallel/ddp.py to launch bucket all-reduces."""
+    global _GRAD_READY_CB
+    _GRAD_READY_CB = fn
+
+
+def _grad_ready(*params):
+    if _GRAD_READY_CB is not None:
+        for p in params:
+            if p is not None:
+                _GRAD_READY_CB(p)
+
+
+def _flat(p):
+    return getattr(p, "_flat_grad", None) if p is not None else None
+
+
+# ---------------------------------------------------------------------------
 # Linear: y = x @ W^T + b, optional fused ReLU epilogue (K1/K7/K8/K12).
 # ---------------------------------------------------------------------------
 
@@ -40,15 +72,44 @@ class _LinearFn(torch.autograd.Function):
         dy = dy.contiguous()
         if ctx.activation == "relu":
             dy = E.relu_bwd(dy, y)  # dz = dy * (y > 0)
-        # dx[M,K] = dy[M,N] @ W[N,K]  -> NT gemm with B = W^T
-        wt = E.transpose2d(w)
-        dx = E.gemm_nt(dy, wt, torch.Tensor(), 0)
-        # dW[N,K] = dy^T[N,M] @ x[M,K] -> NT gemm with A = dy^T, B = x^T
-        dyt = E.transpose2d(dy)
-        xt = E.transpose2d(x)
-        dw = E.gemm_nt(dyt, xt, torch.Tensor(), 0)
+        dx = E.gemm_nn(dy, w)       # dX[M,K] = dY[M,N] @ W[N,K]
+        dw = E.gemm_tn(dy, x)       # dW[N,K] = dY^T @ X
         db = E.colsum(dy) if ctx.has_bias else None
         return dx, dw, db, None
+
+
+class _LinearFlatFn(torch.autograd.Function):
+    """Linear whose weight/bias grads go straight into flat-buffer views.
+
+    w/b arrive hidden inside a list so autograd treats only x as
+    differentiable; backward writes dW/db into `p._flat_grad` (the kernels'
+    `out` argument) and fires the DP readiness callback."""
+
+    @staticmethod
+    def forward(ctx, x, wb, activation):
+        E = ext()
+        w, b = wb
+        y = E.gemm_nt(x, w, b if b is not None else torch.Tensor(),
+                      1 if activation == "relu" else 0)
+        ctx.activation = activation
+        ctx.wb = wb
+        ctx.save_for_backward(x, y if activation == "relu" else torch.Tensor())
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        E = ext()
+        x, y = ctx.saved_tensors
+        w, b = ctx.wb
+        dy = dy.contiguous()
+        if ctx.activation == "relu":
+            dy = E.relu_bwd(dy, y)
+        dx = E.gemm_nn(dy, w)
+        E.gemm_tn(dy, x, _flat(w).view(w.shape[0], -1))
+        if b is not None:
+            E.colsum(dy, _flat(b).view(-1))
+        _grad_ready(w, b)
+        return dx, None, None
 
 
 def linear(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor | None = None,
@@ -56,7 +117,11 @@ def linear(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor | None = None,
     """y = x @ W^T + b (+ReLU). x may have leading batch dims."""
     if x.is_cuda:
         shp = x.shape
-        y = _LinearFn.apply(x.reshape(-1, shp[-1]).contiguous(), w, b, activation)
+        x2 = x.reshape(-1, shp[-1]).contiguous()
+        if _flat(w) is not None:
+            y = _LinearFlatFn.apply(x2, [w, b], activation)
+        else:
+            y = _LinearFn.apply(x2, w, b, activation)
         return y.view(*shp[:-1], w.shape[0])
     y = torch.nn.functional.linear(x, w, b)
     if activation == "relu":
@@ -88,8 +153,85 @@ class _AttentionFn(torch.autograd.Function):
         E = ext()
         q, k, v, o, lse, kv_pad = ctx.saved_tensors
         dq, dk, dv = E.attn_bwd(q, k, v, o, do.contiguous(), lse, kv_pad,
-                                ctx.causal, ctx.scale)
+                                ctx.causal, ctx.scale, 0)
         return dq, dk, dv, None, None, None
+
+
+class _SelfAttnPackedFn(torch.autograd.Function):
+    """Attention straight on the packed (B,S,3,H,dh) QKV tensor: the kernels
+    read the q/k/v slots through strides (no contiguous copies) and backward
+    writes one packed dQKV (no cat/stack kernels)."""
+
+    @staticmethod
+    def forward(ctx, qkv, kv_pad, causal, scale):
+        E = ext()
+        q, k, v = qkv.unbind(dim=2)  # strided views, never materialized
+        o, lse = E.attn_fwd(q, k, v,
+                            kv_pad if kv_pad is not None else torch.Tensor(),
+                            causal, scale)
+        ctx.causal = causal
+        ctx.scale = scale
+        ctx.save_for_backward(qkv, o, lse,
+                              kv_pad if kv_pad is not None else torch.Tensor())
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        E = ext()
+        qkv, o, lse, kv_pad = ctx.saved_tensors
+        q, k, v = qkv.unbind(dim=2)
+        (dqkv,) = E.attn_bwd(q, k, v, o, do.contiguous(), lse, kv_pad,
+                             ctx.causal, ctx.scale, 1)
+        return dqkv, None, None, None
+
+
+class _CrossAttnPackedFn(torch.autograd.Function):
+    """Cross attention: q (B,Sq,H,dh) + packed kv (B,Sk,2,H,dh)."""
+
+    @staticmethod
+    def forward(ctx, q, kv, kv_pad, scale):
+        E = ext()
+        k, v = kv.unbind(dim=2)
+        o, lse = E.attn_fwd(q, k, v,
+                            kv_pad if kv_pad is not None else torch.Tensor(),
+                            False, scale)
+        ctx.scale = scale
+        ctx.save_for_backward(q, kv, o, lse,
+                              kv_pad if kv_pad is not None else torch.Tensor())
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        E = ext()
+        q, kv, o, lse, kv_pad = ctx.saved_tensors
+        k, v = kv.unbind(dim=2)
+        dq, dkv = E.attn_bwd(q, k, v, o, do.contiguous(), lse, kv_pad,
+                             False, ctx.scale, 2)
+        return dq, dkv, None, None
+
+
+def self_attention(qkv, kv_pad=None, causal=False, return_weights=False):
+    """qkv: (B,S,3,H,dh).  GPU path never materializes q/k/v copies."""
+    B, S, _, H, dh = qkv.shape
+    scale = 1.0 / math.sqrt(dh)
+    if qkv.is_cuda and not return_weights:
+        kp = kv_pad.to(torch.uint8).contiguous() if kv_pad is not None else None
+        return _SelfAttnPackedFn.apply(qkv, kp, causal, scale)
+    q, k, v = qkv.unbind(dim=2)
+    return fused_attention(q, k, v, kv_pad=kv_pad, causal=causal,
+                           return_weights=return_weights)
+
+
+def cross_attention(q, kv, kv_pad=None, return_weights=False):
+    """q: (B,Sq,H,dh); kv: (B,Sk,2,H,dh)."""
+    dh = q.shape[-1]
+    scale = 1.0 / math.sqrt(dh)
+    if q.is_cuda and not return_weights:
+        kp = kv_pad.to(torch.uint8).contiguous() if kv_pad is not None else None
+        return _CrossAttnPackedFn.apply(q.contiguous(), kv, kp, scale)
+    k, v = kv.unbind(dim=2)
+    return fused_attention(q, k, v, kv_pad=kv_pad, causal=False,
+                           return_weights=return_weights)
 
 
 def fused_attention(q, k, v, kv_pad=None, causal=False, return_weights=False):
@@ -145,12 +287,37 @@ class _ResidualLNFn(torch.autograd.Function):
         return dx, dx, dgamma, dbeta, None
 
 
+class _ResidualLNFlatFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, res, gb, eps):
+        E = ext()
+        gamma, beta = gb
+        y, s, mean, rstd = E.ln_fwd(x, res, gamma, beta, eps)
+        ctx.gb_params = gb
+        ctx.save_for_backward(s, gamma, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        E = ext()
+        s, gamma, mean, rstd = ctx.saved_tensors
+        g, b = ctx.gb_params
+        dx, _, _ = E.ln_bwd(dy.contiguous(), s, gamma, mean, rstd,
+                            _flat(g).view(-1), _flat(b).view(-1))
+        _grad_ready(g, b)
+        return dx, dx, None, None
+
+
 def residual_layernorm(x, res, gamma, beta, eps: float = 1e-6):
     if x.is_cuda:
         shp = x.shape
         d = shp[-1]
-        y = _ResidualLNFn.apply(x.reshape(-1, d).contiguous(),
-                                res.reshape(-1, d).contiguous(), gamma, beta, eps)
+        x2 = x.reshape(-1, d).contiguous()
+        r2 = res.reshape(-1, d).contiguous()
+        if _flat(gamma) is not None:
+            y = _ResidualLNFlatFn.apply(x2, r2, [gamma, beta], eps)
+        else:
+            y = _ResidualLNFn.apply(x2, r2, gamma, beta, eps)
         return y.view(shp)
     if x.dtype in (torch.bfloat16, torch.float16):
         return R.residual_layernorm(x.float(), res.float(), gamma.float(),
@@ -169,7 +336,6 @@ class _EmbedPEFn(torch.autograd.Function):
         y = E.embed_pe_fwd(tokens, weight, pe)
         ctx.vocab = weight.shape[0]
         ctx.save_for_backward(tokens)
-        ctx.wdtype = weight.dtype
         return y
 
     @staticmethod
@@ -180,9 +346,47 @@ class _EmbedPEFn(torch.autograd.Function):
         return None, dw, None
 
 
+class _EmbedPEFlatFn(torch.autograd.Function):
+    """Embedding fwd with dW scatter-add cast straight into the flat view.
+    Output still requires grad (x flows onward), so a dummy differentiable
+    input threads the graph."""
+
+    @staticmethod
+    def forward(ctx, marker, tokens, wref, pe):
+        E = ext()
+        (weight,) = wref
+        y = E.embed_pe_fwd(tokens, weight, pe)
+        ctx.vocab = weight.shape[0]
+        ctx.wref = wref
+        ctx.save_for_backward(tokens)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        E = ext()
+        (tokens,) = ctx.saved_tensors
+        (weight,) = ctx.wref
+        E.embed_pe_bwd(dy.contiguous(), tokens, ctx.vocab,
+                       _flat(weight).view(-1))
+        _grad_ready(weight)
+        return None, None, None, None
+
+
+def embedding_scale_pe_at(tokens, weight, pe, pos: int):
+    """Embedding + PE for an incremental decode step: positions start at
+    `pos` instead of 0 (KV-cache path)."""
+    return embedding_scale_pe(tokens, weight,
+                              pe[pos:pos + tokens.shape[1]].contiguous())
+
+
 def embedding_scale_pe(tokens, weight, pe):
     """tokens (B,S) int; weight (V,d); pe (P,d) fp32 table, P >= S."""
     if weight.is_cuda:
+        if _flat(weight) is not None and torch.is_grad_enabled():
+            marker = torch.empty(0, device=weight.device,
+                                 dtype=weight.dtype, requires_grad=True)
+            return _EmbedPEFlatFn.apply(marker, tokens.contiguous(),
+                                        [weight], pe)
         return _EmbedPEFn.apply(tokens.contiguous(), weight, pe)
     if weight.dtype in (torch.bfloat16, torch.float16):
         return R.embedding_scale_pe(tokens, weight.float(), pe[None].float()).to(weight.dtype)
@@ -237,7 +441,8 @@ class _CrossEntropyFn(torch.autograd.Function):
     def backward(ctx, dloss):
         E = ext()
         logits, targets, lse = ctx.saved_tensors
-        dlogits = E.ce_bwd(logits, targets, lse, float(dloss.item()),
+        dlogits = E.ce_bwd(logits, targets, lse,
+                           dloss.to(torch.float32).reshape(1).contiguous(),
                            ctx.batch_size, ctx.ls)
         return dlogits, None, None, None
 

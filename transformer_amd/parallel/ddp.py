@@ -84,11 +84,27 @@ class BucketedDataParallel:
         for bi, b in enumerate(self.buckets):
             for p in b["params"]:
                 self._param_bucket[id(p)] = bi
+        # readiness: GPU flat path fires the ops-layer callback (weights are
+        # hidden from autograd there); CPU/plain path fires post-accumulate
+        # hooks.  A param triggers exactly once per step on exactly one path.
         self._hooks = [
             p.register_post_accumulate_grad_hook(self._make_hook(p))
             for p in flat.params
         ]
+        from ..ops import functional as _F
+        _F.set_grad_ready_callback(self._on_param_ready)
         self._reset_step()
+
+    def _on_param_ready(self, p):
+        if self.world <= 1:
+            return
+        bi = self._param_bucket.get(id(p))
+        if bi is None:
+            return
+        b = self.buckets[bi]
+        b["pending"] -= 1
+        if b["pending"] == 0:
+            self._launch(b)
 
     def _push_bucket(self, plist):
         los = [off for _, off in plist]

@@ -107,15 +107,13 @@ class Train:
         tokens = [src_start] + self.src_tokenizer.encode(input_sentence) + [src_end]
         encoder_input = torch.tensor([tokens], dtype=torch.int64,
                                      device=self.device)
-        output = torch.tensor([[tgt_start]], dtype=torch.int64,
-                              device=self.device)
-        for _ in range(self.max_decode_len):
-            predictions, _ = self.transformer((encoder_input, output),
-                                              training=False)
-            predicted_id = ops.argmax_lastdim(predictions[:, -1:, :].float())
-            output = torch.cat([output, predicted_id.view(1, 1)], dim=-1)
-            if int(predicted_id) == tgt_end:
-                break
+        # KV-cached greedy decode: encoder runs once, each step is O(1) in
+        # prefix length (SURVEY.md §3.3 — vs the reference's full re-run per
+        # token, train.py:109-118).
+        from ..models.transformer import greedy_decode
+        output = greedy_decode(self.transformer, encoder_input,
+                               tgt_start, tgt_end,
+                               max_len=self.max_decode_len)
         return output.squeeze(0).cpu()
 
     def load_ckpt(self):
