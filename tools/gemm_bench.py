@@ -1,6 +1,11 @@
-"""Microbenchmark gemm_nt at the model's actual shapes (run on GPU box).
+"""Microbenchmark the GEMM paths at the model's actual shapes (GPU box).
 
     python tools/gemm_bench.py [iters]
+
+For each backward shape it times BOTH implementations:
+  dX:  gemm_nn(dy, w)            vs  transpose2d(w) + gemm_nt(dy, wT)
+  dW:  gemm_tn(dy, x)            vs  transpose2d(dy)+transpose2d(x)+gemm_nt
+so the functional layer's dispatch choices are measurement-driven.
 """
 
 import sys
@@ -11,19 +16,45 @@ import torch
 sys.path.insert(0, ".")
 from transformer_amd.ops import ext  # noqa: E402
 
-# (M, N, K, tag) — transformer-base, B=64, S=256 training shapes
-SHAPES = [
-    (16384, 1536, 512, "enc QKV fwd"),
+
+def timeit(fn, iters):
+    for _ in range(3):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+
+# forward shapes: (M, N, K, tag) — transformer-base B=64 S=256 + big-model
+FWD = [
+    (16384, 1536, 512, "QKV fwd"),
     (16384, 512, 512, "attn O fwd"),
     (16384, 2048, 512, "FFN1 fwd"),
     (16384, 512, 2048, "FFN2 fwd"),
     (16320, 32770, 512, "logits fwd"),
-    (16320, 512, 32770, "logits dx (odd K)"),
-    (32770, 512, 16320, "logits dW"),
-    (512, 512, 16384, "attn O dW"),
-    (2048, 512, 16384, "FFN1 dW"),
-    (16384, 512, 1536, "QKV dx"),
-    (1536, 512, 16384, "QKV dW"),
+    (16384, 3072, 1024, "big QKV fwd"),
+    (16384, 4096, 1024, "big FFN1 fwd"),
+]
+
+# dX shapes: dy (M, N_out) @ w (N_out, K_in): (M, N_out, K_in, tag)
+DX = [
+    (16384, 1536, 512, "QKV dx"),
+    (16384, 512, 512, "attn O dx"),
+    (16384, 512, 2048, "FFN1 dx"),
+    (16384, 2048, 512, "FFN2 dx"),
+    (16320, 32770, 512, "logits dx"),
+]
+
+# dW shapes: dy (Mtok, N_out), x (Mtok, K_in): (Mtok, N_out, K_in, tag)
+DW = [
+    (16384, 1536, 512, "QKV dW"),
+    (16384, 512, 512, "attn O dW"),
+    (16384, 2048, 512, "FFN1 dW"),
+    (16384, 512, 2048, "FFN2 dW"),
+    (16320, 32770, 512, "logits dW"),
 ]
 
 
@@ -31,25 +62,38 @@ def main():
     iters = int(sys.argv[1]) if len(sys.argv) > 1 else 20
     E = ext()
     torch.manual_seed(0)
-    print(f"{'tag':18s} {'M':>6s} {'N':>6s} {'K':>6s} {'ms':>8s} {'TF/s':>7s}")
-    total_t, total_f = 0.0, 0.0
-    for (m, n, k, tag) in SHAPES:
+
+    print("== forward: gemm_nt ==")
+    for (m, n, k, tag) in FWD:
         a = torch.randn(m, k, device="cuda", dtype=torch.bfloat16)
         w = torch.randn(n, k, device="cuda", dtype=torch.bfloat16)
         b = torch.randn(n, device="cuda", dtype=torch.bfloat16)
-        for _ in range(3):
-            E.gemm_nt(a, w, b, 0)
-        torch.cuda.synchronize()
-        t0 = time.perf_counter()
-        for _ in range(iters):
-            E.gemm_nt(a, w, b, 0)
-        torch.cuda.synchronize()
-        dt = (time.perf_counter() - t0) / iters
+        dt = timeit(lambda: E.gemm_nt(a, w, b, 0), iters)
         fl = 2.0 * m * n * k
-        print(f"{tag:18s} {m:6d} {n:6d} {k:6d} {dt*1e3:8.3f} {fl/dt/1e12:7.1f}")
-        total_t += dt
-        total_f += fl
-    print(f"{'TOTAL':18s} {'':21s} {total_t*1e3:8.3f} {total_f/total_t/1e12:7.1f}")
+        print(f"{tag:14s} M{m:6d} N{n:6d} K{k:6d} {dt*1e3:8.3f} ms {fl/dt/1e12:7.1f} TF/s")
+
+    print("== dX: gemm_nn vs transpose(w)+gemm_nt ==")
+    for (m, n, k, tag) in DX:
+        dy = torch.randn(m, n, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(n, k, device="cuda", dtype=torch.bfloat16)
+        t_nn = timeit(lambda: E.gemm_nn(dy, w), iters)
+        t_tr = timeit(lambda: E.gemm_nt(dy, E.transpose2d(w), torch.Tensor(), 0), iters)
+        fl = 2.0 * m * n * k
+        win = "nn" if t_nn < t_tr else "tr+nt"
+        print(f"{tag:14s} nn {t_nn*1e3:8.3f} ms ({fl/t_nn/1e12:6.1f} TF) | "
+              f"tr+nt {t_tr*1e3:8.3f} ms ({fl/t_tr/1e12:6.1f} TF)  -> {win}")
+
+    print("== dW: gemm_tn vs transpose(dy)+transpose(x)+gemm_nt ==")
+    for (mt, n, k, tag) in DW:
+        dy = torch.randn(mt, n, device="cuda", dtype=torch.bfloat16)
+        x = torch.randn(mt, k, device="cuda", dtype=torch.bfloat16)
+        t_tn = timeit(lambda: E.gemm_tn(dy, x), iters)
+        t_tr = timeit(lambda: E.gemm_nt(E.transpose2d(dy), E.transpose2d(x),
+                                        torch.Tensor(), 0), iters)
+        fl = 2.0 * mt * n * k
+        win = "tn" if t_tn < t_tr else "tr+nt"
+        print(f"{tag:14s} tn {t_tn*1e3:8.3f} ms ({fl/t_tn/1e12:6.1f} TF) | "
+              f"tr+nt {t_tr*1e3:8.3f} ms ({fl/t_tr/1e12:6.1f} TF)  -> {win}")
 
 
 if __name__ == "__main__":

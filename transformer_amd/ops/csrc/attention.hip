@@ -89,7 +89,7 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
                      const unsigned char* __restrict__ kv_pad,
                      short* __restrict__ O, float* __restrict__ LSE, int B,
                      int H, int Sq, int Sk, int causal, float scale,
-                     long q_rs, long kv_rs, long o_rs) {
+                     long q_rs, long kv_rs, long o_rs, long q_bs, long kv_bs) {
   constexpr int D32 = DH / 32;   // QK^T MFMA k-steps
   constexpr int D16 = DH / 16;   // O fragments
   __shared__ short k_lds[KVT * DH];
@@ -103,9 +103,9 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
   const int q0 = qb * (WAVES * QW) + wid * QW;  // this wave's first q-row
   const int fr = lane & 15, kg = lane >> 4;
 
-  const short* Qp = Q + b * (Sq * q_rs) + (long)h * DH;
-  const short* Kp = K + b * (Sk * kv_rs) + (long)h * DH;
-  const short* Vp = V + b * (Sk * kv_rs) + (long)h * DH;
+  const short* Qp = Q + b * q_bs + (long)h * DH;
+  const short* Kp = K + b * kv_bs + (long)h * DH;
+  const short* Vp = V + b * kv_bs + (long)h * DH;
   const unsigned char* pad = kv_pad ? kv_pad + (long)b * Sk : nullptr;
 
   // Q fragments in registers: A[row=fr][k=kg*8+j] per 32-chunk
@@ -537,6 +537,12 @@ static void check_attn_view(const torch::Tensor& t, int DH, const char* name) {
               name, " must be 4-D bf16 on GPU");
   TORCH_CHECK(t.stride(3) == 1 && t.stride(2) == DH,
               name, " must have contiguous (head, dh) trailing layout");
+}
+
+// backward assumes batch stride == S * row stride (true for all training
+// layouts: contiguous and packed-QKV/KV views); forward takes explicit batch
+// strides so KV-cache slices (batch stride = S_max * row stride) work.
+static void check_attn_batch(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.stride(0) == t.size(1) * t.stride(1),
               name, " batch stride must equal S * row stride");
 }
@@ -552,6 +558,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   check_attn_view(k, DH, "k");
   check_attn_view(v, DH, "v");
   TORCH_CHECK(k.stride(1) == v.stride(1), "k/v row strides must match");
+  TORCH_CHECK(k.stride(0) == v.stride(0), "k/v batch strides must match");
   auto o = torch::empty({B, Sq, H, DH}, q.options());
   auto lse = torch::empty({B, H, Sq}, q.options().dtype(torch::kFloat32));
   const unsigned char* pad = nullptr;
@@ -566,7 +573,8 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                   (const short*)q.data_ptr(), (const short*)k.data_ptr(),
                   (const short*)v.data_ptr(), pad, (short*)o.data_ptr(),
                   lse.data_ptr<float>(), B, H, Sq, Sk, causal ? 1 : 0,
-                  (float)scale, q.stride(1), k.stride(1), (long)H * DH));
+                  (float)scale, q.stride(1), k.stride(1), (long)H * DH,
+                  q.stride(0), k.stride(0)));
   return {o, lse};
 }
 
@@ -584,6 +592,9 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   check_attn_view(q, DH, "q");
   check_attn_view(k, DH, "k");
   check_attn_view(v, DH, "v");
+  check_attn_batch(q, "q");
+  check_attn_batch(k, "k");
+  check_attn_batch(v, "v");
   TORCH_CHECK(k.stride(1) == v.stride(1), "k/v row strides must match");
   TORCH_CHECK(o.is_contiguous() && dout.is_contiguous(),
               "o/dout must be contiguous");

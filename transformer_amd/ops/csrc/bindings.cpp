@@ -7,7 +7,7 @@
 
 // ---- implemented in the .hip translation units ----------------------------
 torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
-                      int64_t epilogue);
+                      int64_t epilogue, c10::optional<torch::Tensor> out);
 torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b,
                       c10::optional<torch::Tensor> out);
 torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b,
@@ -59,7 +59,10 @@ torch::Tensor argmax_lastdim(torch::Tensor logits);
 std::vector<int64_t> accuracy(torch::Tensor logits, torch::Tensor targets);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("gemm_nt", &gemm_nt, "C[M,N] = A[M,K] @ W[N,K]^T + bias (epilogue: 0=none 1=relu)");
+  m.def("gemm_nt", &gemm_nt,
+        "C[M,N] = A[M,K] @ W[N,K]^T + bias (epilogue: 0=none 1=relu)",
+        pybind11::arg("a"), pybind11::arg("w"), pybind11::arg("bias"),
+        pybind11::arg("epilogue"), pybind11::arg("out") = pybind11::none());
   m.def("gemm_nn", &gemm_nn, "C[M,N] = A[M,K] @ B[K,N] (dX)",
         pybind11::arg("a"), pybind11::arg("b"),
         pybind11::arg("out") = pybind11::none());

@@ -497,13 +497,12 @@ torch::Tensor gemm_launch(torch::Tensor a, torch::Tensor b, torch::Tensor bias,
 
 // C[M,N] = A[M,K] @ W[N,K]^T (+bias)(+ReLU) — forward layout.
 torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
-                      int64_t epilogue) {
+                      int64_t epilogue, c10::optional<torch::Tensor> out) {
   CHECK_BF16_2D(a);
   CHECK_BF16_2D(w);
   const int M = a.size(0), K = a.size(1), N = w.size(0);
   TORCH_CHECK(w.size(1) == K, "gemm_nt: K mismatch");
-  return gemm_launch<false, false>(a, w, bias, epilogue, M, N, K, K, K,
-                                   c10::nullopt);
+  return gemm_launch<false, false>(a, w, bias, epilogue, M, N, K, K, K, out);
 }
 
 // C[M,N] = A[M,K] @ B[K,N] — dX = dY @ W without transposing W.

@@ -72,8 +72,12 @@ class _LinearFn(torch.autograd.Function):
         dy = dy.contiguous()
         if ctx.activation == "relu":
             dy = E.relu_bwd(dy, y)  # dz = dy * (y > 0)
-        dx = E.gemm_nn(dy, w)       # dX[M,K] = dY[M,N] @ W[N,K]
-        dw = E.gemm_tn(dy, x)       # dW[N,K] = dY^T @ X
+        # Measured on MI355X (tools/gemm_bench.py): explicit transpose2d +
+        # the NT kernel beats the strided NN/TN variants at every model
+        # shape (the LDS-bounce staging of NN/TN loses more than the
+        # transpose costs), so backward materializes the transposes.
+        dx = E.gemm_nt(dy, E.transpose2d(w), torch.Tensor(), 0)
+        dw = E.gemm_nt(E.transpose2d(dy), E.transpose2d(x), torch.Tensor(), 0)
         db = E.colsum(dy) if ctx.has_bias else None
         return dx, dw, db, None
 
@@ -104,8 +108,10 @@ class _LinearFlatFn(torch.autograd.Function):
         dy = dy.contiguous()
         if ctx.activation == "relu":
             dy = E.relu_bwd(dy, y)
-        dx = E.gemm_nn(dy, w)
-        E.gemm_tn(dy, x, _flat(w).view(w.shape[0], -1))
+        # transpose2d + NT beats NN/TN at every model shape (see _LinearFn).
+        dx = E.gemm_nt(dy, E.transpose2d(w), torch.Tensor(), 0)
+        E.gemm_nt(E.transpose2d(dy), E.transpose2d(x), torch.Tensor(), 0,
+                  _flat(w).view(w.shape[0], -1))
         if b is not None:
             E.colsum(dy, _flat(b).view(-1))
         _grad_ready(w, b)
