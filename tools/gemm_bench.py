@@ -63,6 +63,24 @@ def main():
     E = ext()
     torch.manual_seed(0)
 
+    print("== 256-template vs 128-kernel vs torch.matmul (hipBLASLt) ==")
+    for (m, n, k, tag) in FWD + [(16384, 512, 2048, "FFN2 fwd"),
+                                 (1536, 512, 16384, "QKV dW"),
+                                 (32770, 512, 16320, "logits dW")]:
+        a = torch.randn(m, k, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(n, k, device="cuda", dtype=torch.bfloat16)
+        nob = torch.Tensor()
+        ref = (a[:512].float() @ w.t().float()[:, :512])
+        got = E.gemm256_nt(a, w, nob, 0)[:512, :512].float()
+        err = (got - ref).abs().max().item() / max(ref.abs().max().item(), 1)
+        t256 = timeit(lambda: E.gemm256_nt(a, w, nob, 0), iters)
+        t128 = timeit(lambda: E.gemm128_nt(a, w, nob, 0), iters)
+        tbl = timeit(lambda: torch.matmul(a, w.t()), iters)
+        fl = 2.0 * m * n * k
+        print(f"{tag:14s} 256 {t256*1e3:8.3f} ms ({fl/t256/1e12:6.1f} TF) | "
+              f"128 {t128*1e3:8.3f} ({fl/t128/1e12:6.1f}) | "
+              f"blaslt {tbl*1e3:8.3f} ({fl/tbl/1e12:6.1f})  relerr {err:.2e}")
+
     print("== forward: gemm_nt ==")
     for (m, n, k, tag) in FWD:
         a = torch.randn(m, k, device="cuda", dtype=torch.bfloat16)

@@ -8,6 +8,10 @@
 // ---- implemented in the .hip translation units ----------------------------
 torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
                       int64_t epilogue, c10::optional<torch::Tensor> out);
+torch::Tensor gemm256_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
+                         int64_t epilogue, c10::optional<torch::Tensor> out);
+torch::Tensor gemm128_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
+                         int64_t epilogue, c10::optional<torch::Tensor> out);
 torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b,
                       c10::optional<torch::Tensor> out);
 torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b,
@@ -61,6 +65,13 @@ std::vector<int64_t> accuracy(torch::Tensor logits, torch::Tensor targets);
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_nt", &gemm_nt,
         "C[M,N] = A[M,K] @ W[N,K]^T + bias (epilogue: 0=none 1=relu)",
+        pybind11::arg("a"), pybind11::arg("w"), pybind11::arg("bias"),
+        pybind11::arg("epilogue"), pybind11::arg("out") = pybind11::none());
+  m.def("gemm128_nt", &gemm128_nt, "128x128 NT path, no 256 dispatch (A/B)",
+        pybind11::arg("a"), pybind11::arg("w"), pybind11::arg("bias"),
+        pybind11::arg("epilogue"), pybind11::arg("out") = pybind11::none());
+  m.def("gemm256_nt", &gemm256_nt,
+        "256x256 deep-pipelined NT GEMM (K%32==0; M/N any, edge-clamped)",
         pybind11::arg("a"), pybind11::arg("w"), pybind11::arg("bias"),
         pybind11::arg("epilogue"), pybind11::arg("out") = pybind11::none());
   m.def("gemm_nn", &gemm_nn, "C[M,N] = A[M,K] @ B[K,N] (dX)",

@@ -495,9 +495,26 @@ torch::Tensor gemm_launch(torch::Tensor a, torch::Tensor b, torch::Tensor bias,
 
 }  // namespace
 
+// Implemented in gemm256.hip: deep-pipelined 256x256 kernel + viability.
+bool gemm256_viable(int M, int N, int K, int lda, int ldb);
+torch::Tensor gemm256_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
+                         int64_t epilogue, c10::optional<torch::Tensor> out);
+
 // C[M,N] = A[M,K] @ W[N,K]^T (+bias)(+ReLU) — forward layout.
 torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
                       int64_t epilogue, c10::optional<torch::Tensor> out) {
+  CHECK_BF16_2D(a);
+  CHECK_BF16_2D(w);
+  const int M = a.size(0), K = a.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "gemm_nt: K mismatch");
+  if (gemm256_viable(M, N, K, K, K))
+    return gemm256_nt(a, w, bias, epilogue, out);
+  return gemm_launch<false, false>(a, w, bias, epilogue, M, N, K, K, K, out);
+}
+
+// The 128x128 path with no 256-template dispatch (A/B benchmarking).
+torch::Tensor gemm128_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
+                         int64_t epilogue, c10::optional<torch::Tensor> out) {
   CHECK_BF16_2D(a);
   CHECK_BF16_2D(w);
   const int M = a.size(0), K = a.size(1), N = w.size(0);
