@@ -96,10 +96,11 @@ def main():
         w = torch.randn(n, k, device="cuda", dtype=torch.bfloat16)
         t_nn = timeit(lambda: E.gemm_nn(dy, w), iters)
         t_tr = timeit(lambda: E.gemm_nt(dy, E.transpose2d(w), torch.Tensor(), 0), iters)
+        t_bl = timeit(lambda: torch.matmul(dy, w), iters)
         fl = 2.0 * m * n * k
-        win = "nn" if t_nn < t_tr else "tr+nt"
         print(f"{tag:14s} nn {t_nn*1e3:8.3f} ms ({fl/t_nn/1e12:6.1f} TF) | "
-              f"tr+nt {t_tr*1e3:8.3f} ms ({fl/t_tr/1e12:6.1f} TF)  -> {win}")
+              f"tr+nt {t_tr*1e3:8.3f} ms ({fl/t_tr/1e12:6.1f} TF) | "
+              f"blaslt {t_bl*1e3:8.3f} ({fl/t_bl/1e12:6.1f})")
 
     print("== dW: gemm_tn vs transpose(dy)+transpose(x)+gemm_nt ==")
     for (mt, n, k, tag) in DW:
@@ -108,10 +109,11 @@ def main():
         t_tn = timeit(lambda: E.gemm_tn(dy, x), iters)
         t_tr = timeit(lambda: E.gemm_nt(E.transpose2d(dy), E.transpose2d(x),
                                         torch.Tensor(), 0), iters)
+        t_bl = timeit(lambda: torch.matmul(dy.t(), x), iters)
         fl = 2.0 * mt * n * k
-        win = "tn" if t_tn < t_tr else "tr+nt"
         print(f"{tag:14s} tn {t_tn*1e3:8.3f} ms ({fl/t_tn/1e12:6.1f} TF) | "
-              f"tr+nt {t_tr*1e3:8.3f} ms ({fl/t_tr/1e12:6.1f} TF)  -> {win}")
+              f"tr+nt {t_tr*1e3:8.3f} ms ({fl/t_tr/1e12:6.1f} TF) | "
+              f"blaslt {t_bl*1e3:8.3f} ({fl/t_bl/1e12:6.1f})")
 
 
 if __name__ == "__main__":

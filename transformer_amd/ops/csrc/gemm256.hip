@@ -220,7 +220,10 @@ bool gemm256_viable(int M, int N, int K, int lda, int ldb) {
   if (K % G_BK != 0 || K < 3 * G_BK) return false;
   if (lda % 8 != 0 || ldb % 8 != 0) return false;
   long nwg = (long)cdiv(M, G_BM) * cdiv(N, G_BN);
-  return nwg >= 224;  // fills the 256-CU chip (1 WG/CU at 128 KiB LDS)
+  // Measured rule (tools/gemm_bench.py on MI355X): the pipelined kernel
+  // needs the chip full (>=224 WGs at 1 WG/CU) and either deep K or a
+  // grid big enough to amortize the 3-tile prologue per WG.
+  return nwg >= 224 && (K >= 1024 || nwg >= 512);
 }
 
 torch::Tensor gemm256_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
