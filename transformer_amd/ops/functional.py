@@ -53,6 +53,20 @@ def _flat(p):
 # Linear: y = x @ W^T + b, optional fused ReLU epilogue (K1/K7/K8/K12).
 # ---------------------------------------------------------------------------
 
+def _dw_gemm(dy, x, out=None):
+    """dW[N,K] = dY^T @ X — plain GEMM; backend measured per shape on
+    MI355X (tools/gemm_bench.py): hipBLASLt TN wins at the wide logits
+    head (N>=8k: 2x), the hand-written transpose2d+NT path wins at the
+    d_model-sized heads."""
+    E = ext()
+    if dy.shape[1] >= 8192:
+        if out is not None:
+            return torch.matmul(dy.t(), x, out=out)
+        return torch.matmul(dy.t(), x)
+    return E.gemm_nt(E.transpose2d(dy), E.transpose2d(x), torch.Tensor(), 0,
+                     out)
+
+
 class _LinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, activation):
@@ -77,7 +91,7 @@ class _LinearFn(torch.autograd.Function):
         # physical transposes, and measured faster than both hand-written
         # paths at every model shape (tools/gemm_bench.py on MI355X).
         dx = torch.matmul(dy, w)        # dX[M,K] = dY[M,N] @ W[N,K]
-        dw = torch.matmul(dy.t(), x)    # dW[N,K] = dY^T @ X
+        dw = _dw_gemm(dy, x)
         db = E.colsum(dy) if ctx.has_bias else None
         return dx, dw, db, None
 
@@ -111,7 +125,7 @@ class _LinearFlatFn(torch.autograd.Function):
         # Plain GEMMs -> hipBLASLt, straight into the flat-grad view
         # (see _LinearFn.backward for the measurement rationale).
         dx = torch.matmul(dy, w)
-        torch.matmul(dy.t(), x, out=_flat(w).view(w.shape[0], -1))
+        _dw_gemm(dy, x, out=_flat(w).view(w.shape[0], -1))
         if b is not None:
             E.colsum(dy, _flat(b).view(-1))
         _grad_ready(w, b)
