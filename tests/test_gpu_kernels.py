@@ -48,6 +48,37 @@ def test_gemm_nt(m, n, k):
     assert_close(c, ref, 0.03, "gemm_nt")
 
 
+@pytest.mark.parametrize("m,n,k,epi", [
+    (512, 512, 96, 0),        # minimum-depth pipeline (3 K-tiles)
+    (777, 300, 512, 0),       # ragged M/N edge clamping
+    (2048, 1536, 512, 1),     # QKV-like + ReLU epilogue
+    (300, 1000, 1024, 0),     # partial tail tiles both dims
+])
+def test_gemm256_nt(m, n, k, epi):
+    """Deep-pipelined 256x256 kernel (gemm256.hip) against fp32 reference,
+    invoked directly so small shapes exercise it regardless of the
+    dispatch heuristic in gemm_nt."""
+    torch.manual_seed(2)
+    a = torch.randn(m, k, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(n, k, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    c = _ext().gemm256_nt(a, w, b, epi)
+    ref = a.float() @ w.float().T + b.float()
+    if epi == 1:
+        ref = torch.relu(ref)
+    assert_close(c, ref, 0.03, "gemm256_nt")
+
+
+def test_gemm256_nt_out_destination():
+    torch.manual_seed(3)
+    a = torch.randn(512, 256, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(512, 256, device="cuda", dtype=torch.bfloat16)
+    out = torch.empty(512, 512, device="cuda", dtype=torch.bfloat16)
+    r = _ext().gemm256_nt(a, w, torch.Tensor(), 0, out)
+    assert r.data_ptr() == out.data_ptr()
+    assert_close(out, a.float() @ w.float().T, 0.03, "gemm256_out")
+
+
 def test_gemm_nt_relu():
     torch.manual_seed(1)
     a = torch.randn(200, 512, device="cuda", dtype=torch.bfloat16)
