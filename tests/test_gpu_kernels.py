@@ -69,6 +69,32 @@ def test_gemm256_nt(m, n, k, epi):
     assert_close(c, ref, 0.03, "gemm256_nt")
 
 
+@pytest.mark.parametrize("m,n,k", [
+    (16384, 512, 512),    # split-M accumulation path
+    (4096, 1536, 512),
+    (200, 130, 70),       # ragged every dim, single-slice path
+    (64, 512, 512),
+])
+def test_gemm_dw(m, n, k):
+    """tr16 transpose-read dW kernel (gemm_dw.hip) vs fp32 reference."""
+    torch.manual_seed(4)
+    dy = torch.randn(m, n, device="cuda", dtype=torch.bfloat16)
+    x = torch.randn(m, k, device="cuda", dtype=torch.bfloat16)
+    c = _ext().gemm_dw(dy, x)
+    ref = dy.float().t() @ x.float()
+    assert_close(c, ref, 0.03, "gemm_dw")
+
+
+def test_gemm_dw_out_destination():
+    torch.manual_seed(5)
+    dy = torch.randn(1024, 256, device="cuda", dtype=torch.bfloat16)
+    x = torch.randn(1024, 384, device="cuda", dtype=torch.bfloat16)
+    out = torch.empty(256, 384, device="cuda", dtype=torch.bfloat16)
+    r = _ext().gemm_dw(dy, x, out)
+    assert r.data_ptr() == out.data_ptr()
+    assert_close(out, dy.float().t() @ x.float(), 0.03, "gemm_dw_out")
+
+
 def test_gemm256_nt_out_destination():
     torch.manual_seed(3)
     a = torch.randn(512, 256, device="cuda", dtype=torch.bfloat16)

@@ -110,10 +110,12 @@ def main():
         t_tr = timeit(lambda: E.gemm_nt(E.transpose2d(dy), E.transpose2d(x),
                                         torch.Tensor(), 0), iters)
         t_bl = timeit(lambda: torch.matmul(dy.t(), x), iters)
+        t_dw = timeit(lambda: E.gemm_dw(dy, x), iters)
         fl = 2.0 * mt * n * k
         print(f"{tag:14s} tn {t_tn*1e3:8.3f} ms ({fl/t_tn/1e12:6.1f} TF) | "
               f"tr+nt {t_tr*1e3:8.3f} ms ({fl/t_tr/1e12:6.1f} TF) | "
-              f"blaslt {t_bl*1e3:8.3f} ({fl/t_bl/1e12:6.1f})")
+              f"blaslt {t_bl*1e3:8.3f} ({fl/t_bl/1e12:6.1f}) | "
+              f"dw {t_dw*1e3:8.3f} ({fl/t_dw/1e12:6.1f})")
 
 
 if __name__ == "__main__":

@@ -1,0 +1,228 @@
+// Deep-contraction dW GEMM for CDNA4 (gfx950): C[N,K] = dY^T @ X.
+//
+// The linear-backward weight gradient contracts over tokens (M = 16384 for
+// the base model) with small N,K (512..2048).  Both operands are stored
+// token-major, so the MFMA fragments need contraction-contiguous (column)
+// reads — the transpose problem.  Previous approaches measured on MI355X
+// (tools/gemm_bench.py): physical transpose2d + NT kernel ~340 TF, LDS
+// scratch-bounce TN ~230 TF, hipBLASLt TN ~240-330 TF at these shapes.
+//
+// This kernel instead stages dY/X blocks token-major (coalesced HBM reads)
+// into [32 m][16 col] LDS subtiles and reads MFMA fragments directly with
+// gfx950's ds_read_b64_tr_b16 hardware transpose-read (guide T10; lane
+// semantics verified by tools/tr16_probe.hip):
+//   per 16-lane group, the lane with local index m' supplies the address
+//   &T[mb + m'/4][nb + 4*(m'%4)] and the lane with local index n receives
+//   {T[mb+j][nb+n]}_{j=0..3} — a free 4x4 transpose, no scratch pass.
+//
+// Tiling: output 128(n) x 128(k) per 256-thread WG (4 waves, 64x64 each,
+// acc[4][4] 16x16 frags); contraction in 64-token blocks (2 MFMA k-steps);
+// the M axis is split across blockIdx.y slices with fp32 atomic
+// accumulation + a finalize pass so the 256-CU chip stays full even when
+// N=K=512 (16 output tiles).
+#include "common.h"
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+typedef __bf16 bf16x8d __attribute__((ext_vector_type(8)));
+typedef short s16x4d __attribute__((ext_vector_type(4)));
+
+namespace {
+
+#define DW_BN 128   // output rows (dY columns)
+#define DW_BK 128   // output cols (X columns)
+#define DW_BM 64    // contraction (token) block
+#define DW_THREADS 256
+
+// LDS image: [col-subtile][m-subtile][32 m][16 col] bf16; one operand block
+// (64 m x 128 col) = 8 col-subtiles x 2 m-subtiles x 512 shorts = 16 KiB.
+// Within a subtile the m-row is stored permuted (pr = low-3-bits<<2 | top-2)
+// so the four 16-lane tr-read groups (m windows 8 apart) interleave across
+// the 512-B subtile instead of stacking on the same 256-B bank row.
+DEV_INLINE int dw_img(int m, int col) {
+  const int r = m & 31;
+  const int pr = ((r & 7) << 2) | (r >> 3);
+  return ((col >> 4) * 2 + (m >> 5)) * 512 + pr * 16 + (col & 15);
+}
+
+// Stage a [64 m][128 col] block from global (row-major, ldg elems/row) into
+// the subtile image.  Each of 256 threads: 4 x s16x8 loads (16 B, coalesced
+// along col) + 4 ds_writes.  Rows clamped to mrows (garbage cols guarded at
+// epilogue); col guard zero-fills.
+DEV_INLINE void dw_stage(const short* __restrict__ g, long ldg, long m0,
+                         long mmax, int col0, int ncols, short* lds) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    int idx = p * DW_THREADS + t;      // 1024 chunks of 8 shorts
+    int m = idx >> 4;                  // 16 chunks per 128-col row
+    int c8 = (idx & 15) << 3;
+    long gm = m0 + m;
+    if (gm > mmax - 1) gm = mmax - 1;
+    s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (col0 + c8 + 8 <= ncols) {
+      v = *(const s16x8*)(g + gm * ldg + col0 + c8);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (col0 + c8 + j < ncols) v[j] = g[gm * ldg + col0 + c8 + j];
+    }
+    // image dest: subtile row-half [m][c8..c8+8) — contiguous 16 B
+    *(s16x4*)&lds[dw_img(m, c8)] = {v[0], v[1], v[2], v[3]};
+    *(s16x4*)&lds[dw_img(m, c8 + 4)] = {v[4], v[5], v[6], v[7]};
+  }
+}
+
+// Transpose-read one 4-deep fragment slice: returns T[mb..mb+4)[colb + n]
+// for this lane's n-role (n = lane & 15).  mb must be a multiple of 4
+// within one 32-m subtile.
+DEV_INLINE s16x4d dw_tr4(const short* lds, int mb, int colb) {
+  const int mp = threadIdx.x & 15;   // this lane's address role
+  const int off = dw_img(mb + (mp >> 2), colb + 4 * (mp & 3));
+  return __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (__attribute__((address_space(3))) s16x4d*)(
+          const_cast<short*>(&lds[off])));
+}
+
+// Full 8-deep bf16x8 fragment: contraction ms..ms+8, column colb + (l&15).
+DEV_INLINE bf16x8d dw_frag(const short* lds, int ms, int colb) {
+  s16x4d lo = dw_tr4(lds, ms, colb);
+  s16x4d hi = dw_tr4(lds, ms + 4, colb);
+  s16x8 v = {lo[0], lo[1], lo[2], lo[3], hi[0], hi[1], hi[2], hi[3]};
+  return (bf16x8d)v;
+}
+
+template <bool SPLIT>
+__global__ __launch_bounds__(DW_THREADS)
+void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
+                    short* __restrict__ C, float* __restrict__ CW, int M,
+                    int N, int K, long m_per_slice, int nbk) {
+  __shared__ short a_img[2][DW_BM * DW_BN];  // dY block, double-buffered
+  __shared__ short b_img[2][DW_BM * DW_BK];  // X block
+
+  const int wg = blockIdx.x;
+  const int bn0 = (wg / nbk) * DW_BN;
+  const int bk0 = (wg % nbk) * DW_BK;
+  const long m_lo = (long)blockIdx.y * m_per_slice;
+  const long m_hi = min((long)M, m_lo + m_per_slice);
+
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int wn = (wid >> 1) * 64;    // wave rows (n) in tile
+  const int wk = (wid & 1) * 64;     // wave cols (k)
+  const int fr = lane & 15;
+  const int kg = lane >> 4;          // contraction group
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  int buf = 0;
+  dw_stage(dY, N, m_lo, m_hi, bn0, N, a_img[0]);
+  dw_stage(X, K, m_lo, m_hi, bk0, K, b_img[0]);
+  __syncthreads();
+
+  for (long m0 = m_lo; m0 < m_hi; m0 += DW_BM) {
+    if (m0 + DW_BM < m_hi) {
+      dw_stage(dY, N, m0 + DW_BM, m_hi, bn0, N, a_img[buf ^ 1]);
+      dw_stage(X, K, m0 + DW_BM, m_hi, bk0, K, b_img[buf ^ 1]);
+    }
+    // 2 MFMA k-steps of 32 tokens each
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      const int ms = ks * 32 + kg * 8;
+      bf16x8d af[4], bf_[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        af[i] = dw_frag(a_img[buf], ms, wn + i * 16);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        bf_[j] = dw_frag(b_img[buf], ms, wk + j * 16);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf_[j], acc[i][j], 0, 0, 0);
+    }
+    buf ^= 1;
+    __syncthreads();
+  }
+
+  // Epilogue: D lane map col = lane&15 -> k, row = (lane>>4)*4 + r -> n.
+  // The fragment A row role is n (tr-read result lane = n), so output rows
+  // follow the standard C/D mapping with rows = n, cols = k.
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int nrow_base = bn0 + wn + i * 16 + kg * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int kcol = bk0 + wk + j * 16 + fr;
+      if (kcol >= K) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int nrow = nrow_base + r;
+        if (nrow >= N) continue;
+        if (SPLIT)
+          atomicAdd(&CW[(long)nrow * K + kcol], acc[i][j][r]);
+        else
+          C[(long)nrow * K + kcol] = f2bfbits(acc[i][j][r]);
+      }
+    }
+  }
+}
+
+__global__ void dw_finalize_kernel(const float* __restrict__ cw,
+                                   short* __restrict__ c, long nk) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < nk) c[i] = f2bfbits(cw[i]);
+}
+
+}  // namespace
+
+// C[N,K] = dY[M,N]^T @ X[M,K], bf16 in/out, fp32 accumulate.
+torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x,
+                      c10::optional<torch::Tensor> out) {
+  TORCH_CHECK(dy.is_cuda() && dy.dtype() == torch::kBFloat16 &&
+              dy.dim() == 2 && dy.is_contiguous(), "gemm_dw: bad dy");
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.dim() == 2 &&
+              x.is_contiguous(), "gemm_dw: bad x");
+  const long M = dy.size(0);
+  const int N = dy.size(1), K = x.size(1);
+  TORCH_CHECK(x.size(0) == M, "gemm_dw: contraction mismatch");
+  torch::Tensor c;
+  if (out.has_value()) {
+    c = *out;
+    TORCH_CHECK(c.is_cuda() && c.dtype() == torch::kBFloat16 &&
+                c.is_contiguous() && c.numel() == (long)N * K,
+                "gemm_dw: bad out");
+  } else {
+    c = torch::empty({N, K}, dy.options());
+  }
+  const int nbn = cdiv(N, DW_BN), nbk = cdiv(K, DW_BK);
+  const int ntiles = nbn * nbk;
+  // slices: fill >=512 WGs (2 WG/CU at 32-wave occupancy), 64-token quanta
+  int nslices = (int)min((M + DW_BM - 1) / DW_BM, (long)cdiv(512, ntiles));
+  long m_per_slice = (M + nslices - 1) / nslices;
+  m_per_slice = (m_per_slice + DW_BM - 1) / DW_BM * DW_BM;
+  nslices = (int)((M + m_per_slice - 1) / m_per_slice);
+  dim3 grid(ntiles, nslices);
+  auto stream = at::hip::getCurrentHIPStream();
+  if (nslices == 1) {
+    gemm_dw_kernel<false><<<grid, DW_THREADS, 0, stream>>>(
+        (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
+        (short*)c.data_ptr(), nullptr, (int)M, N, K, m_per_slice, nbk);
+  } else {
+    auto cw = torch::zeros({N, K},
+                           dy.options().dtype(torch::kFloat32));
+    gemm_dw_kernel<true><<<grid, DW_THREADS, 0, stream>>>(
+        (const short*)dy.data_ptr(), (const short*)x.data_ptr(), nullptr,
+        cw.data_ptr<float>(), (int)M, N, K, m_per_slice, nbk);
+    long nk = (long)N * K;
+    dw_finalize_kernel<<<(nk + 1023) / 1024, 1024, 0, stream>>>(
+        cw.data_ptr<float>(), (short*)c.data_ptr(), nk);
+  }
+  return c;
+}

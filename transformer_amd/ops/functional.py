@@ -63,8 +63,7 @@ def _dw_gemm(dy, x, out=None):
         if out is not None:
             return torch.matmul(dy.t(), x, out=out)
         return torch.matmul(dy.t(), x)
-    return E.gemm_nt(E.transpose2d(dy), E.transpose2d(x), torch.Tensor(), 0,
-                     out)
+    return E.gemm_dw(dy, x, out)
 
 
 class _LinearFn(torch.autograd.Function):
