@@ -48,8 +48,9 @@ DEV_INLINE int dw_img(int m, int col) {
 
 // Stage a [64 m][128 col] block from global (row-major, ldg elems/row) into
 // the subtile image.  Each of 256 threads: 4 x s16x8 loads (16 B, coalesced
-// along col) + 4 ds_writes.  Rows clamped to mrows (garbage cols guarded at
-// epilogue); col guard zero-fills.
+// along col) + 4 ds_writes.  Out-of-range token rows and columns are
+// ZERO-filled — rows are contraction terms here, so clamping (as the NT
+// kernels do for discarded output rows) would add duplicate contributions.
 DEV_INLINE void dw_stage(const short* __restrict__ g, long ldg, long m0,
                          long mmax, int col0, int ncols, short* lds) {
   const int t = threadIdx.x;
@@ -59,14 +60,15 @@ DEV_INLINE void dw_stage(const short* __restrict__ g, long ldg, long m0,
     int m = idx >> 4;                  // 16 chunks per 128-col row
     int c8 = (idx & 15) << 3;
     long gm = m0 + m;
-    if (gm > mmax - 1) gm = mmax - 1;
     s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-    if (col0 + c8 + 8 <= ncols) {
-      v = *(const s16x8*)(g + gm * ldg + col0 + c8);
-    } else {
+    if (gm < mmax) {
+      if (col0 + c8 + 8 <= ncols) {
+        v = *(const s16x8*)(g + gm * ldg + col0 + c8);
+      } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        if (col0 + c8 + j < ncols) v[j] = g[gm * ldg + col0 + c8 + j];
+        for (int j = 0; j < 8; ++j)
+          if (col0 + c8 + j < ncols) v[j] = g[gm * ldg + col0 + c8 + j];
+      }
     }
     // image dest: subtile row-half [m][c8..c8+8) — contiguous 16 B
     *(s16x4*)&lds[dw_img(m, c8)] = {v[0], v[1], v[2], v[3]};
@@ -93,13 +95,39 @@ DEV_INLINE bf16x8d dw_frag(const short* lds, int ms, int colb) {
   return (bf16x8d)v;
 }
 
+// glds staging for FULL blocks (64 valid m rows, 128 in-range cols): the
+// image mapping is bijective, so the destination stays lane-linear and the
+// inverse row permutation is applied to the per-lane SOURCE address — no
+// register bounce, no ds_write pass (guide §5 rule 1).  Each wave stages 4
+// of the 16 1-KiB chunks.
+DEV_INLINE void dw_stage_glds(const short* __restrict__ g, long ldg, long m0,
+                              int col0, short* lds) {
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    const int chunk = p * 4 + wid;
+    const int d = chunk * 1024 + lane * 16;  // dest byte in image
+    const int e = d >> 1;                    // dest element
+    const int sub = e >> 9;
+    const int we = e & 511;
+    const int pr = we >> 4;
+    const int r = ((pr & 3) << 3) | (pr >> 2);     // inverse permutation
+    const int m = (sub & 1) * 32 + r;
+    const int col = (sub >> 1) * 16 + (we & 15);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)(g + (m0 + m) * ldg +
+                                                        col0 + col),
+        (__attribute__((address_space(3))) void*)((char*)lds + d), 16, 0, 0);
+  }
+}
+
 template <bool SPLIT>
 __global__ __launch_bounds__(DW_THREADS)
 void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
                     short* __restrict__ C, float* __restrict__ CW, int M,
                     int N, int K, long m_per_slice, int nbk) {
-  __shared__ short a_img[2][DW_BM * DW_BN];  // dY block, double-buffered
-  __shared__ short b_img[2][DW_BM * DW_BK];  // X block
+  __shared__ short a_img[DW_BM * DW_BN];  // dY block
+  __shared__ short b_img[DW_BM * DW_BK];  // X block
 
   const int wg = blockIdx.x;
   const int bn0 = (wg / nbk) * DW_BN;
@@ -119,16 +147,21 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  int buf = 0;
-  dw_stage(dY, N, m_lo, m_hi, bn0, N, a_img[0]);
-  dw_stage(X, K, m_lo, m_hi, bk0, K, b_img[0]);
-  __syncthreads();
+  // Full-block fast path: no m tail in range, both col blocks in range.
+  const bool a_full = bn0 + DW_BN <= N;
+  const bool b_full = bk0 + DW_BK <= K;
 
   for (long m0 = m_lo; m0 < m_hi; m0 += DW_BM) {
-    if (m0 + DW_BM < m_hi) {
-      dw_stage(dY, N, m0 + DW_BM, m_hi, bn0, N, a_img[buf ^ 1]);
-      dw_stage(X, K, m0 + DW_BM, m_hi, bk0, K, b_img[buf ^ 1]);
-    }
+    const bool m_full = m0 + DW_BM <= m_hi;
+    if (m_full && a_full)
+      dw_stage_glds(dY, N, m0, bn0, a_img);
+    else
+      dw_stage(dY, N, m0, m_hi, bn0, N, a_img);
+    if (m_full && b_full)
+      dw_stage_glds(X, K, m0, bk0, b_img);
+    else
+      dw_stage(X, K, m0, m_hi, bk0, K, b_img);
+    __syncthreads();  // drains the glds queue (vmcnt 0) + ds_writes
     // 2 MFMA k-steps of 32 tokens each
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
@@ -136,10 +169,10 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
       bf16x8d af[4], bf_[4];
 #pragma unroll
       for (int i = 0; i < 4; ++i)
-        af[i] = dw_frag(a_img[buf], ms, wn + i * 16);
+        af[i] = dw_frag(a_img, ms, wn + i * 16);
 #pragma unroll
       for (int j = 0; j < 4; ++j)
-        bf_[j] = dw_frag(b_img[buf], ms, wk + j * 16);
+        bf_[j] = dw_frag(b_img, ms, wk + j * 16);
 #pragma unroll
       for (int i = 0; i < 4; ++i)
 #pragma unroll
@@ -147,7 +180,6 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[i], bf_[j], acc[i][j], 0, 0, 0);
     }
-    buf ^= 1;
     __syncthreads();
   }
 
