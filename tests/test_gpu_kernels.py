@@ -85,6 +85,17 @@ def test_gemm_dw(m, n, k):
     assert_close(c, ref, 0.03, "gemm_dw")
 
 
+@pytest.mark.parametrize("m,n,k", [(16384, 512, 512), (200, 130, 70)])
+def test_gemm_dw_fused_bias_grad(m, n, k):
+    torch.manual_seed(6)
+    dy = torch.randn(m, n, device="cuda", dtype=torch.bfloat16)
+    x = torch.randn(m, k, device="cuda", dtype=torch.bfloat16)
+    db = torch.empty(n, device="cuda", dtype=torch.bfloat16)
+    c = _ext().gemm_dw(dy, x, None, db)
+    assert_close(c, dy.float().t() @ x.float(), 0.03, "gemm_dw+db dW")
+    assert_close(db, dy.float().sum(0), 0.03, "gemm_dw+db db")
+
+
 def test_gemm_dw_out_destination():
     torch.manual_seed(5)
     dy = torch.randn(1024, 256, device="cuda", dtype=torch.bfloat16)

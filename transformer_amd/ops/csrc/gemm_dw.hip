@@ -121,10 +121,16 @@ DEV_INLINE void dw_stage_glds(const short* __restrict__ g, long ldg, long m0,
   }
 }
 
-template <bool SPLIT>
+// DB: k-block-0 workgroups additionally accumulate the bias gradient
+// db[n] = sum_m dY[m][n] into DBW (fp32, atomic) — dY is already streaming
+// through this kernel, so the separate colsum pass (a full extra HBM read
+// of dY per linear) disappears.  The re-read here hits L2: the same rows
+// were just fetched by the glds staging.
+template <bool SPLIT, bool DB>
 __global__ __launch_bounds__(DW_THREADS)
 void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
-                    short* __restrict__ C, float* __restrict__ CW, int M,
+                    short* __restrict__ C, float* __restrict__ CW,
+                    float* __restrict__ DBW, int M,
                     int N, int K, long m_per_slice, int nbk) {
   __shared__ short a_img[DW_BM * DW_BN];  // dY block
   __shared__ short b_img[DW_BM * DW_BK];  // X block
@@ -151,7 +157,31 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
   const bool a_full = bn0 + DW_BN <= N;
   const bool b_full = bk0 + DW_BK <= K;
 
+  // bias-grad accumulators: thread t covers dY cols bn0+c8..+8, rows
+  // p*16 + t/16 of each 64-m block (the dw_stage access pattern).
+  const bool do_db = DB && bk0 == 0;
+  const int db_c8 = (threadIdx.x & 15) << 3;
+  float db_acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) db_acc[j] = 0.f;
+
   for (long m0 = m_lo; m0 < m_hi; m0 += DW_BM) {
+    if (do_db) {
+#pragma unroll
+      for (int p = 0; p < 4; ++p) {
+        const long gm = m0 + p * 16 + (threadIdx.x >> 4);
+        if (gm < m_hi && bn0 + db_c8 + 8 <= N) {
+          s16x8 v = *(const s16x8*)(dY + gm * N + bn0 + db_c8);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) db_acc[j] += bfbits2f(v[j]);
+        } else if (gm < m_hi) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (bn0 + db_c8 + j < N)
+              db_acc[j] += bfbits2f(dY[gm * N + bn0 + db_c8 + j]);
+        }
+      }
+    }
     const bool m_full = m0 + DW_BM <= m_hi;
     if (m_full && a_full)
       dw_stage_glds(dY, N, m0, bn0, a_img);
@@ -181,6 +211,12 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
               af[i], bf_[j], acc[i][j], 0, 0, 0);
     }
     __syncthreads();
+  }
+
+  if (do_db) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      if (bn0 + db_c8 + j < N) atomicAdd(&DBW[bn0 + db_c8 + j], db_acc[j]);
   }
 
   // Epilogue: D lane map col = lane&15 -> k, row = (lane>>4)*4 + r -> n.
@@ -214,9 +250,12 @@ __global__ void dw_finalize_kernel(const float* __restrict__ cw,
 
 }  // namespace
 
-// C[N,K] = dY[M,N]^T @ X[M,K], bf16 in/out, fp32 accumulate.
+// C[N,K] = dY[M,N]^T @ X[M,K], bf16 in/out, fp32 accumulate.  If `db` is
+// given (bf16 [N]), the bias gradient sum_m dY[m][n] is produced in the
+// same pass (no separate colsum read of dY).
 torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x,
-                      c10::optional<torch::Tensor> out) {
+                      c10::optional<torch::Tensor> out,
+                      c10::optional<torch::Tensor> db) {
   TORCH_CHECK(dy.is_cuda() && dy.dtype() == torch::kBFloat16 &&
               dy.dim() == 2 && dy.is_contiguous(), "gemm_dw: bad dy");
   TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.dim() == 2 &&
@@ -242,19 +281,36 @@ torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x,
   nslices = (int)((M + m_per_slice - 1) / m_per_slice);
   dim3 grid(ntiles, nslices);
   auto stream = at::hip::getCurrentHIPStream();
+  torch::Tensor dbw;
+  float* dbw_p = nullptr;
+  const bool has_db = db.has_value();
+  if (has_db) {
+    TORCH_CHECK(db->is_cuda() && db->dtype() == torch::kBFloat16 &&
+                db->is_contiguous() && db->numel() == N, "gemm_dw: bad db");
+    dbw = torch::zeros({N}, dy.options().dtype(torch::kFloat32));
+    dbw_p = dbw.data_ptr<float>();
+  }
+  auto launch = [&](auto split, auto dbc, float* cwp) {
+    gemm_dw_kernel<decltype(split)::value, decltype(dbc)::value>
+        <<<grid, DW_THREADS, 0, stream>>>(
+            (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
+            (short*)c.data_ptr(), cwp, dbw_p, (int)M, N, K, m_per_slice,
+            nbk);
+  };
+  using T = std::true_type;
+  using F = std::false_type;
   if (nslices == 1) {
-    gemm_dw_kernel<false><<<grid, DW_THREADS, 0, stream>>>(
-        (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
-        (short*)c.data_ptr(), nullptr, (int)M, N, K, m_per_slice, nbk);
+    if (has_db) launch(F{}, T{}, nullptr); else launch(F{}, F{}, nullptr);
   } else {
-    auto cw = torch::zeros({N, K},
-                           dy.options().dtype(torch::kFloat32));
-    gemm_dw_kernel<true><<<grid, DW_THREADS, 0, stream>>>(
-        (const short*)dy.data_ptr(), (const short*)x.data_ptr(), nullptr,
-        cw.data_ptr<float>(), (int)M, N, K, m_per_slice, nbk);
+    auto cw = torch::zeros({N, K}, dy.options().dtype(torch::kFloat32));
+    if (has_db) launch(T{}, T{}, cw.data_ptr<float>());
+    else launch(T{}, F{}, cw.data_ptr<float>());
     long nk = (long)N * K;
     dw_finalize_kernel<<<(nk + 1023) / 1024, 1024, 0, stream>>>(
         cw.data_ptr<float>(), (short*)c.data_ptr(), nk);
   }
+  if (has_db)
+    dw_finalize_kernel<<<cdiv(N, 256), 256, 0, stream>>>(
+        dbw.data_ptr<float>(), (short*)db->data_ptr(), N);
   return c;
 }

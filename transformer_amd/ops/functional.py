@@ -63,7 +63,23 @@ def _dw_gemm(dy, x, out=None):
         if out is not None:
             return torch.matmul(dy.t(), x, out=out)
         return torch.matmul(dy.t(), x)
-    return E.gemm_dw(dy, x, out)
+    return E.gemm_dw(dy, x, out, None)
+
+
+def _dw_db_gemm(dy, x, has_bias, w_out=None, b_out=None):
+    """dW and (optionally) db together.  On the gemm_dw path the bias
+    gradient is fused into the dW kernel's dY stream; on the hipBLASLt
+    path it stays a separate colsum pass."""
+    E = ext()
+    if dy.shape[1] >= 8192:
+        dw = _dw_gemm(dy, x, out=w_out)
+        db = E.colsum(dy, b_out) if has_bias else None
+        return dw, db
+    if has_bias:
+        db = b_out if b_out is not None else \
+            torch.empty(dy.shape[1], device=dy.device, dtype=dy.dtype)
+        return E.gemm_dw(dy, x, w_out, db), db
+    return E.gemm_dw(dy, x, w_out, None), None
 
 
 class _LinearFn(torch.autograd.Function):
@@ -90,8 +106,7 @@ class _LinearFn(torch.autograd.Function):
         # physical transposes, and measured faster than both hand-written
         # paths at every model shape (tools/gemm_bench.py on MI355X).
         dx = torch.matmul(dy, w)        # dX[M,K] = dY[M,N] @ W[N,K]
-        dw = _dw_gemm(dy, x)
-        db = E.colsum(dy) if ctx.has_bias else None
+        dw, db = _dw_db_gemm(dy, x, ctx.has_bias)
         return dx, dw, db, None
 
 
@@ -124,9 +139,9 @@ class _LinearFlatFn(torch.autograd.Function):
         # Plain GEMMs -> hipBLASLt, straight into the flat-grad view
         # (see _LinearFn.backward for the measurement rationale).
         dx = torch.matmul(dy, w)
-        _dw_gemm(dy, x, out=_flat(w).view(w.shape[0], -1))
-        if b is not None:
-            E.colsum(dy, _flat(b).view(-1))
+        _dw_db_gemm(dy, x, b is not None,
+                    w_out=_flat(w).view(w.shape[0], -1),
+                    b_out=_flat(b).view(-1) if b is not None else None)
         _grad_ready(w, b)
         return dx, None, None
 
