@@ -126,6 +126,9 @@ DEV_INLINE void dw_stage_glds(const short* __restrict__ g, long ldg, long m0,
 // through this kernel, so the separate colsum pass (a full extra HBM read
 // of dY per linear) disappears.  The re-read here hits L2: the same rows
 // were just fetched by the glds staging.
+// second launch_bounds arg pins >=4 waves/SIMD (VGPR cap 128): the DB
+// variant otherwise allocates 134 VGPRs and drops a whole workgroup of
+// block-level overlap per CU (measured -20% end-to-end).
 template <bool SPLIT, bool DB>
 __global__ __launch_bounds__(DW_THREADS)
 void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
@@ -157,28 +160,27 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
   const bool a_full = bn0 + DW_BN <= N;
   const bool b_full = bk0 + DW_BK <= K;
 
-  // bias-grad accumulators: thread t covers dY cols bn0+c8..+8, rows
-  // p*16 + t/16 of each 64-m block (the dw_stage access pattern).
+  // bias-grad accumulators: thread t covers 4 dY cols (bn0 + (t&31)*4),
+  // 8 token rows per 64-m block — 4 fp32 regs so the DB variant stays
+  // under the 4-workgroup VGPR budget (8 regs spilled or dropped a WG).
   const bool do_db = DB && bk0 == 0;
-  const int db_c8 = (threadIdx.x & 15) << 3;
-  float db_acc[8];
-#pragma unroll
-  for (int j = 0; j < 8; ++j) db_acc[j] = 0.f;
+  const int db_c4 = (threadIdx.x & 31) << 2;
+  float db_acc[4] = {0.f, 0.f, 0.f, 0.f};
 
   for (long m0 = m_lo; m0 < m_hi; m0 += DW_BM) {
     if (do_db) {
 #pragma unroll
-      for (int p = 0; p < 4; ++p) {
-        const long gm = m0 + p * 16 + (threadIdx.x >> 4);
-        if (gm < m_hi && bn0 + db_c8 + 8 <= N) {
-          s16x8 v = *(const s16x8*)(dY + gm * N + bn0 + db_c8);
+      for (int p = 0; p < 8; ++p) {
+        const long gm = m0 + p * 8 + (threadIdx.x >> 5);
+        if (gm < m_hi && bn0 + db_c4 + 4 <= N) {
+          s16x4 v = *(const s16x4*)(dY + gm * N + bn0 + db_c4);
 #pragma unroll
-          for (int j = 0; j < 8; ++j) db_acc[j] += bfbits2f(v[j]);
+          for (int j = 0; j < 4; ++j) db_acc[j] += bfbits2f(v[j]);
         } else if (gm < m_hi) {
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            if (bn0 + db_c8 + j < N)
-              db_acc[j] += bfbits2f(dY[gm * N + bn0 + db_c8 + j]);
+          for (int j = 0; j < 4; ++j)
+            if (bn0 + db_c4 + j < N)
+              db_acc[j] += bfbits2f(dY[gm * N + bn0 + db_c4 + j]);
         }
       }
     }
@@ -215,8 +217,8 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
 
   if (do_db) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j)
-      if (bn0 + db_c8 + j < N) atomicAdd(&DBW[bn0 + db_c8 + j], db_acc[j]);
+    for (int j = 0; j < 4; ++j)
+      if (bn0 + db_c4 + j < N) atomicAdd(&DBW[bn0 + db_c4 + j], db_acc[j]);
   }
 
   // Epilogue: D lane map col = lane&15 -> k, row = (lane>>4)*4 + r -> n.
