@@ -117,19 +117,56 @@ void ln_gb_kernel(const short* __restrict__ dy, const short* __restrict__ s,
                   const float* __restrict__ mean,
                   const float* __restrict__ rstd, float* __restrict__ acc_g,
                   float* __restrict__ acc_b, int R, int D) {
-  const int c = blockIdx.x * 256 + threadIdx.x;
+  // 2 columns per thread (s16x2 loads) x 4-row unroll: the scalar
+  // one-col-per-thread version was latency-bound at ~0.9 TB/s.
+  const int c = (blockIdx.x * 64 + threadIdx.x) * 2;
   if (c >= D) return;
   const long r0 = (long)blockIdx.y * LNGB_ROWS;
   const long r1 = min((long)R, r0 + LNGB_ROWS);
-  float sg = 0.f, sb = 0.f;
-  for (long r = r0; r < r1; ++r) {
-    float dyv = bfbits2f(dy[r * D + c]);
-    float xh = (bfbits2f(s[r * D + c]) - mean[r]) * rstd[r];
-    sg += dyv * xh;
-    sb += dyv;
+  float sg0 = 0.f, sg1 = 0.f, sb0 = 0.f, sb1 = 0.f;
+  const bool pair = (c + 1 < D) && (D % 2 == 0);
+  if (pair) {
+    long r = r0;
+    for (; r + 4 <= r1; r += 4) {
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        s16x2 dv = *(const s16x2*)(dy + (r + u) * D + c);
+        s16x2 sv = *(const s16x2*)(s + (r + u) * D + c);
+        float mu = mean[r + u], rs = rstd[r + u];
+        float d0 = bfbits2f(dv[0]), d1 = bfbits2f(dv[1]);
+        sg0 += d0 * (bfbits2f(sv[0]) - mu) * rs;
+        sg1 += d1 * (bfbits2f(sv[1]) - mu) * rs;
+        sb0 += d0;
+        sb1 += d1;
+      }
+    }
+    for (; r < r1; ++r) {
+      s16x2 dv = *(const s16x2*)(dy + r * D + c);
+      s16x2 sv = *(const s16x2*)(s + r * D + c);
+      float mu = mean[r], rs = rstd[r];
+      float d0 = bfbits2f(dv[0]), d1 = bfbits2f(dv[1]);
+      sg0 += d0 * (bfbits2f(sv[0]) - mu) * rs;
+      sg1 += d1 * (bfbits2f(sv[1]) - mu) * rs;
+      sb0 += d0;
+      sb1 += d1;
+    }
+    atomicAdd(&acc_g[c], sg0);
+    atomicAdd(&acc_b[c], sb0);
+    atomicAdd(&acc_g[c + 1], sg1);
+    atomicAdd(&acc_b[c + 1], sb1);
+  } else {
+    for (int cc = c; cc < min(c + 2, D); ++cc) {
+      float sg = 0.f, sb = 0.f;
+      for (long r = r0; r < r1; ++r) {
+        float dyv = bfbits2f(dy[r * D + cc]);
+        float xh = (bfbits2f(s[r * D + cc]) - mean[r]) * rstd[r];
+        sg += dyv * xh;
+        sb += dyv;
+      }
+      atomicAdd(&acc_g[cc], sg);
+      atomicAdd(&acc_b[cc], sb);
+    }
   }
-  atomicAdd(&acc_g[c], sg);
-  atomicAdd(&acc_b[c], sb);
 }
 
 __global__ void ln_gb_cast_kernel(const float* __restrict__ ag,
@@ -187,8 +224,10 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor s,
       rstd.data_ptr<float>(), (short*)dx.data_ptr(), R, D);
   auto acc_g = torch::zeros({D}, dy.options().dtype(torch::kFloat32));
   auto acc_b = torch::zeros({D}, dy.options().dtype(torch::kFloat32));
-  dim3 gbgrid(cdiv(D, 256), cdiv(R, LNGB_ROWS));
-  ln_gb_kernel<<<gbgrid, 256, 0, stream>>>(
+  // 64-thread blocks: at d_model=512 a 256-thread block grid is only 128
+  // workgroups — half the 256-CU chip idle.
+  dim3 gbgrid(cdiv(cdiv(D, 2), 64), cdiv(R, LNGB_ROWS));
+  ln_gb_kernel<<<gbgrid, 64, 0, stream>>>(
       (const short*)dy.data_ptr(), (const short*)s.data_ptr(),
       mean.data_ptr<float>(), rstd.data_ptr<float>(),
       acc_g.data_ptr<float>(), acc_b.data_ptr<float>(), R, D);
