@@ -26,6 +26,21 @@ __global__ void dropout_fwd_kernel(const short* __restrict__ x,
   long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
   if (i >= n) return;
   const unsigned int thr = (unsigned int)(p * 4294967296.0f);
+  if (i + 4 <= n) {                        // vector path (s16x4 + b32 mask)
+    s16x4 xv = *(const s16x4*)(x + i);
+    s16x4 yv;
+    unsigned int mbits = 0;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      unsigned int r = pcg_hash(seed * 0x9E3779B97F4A7C15ull + (i + j));
+      unsigned char keep = r >= thr;
+      mbits |= (unsigned int)keep << (8 * j);
+      yv[j] = keep ? f2bfbits(bfbits2f(xv[j]) * inv_keep) : (short)0;
+    }
+    *(s16x4*)(y + i) = yv;
+    *(unsigned int*)(mask + i) = mbits;
+    return;
+  }
 #pragma unroll
   for (int j = 0; j < 4; ++j) {
     if (i + j >= n) break;
@@ -42,6 +57,18 @@ __global__ void dropout_bwd_kernel(const short* __restrict__ dy,
                                    float inv_keep) {
   long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
   if (i >= n) return;
+  if (i + 4 <= n) {
+    s16x4 dv = *(const s16x4*)(dy + i);
+    unsigned int mbits = *(const unsigned int*)(mask + i);
+    s16x4 xv;
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      xv[j] = ((mbits >> (8 * j)) & 0xff)
+                  ? f2bfbits(bfbits2f(dv[j]) * inv_keep)
+                  : (short)0;
+    *(s16x4*)(dx + i) = xv;
+    return;
+  }
 #pragma unroll
   for (int j = 0; j < 4; ++j) {
     if (i + j >= n) break;
