@@ -43,6 +43,9 @@ def main():
     p.add_argument("--vocab", type=int, default=32768)
     p.add_argument("--label_smoothing", type=float, default=0.1)
     p.add_argument("--dropout", type=float, default=0.1)
+    p.add_argument("--graph", action="store_true",
+                   help="capture the step into a hipGraph and replay "
+                        "(single-GPU only)")
     args = p.parse_args()
 
     from transformer_amd.models import Transformer
@@ -85,8 +88,21 @@ def main():
 
     batches = [(make_batch(), make_batch()) for _ in range(n_distinct)]
 
+    captured = None
+    if args.graph and world == 1 and use_cuda:
+        from transformer_amd.runtime.graph import CapturedTrainStep
+
+        def loss_fn(real, pred):
+            return ops.masked_cross_entropy(pred, real, global_batch,
+                                            args.label_smoothing)
+
+        captured = CapturedTrainStep(model, opt, loss_fn, (B, S), (B, S),
+                                     device)
+
     def step(i):
         src, tar = batches[i % n_distinct]
+        if captured is not None:
+            return captured(src, tar)
         tar_inp = tar[:, :-1].contiguous()
         tar_real = tar[:, 1:].contiguous()
         logits, _ = model((src, tar_inp), training=True)

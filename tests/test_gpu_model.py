@@ -119,3 +119,33 @@ def test_predict_greedy_gpu(tmp_path):
                ckpt_path=str(tmp_path), d_model=128)
     out = tr.predict("hello world")
     assert out.dim() == 1 and out.shape[0] >= 2
+
+
+def test_graph_captured_step():
+    """Q12: enable_function -> hipGraph capture.  A captured step must (a)
+    run and reduce loss over replays, (b) advance the Noam schedule on
+    device, (c) vary dropout masks across replays (losses not bitwise
+    frozen after the first replay)."""
+    import torch
+    from transformer_amd.models import Transformer
+    from transformer_amd.runtime import NoamAdam
+    from transformer_amd.runtime.graph import CapturedTrainStep
+    from transformer_amd import ops
+
+    torch.manual_seed(7)
+    model = Transformer(num_layers=2, d_model=64, num_heads=2, dff=128,
+                        input_vocab_size=130, target_vocab_size=130,
+                        rate=0.1, max_position=32).cuda().bfloat16()
+    opt = NoamAdam(model, 64, warmup_steps=100, use_flat=True)
+    B, S = 4, 16
+    cap = CapturedTrainStep(model, opt, lambda real, pred:
+                            ops.masked_cross_entropy(pred, real, B, 0.0),
+                            (B, S), (B, S), torch.device("cuda"))
+    src = torch.randint(1, 120, (B, S), device="cuda")
+    tar = torch.randint(1, 120, (B, S), device="cuda")
+    losses = [cap(src, tar).item() for _ in range(8)]
+    assert losses[-1] < losses[0], losses
+    step_t, _ = opt.graph_state()
+    assert int(step_t.item()) == 8 == opt.step_count
+    # same inputs but stochastic dropout: consecutive losses not identical
+    assert len({round(l, 6) for l in losses[2:]}) > 1, losses

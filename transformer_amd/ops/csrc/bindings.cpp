@@ -49,7 +49,8 @@ torch::Tensor embed_pe_bwd(torch::Tensor dy, torch::Tensor tokens,
                            c10::optional<torch::Tensor> out);
 
 std::vector<torch::Tensor> dropout_fwd(torch::Tensor x, double p,
-                                       int64_t seed);
+                                        int64_t seed,
+                                        c10::optional<torch::Tensor> seed_t);
 torch::Tensor dropout_bwd(torch::Tensor dy, torch::Tensor mask, double p);
 
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets,
@@ -58,6 +59,10 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets,
                      torch::Tensor lse, torch::Tensor dloss,
                      double batch_size, double label_smoothing);
 
+void adam_fused_dev(torch::Tensor master, torch::Tensor m, torch::Tensor v,
+                    torch::Tensor grad, torch::Tensor param,
+                    torch::Tensor step, torch::Tensor coefs, double d_model,
+                    double warmup, double beta1, double beta2, double eps);
 void adam_fused(torch::Tensor master, torch::Tensor m, torch::Tensor v,
                 torch::Tensor grad, torch::Tensor param, double lr,
                 double beta1, double beta2, double eps, int64_t step);
@@ -108,11 +113,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embed_pe_bwd", &embed_pe_bwd, pybind11::arg("dy"),
         pybind11::arg("tokens"), pybind11::arg("vocab"),
         pybind11::arg("out") = pybind11::none());
-  m.def("dropout_fwd", &dropout_fwd);
+  m.def("dropout_fwd", &dropout_fwd, pybind11::arg("x"), pybind11::arg("p"),
+        pybind11::arg("seed"), pybind11::arg("seed_t") = pybind11::none());
   m.def("dropout_bwd", &dropout_bwd);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
   m.def("adam_fused", &adam_fused);
+  m.def("adam_fused_dev", &adam_fused_dev,
+        "graph-capturable Adam: device step tensor + in-kernel Noam lr");
   m.def("argmax_lastdim", &argmax_lastdim);
   m.def("accuracy", &accuracy);
 }

@@ -18,13 +18,21 @@ DEV_INLINE unsigned int pcg_hash(unsigned long long key) {
   return x;
 }
 
+// seed_ptr (optional): device step/epoch counter — under HIP-graph capture
+// the host seed would be baked into the graph and every replay would
+// reuse the same mask; reading the counter (advanced in-graph by
+// adam_coefs_kernel) keeps replays stochastic.  `seed` doubles as a
+// per-call-site salt on that path.
 __global__ void dropout_fwd_kernel(const short* __restrict__ x,
                                    short* __restrict__ y,
                                    unsigned char* __restrict__ mask, long n,
                                    float p, float inv_keep,
-                                   unsigned long long seed) {
+                                   unsigned long long seed,
+                                   const long long* __restrict__ seed_ptr) {
   long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
   if (i >= n) return;
+  if (seed_ptr)
+    seed = seed * 0xD1342543DE82EF95ull + (unsigned long long)(*seed_ptr);
   const unsigned int thr = (unsigned int)(p * 4294967296.0f);
   if (i + 4 <= n) {                        // vector path (s16x4 + b32 mask)
     s16x4 xv = *(const s16x4*)(x + i);
@@ -78,17 +86,24 @@ __global__ void dropout_bwd_kernel(const short* __restrict__ dy,
 }
 
 std::vector<torch::Tensor> dropout_fwd(torch::Tensor x, double p,
-                                       int64_t seed) {
+                                       int64_t seed,
+                                       c10::optional<torch::Tensor> seed_t) {
   TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 &&
               x.is_contiguous());
   auto y = torch::empty_like(x);
   auto mask = torch::empty({x.numel()}, x.options().dtype(torch::kUInt8));
   long n = x.numel();
+  const long long* sp = nullptr;
+  if (seed_t.has_value()) {
+    TORCH_CHECK(seed_t->is_cuda() && seed_t->dtype() == torch::kInt64 &&
+                seed_t->numel() == 1);
+    sp = (const long long*)seed_t->data_ptr();
+  }
   auto stream = at::hip::getCurrentHIPStream();
   dropout_fwd_kernel<<<((n + 3) / 4 + 255) / 256, 256, 0, stream>>>(
       (const short*)x.data_ptr(), (short*)y.data_ptr(),
       mask.data_ptr<unsigned char>(), n, (float)p, 1.0f / (1.0f - (float)p),
-      (unsigned long long)seed);
+      (unsigned long long)seed, sp);
   return {y, mask};
 }
 

@@ -426,12 +426,33 @@ def embedding_scale_pe(tokens, weight, pe):
 # Dropout (K11): mask saved for exact backward; seeds from torch RNG.
 # ---------------------------------------------------------------------------
 
+# HIP-graph capture support: while a step is being
+# captured, dropout seeds must come from a DEVICE counter (advanced
+# in-graph by the optimizer's adam_coefs kernel) — a host seed would be
+# baked into the graph and every replay would reuse the same mask.  Each
+# call site gets a distinct salt (capture-order is deterministic).
+_GRAPH_SEED_T = None
+_GRAPH_SALT = [0]
+
+
+def set_graph_rng(seed_tensor):
+    """Install (or clear, with None) the device seed tensor used by dropout
+    during graph capture."""
+    global _GRAPH_SEED_T
+    _GRAPH_SEED_T = seed_tensor
+    _GRAPH_SALT[0] = 0
+
+
 class _DropoutFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, p):
         E = ext()
-        seed = int(torch.randint(0, 2**31 - 1, (1,)).item())
-        y, mask = E.dropout_fwd(x, p, seed)
+        if _GRAPH_SEED_T is not None:
+            _GRAPH_SALT[0] += 1
+            y, mask = E.dropout_fwd(x, p, _GRAPH_SALT[0], _GRAPH_SEED_T)
+        else:
+            seed = int(torch.randint(0, 2**31 - 1, (1,)).item())
+            y, mask = E.dropout_fwd(x, p, seed)
         ctx.p = p
         ctx.save_for_backward(mask)
         return y

@@ -118,6 +118,34 @@ class NoamAdam:
     def lr(self) -> float:
         return self.schedule(max(self.step_count, 1))
 
+    # ---- HIP-graph capture support (Q12) ---------------------------------
+    def graph_state(self):
+        """Device-side (step, coefs) tensors for captured steps.  The step
+        tensor doubles as the dropout seed epoch (ops.set_graph_rng)."""
+        assert self.flat is not None and self.is_cuda
+        if not hasattr(self, "_step_t"):
+            dev = self.flat.flat_w.device
+            self._step_t = torch.tensor([self.step_count], dtype=torch.int64,
+                                        device=dev)
+            self._coefs = torch.zeros(3, dtype=torch.float32, device=dev)
+        return self._step_t, self._coefs
+
+    @torch.no_grad()
+    def step_captured(self):
+        """Graph-capturable optimizer step: the Noam lr and Adam bias
+        corrections are derived ON DEVICE from a step tensor advanced
+        in-graph, so a replayed graph keeps the schedule moving.  The host
+        `step_count` must be advanced by the replay wrapper."""
+        from ..ops import ext as _ext
+        st, cf = self.graph_state()
+        b1, b2 = self.betas
+        self.flat.check()
+        _ext.ext().adam_fused_dev(self.master, self.m, self.v,
+                                  self.flat.flat_g, self.flat.flat_w, st, cf,
+                                  self.schedule.d_model,
+                                  self.schedule.warmup_steps, b1, b2,
+                                  self.eps)
+
     def zero_grad(self, set_to_none: bool = False):
         if self.flat is not None:
             self.flat.zero_grad()

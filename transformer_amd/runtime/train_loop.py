@@ -66,10 +66,35 @@ class Train:
     def _grad_sync(self):
         """Hook point for the DP subclass (bucket finalize)."""
 
+    def _graph_eligible(self) -> bool:
+        """enable_function (Q12) -> HIP-graph capture of the whole step.
+        DP>1 keeps the eager step (collectives are not captured)."""
+        import torch.distributed as dist
+        return (self.enable_function and self.device.type == "cuda"
+                and self.optimizer.flat is not None
+                and not (dist.is_available() and dist.is_initialized()
+                         and dist.get_world_size() > 1))
+
+    def _captured_step(self, src, tar):
+        cap = getattr(self, "_captured", None)
+        if cap is None or not cap.fits(src, tar):
+            from .graph import CapturedTrainStep
+            cap = CapturedTrainStep(self.transformer, self.optimizer,
+                                    self.loss_function, tuple(src.shape),
+                                    tuple(tar.shape), self.device)
+            self._captured = cap
+        loss = cap(src, tar)
+        self.train_loss.update(loss.item())
+        self.train_accuracy.update(ops.masked_accuracy(cap.logits,
+                                                       cap.tar_real))
+        return loss
+
     def train_step(self, inputs):
         src, tar = inputs
         src = src.to(self.device, non_blocking=True)
         tar = tar.to(self.device, non_blocking=True)
+        if self._graph_eligible():
+            return self._captured_step(src, tar)
         tar_inp, tar_real = tar[:, :-1].contiguous(), tar[:, 1:].contiguous()
 
         predictions, _ = self.transformer((src, tar_inp), training=True)
