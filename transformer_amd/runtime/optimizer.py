@@ -136,11 +136,11 @@ class NoamAdam:
         corrections are derived ON DEVICE from a step tensor advanced
         in-graph, so a replayed graph keeps the schedule moving.  The host
         `step_count` must be advanced by the replay wrapper."""
-        from ..ops import ext as _ext
+        from ..ops import ext as _ext_fn
         st, cf = self.graph_state()
         b1, b2 = self.betas
         self.flat.check()
-        _ext.ext().adam_fused_dev(self.master, self.m, self.v,
+        _ext_fn().adam_fused_dev(self.master, self.m, self.v,
                                   self.flat.flat_g, self.flat.flat_w, st, cf,
                                   self.schedule.d_model,
                                   self.schedule.warmup_steps, b1, b2,
