@@ -56,6 +56,29 @@ DEV_INLINE void lds_write1(short* base, int row, int col, short v) {
   *(short*)((char*)base + (row * ROWB + ((col * 2) ^ ((ROWB >= 128 ? (row & 7) : (row & 3)) << 4)))) = v;
 }
 
+
+// Read an MFMA B-fragment COLUMN-wise from a natural [rows][DH] image via
+// 2x ds_read_b64_tr_b16 (hardware transpose read, guide T10; lane
+// semantics verified by tools/tr16_probe.hip): lane l receives
+// T[kb .. kb+8)[col0 + (l&15)].  Replaces the scalar-write transposed
+// LDS images (per-element ds_write_b16 scatter) entirely.
+typedef short s16x4t __attribute__((ext_vector_type(4)));
+
+template <int ROWB>
+DEV_INLINE bf16x8 lds_read8_tr(const short* base, int kb, int col0) {
+  const int mp = threadIdx.x & 15;
+  const int row = kb + (mp >> 2);
+  const int cb = (col0 + 4 * (mp & 3)) * 2;
+  s16x4t lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (__attribute__((address_space(3))) s16x4t*)(
+          const_cast<char*>((const char*)base) + lds_swz<ROWB>(row, cb)));
+  s16x4t hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (__attribute__((address_space(3))) s16x4t*)(
+          const_cast<char*>((const char*)base) + lds_swz<ROWB>(row + 4, cb)));
+  s16x8 v = {lo[0], lo[1], lo[2], lo[3], hi[0], hi[1], hi[2], hi[3]};
+  return (bf16x8)v;
+}
+
 // Stage a [KVT][DH] bf16 tile from global rows (stride row_stride elems)
 // into swizzled LDS: natural layout into lds_n (if WRITE_N) and/or the
 // transpose [DH][KVT] into lds_t (if WRITE_T).
@@ -93,7 +116,7 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
   constexpr int D32 = DH / 32;   // QK^T MFMA k-steps
   constexpr int D16 = DH / 16;   // O fragments
   __shared__ short k_lds[KVT * DH];
-  __shared__ short vt_lds[DH * KVT];
+  __shared__ short v_lds[KVT * DH];
   __shared__ short p_lds[WAVES][QW * KVT];
 
   const int bh = blockIdx.x;      // b*H + h
@@ -132,8 +155,8 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
     const int kc = min(KVT, Sk - k0);
     stage_kv<DH, true, false>(Kp + (long)k0 * kv_rs, kv_rs, kc,
                               k_lds, nullptr);
-    stage_kv<DH, false, true>(Vp + (long)k0 * kv_rs, kv_rs, kc,
-                              nullptr, vt_lds);
+    stage_kv<DH, true, false>(Vp + (long)k0 * kv_rs, kv_rs, kc,
+                              v_lds, nullptr);
     __syncthreads();
 
     // ---- S = scale*(Q K^T) + mask, two 16-key halves -----------------
@@ -209,14 +232,16 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
         lds_write1<KVT * 2>(p_lds[wid], kg * 4 + r, half * 16 + fr,
                             f2bfbits(p_raw[half][r]));
     }
-    __syncthreads();  // vt_lds ready + p_lds visible to own wave
+    // p_lds is per-wave: a wave barrier (not a workgroup barrier) orders
+    // the scalar P writes against this wave's own PV reads.
+    __builtin_amdgcn_wave_barrier();
 
     // ---- O += P · V  (A = P[q][key] from LDS, B = V^T[d][key]) -------
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int i = 0; i < D16; ++i) {
       bf16x8 pa = lds_read8<KVT * 2>(p_lds[wid], fr, kg * 8);
-      bf16x8 vb = lds_read8<KVT * 2>(vt_lds, i * 16 + fr, kg * 8);
+      bf16x8 vb = lds_read8_tr<DH * 2>(v_lds, kg * 8, i * 16);
       acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, acc[i], 0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
@@ -292,8 +317,6 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
   constexpr int D16 = DH / 16;
   __shared__ short q_lds[KVT * DH];    // q-tile natural [32][DH]
   __shared__ short do_lds[KVT * DH];   // dO-tile natural
-  __shared__ short qt_lds[DH * KVT];   // q-tile transposed
-  __shared__ short dot_lds[DH * KVT];  // dO-tile transposed
   __shared__ short x_lds[WAVES][QW * KVT];  // per-wave P^T / dS^T scratch
 
   const int bh = blockIdx.x;
@@ -339,10 +362,10 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
 
   for (int j0 = q_start; j0 < Sq; j0 += KVT) {
     const int jc = min(KVT, Sq - j0);
-    stage_kv<DH, true, true>(Qp + (long)j0 * q_rs, q_rs, jc,
-                             q_lds, qt_lds);
-    stage_kv<DH, true, true>(dOp + (long)j0 * do_rs, do_rs, jc,
-                             do_lds, dot_lds);
+    stage_kv<DH, true, false>(Qp + (long)j0 * q_rs, q_rs, jc,
+                              q_lds, nullptr);
+    stage_kv<DH, true, false>(dOp + (long)j0 * do_rs, do_rs, jc,
+                              do_lds, nullptr);
     __syncthreads();
 
     // pass 1: both halves of P^T into x_lds; dS^T kept in registers
@@ -381,7 +404,7 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
 #pragma unroll
     for (int i = 0; i < D16; ++i) {
       bf16x8 pa = lds_read8<KVT * 2>(x_lds[wid], fr, kg * 8);
-      bf16x8 db = lds_read8<KVT * 2>(dot_lds, i * 16 + fr, kg * 8);
+      bf16x8 db = lds_read8_tr<DH * 2>(do_lds, kg * 8, i * 16);
       acc_dv[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, db, acc_dv[i], 0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
@@ -397,7 +420,7 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
 #pragma unroll
     for (int i = 0; i < D16; ++i) {
       bf16x8 sa = lds_read8<KVT * 2>(x_lds[wid], fr, kg * 8);
-      bf16x8 qb2 = lds_read8<KVT * 2>(qt_lds, i * 16 + fr, kg * 8);
+      bf16x8 qb2 = lds_read8_tr<DH * 2>(q_lds, kg * 8, i * 16);
       acc_dk[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(sa, qb2, acc_dk[i], 0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
@@ -437,7 +460,6 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
   constexpr int D16 = DH / 16;
   __shared__ short k_lds[KVT * DH];
   __shared__ short v_lds[KVT * DH];
-  __shared__ short kt_lds[DH * KVT];
   __shared__ short x_lds[WAVES][QW * KVT];
 
   const int bh = blockIdx.x;
@@ -477,8 +499,8 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
 
   for (int k0 = 0; k0 < kend; k0 += KVT) {
     const int kc = min(KVT, Sk - k0);
-    stage_kv<DH, true, true>(Kp + (long)k0 * kv_rs, kv_rs, kc,
-                             k_lds, kt_lds);
+    stage_kv<DH, true, false>(Kp + (long)k0 * kv_rs, kv_rs, kc,
+                              k_lds, nullptr);
     stage_kv<DH, true, false>(Vp + (long)k0 * kv_rs, kv_rs, kc,
                               v_lds, nullptr);
     __syncthreads();
@@ -516,7 +538,7 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
 #pragma unroll
     for (int i = 0; i < D16; ++i) {
       bf16x8 sa = lds_read8<KVT * 2>(x_lds[wid], fr, kg * 8);
-      bf16x8 kb2 = lds_read8<KVT * 2>(kt_lds, i * 16 + fr, kg * 8);
+      bf16x8 kb2 = lds_read8_tr<DH * 2>(k_lds, kg * 8, i * 16);
       acc_dq[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(sa, kb2, acc_dq[i], 0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
