@@ -105,7 +105,11 @@ DEV_INLINE void stage_kv(const short* __restrict__ g, long row_stride,
 }
 
 // ---- forward ---------------------------------------------------------------
-template <int DH>
+// TRV: read the PV B-operand from the NATURAL V image via
+// ds_read_b64_tr_b16 (no transposed staging); else build a transposed V
+// image with scalar writes and read it row-wise.  Same buffer either way;
+// the faster variant is chosen by measurement (tools/attn_bench.py).
+template <int DH, bool TRV>
 __global__ __launch_bounds__(256)
 void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
                      const short* __restrict__ V,
@@ -116,7 +120,7 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
   constexpr int D32 = DH / 32;   // QK^T MFMA k-steps
   constexpr int D16 = DH / 16;   // O fragments
   __shared__ short k_lds[KVT * DH];
-  __shared__ short v_lds[KVT * DH];
+  __shared__ short v_lds[KVT * DH];   // natural [KVT][DH] or transposed
   __shared__ short p_lds[WAVES][QW * KVT];
 
   const int bh = blockIdx.x;      // b*H + h
@@ -155,8 +159,12 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
     const int kc = min(KVT, Sk - k0);
     stage_kv<DH, true, false>(Kp + (long)k0 * kv_rs, kv_rs, kc,
                               k_lds, nullptr);
-    stage_kv<DH, true, false>(Vp + (long)k0 * kv_rs, kv_rs, kc,
-                              v_lds, nullptr);
+    if (TRV)
+      stage_kv<DH, true, false>(Vp + (long)k0 * kv_rs, kv_rs, kc,
+                                v_lds, nullptr);
+    else
+      stage_kv<DH, false, true>(Vp + (long)k0 * kv_rs, kv_rs, kc,
+                                nullptr, v_lds);
     __syncthreads();
 
     // ---- S = scale*(Q K^T) + mask, two 16-key halves -----------------
@@ -241,7 +249,8 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
 #pragma unroll
     for (int i = 0; i < D16; ++i) {
       bf16x8 pa = lds_read8<KVT * 2>(p_lds[wid], fr, kg * 8);
-      bf16x8 vb = lds_read8_tr<DH * 2>(v_lds, kg * 8, i * 16);
+      bf16x8 vb = TRV ? lds_read8_tr<DH * 2>(v_lds, kg * 8, i * 16)
+                      : lds_read8<KVT * 2>(v_lds, i * 16 + fr, kg * 8);
       acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, acc[i], 0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
@@ -587,7 +596,8 @@ static void check_attn_batch(const torch::Tensor& t, const char* name) {
 // packed (B,S,3,H,dh) QKV tensor); k and v must share their row stride.
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor kv_pad,
-                                    bool causal, double scale) {
+                                    bool causal, double scale,
+                                    int64_t trv) {
   const int B = q.size(0), Sq = q.size(1), H = q.size(2), DH = q.size(3);
   const int Sk = k.size(1);
   check_attn_view(q, DH, "q");
@@ -605,12 +615,22 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   }
   dim3 grid(B * H, cdiv(Sq, WAVES * QW));
   auto stream = at::hip::getCurrentHIPStream();
-  DISPATCH_DH(DH, attn_fwd_kernel<DHC><<<grid, 256, 0, stream>>>(
+  DISPATCH_DH(DH, {
+    if (trv)
+      attn_fwd_kernel<DHC, true><<<grid, 256, 0, stream>>>(
+          (const short*)q.data_ptr(), (const short*)k.data_ptr(),
+          (const short*)v.data_ptr(), pad, (short*)o.data_ptr(),
+          lse.data_ptr<float>(), B, H, Sq, Sk, causal ? 1 : 0,
+          (float)scale, q.stride(1), k.stride(1), (long)H * DH,
+          q.stride(0), k.stride(0));
+    else
+      attn_fwd_kernel<DHC, false><<<grid, 256, 0, stream>>>(
                   (const short*)q.data_ptr(), (const short*)k.data_ptr(),
                   (const short*)v.data_ptr(), pad, (short*)o.data_ptr(),
                   lse.data_ptr<float>(), B, H, Sq, Sk, causal ? 1 : 0,
                   (float)scale, q.stride(1), k.stride(1), (long)H * DH,
-                  q.stride(0), k.stride(0)));
+                  q.stride(0), k.stride(0));
+  });
   return {o, lse};
 }
 

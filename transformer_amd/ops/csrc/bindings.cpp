@@ -35,7 +35,7 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor s,
 
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor kv_pad,
-                                    bool causal, double scale);
+                                    bool causal, double scale, int64_t trv);
 std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor o,
                                     torch::Tensor dout, torch::Tensor lse,
@@ -104,7 +104,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("gamma"), pybind11::arg("mean"), pybind11::arg("rstd"),
         pybind11::arg("dgamma_out") = pybind11::none(),
         pybind11::arg("dbeta_out") = pybind11::none());
-  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_fwd", &attn_fwd, pybind11::arg("q"), pybind11::arg("k"),
+        pybind11::arg("v"), pybind11::arg("kv_pad"), pybind11::arg("causal"),
+        pybind11::arg("scale"), pybind11::arg("trv") = 1);
   m.def("attn_bwd", &attn_bwd, pybind11::arg("q"), pybind11::arg("k"),
         pybind11::arg("v"), pybind11::arg("o"), pybind11::arg("dout"),
         pybind11::arg("lse"), pybind11::arg("kv_pad"), pybind11::arg("causal"),
