@@ -138,7 +138,16 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
   __shared__ short a_img[DW_BM * DW_BN];  // dY block
   __shared__ short b_img[DW_BM * DW_BK];  // X block
 
-  const int wg = blockIdx.x;
+  // XCD-aware bijective remap: consecutive logical tiles (same n-block,
+  // varying k) land on the SAME XCD, so a dY slice is read into one XCD's
+  // L2 once instead of once per k-tile (dY is the larger operand; the
+  // blocks would otherwise round-robin across all 8 XCDs).
+  int wg = blockIdx.x;
+  {
+    const int nwg = gridDim.x;
+    const int q = nwg / 8, r = nwg % 8, x = wg % 8, o = wg / 8;
+    wg = (x < r ? x * (q + 1) : r * (q + 1) + (x - r) * q) + o;
+  }
   const int bn0 = (wg / nbk) * DW_BN;
   const int bk0 = (wg % nbk) * DW_BK;
   const long m_lo = (long)blockIdx.y * m_per_slice;
