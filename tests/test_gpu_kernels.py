@@ -70,8 +70,9 @@ def test_gemm256_nt(m, n, k, epi):
 
 
 @pytest.mark.parametrize("m,n,k", [
-    (16384, 512, 512),    # split-M accumulation path
-    (4096, 1536, 512),
+    (16384, 512, 512),    # split-M accumulation path (128-tile)
+    (4096, 1536, 512),    # 256-tile path
+    (1000, 2049, 300),    # 256-tile path, ragged n/k/m edges
     (200, 130, 70),       # ragged every dim, single-slice path
     (64, 512, 512),
 ])
