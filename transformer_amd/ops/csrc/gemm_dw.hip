@@ -253,145 +253,6 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
   }
 }
 
-// 256x256 output-tile variant (8 waves, 512 threads, per-wave 128n x 64k,
-// acc[8][4]): halves the cross-tile re-reads of dY/X — the 128-tile kernel
-// measures at ~90% of its HBM traffic bound, so traffic IS the lever for
-// the bigger heads (FFN/QKV/logits dW).  Same LDS image + tr16 reads.
-#define DW2_BN 256
-#define DW2_BK 256
-#define DW2_THREADS 512
-
-DEV_INLINE void dw2_stage_glds(const short* __restrict__ g, long ldg, long m0,
-                               int col0, short* lds) {
-  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
-#pragma unroll
-  for (int p = 0; p < 4; ++p) {
-    const int chunk = p * 8 + wid;           // 32 x 1 KiB = [64m][256col]
-    const int d = chunk * 1024 + lane * 16;
-    const int e = d >> 1;
-    const int sub = e >> 9;
-    const int we = e & 511;
-    const int pr = we >> 4;
-    const int r = ((pr & 3) << 3) | (pr >> 2);
-    const int m = (sub & 1) * 32 + r;
-    const int col = (sub >> 1) * 16 + (we & 15);
-    __builtin_amdgcn_global_load_lds(
-        (const __attribute__((address_space(1))) void*)(g + (m0 + m) * ldg +
-                                                        col0 + col),
-        (__attribute__((address_space(3))) void*)((char*)lds + d), 16, 0, 0);
-  }
-}
-
-// guarded fallback staging for edge blocks (zero-fill)
-DEV_INLINE void dw2_stage(const short* __restrict__ g, long ldg, long m0,
-                          long mmax, int col0, int ncols, short* lds) {
-  const int t = threadIdx.x;
-#pragma unroll
-  for (int p = 0; p < 4; ++p) {
-    int idx = p * DW2_THREADS + t;      // 2048 chunks of 8 shorts
-    int m = idx >> 5;                   // 32 chunks per 256-col row
-    int c8 = (idx & 31) << 3;
-    long gm = m0 + m;
-    s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-    if (gm < mmax) {
-      if (col0 + c8 + 8 <= ncols) {
-        v = *(const s16x8*)(g + gm * ldg + col0 + c8);
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          if (col0 + c8 + j < ncols) v[j] = g[gm * ldg + col0 + c8 + j];
-      }
-    }
-    *(s16x4*)&lds[dw_img(m, c8)] = {v[0], v[1], v[2], v[3]};
-    *(s16x4*)&lds[dw_img(m, c8 + 4)] = {v[4], v[5], v[6], v[7]};
-  }
-}
-
-template <bool SPLIT>
-__global__ __launch_bounds__(DW2_THREADS)
-void gemm_dw2_kernel(const short* __restrict__ dY, const short* __restrict__ X,
-                     short* __restrict__ C, float* __restrict__ CW, int M,
-                     int N, int K, long m_per_slice, int nbk) {
-  __shared__ short a_img[DW_BM * DW2_BN];  // 32 KiB
-  __shared__ short b_img[DW_BM * DW2_BK];
-
-  int wg = blockIdx.x;
-  {
-    const int nwg = gridDim.x;
-    const int q = nwg / 8, r = nwg % 8, x = wg % 8, o = wg / 8;
-    wg = (x < r ? x * (q + 1) : r * (q + 1) + (x - r) * q) + o;
-  }
-  const int bn0 = (wg / nbk) * DW2_BN;
-  const int bk0 = (wg % nbk) * DW2_BK;
-  const long m_lo = (long)blockIdx.y * m_per_slice;
-  const long m_hi = min((long)M, m_lo + m_per_slice);
-
-  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  const int wn = (wid >> 2) * 128;   // wave rows (n): 2 x 128
-  const int wk = (wid & 3) * 64;     // wave cols (k): 4 x 64
-  const int fr = lane & 15;
-  const int kg = lane >> 4;
-
-  f32x4 acc[8][4];
-#pragma unroll
-  for (int i = 0; i < 8; ++i)
-#pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
-
-  const bool a_full = bn0 + DW2_BN <= N;
-  const bool b_full = bk0 + DW2_BK <= K;
-
-  for (long m0 = m_lo; m0 < m_hi; m0 += DW_BM) {
-    const bool m_full = m0 + DW_BM <= m_hi;
-    if (m_full && a_full)
-      dw2_stage_glds(dY, N, m0, bn0, a_img);
-    else
-      dw2_stage(dY, N, m0, m_hi, bn0, N, a_img);
-    if (m_full && b_full)
-      dw2_stage_glds(X, K, m0, bk0, b_img);
-    else
-      dw2_stage(X, K, m0, m_hi, bk0, K, b_img);
-    __syncthreads();
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      const int ms = ks * 32 + kg * 8;
-      bf16x8d af[8], bf_[4];
-#pragma unroll
-      for (int i = 0; i < 8; ++i)
-        af[i] = dw_frag(a_img, ms, wn + i * 16);
-#pragma unroll
-      for (int j = 0; j < 4; ++j)
-        bf_[j] = dw_frag(b_img, ms, wk + j * 16);
-#pragma unroll
-      for (int i = 0; i < 8; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[i], bf_[j], acc[i][j], 0, 0, 0);
-    }
-    __syncthreads();
-  }
-
-#pragma unroll
-  for (int i = 0; i < 8; ++i) {
-    const int nrow_base = bn0 + wn + i * 16 + kg * 4;
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const int kcol = bk0 + wk + j * 16 + fr;
-      if (kcol >= K) continue;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int nrow = nrow_base + r;
-        if (nrow >= N) continue;
-        if (SPLIT)
-          atomicAdd(&CW[(long)nrow * K + kcol], acc[i][j][r]);
-        else
-          C[(long)nrow * K + kcol] = f2bfbits(acc[i][j][r]);
-      }
-    }
-  }
-}
-
 __global__ void dw_finalize_kernel(const float* __restrict__ cw,
                                    short* __restrict__ c, long nk) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -422,12 +283,12 @@ torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x,
   } else {
     c = torch::empty({N, K}, dy.options());
   }
-  // Tile-size choice: the 256x256 8-wave kernel halves dY/X re-reads
-  // (these shapes are HBM-traffic-bound) but quarters the tile count, so
-  // it needs enough 256-tiles to still fill the chip via M-slices.
-  const bool big = (long)cdiv(N, DW2_BN) * cdiv(K, DW2_BK) >= 8;
-  const int BN_ = big ? DW2_BN : DW_BN, BK_ = big ? DW2_BK : DW_BK;
-  const int nbn = cdiv(N, BN_), nbk = cdiv(K, BK_);
+  // Tile-size choice: the 256x256 8-wave kernel halves dY/X re-reads but
+  // MEASURED SLOWER at every model shape (tools/gemm_bench.py: QKV dW
+  // 0.155 vs 0.081 ms, FFN1 0.149 vs 0.076) — the extra split-M slices'
+  // fp32 atomic traffic and the 8-wave barrier cost exceed the re-read
+  // savings, so the 128-tile kernel is the only dispatch target.
+  const int nbn = cdiv(N, DW_BN), nbk = cdiv(K, DW_BK);
   const int ntiles = nbn * nbk;
   // slices: fill >=512 WGs, 64-token quanta
   int nslices = (int)min((M + DW_BM - 1) / DW_BM, (long)cdiv(512, ntiles));
@@ -446,22 +307,14 @@ torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x,
     dbw_p = dbw.data_ptr<float>();
   }
   auto launch = [&](auto split, auto dbc, float* cwp) {
-    if (big)
-      gemm_dw2_kernel<decltype(split)::value>
-          <<<grid, DW2_THREADS, 0, stream>>>(
-              (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
-              (short*)c.data_ptr(), cwp, (int)M, N, K, m_per_slice, nbk);
-    else
-      gemm_dw_kernel<decltype(split)::value, decltype(dbc)::value>
-          <<<grid, DW_THREADS, 0, stream>>>(
-              (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
-              (short*)c.data_ptr(), cwp, dbw_p, (int)M, N, K, m_per_slice,
-              nbk);
+    gemm_dw_kernel<decltype(split)::value, decltype(dbc)::value>
+        <<<grid, DW_THREADS, 0, stream>>>(
+            (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
+            (short*)c.data_ptr(), cwp, dbw_p, (int)M, N, K, m_per_slice,
+            nbk);
   };
   using T = std::true_type;
   using F = std::false_type;
-  TORCH_CHECK(!(big && has_db),
-              "gemm_dw: fused db not supported on the 256-tile path");
   if (nslices == 1) {
     if (has_db) launch(F{}, T{}, nullptr); else launch(F{}, F{}, nullptr);
   } else {
