@@ -14,7 +14,8 @@ torch::Tensor gemm128_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
                          int64_t epilogue, c10::optional<torch::Tensor> out);
 torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x,
                       c10::optional<torch::Tensor> out,
-                      c10::optional<torch::Tensor> db);
+                      c10::optional<torch::Tensor> db,
+                      c10::optional<torch::Tensor> cw_cached);
 torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b,
                       c10::optional<torch::Tensor> out);
 torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b,
@@ -80,7 +81,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "optional fused bias grad db[N]",
         pybind11::arg("dy"), pybind11::arg("x"),
         pybind11::arg("out") = pybind11::none(),
-        pybind11::arg("db") = pybind11::none());
+        pybind11::arg("db") = pybind11::none(),
+        pybind11::arg("cw") = pybind11::none());
   m.def("gemm128_nt", &gemm128_nt, "128x128 NT path, no 256 dispatch (A/B)",
         pybind11::arg("a"), pybind11::arg("w"), pybind11::arg("bias"),
         pybind11::arg("epilogue"), pybind11::arg("out") = pybind11::none());

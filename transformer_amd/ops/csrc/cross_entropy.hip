@@ -36,57 +36,36 @@ void ce_fwd_kernel(const short* __restrict__ logits,
   const short* lrow = logits + row * V;
   const int t = threadIdx.x;
 
-  // ONE pass over the 32k-vocab row (it was the whole row twice — ~2x the
-  // HBM traffic of this logits-bound kernel): per-thread online max/sum
-  // with per-8-chunk rescale, then an (m, s) pair merge across lanes/waves.
   float mx = -1e30f, sume = 0.f, sumx = 0.f;
   for (int c = t * 8; c < V; c += 256 * 8) {
-    float xs[8];
-    int nv = 0;
     if (c + 8 <= V) {
       s16x8 v = *(const s16x8*)(lrow + c);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) xs[j] = bfbits2f(v[j]);
-      nv = 8;
+      for (int j = 0; j < 8; ++j) mx = fmaxf(mx, bfbits2f(v[j]));
     } else {
-      for (int j = 0; c + j < V; ++j) xs[nv++] = bfbits2f(lrow[c + j]);
-    }
-    float cm = -1e30f;
-    for (int j = 0; j < nv; ++j) cm = fmaxf(cm, xs[j]);
-    if (cm > mx) {
-      sume *= __expf(mx - cm);
-      mx = cm;
-    }
-    for (int j = 0; j < nv; ++j) {
-      sume += __expf(xs[j] - mx);
-      sumx += xs[j];
+      for (int j = 0; c + j < V; ++j) mx = fmaxf(mx, bfbits2f(lrow[c + j]));
     }
   }
-  // merge (mx, sume) across the wave, then across the 4 waves
-  {
-    const int lane = threadIdx.x & 63;
-    (void)lane;
+  mx = block_reduce(mx, scratch, 0);
+  for (int c = t * 8; c < V; c += 256 * 8) {
+    if (c + 8 <= V) {
+      s16x8 v = *(const s16x8*)(lrow + c);
 #pragma unroll
-    for (int off = 1; off < 64; off <<= 1) {
-      float m2 = __shfl_xor(mx, off);
-      float s2 = __shfl_xor(sume, off);
-      float mn = fmaxf(mx, m2);
-      sume = sume * __expf(mx - mn) + s2 * __expf(m2 - mn);
-      mx = mn;
+      for (int j = 0; j < 8; ++j) {
+        float x = bfbits2f(v[j]);
+        sume += __expf(x - mx);
+        sumx += x;
+      }
+    } else {
+      for (int j = 0; c + j < V; ++j) {
+        float x = bfbits2f(lrow[c + j]);
+        sume += __expf(x - mx);
+        sumx += x;
+      }
     }
-    __shared__ float sm[4], ss[4];
-    const int wid = threadIdx.x >> 6;
-    if ((threadIdx.x & 63) == 0) {
-      sm[wid] = mx;
-      ss[wid] = sume;
-    }
-    __syncthreads();
-    float M = fmaxf(fmaxf(sm[0], sm[1]), fmaxf(sm[2], sm[3]));
-    sume = ss[0] * __expf(sm[0] - M) + ss[1] * __expf(sm[1] - M) +
-           ss[2] * __expf(sm[2] - M) + ss[3] * __expf(sm[3] - M);
-    mx = M;
-    __syncthreads();
   }
+  sume = block_reduce(sume, scratch, 1);
+  __syncthreads();
   sumx = block_reduce(sumx, scratch, 1);
   const float lse = mx + __logf(sume);
   if (t == 0) {

@@ -354,8 +354,10 @@ void transpose2d_kernel(const short* __restrict__ in, short* __restrict__ out,
 }
 
 // colsum: db[N] = sum_m dy[M,N] (fp32 accumulate, bf16 out).
-// Two-stage: blocks cover (col-chunk, row-chunk) so the whole chip
-// participates; fp32 atomics into a zeroed workspace, then cast.
+// Two-stage WITHOUT a zeroed workspace: blocks cover (col-chunk,
+// row-chunk) and WRITE disjoint partials [gy][N]; the cast pass sums the
+// gy partials.  (The previous atomic-into-zeros scheme cost a fill_
+// launch per call — ~200 pure-zeroing launches per training step.)
 #define COLSUM_ROWS 128
 __global__ __launch_bounds__(256)
 void colsum_part_kernel(const short* __restrict__ dy, float* __restrict__ acc,
@@ -374,13 +376,16 @@ void colsum_part_kernel(const short* __restrict__ dy, float* __restrict__ acc,
     s3 += bfbits2f(dy[(m + 3) * N + n]);
   }
   for (; m < m1; ++m) s0 += bfbits2f(dy[m * N + n]);
-  atomicAdd(&acc[n], (s0 + s1) + (s2 + s3));
+  acc[(long)blockIdx.y * N + n] = (s0 + s1) + (s2 + s3);
 }
 
 __global__ void cast_colsum_kernel(const float* __restrict__ in,
-                                   short* __restrict__ out, int n) {
+                                   short* __restrict__ out, int n, int gy) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < n) out[i] = f2bfbits(in[i]);
+  if (i >= n) return;
+  float s = 0.f;
+  for (int g = 0; g < gy; ++g) s += in[(long)g * n + i];
+  out[i] = f2bfbits(s);
 }
 
 // relu_bwd: dz = dy * (y > 0)
@@ -558,7 +563,9 @@ torch::Tensor transpose2d(torch::Tensor a) {
 torch::Tensor colsum(torch::Tensor a, c10::optional<torch::Tensor> out_opt) {
   CHECK_BF16_2D(a);
   int M = a.size(0), N = a.size(1);
-  auto acc = torch::zeros({N}, a.options().dtype(torch::kFloat32));
+  const int gy = cdiv(M, COLSUM_ROWS);
+  auto acc = torch::empty({(long)gy * N},
+                          a.options().dtype(torch::kFloat32));
   torch::Tensor out;
   if (out_opt.has_value()) {
     out = *out_opt;
@@ -568,11 +575,11 @@ torch::Tensor colsum(torch::Tensor a, c10::optional<torch::Tensor> out_opt) {
     out = torch::empty({N}, a.options());
   }
   auto stream = at::hip::getCurrentHIPStream();
-  dim3 grid(cdiv(N, 256), cdiv(M, COLSUM_ROWS));
+  dim3 grid(cdiv(N, 256), gy);
   colsum_part_kernel<<<grid, 256, 0, stream>>>(
       (const short*)a.data_ptr(), acc.data_ptr<float>(), M, N);
   cast_colsum_kernel<<<cdiv(N, 256), 256, 0, stream>>>(
-      acc.data_ptr<float>(), (short*)out.data_ptr(), N);
+      acc.data_ptr<float>(), (short*)out.data_ptr(), N, gy);
   return out;
 }
 

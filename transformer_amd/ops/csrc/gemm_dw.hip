@@ -253,10 +253,16 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
   }
 }
 
-__global__ void dw_finalize_kernel(const float* __restrict__ cw,
+// CLEAR: re-zero the workspace after reading so a cached workspace is
+// ready for the next same-shape call without a separate fill_ launch.
+template <bool CLEAR>
+__global__ void dw_finalize_kernel(float* __restrict__ cw,
                                    short* __restrict__ c, long nk) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < nk) c[i] = f2bfbits(cw[i]);
+  if (i < nk) {
+    c[i] = f2bfbits(cw[i]);
+    if (CLEAR) cw[i] = 0.f;
+  }
 }
 
 }  // namespace
@@ -266,7 +272,8 @@ __global__ void dw_finalize_kernel(const float* __restrict__ cw,
 // same pass (no separate colsum read of dY).
 torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x,
                       c10::optional<torch::Tensor> out,
-                      c10::optional<torch::Tensor> db) {
+                      c10::optional<torch::Tensor> db,
+                      c10::optional<torch::Tensor> cw_cached) {
   TORCH_CHECK(dy.is_cuda() && dy.dtype() == torch::kBFloat16 &&
               dy.dim() == 2 && dy.is_contiguous(), "gemm_dw: bad dy");
   TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.dim() == 2 &&
@@ -318,15 +325,29 @@ torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x,
   if (nslices == 1) {
     if (has_db) launch(F{}, T{}, nullptr); else launch(F{}, F{}, nullptr);
   } else {
-    auto cw = torch::zeros({N, K}, dy.options().dtype(torch::kFloat32));
+    // caller may pass a cached ALREADY-ZERO workspace (the finalize pass
+    // below re-zeroes it), killing the per-call fill_ launch+write.
+    torch::Tensor cw;
+    bool cached = cw_cached.has_value();
+    if (cached) {
+      cw = *cw_cached;
+      TORCH_CHECK(cw.dtype() == torch::kFloat32 && cw.is_cuda() &&
+                  cw.numel() >= (long)N * K, "gemm_dw: bad cw workspace");
+    } else {
+      cw = torch::zeros({N, K}, dy.options().dtype(torch::kFloat32));
+    }
     if (has_db) launch(T{}, T{}, cw.data_ptr<float>());
     else launch(T{}, F{}, cw.data_ptr<float>());
     long nk = (long)N * K;
-    dw_finalize_kernel<<<(nk + 1023) / 1024, 1024, 0, stream>>>(
-        cw.data_ptr<float>(), (short*)c.data_ptr(), nk);
+    if (cached)
+      dw_finalize_kernel<true><<<(nk + 1023) / 1024, 1024, 0, stream>>>(
+          cw.data_ptr<float>(), (short*)c.data_ptr(), nk);
+    else
+      dw_finalize_kernel<false><<<(nk + 1023) / 1024, 1024, 0, stream>>>(
+          cw.data_ptr<float>(), (short*)c.data_ptr(), nk);
   }
   if (has_db)
-    dw_finalize_kernel<<<cdiv(N, 256), 256, 0, stream>>>(
+    dw_finalize_kernel<false><<<cdiv(N, 256), 256, 0, stream>>>(
         dbw.data_ptr<float>(), (short*)db->data_ptr(), N);
   return c;
 }

@@ -150,10 +150,11 @@ void ln_gb_kernel(const short* __restrict__ dy, const short* __restrict__ s,
       sb0 += d0;
       sb1 += d1;
     }
-    atomicAdd(&acc_g[c], sg0);
-    atomicAdd(&acc_b[c], sb0);
-    atomicAdd(&acc_g[c + 1], sg1);
-    atomicAdd(&acc_b[c + 1], sb1);
+    const long po = (long)blockIdx.y * D;
+    acc_g[po + c] = sg0;
+    acc_b[po + c] = sb0;
+    acc_g[po + c + 1] = sg1;
+    acc_b[po + c + 1] = sb1;
   } else {
     for (int cc = c; cc < min(c + 2, D); ++cc) {
       float sg = 0.f, sb = 0.f;
@@ -163,21 +164,27 @@ void ln_gb_kernel(const short* __restrict__ dy, const short* __restrict__ s,
         sg += dyv * xh;
         sb += dyv;
       }
-      atomicAdd(&acc_g[cc], sg);
-      atomicAdd(&acc_b[cc], sb);
+      acc_g[(long)blockIdx.y * D + cc] = sg;
+      acc_b[(long)blockIdx.y * D + cc] = sb;
     }
   }
 }
 
+// sums the gy disjoint partials (no zeroed workspace / atomics — the
+// fill_ launches were ~200 per step across colsum/ln_gb/dW).
 __global__ void ln_gb_cast_kernel(const float* __restrict__ ag,
                                   const float* __restrict__ ab,
                                   short* __restrict__ dgamma,
-                                  short* __restrict__ dbeta, int D) {
+                                  short* __restrict__ dbeta, int D, int gy) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < D) {
-    dgamma[i] = f2bfbits(ag[i]);
-    dbeta[i] = f2bfbits(ab[i]);
+  if (i >= D) return;
+  float sg = 0.f, sb = 0.f;
+  for (int g = 0; g < gy; ++g) {
+    sg += ag[(long)g * D + i];
+    sb += ab[(long)g * D + i];
   }
+  dgamma[i] = f2bfbits(sg);
+  dbeta[i] = f2bfbits(sb);
 }
 
 // ---------------------------------------------------------------------------
@@ -222,17 +229,20 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor s,
       (const short*)dy.data_ptr(), (const short*)s.data_ptr(),
       (const short*)gamma.data_ptr(), mean.data_ptr<float>(),
       rstd.data_ptr<float>(), (short*)dx.data_ptr(), R, D);
-  auto acc_g = torch::zeros({D}, dy.options().dtype(torch::kFloat32));
-  auto acc_b = torch::zeros({D}, dy.options().dtype(torch::kFloat32));
+  const int gy = cdiv(R, LNGB_ROWS);
+  auto acc_g = torch::empty({(long)gy * D},
+                            dy.options().dtype(torch::kFloat32));
+  auto acc_b = torch::empty({(long)gy * D},
+                            dy.options().dtype(torch::kFloat32));
   // 64-thread blocks: at d_model=512 a 256-thread block grid is only 128
   // workgroups — half the 256-CU chip idle.
-  dim3 gbgrid(cdiv(cdiv(D, 2), 64), cdiv(R, LNGB_ROWS));
+  dim3 gbgrid(cdiv(cdiv(D, 2), 64), gy);
   ln_gb_kernel<<<gbgrid, 64, 0, stream>>>(
       (const short*)dy.data_ptr(), (const short*)s.data_ptr(),
       mean.data_ptr<float>(), rstd.data_ptr<float>(),
       acc_g.data_ptr<float>(), acc_b.data_ptr<float>(), R, D);
   ln_gb_cast_kernel<<<cdiv(D, 256), 256, 0, stream>>>(
       acc_g.data_ptr<float>(), acc_b.data_ptr<float>(),
-      (short*)dgamma.data_ptr(), (short*)dbeta.data_ptr(), D);
+      (short*)dgamma.data_ptr(), (short*)dbeta.data_ptr(), D, gy);
   return {dx, dgamma, dbeta};
 }
