@@ -7,6 +7,8 @@
 // workspaces with atomics, then cast to bf16.
 #include "common.h"
 
+#include <map>
+
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
@@ -150,11 +152,10 @@ void ln_gb_kernel(const short* __restrict__ dy, const short* __restrict__ s,
       sb0 += d0;
       sb1 += d1;
     }
-    const long po = (long)blockIdx.y * D;
-    acc_g[po + c] = sg0;
-    acc_b[po + c] = sb0;
-    acc_g[po + c + 1] = sg1;
-    acc_b[po + c + 1] = sb1;
+    atomicAdd(&acc_g[c], sg0);
+    atomicAdd(&acc_b[c], sb0);
+    atomicAdd(&acc_g[c + 1], sg1);
+    atomicAdd(&acc_b[c + 1], sb1);
   } else {
     for (int cc = c; cc < min(c + 2, D); ++cc) {
       float sg = 0.f, sb = 0.f;
@@ -164,27 +165,25 @@ void ln_gb_kernel(const short* __restrict__ dy, const short* __restrict__ s,
         sg += dyv * xh;
         sb += dyv;
       }
-      acc_g[(long)blockIdx.y * D + cc] = sg;
-      acc_b[(long)blockIdx.y * D + cc] = sb;
+      atomicAdd(&acc_g[cc], sg);
+      atomicAdd(&acc_b[cc], sb);
     }
   }
 }
 
-// sums the gy disjoint partials (no zeroed workspace / atomics — the
-// fill_ launches were ~200 per step across colsum/ln_gb/dW).
-__global__ void ln_gb_cast_kernel(const float* __restrict__ ag,
-                                  const float* __restrict__ ab,
+// cast pass also re-zeroes the cached fp32 accumulators so the next
+// same-shape call skips the zero-fill launch.
+__global__ void ln_gb_cast_kernel(float* __restrict__ ag,
+                                  float* __restrict__ ab,
                                   short* __restrict__ dgamma,
-                                  short* __restrict__ dbeta, int D, int gy) {
+                                  short* __restrict__ dbeta, int D) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= D) return;
-  float sg = 0.f, sb = 0.f;
-  for (int g = 0; g < gy; ++g) {
-    sg += ag[(long)g * D + i];
-    sb += ab[(long)g * D + i];
+  if (i < D) {
+    dgamma[i] = f2bfbits(ag[i]);
+    dbeta[i] = f2bfbits(ab[i]);
+    ag[i] = 0.f;
+    ab[i] = 0.f;
   }
-  dgamma[i] = f2bfbits(sg);
-  dbeta[i] = f2bfbits(sb);
 }
 
 // ---------------------------------------------------------------------------
@@ -229,20 +228,23 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor s,
       (const short*)dy.data_ptr(), (const short*)s.data_ptr(),
       (const short*)gamma.data_ptr(), mean.data_ptr<float>(),
       rstd.data_ptr<float>(), (short*)dx.data_ptr(), R, D);
-  const int gy = cdiv(R, LNGB_ROWS);
-  auto acc_g = torch::empty({(long)gy * D},
-                            dy.options().dtype(torch::kFloat32));
-  auto acc_b = torch::empty({(long)gy * D},
-                            dy.options().dtype(torch::kFloat32));
+  static std::map<std::pair<int, int>, torch::Tensor> ws_cache;
+  auto wkey = std::make_pair((int)dy.get_device(), D);
+  auto wit = ws_cache.find(wkey);
+  if (wit == ws_cache.end())
+    wit = ws_cache.emplace(wkey, torch::zeros(
+        {2L * D}, dy.options().dtype(torch::kFloat32))).first;
+  auto acc_g = wit->second.narrow(0, 0, D);
+  auto acc_b = wit->second.narrow(0, D, D);
   // 64-thread blocks: at d_model=512 a 256-thread block grid is only 128
   // workgroups — half the 256-CU chip idle.
-  dim3 gbgrid(cdiv(cdiv(D, 2), 64), gy);
+  dim3 gbgrid(cdiv(cdiv(D, 2), 64), cdiv(R, LNGB_ROWS));
   ln_gb_kernel<<<gbgrid, 64, 0, stream>>>(
       (const short*)dy.data_ptr(), (const short*)s.data_ptr(),
       mean.data_ptr<float>(), rstd.data_ptr<float>(),
       acc_g.data_ptr<float>(), acc_b.data_ptr<float>(), R, D);
   ln_gb_cast_kernel<<<cdiv(D, 256), 256, 0, stream>>>(
       acc_g.data_ptr<float>(), acc_b.data_ptr<float>(),
-      (short*)dgamma.data_ptr(), (short*)dbeta.data_ptr(), D, gy);
+      (short*)dgamma.data_ptr(), (short*)dbeta.data_ptr(), D);
   return {dx, dgamma, dbeta};
 }
