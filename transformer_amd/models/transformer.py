@@ -135,11 +135,13 @@ class EncoderLayer(nn.Module):
 
     def forward(self, x, kv_pad, training):
         attn, _ = self.mha(x, kv_pad=kv_pad, causal=False)
-        attn = ops.dropout(attn, self.rate, training)
-        out1 = ops.residual_layernorm(attn, x, self.ln1.gamma, self.ln1.beta, self.ln1.eps)
+        out1 = ops.dropout_residual_layernorm(attn, x, self.ln1.gamma,
+                                              self.ln1.beta, self.rate,
+                                              training, self.ln1.eps)
         ffn = self.ffn(out1)
-        ffn = ops.dropout(ffn, self.rate, training)
-        return ops.residual_layernorm(ffn, out1, self.ln2.gamma, self.ln2.beta, self.ln2.eps)
+        return ops.dropout_residual_layernorm(ffn, out1, self.ln2.gamma,
+                                              self.ln2.beta, self.rate,
+                                              training, self.ln2.eps)
 
 
 class DecoderLayer(nn.Module):
@@ -160,15 +162,18 @@ class DecoderLayer(nn.Module):
                 return_weights=False):
         attn1, w1 = self.mha1(x, kv_pad=tgt_pad, causal=True,
                               return_weights=return_weights)
-        attn1 = ops.dropout(attn1, self.rate, training)
-        out1 = ops.residual_layernorm(attn1, x, self.ln1.gamma, self.ln1.beta, self.ln1.eps)
+        out1 = ops.dropout_residual_layernorm(attn1, x, self.ln1.gamma,
+                                              self.ln1.beta, self.rate,
+                                              training, self.ln1.eps)
         attn2, w2 = self.mha2(out1, enc_output, kv_pad=src_pad,
                               return_weights=return_weights)
-        attn2 = ops.dropout(attn2, self.rate, training)
-        out2 = ops.residual_layernorm(attn2, out1, self.ln2.gamma, self.ln2.beta, self.ln2.eps)
+        out2 = ops.dropout_residual_layernorm(attn2, out1, self.ln2.gamma,
+                                               self.ln2.beta, self.rate,
+                                               training, self.ln2.eps)
         ffn = self.ffn(out2)
-        ffn = ops.dropout(ffn, self.rate, training)
-        out3 = ops.residual_layernorm(ffn, out2, self.ln3.gamma, self.ln3.beta, self.ln3.eps)
+        out3 = ops.dropout_residual_layernorm(ffn, out2, self.ln3.gamma,
+                                               self.ln3.beta, self.rate,
+                                               training, self.ln3.eps)
         return out3, w1, w2
 
 

@@ -53,6 +53,7 @@ from .functional import (  # noqa: E402
     self_attention,
     cross_attention,
     residual_layernorm,
+    dropout_residual_layernorm,
     embedding_scale_pe,
     embedding_scale_pe_at,
     dropout,

@@ -27,12 +27,15 @@ torch::Tensor smoke_add(torch::Tensor a, torch::Tensor b);
 
 std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor res,
                                   torch::Tensor gamma, torch::Tensor beta,
-                                  double eps);
+                                  double eps, double p, int64_t seed,
+                                  c10::optional<torch::Tensor> seed_t);
 std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor s,
                                   torch::Tensor gamma, torch::Tensor mean,
                                   torch::Tensor rstd,
                                   c10::optional<torch::Tensor> dgamma_out,
-                                  c10::optional<torch::Tensor> dbeta_out);
+                                  c10::optional<torch::Tensor> dbeta_out,
+                                  c10::optional<torch::Tensor> mask,
+                                  double p);
 
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor kv_pad,
@@ -101,11 +104,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("out") = pybind11::none());
   m.def("relu_bwd", &relu_bwd);
   m.def("smoke_add", &smoke_add);
-  m.def("ln_fwd", &ln_fwd);
+  m.def("ln_fwd", &ln_fwd, pybind11::arg("x"), pybind11::arg("res"),
+        pybind11::arg("gamma"), pybind11::arg("beta"), pybind11::arg("eps"),
+        pybind11::arg("p") = 0.0, pybind11::arg("seed") = 0,
+        pybind11::arg("seed_t") = pybind11::none());
   m.def("ln_bwd", &ln_bwd, pybind11::arg("dy"), pybind11::arg("s"),
         pybind11::arg("gamma"), pybind11::arg("mean"), pybind11::arg("rstd"),
         pybind11::arg("dgamma_out") = pybind11::none(),
-        pybind11::arg("dbeta_out") = pybind11::none());
+        pybind11::arg("dbeta_out") = pybind11::none(),
+        pybind11::arg("mask") = pybind11::none(),
+        pybind11::arg("p") = 0.0);
   m.def("attn_fwd", &attn_fwd, pybind11::arg("q"), pybind11::arg("k"),
         pybind11::arg("v"), pybind11::arg("kv_pad"), pybind11::arg("causal"),
         pybind11::arg("scale"), pybind11::arg("trv") = 1);

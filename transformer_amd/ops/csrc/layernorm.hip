@@ -12,37 +12,72 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
-// forward: y = LN(x+res)*gamma+beta; saves s=x+res (bf16), mean & rstd (f32)
+DEV_INLINE unsigned int ln_pcg(unsigned long long key) {
+  key = key * 6364136223846793005ull + 1442695040888963407ull;
+  unsigned int h = (unsigned int)((key ^ (key >> 33)) >> 11);
+  h ^= h >> 16; h *= 0x7feb352dU; h ^= h >> 15; h *= 0x846ca68bU;
+  h ^= h >> 16;
+  return h;
+}
+
+// forward: y = LN(drop(x)+res)*gamma+beta; saves s=drop(x)+res (bf16),
+// mean & rstd (f32).  DROP fuses the sublayer dropout (reference
+// Encoder.py:22 `dropout(attn)` feeding `layernorm(x + .)`) — one byte
+// mask per element for the exact backward; DROP=false is plain
+// residual+LN (inference / rate 0).
+template <bool DROP>
 __global__ __launch_bounds__(256)
 void ln_fwd_kernel(const short* __restrict__ x, const short* __restrict__ res,
                    const short* __restrict__ gamma,
                    const short* __restrict__ beta, short* __restrict__ y,
                    short* __restrict__ s, float* __restrict__ mean_out,
-                   float* __restrict__ rstd_out, int R, int D, float eps) {
+                   float* __restrict__ rstd_out, int R, int D, float eps,
+                   unsigned char* __restrict__ mask, float p, float inv_keep,
+                   unsigned long long seed,
+                   const long long* __restrict__ seed_ptr) {
+  unsigned int thr = 0;
+  if (DROP) {
+    if (seed_ptr)
+      seed = seed * 0xD1342543DE82EF95ull + (unsigned long long)(*seed_ptr);
+    thr = (unsigned int)(p * 4294967296.0f);
+  }
   const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
   if (row >= R) return;
   const int lane = threadIdx.x & 63;
   const long base = (long)row * D;
 
   float sum = 0.f, sumsq = 0.f;
-  // pass 1: s = x + res, accumulate stats
+  // pass 1: s = drop(x) + res, accumulate stats
   for (int c = lane * 8; c < D; c += WAVE * 8) {
-    float v[8];
     if (c + 8 <= D) {
       s16x8 xv = *(const s16x8*)(x + base + c);
       s16x8 rv = *(const s16x8*)(res + base + c);
       s16x8 sv;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        v[j] = bfbits2f(xv[j]) + bfbits2f(rv[j]);
-        sv[j] = f2bfbits(v[j]);
-        sum += v[j];
-        sumsq += v[j] * v[j];
+        float xj = bfbits2f(xv[j]);
+        if (DROP) {
+          unsigned char keep =
+              ln_pcg(seed * 0x9E3779B97F4A7C15ull + (base + c + j)) >= thr;
+          mask[base + c + j] = keep;
+          xj = keep ? xj * inv_keep : 0.f;
+        }
+        float t = xj + bfbits2f(rv[j]);
+        sv[j] = f2bfbits(t);
+        sum += t;
+        sumsq += t * t;
       }
       *(s16x8*)(s + base + c) = sv;
     } else {
       for (int j = 0; c + j < D; ++j) {
-        float t = bfbits2f(x[base + c + j]) + bfbits2f(res[base + c + j]);
+        float xj = bfbits2f(x[base + c + j]);
+        if (DROP) {
+          unsigned char keep =
+              ln_pcg(seed * 0x9E3779B97F4A7C15ull + (base + c + j)) >= thr;
+          mask[base + c + j] = keep;
+          xj = keep ? xj * inv_keep : 0.f;
+        }
+        float t = xj + bfbits2f(res[base + c + j]);
         s[base + c + j] = f2bfbits(t);
         sum += t;
         sumsq += t * t;
@@ -80,13 +115,17 @@ void ln_fwd_kernel(const short* __restrict__ x, const short* __restrict__ res,
 }
 
 // backward: dx = rstd*(g - mean(g) - xhat*mean(g*xhat)), g = dy*gamma;
-// dgamma += dy*xhat, dbeta += dy (atomics into fp32 workspace)
+// dgamma += dy*xhat, dbeta += dy (atomics into fp32 workspace).
+// DROP: additionally writes dxm = dx*mask*inv_keep — the gradient through
+// the fused sublayer dropout (dx itself is the residual-branch gradient).
+template <bool DROP>
 __global__ __launch_bounds__(256)
 void ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ s,
                    const short* __restrict__ gamma,
                    const float* __restrict__ mean,
                    const float* __restrict__ rstd, short* __restrict__ dx,
-                   int R, int D) {
+                   int R, int D, const unsigned char* __restrict__ mask,
+                   short* __restrict__ dxm, float inv_keep) {
   const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
   if (row >= R) return;
   const int lane = threadIdx.x & 63;
@@ -106,7 +145,10 @@ void ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ s,
   for (int c = lane; c < D; c += WAVE) {
     float xh = (bfbits2f(s[base + c]) - mu) * rs;
     float g = bfbits2f(dy[base + c]) * bfbits2f(gamma[c]);
-    dx[base + c] = f2bfbits(rs * (g - sg - xh * sgx));
+    float d = rs * (g - sg - xh * sgx);
+    dx[base + c] = f2bfbits(d);
+    if (DROP)
+      dxm[base + c] = mask[base + c] ? f2bfbits(d * inv_keep) : (short)0;
   }
 }
 
@@ -187,9 +229,13 @@ __global__ void ln_gb_cast_kernel(float* __restrict__ ag,
 }
 
 // ---------------------------------------------------------------------------
+// p > 0: dropout on x fused before the residual add; returns an extra
+// byte mask tensor.  seed_t (optional int64[1] device counter) keeps
+// HIP-graph replays stochastic, as in dropout_fwd.
 std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor res,
                                   torch::Tensor gamma, torch::Tensor beta,
-                                  double eps) {
+                                  double eps, double p, int64_t seed,
+                                  c10::optional<torch::Tensor> seed_t) {
   TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.dim() == 2 &&
               x.is_contiguous() && res.is_contiguous());
   const int R = x.size(0), D = x.size(1);
@@ -198,19 +244,37 @@ std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor res,
   auto mean = torch::empty({R}, x.options().dtype(torch::kFloat32));
   auto rstd = torch::empty({R}, x.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
-  ln_fwd_kernel<<<cdiv(R, 4), 256, 0, stream>>>(
+  if (p > 0.0) {
+    auto mask = torch::empty({(long)R * D}, x.options().dtype(torch::kUInt8));
+    const long long* sp = nullptr;
+    if (seed_t.has_value()) sp = (const long long*)seed_t->data_ptr();
+    ln_fwd_kernel<true><<<cdiv(R, 4), 256, 0, stream>>>(
+        (const short*)x.data_ptr(), (const short*)res.data_ptr(),
+        (const short*)gamma.data_ptr(), (const short*)beta.data_ptr(),
+        (short*)y.data_ptr(), (short*)s.data_ptr(), mean.data_ptr<float>(),
+        rstd.data_ptr<float>(), R, D, (float)eps,
+        mask.data_ptr<unsigned char>(), (float)p,
+        1.0f / (1.0f - (float)p), (unsigned long long)seed, sp);
+    return {y, s, mean, rstd, mask};
+  }
+  ln_fwd_kernel<false><<<cdiv(R, 4), 256, 0, stream>>>(
       (const short*)x.data_ptr(), (const short*)res.data_ptr(),
       (const short*)gamma.data_ptr(), (const short*)beta.data_ptr(),
       (short*)y.data_ptr(), (short*)s.data_ptr(), mean.data_ptr<float>(),
-      rstd.data_ptr<float>(), R, D, (float)eps);
+      rstd.data_ptr<float>(), R, D, (float)eps, nullptr, 0.f, 1.f, 0,
+      nullptr);
   return {y, s, mean, rstd};
 }
 
+// mask given: additionally returns dxm = dx*mask*inv_keep (gradient of
+// the fused sublayer dropout input); dx is the residual-branch gradient.
 std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor s,
                                   torch::Tensor gamma, torch::Tensor mean,
                                   torch::Tensor rstd,
                                   c10::optional<torch::Tensor> dgamma_out,
-                                  c10::optional<torch::Tensor> dbeta_out) {
+                                  c10::optional<torch::Tensor> dbeta_out,
+                                  c10::optional<torch::Tensor> mask,
+                                  double p) {
   const int R = dy.size(0), D = dy.size(1);
   auto dx = torch::empty_like(dy);
   auto dest = [&](c10::optional<torch::Tensor>& o) {
@@ -224,10 +288,22 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor s,
   auto dgamma = dest(dgamma_out);
   auto dbeta = dest(dbeta_out);
   auto stream = at::hip::getCurrentHIPStream();
-  ln_bwd_kernel<<<cdiv(R, 4), 256, 0, stream>>>(
-      (const short*)dy.data_ptr(), (const short*)s.data_ptr(),
-      (const short*)gamma.data_ptr(), mean.data_ptr<float>(),
-      rstd.data_ptr<float>(), (short*)dx.data_ptr(), R, D);
+  torch::Tensor dxm;
+  if (mask.has_value()) {
+    dxm = torch::empty_like(dy);
+    ln_bwd_kernel<true><<<cdiv(R, 4), 256, 0, stream>>>(
+        (const short*)dy.data_ptr(), (const short*)s.data_ptr(),
+        (const short*)gamma.data_ptr(), mean.data_ptr<float>(),
+        rstd.data_ptr<float>(), (short*)dx.data_ptr(), R, D,
+        mask->data_ptr<unsigned char>(), (short*)dxm.data_ptr(),
+        1.0f / (1.0f - (float)p));
+  } else {
+    ln_bwd_kernel<false><<<cdiv(R, 4), 256, 0, stream>>>(
+        (const short*)dy.data_ptr(), (const short*)s.data_ptr(),
+        (const short*)gamma.data_ptr(), mean.data_ptr<float>(),
+        rstd.data_ptr<float>(), (short*)dx.data_ptr(), R, D, nullptr,
+        nullptr, 1.f);
+  }
   static std::map<std::pair<int, int>, torch::Tensor> ws_cache;
   auto wkey = std::make_pair((int)dy.get_device(), D);
   auto wit = ws_cache.find(wkey);
@@ -246,5 +322,6 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor s,
   ln_gb_cast_kernel<<<cdiv(D, 256), 256, 0, stream>>>(
       acc_g.data_ptr<float>(), acc_b.data_ptr<float>(),
       (short*)dgamma.data_ptr(), (short*)dbeta.data_ptr(), D);
+  if (mask.has_value()) return {dx, dgamma, dbeta, dxm};
   return {dx, dgamma, dbeta};
 }

@@ -349,6 +349,56 @@ class _ResidualLNFlatFn(torch.autograd.Function):
         return dx, dx, None, None
 
 
+class _DropResidualLNFn(torch.autograd.Function):
+    """y = LN(dropout(x) + res): the sublayer dropout fused into the LN
+    kernels — no standalone dropout read/write of the (B,S,d) tensor in
+    either direction."""
+
+    @staticmethod
+    def forward(ctx, x, res, gamma, beta, p, eps):
+        E = ext()
+        if _GRAPH_SEED_T is not None:
+            _GRAPH_SALT[0] += 1
+            y, s, mean, rstd, mask = E.ln_fwd(x, res, gamma, beta, eps, p,
+                                              _GRAPH_SALT[0], _GRAPH_SEED_T)
+        else:
+            seed = int(torch.randint(0, 2**31 - 1, (1,)).item())
+            y, s, mean, rstd, mask = E.ln_fwd(x, res, gamma, beta, eps, p,
+                                              seed)
+        ctx.p = p
+        ctx.flat_gb = None
+        ctx.save_for_backward(s, gamma, mean, rstd, mask)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        E = ext()
+        s, gamma, mean, rstd, mask = ctx.saved_tensors
+        if ctx.flat_gb is not None:
+            g, b = ctx.flat_gb
+            dres, _, _, dxm = E.ln_bwd(dy.contiguous(), s, gamma, mean, rstd,
+                                       _flat(g).view(-1), _flat(b).view(-1),
+                                       mask, ctx.p)
+            _grad_ready(g, b)
+            return dxm, dres, None, None, None, None
+        dres, dgamma, dbeta, dxm = E.ln_bwd(dy.contiguous(), s, gamma, mean,
+                                            rstd, None, None, mask, ctx.p)
+        return dxm, dres, dgamma, dbeta, None, None
+
+
+class _DropResidualLNFlatFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, res, gb, p, eps):
+        y = _DropResidualLNFn.forward(ctx, x, res, gb[0], gb[1], p, eps)
+        ctx.flat_gb = gb
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        dxm, dres, _, _, _, _ = _DropResidualLNFn.backward(ctx, dy)
+        return dxm, dres, None, None, None
+
+
 def residual_layernorm(x, res, gamma, beta, eps: float = 1e-6):
     if x.is_cuda:
         shp = x.shape
@@ -364,6 +414,27 @@ def residual_layernorm(x, res, gamma, beta, eps: float = 1e-6):
         return R.residual_layernorm(x.float(), res.float(), gamma.float(),
                                     beta.float(), eps).to(x.dtype)
     return R.residual_layernorm(x, res, gamma, beta, eps)
+
+
+def dropout_residual_layernorm(x, res, gamma, beta, rate: float,
+                               training: bool, eps: float = 1e-6):
+    """LN(dropout(x) + res) — the post-sublayer pattern of every encoder/
+    decoder sublayer (reference Encoder.py:22-23).  Fused on GPU when
+    training with rate>0; otherwise plain residual+LN / composed ops."""
+    if not training or rate == 0.0:
+        return residual_layernorm(x, res, gamma, beta, eps)
+    if x.is_cuda:
+        shp = x.shape
+        d = shp[-1]
+        x2 = x.reshape(-1, d).contiguous()
+        r2 = res.reshape(-1, d).contiguous()
+        if _flat(gamma) is not None:
+            y = _DropResidualLNFlatFn.apply(x2, r2, [gamma, beta], rate, eps)
+        else:
+            y = _DropResidualLNFn.apply(x2, r2, gamma, beta, rate, eps)
+        return y.view(shp)
+    xd = torch.nn.functional.dropout(x, p=rate, training=True)
+    return residual_layernorm(xd, res, gamma, beta, eps)
 
 
 # ---------------------------------------------------------------------------
