@@ -289,24 +289,25 @@ __global__ __launch_bounds__(256)
 void attn_bwd_dot_kernel(const short* __restrict__ dO,
                          const short* __restrict__ O, float* __restrict__ Dl,
                          int H, int S, int DH, long n_rows) {
-  const long row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  // ONE THREAD per (b,s,h) row with s16x8 vector loads: a wave-per-row
+  // version (1 scalar load per lane + shuffle reduce) measured 7x off the
+  // HBM bound — too little ILP per wave.
+  const long row = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (row >= n_rows) return;   // row = ((b*Sq)+s) over (B,Sq) with H inner
-  const int lane = threadIdx.x & 63;
-  // rows iterate (b, s, h): dO layout (B,S,H,DH) flattened rows b*S*H+s*H+h
   const short* dop = dO + row * DH;
   const short* op = O + row * DH;
   float acc = 0.f;
-  for (int c = lane; c < DH; c += WAVE)
-    acc += bfbits2f(dop[c]) * bfbits2f(op[c]);
-  acc = wave_sum(acc);
-  if (lane == 0) {
-    // remap (b,s,h) -> (b,h,s) to match LSE layout
-    long bsh = row;
-    long h = bsh % H;
-    long s = (bsh / H) % S;
-    long b = bsh / ((long)H * S);
-    Dl[((b * H) + h) * S + s] = acc;
+  for (int c = 0; c + 8 <= DH; c += 8) {
+    s16x8 dv = *(const s16x8*)(dop + c);
+    s16x8 ov = *(const s16x8*)(op + c);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc += bfbits2f(dv[j]) * bfbits2f(ov[j]);
   }
+  // remap (b,s,h) -> (b,h,s) to match LSE layout
+  long h = row % H;
+  long s = (row / H) % S;
+  long b = row / ((long)H * S);
+  Dl[((b * H) + h) * S + s] = acc;
 }
 
 // dK/dV kernel: one 4-wave block per 64-key tile; loops q-tiles of 32.
@@ -692,7 +693,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
     pad = kv_pad.data_ptr<unsigned char>();
   auto stream = at::hip::getCurrentHIPStream();
   long n_rows = (long)B * Sq * H;
-  attn_bwd_dot_kernel<<<cdiv(n_rows, 4), 256, 0, stream>>>(
+  attn_bwd_dot_kernel<<<cdiv(n_rows, 256), 256, 0, stream>>>(
       (const short*)dout.data_ptr(), (const short*)o.data_ptr(),
       dl.data_ptr<float>(), H, Sq, DH, n_rows);
   dim3 grid_kv(B * H, cdiv(Sk, WAVES * QW));
