@@ -68,3 +68,29 @@ def test_restore_none_when_empty(tmp_path):
     m = _model()
     cm = CheckpointManager(m, None, str(tmp_path / "empty"))
     assert cm.restore() is None
+
+
+def test_export_and_reload_forward_parity(tmp_path):
+    """C24: export (SavedModel-equivalent dir) -> load -> same logits."""
+    import torch
+    from transformer_amd.models import Transformer
+    from transformer_amd.runtime import export_model
+    from transformer_amd.runtime.export import load_exported
+
+    torch.manual_seed(3)
+    m = Transformer(num_layers=1, d_model=16, num_heads=2, dff=32,
+                    input_vocab_size=40, target_vocab_size=44, rate=0.0,
+                    max_position=32)
+    cfg = dict(num_layers=1, d_model=16, num_heads=2, dff=32,
+               input_vocab_size=40, target_vocab_size=44, dropout_rate=0.0,
+               max_position=32)
+    d = str(tmp_path / "model")
+    export_model(m, d, cfg)
+    m2, cfg2 = load_exported(d)
+    assert cfg2["d_model"] == 16
+    src = torch.randint(1, 39, (2, 7))
+    tar = torch.randint(1, 43, (2, 5))
+    with torch.no_grad():
+        a, _ = m((src, tar), training=False)
+        b, _ = m2((src, tar), training=False)
+    assert torch.equal(a, b)
