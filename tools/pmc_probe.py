@@ -2,7 +2,7 @@
 --pmc, which must not be combined with tracing on this pool)."""
 import sys
 import torch
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__))))
 from transformer_amd.ops import ext
 
 E = ext()
