@@ -83,11 +83,11 @@ DEV_INLINE bf16x8 lds_read8_tr(const short* base, int kb, int col0) {
 // into swizzled LDS: natural layout into lds_n (if WRITE_N) and/or the
 // transpose [DH][KVT] into lds_t (if WRITE_T).
 // 256 threads cooperate; guards rows >= nrows with zeros.
-template <int DH, bool WRITE_N, bool WRITE_T>
+template <int DH, bool WRITE_N, bool WRITE_T, int NR = KVT>
 DEV_INLINE void stage_kv(const short* __restrict__ g, long row_stride,
                          int nrows, short* lds_n, short* lds_t) {
   const int t = threadIdx.x;
-  constexpr int CH = KVT * DH / 8;  // 16-byte chunks
+  constexpr int CH = NR * DH / 8;  // 16-byte chunks
 #pragma unroll
   for (int p = 0; p < (CH + 255) / 256; ++p) {
     int idx = p * 256 + t;
@@ -99,7 +99,7 @@ DEV_INLINE void stage_kv(const short* __restrict__ g, long row_stride,
     if (WRITE_N) lds_write8<DH * 2>(lds_n, row, c8, v);
     if (WRITE_T) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) lds_write1<KVT * 2>(lds_t, c8 + j, row, v[j]);
+      for (int j = 0; j < 8; ++j) lds_write1<NR * 2>(lds_t, c8 + j, row, v[j]);
     }
   }
 }
@@ -119,9 +119,13 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
                      long q_rs, long kv_rs, long o_rs, long q_bs, long kv_bs) {
   constexpr int D32 = DH / 32;   // QK^T MFMA k-steps
   constexpr int D16 = DH / 16;   // O fragments
-  __shared__ short k_lds[KVT * DH];
-  __shared__ short v_lds[KVT * DH];   // natural [KVT][DH] or transposed
-  __shared__ short p_lds[WAVES][QW * KVT];
+  constexpr int FKVT = 64;       // forward key tile (wider than bwd: the
+                                 // kernel is issue-bound, so per-key fixed
+                                 // costs are amortized over 2x keys)
+  constexpr int NHALF = FKVT / 16;
+  __shared__ short k_lds[FKVT * DH];
+  __shared__ short v_lds[FKVT * DH];   // natural [FKVT][DH] or transposed
+  __shared__ short p_lds[WAVES][QW * FKVT];
 
   const int bh = blockIdx.x;      // b*H + h
   const int qb = blockIdx.y;      // q-block of 64
@@ -155,26 +159,26 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
   // causal: keys beyond this block's last row are fully masked
   const int kend = causal ? min(Sk, qb * (WAVES * QW) + WAVES * QW) : Sk;
 
-  for (int k0 = 0; k0 < kend; k0 += KVT) {
-    const int kc = min(KVT, Sk - k0);
-    stage_kv<DH, true, false>(Kp + (long)k0 * kv_rs, kv_rs, kc,
-                              k_lds, nullptr);
+  for (int k0 = 0; k0 < kend; k0 += FKVT) {
+    const int kc = min(FKVT, Sk - k0);
+    stage_kv<DH, true, false, FKVT>(Kp + (long)k0 * kv_rs, kv_rs, kc,
+                                    k_lds, nullptr);
     if (TRV)
-      stage_kv<DH, true, false>(Vp + (long)k0 * kv_rs, kv_rs, kc,
-                                v_lds, nullptr);
+      stage_kv<DH, true, false, FKVT>(Vp + (long)k0 * kv_rs, kv_rs, kc,
+                                      v_lds, nullptr);
     else
-      stage_kv<DH, false, true>(Vp + (long)k0 * kv_rs, kv_rs, kc,
-                                nullptr, v_lds);
+      stage_kv<DH, false, true, FKVT>(Vp + (long)k0 * kv_rs, kv_rs, kc,
+                                      nullptr, v_lds);
     __syncthreads();
 
-    // ---- S = scale*(Q K^T) + mask, two 16-key halves -----------------
+    // ---- S = scale*(Q K^T) + mask, NHALF 16-key halves ---------------
     // A = Q rows, B = K rows (NT): C[row=q][col=key].
-    float p_raw[2][4];   // C-layout: [half][reg r] value for (row kg*4+r, col fr)
+    float p_raw[NHALF][4];  // C-layout: [half][reg r] for (row kg*4+r, col fr)
     float tile_pmax[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) tile_pmax[r] = -1e30f;
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
+    for (int half = 0; half < NHALF; ++half) {
       f32x4 s2 = {0, 0, 0, 0};
       __builtin_amdgcn_s_setprio(1);  // +4-7% on attn MFMA bursts (guide)
 #pragma unroll
@@ -217,7 +221,7 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
 
     // P = exp(S - m), row-sum into l, and park bf16 P in LDS (A-layout)
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
+    for (int half = 0; half < NHALF; ++half) {
       float psum[4];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -230,15 +234,13 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
 #pragma unroll
         for (int off = 1; off < 16; off <<= 1)
           psum[r] += __shfl_xor(psum[r], off);
+        // first half applies alpha to l; later halves add directly
         l_run[r] = l_run[r] * (half == 0 ? alpha[r] : 1.f) + psum[r];
-      }
-      if (half == 0) {
-        // first half applied alpha to l above; second half adds directly
       }
 #pragma unroll
       for (int r = 0; r < 4; ++r)
-        lds_write1<KVT * 2>(p_lds[wid], kg * 4 + r, half * 16 + fr,
-                            f2bfbits(p_raw[half][r]));
+        lds_write1<FKVT * 2>(p_lds[wid], kg * 4 + r, half * 16 + fr,
+                             f2bfbits(p_raw[half][r]));
     }
     // p_lds is per-wave: a wave barrier (not a workgroup barrier) orders
     // the scalar P writes against this wave's own PV reads.
@@ -247,12 +249,16 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
     // ---- O += P · V  (A = P[q][key] from LDS, B = V^T[d][key]) -------
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int i = 0; i < D16; ++i) {
-      bf16x8 pa = lds_read8<KVT * 2>(p_lds[wid], fr, kg * 8);
-      bf16x8 vb = TRV ? lds_read8_tr<DH * 2>(v_lds, kg * 8, i * 16)
-                      : lds_read8<KVT * 2>(v_lds, i * 16 + fr, kg * 8);
-      acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, acc[i], 0, 0, 0);
-    }
+    for (int ks = 0; ks < FKVT / 32; ++ks)
+#pragma unroll
+      for (int i = 0; i < D16; ++i) {
+        bf16x8 pa = lds_read8<FKVT * 2>(p_lds[wid], fr, ks * 32 + kg * 8);
+        bf16x8 vb = TRV
+            ? lds_read8_tr<DH * 2>(v_lds, ks * 32 + kg * 8, i * 16)
+            : lds_read8<FKVT * 2>(v_lds, i * 16 + fr, ks * 32 + kg * 8);
+        acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, acc[i],
+                                                         0, 0, 0);
+      }
     __builtin_amdgcn_s_setprio(0);
     __syncthreads();
   }
