@@ -361,6 +361,22 @@ def test_attn_fwd(dh, causal):
     assert_close(o, ref, 0.04, f"attn_fwd dh{dh} causal{causal}")
 
 
+@pytest.mark.parametrize("causal", [False, True])
+def test_attn_fwd_long_seq_rf2(causal):
+    """Sk >= 2048 dispatches the RF=2 (32 q-rows/wave) forward variant;
+    odd length exercises its edge guards."""
+    torch.manual_seed(11)
+    B, S, H, dh = 1, 2051, 2, 64
+    q = torch.randn(B, S, H, dh, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, S, H, dh, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, S, H, dh, device="cuda", dtype=torch.bfloat16)
+    pad = torch.zeros(B, S, device="cuda", dtype=torch.uint8)
+    pad[0, 2000:] = 1
+    o, lse = _ext().attn_fwd(q, k, v, pad, causal, 1.0 / math.sqrt(dh))
+    ref = _attn_ref(q, k, v, pad, causal)
+    assert_close(o, ref, 0.05, f"attn_fwd_rf2 causal{causal}")
+
+
 def test_attn_fwd_cross_shapes():
     torch.manual_seed(1)
     B, Sq, Sk, H, dh = 2, 37, 75, 4, 64

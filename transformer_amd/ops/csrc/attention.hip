@@ -87,19 +87,31 @@ template <int DH, bool WRITE_N, bool WRITE_T, int NR = KVT>
 DEV_INLINE void stage_kv(const short* __restrict__ g, long row_stride,
                          int nrows, short* lds_n, short* lds_t) {
   const int t = threadIdx.x;
-  constexpr int CH = NR * DH / 8;  // 16-byte chunks
+  constexpr int CH = NR * DH / 8;   // 16-byte chunks
+  constexpr int NP = (CH + 255) / 256;
+  // issue ALL global loads before any LDS write: at reduced occupancy the
+  // staging must keep several loads in flight per lane to reach HBM rate
+  s16x8 v[NP];
 #pragma unroll
-  for (int p = 0; p < (CH + 255) / 256; ++p) {
+  for (int p = 0; p < NP; ++p) {
+    int idx = p * 256 + t;
+    int row = idx / (DH / 8);
+    int c8 = (idx % (DH / 8)) * 8;
+    v[p] = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (idx < CH && row < nrows)
+      v[p] = *(const s16x8*)(g + row * row_stride + c8);
+  }
+#pragma unroll
+  for (int p = 0; p < NP; ++p) {
     int idx = p * 256 + t;
     if (idx >= CH) break;
     int row = idx / (DH / 8);
     int c8 = (idx % (DH / 8)) * 8;
-    s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-    if (row < nrows) v = *(const s16x8*)(g + row * row_stride + c8);
-    if (WRITE_N) lds_write8<DH * 2>(lds_n, row, c8, v);
+    if (WRITE_N) lds_write8<DH * 2>(lds_n, row, c8, v[p]);
     if (WRITE_T) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) lds_write1<NR * 2>(lds_t, c8 + j, row, v[j]);
+      for (int j = 0; j < 8; ++j)
+        lds_write1<NR * 2>(lds_t, c8 + j, row, v[p][j]);
     }
   }
 }
@@ -109,7 +121,13 @@ DEV_INLINE void stage_kv(const short* __restrict__ g, long row_stride,
 // ds_read_b64_tr_b16 (no transposed staging); else build a transposed V
 // image with scalar writes and read it row-wise.  Same buffer either way;
 // the faster variant is chosen by measurement (tools/attn_bench.py).
-template <int DH, bool TRV>
+//
+// Each wave owns FQW=32 q-rows as RF=2 16-row fragments, so one workgroup
+// covers 128 q-rows: at seq 4096 the kernel is bound by re-staging K/V
+// once per q-block (16.4 GB per call at 64-row blocks), and doubling the
+// block height halves that traffic.  Softmax temporaries live per
+// row-fragment iteration; only qf/acc/m/l are duplicated.
+template <int DH, bool TRV, int RF>
 __global__ __launch_bounds__(256)
 void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
                      const short* __restrict__ V,
@@ -119,19 +137,18 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
                      long q_rs, long kv_rs, long o_rs, long q_bs, long kv_bs) {
   constexpr int D32 = DH / 32;   // QK^T MFMA k-steps
   constexpr int D16 = DH / 16;   // O fragments
-  constexpr int FKVT = 64;       // forward key tile (wider than bwd: the
-                                 // kernel is issue-bound, so per-key fixed
-                                 // costs are amortized over 2x keys)
+  constexpr int FKVT = 64;       // key tile
   constexpr int NHALF = FKVT / 16;
+  constexpr int FQW = 16 * RF;   // q-rows per wave (RF template)
   __shared__ short k_lds[FKVT * DH];
   __shared__ short v_lds[FKVT * DH];   // natural [FKVT][DH] or transposed
-  __shared__ short p_lds[WAVES][QW * FKVT];
+  __shared__ short p_lds[WAVES][QW * FKVT];  // one 16-row rf at a time
 
   const int bh = blockIdx.x;      // b*H + h
-  const int qb = blockIdx.y;      // q-block of 64
+  const int qb = blockIdx.y;      // q-block of WAVES*FQW rows
   const int b = bh / H, h = bh % H;
   const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  const int q0 = qb * (WAVES * QW) + wid * QW;  // this wave's first q-row
+  const int q0 = qb * (WAVES * FQW) + wid * FQW;  // wave's first q-row
   const int fr = lane & 15, kg = lane >> 4;
 
   const short* Qp = Q + b * q_bs + (long)h * DH;
@@ -139,25 +156,35 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
   const short* Vp = V + b * kv_bs + (long)h * DH;
   const unsigned char* pad = kv_pad ? kv_pad + (long)b * Sk : nullptr;
 
-  // Q fragments in registers: A[row=fr][k=kg*8+j] per 32-chunk
-  bf16x8 qf[D32];
+  // Q fragments in registers: A[row=fr][k=kg*8+j] per 32-chunk, per rf
+  bf16x8 qf[RF][D32];
 #pragma unroll
-  for (int d = 0; d < D32; ++d) {
-    s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-    int row = q0 + fr;
-    if (row < Sq) v = *(const s16x8*)(Qp + (long)row * q_rs + d * 32 + kg * 8);
-    qf[d] = (bf16x8)v;
-  }
+  for (int rf = 0; rf < RF; ++rf)
+#pragma unroll
+    for (int d = 0; d < D32; ++d) {
+      s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      int row = q0 + rf * 16 + fr;
+      if (row < Sq)
+        v = *(const s16x8*)(Qp + (long)row * q_rs + d * 32 + kg * 8);
+      qf[rf][d] = (bf16x8)v;
+    }
 
-  f32x4 acc[D16];
+  f32x4 acc[RF][D16];
 #pragma unroll
-  for (int i = 0; i < D16; ++i) acc[i] = {0, 0, 0, 0};
-  float m_run[4], l_run[4];
+  for (int rf = 0; rf < RF; ++rf)
 #pragma unroll
-  for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.f; }
+    for (int i = 0; i < D16; ++i) acc[rf][i] = {0, 0, 0, 0};
+  float m_run[RF][4], l_run[RF][4];
+#pragma unroll
+  for (int rf = 0; rf < RF; ++rf)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      m_run[rf][r] = -1e30f;
+      l_run[rf][r] = 0.f;
+    }
 
   // causal: keys beyond this block's last row are fully masked
-  const int kend = causal ? min(Sk, qb * (WAVES * QW) + WAVES * QW) : Sk;
+  const int kend = causal ? min(Sk, (qb + 1) * (WAVES * FQW)) : Sk;
 
   for (int k0 = 0; k0 < kend; k0 += FKVT) {
     const int kc = min(FKVT, Sk - k0);
@@ -171,119 +198,130 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
                                       nullptr, v_lds);
     __syncthreads();
 
-    // ---- S = scale*(Q K^T) + mask, NHALF 16-key halves ---------------
-    // A = Q rows, B = K rows (NT): C[row=q][col=key].
-    float p_raw[NHALF][4];  // C-layout: [half][reg r] for (row kg*4+r, col fr)
-    float tile_pmax[4];
 #pragma unroll
-    for (int r = 0; r < 4; ++r) tile_pmax[r] = -1e30f;
+    for (int rf = 0; rf < RF; ++rf) {
+      const int qr0 = q0 + rf * 16;
+      // ---- S = scale*(Q K^T) + mask, NHALF 16-key halves -------------
+      float p_raw[NHALF][4];
+      float tile_pmax[4];
 #pragma unroll
-    for (int half = 0; half < NHALF; ++half) {
-      f32x4 s2 = {0, 0, 0, 0};
-      __builtin_amdgcn_s_setprio(1);  // +4-7% on attn MFMA bursts (guide)
+      for (int r = 0; r < 4; ++r) tile_pmax[r] = -1e30f;
 #pragma unroll
-      for (int d = 0; d < D32; ++d) {
-        bf16x8 kf = lds_read8<DH * 2>(k_lds, half * 16 + fr, d * 32 + kg * 8);
-        s2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[d], kf, s2, 0, 0, 0);
+      for (int half = 0; half < NHALF; ++half) {
+        f32x4 s2 = {0, 0, 0, 0};
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int d = 0; d < D32; ++d) {
+          bf16x8 kf =
+              lds_read8<DH * 2>(k_lds, half * 16 + fr, d * 32 + kg * 8);
+          s2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[rf][d], kf, s2,
+                                                       0, 0, 0);
+        }
+        __builtin_amdgcn_s_setprio(0);
+        const int kcol = k0 + half * 16 + fr;   // C col = lane&15
+        const bool col_pad = (kcol >= Sk) || (pad && pad[min(kcol, Sk - 1)]);
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int qrow = qr0 + kg * 4 + r;    // C row = (lane>>4)*4+r
+          float x = s2[r] * scale;
+          if (col_pad) x += NEG_BIG;
+          if (causal && kcol > qrow) x += NEG_BIG;
+          p_raw[half][r] = x;
+          tile_pmax[r] = fmaxf(tile_pmax[r], x);
+        }
       }
-      __builtin_amdgcn_s_setprio(0);
-      const int kcol = k0 + half * 16 + fr;   // C col = lane&15
-      const bool col_pad = (kcol >= Sk) || (pad && pad[min(kcol, Sk - 1)]);
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int qrow = q0 + kg * 4 + r;     // C row = (lane>>4)*4+r
-        float x = s2[r] * scale;
-        if (col_pad) x += NEG_BIG;
-        if (causal && kcol > qrow) x += NEG_BIG;
-        p_raw[half][r] = x;
-        tile_pmax[r] = fmaxf(tile_pmax[r], x);
-      }
-    }
-    // row max across the 16 cols held by the 16-lane group
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1)
-        tile_pmax[r] = fmaxf(tile_pmax[r], __shfl_xor(tile_pmax[r], off));
-    }
-    // online rescale
-    float alpha[4];
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float mn = fmaxf(m_run[r], tile_pmax[r]);
-      alpha[r] = __expf(m_run[r] - mn);
-      m_run[r] = mn;
-    }
-#pragma unroll
-    for (int i = 0; i < D16; ++i)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) acc[i][r] *= alpha[r];
-
-    // P = exp(S - m), row-sum into l, and park bf16 P in LDS (A-layout)
-#pragma unroll
-    for (int half = 0; half < NHALF; ++half) {
-      float psum[4];
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float p = __expf(p_raw[half][r] - m_run[r]);
-        p_raw[half][r] = p;
-        psum[r] = p;
-      }
+      // row max across the 16 cols held by the 16-lane group
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
 #pragma unroll
         for (int off = 1; off < 16; off <<= 1)
-          psum[r] += __shfl_xor(psum[r], off);
-        // first half applies alpha to l; later halves add directly
-        l_run[r] = l_run[r] * (half == 0 ? alpha[r] : 1.f) + psum[r];
+          tile_pmax[r] = fmaxf(tile_pmax[r], __shfl_xor(tile_pmax[r], off));
+      }
+      // online rescale
+      float alpha[4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float mn = fmaxf(m_run[rf][r], tile_pmax[r]);
+        alpha[r] = __expf(m_run[rf][r] - mn);
+        m_run[rf][r] = mn;
       }
 #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        lds_write1<FKVT * 2>(p_lds[wid], kg * 4 + r, half * 16 + fr,
-                             f2bfbits(p_raw[half][r]));
-    }
-    // p_lds is per-wave: a wave barrier (not a workgroup barrier) orders
-    // the scalar P writes against this wave's own PV reads.
-    __builtin_amdgcn_wave_barrier();
+      for (int i = 0; i < D16; ++i)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) acc[rf][i][r] *= alpha[r];
 
-    // ---- O += P · V  (A = P[q][key] from LDS, B = V^T[d][key]) -------
-    __builtin_amdgcn_s_setprio(1);
+      // P = exp(S - m), row-sum into l, park bf16 P in LDS (A-layout)
 #pragma unroll
-    for (int ks = 0; ks < FKVT / 32; ++ks)
+      for (int half = 0; half < NHALF; ++half) {
+        float psum[4];
 #pragma unroll
-      for (int i = 0; i < D16; ++i) {
-        bf16x8 pa = lds_read8<FKVT * 2>(p_lds[wid], fr, ks * 32 + kg * 8);
-        bf16x8 vb = TRV
-            ? lds_read8_tr<DH * 2>(v_lds, ks * 32 + kg * 8, i * 16)
-            : lds_read8<FKVT * 2>(v_lds, i * 16 + fr, ks * 32 + kg * 8);
-        acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, acc[i],
-                                                         0, 0, 0);
+        for (int r = 0; r < 4; ++r) {
+          float pv = __expf(p_raw[half][r] - m_run[rf][r]);
+          p_raw[half][r] = pv;
+          psum[r] = pv;
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+#pragma unroll
+          for (int off = 1; off < 16; off <<= 1)
+            psum[r] += __shfl_xor(psum[r], off);
+          // first half applies alpha to l; later halves add directly
+          l_run[rf][r] = l_run[rf][r] * (half == 0 ? alpha[r] : 1.f) +
+                         psum[r];
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          lds_write1<FKVT * 2>(p_lds[wid], kg * 4 + r, half * 16 + fr,
+                               f2bfbits(p_raw[half][r]));
       }
-    __builtin_amdgcn_s_setprio(0);
+      // p_lds is per-wave: a wave barrier orders the scalar P writes
+      // against this wave's own PV reads.
+      __builtin_amdgcn_wave_barrier();
+
+      // ---- O += P · V  (A = P[q][key] from LDS, B = V^T[d][key]) -----
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ks = 0; ks < FKVT / 32; ++ks)
+#pragma unroll
+        for (int i = 0; i < D16; ++i) {
+          bf16x8 pa =
+              lds_read8<FKVT * 2>(p_lds[wid], fr, ks * 32 + kg * 8);
+          bf16x8 vb = TRV
+              ? lds_read8_tr<DH * 2>(v_lds, ks * 32 + kg * 8, i * 16)
+              : lds_read8<FKVT * 2>(v_lds, i * 16 + fr, ks * 32 + kg * 8);
+          acc[rf][i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              pa, vb, acc[rf][i], 0, 0, 0);
+        }
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_wave_barrier();  // p_lds reused by next rf
+    }
     __syncthreads();
   }
 
   // epilogue: O = acc / l ; LSE = m + log(l)
   short* Op = O + b * (Sq * o_rs) + (long)h * DH;
 #pragma unroll
-  for (int i = 0; i < D16; ++i) {
-    const int gcol = i * 16 + fr;
+  for (int rf = 0; rf < RF; ++rf) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int qrow = q0 + kg * 4 + r;
-      if (qrow >= Sq) continue;
-      float l = l_run[r];
-      float o = (l > 0.f) ? acc[i][r] / l : 0.f;
-      Op[(long)qrow * o_rs + gcol] = f2bfbits(o);
+    for (int i = 0; i < D16; ++i) {
+      const int gcol = i * 16 + fr;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qrow = q0 + rf * 16 + kg * 4 + r;
+        if (qrow >= Sq) continue;
+        float l = l_run[rf][r];
+        float o = (l > 0.f) ? acc[rf][i][r] / l : 0.f;
+        Op[(long)qrow * o_rs + gcol] = f2bfbits(o);
+      }
     }
-  }
-  if (fr == 0) {
+    if (fr == 0) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int qrow = q0 + kg * 4 + r;
-      if (qrow < Sq)
-        LSE[((long)b * H + h) * Sq + qrow] =
-            m_run[r] + __logf(fmaxf(l_run[r], 1e-30f));
+      for (int r = 0; r < 4; ++r) {
+        const int qrow = q0 + rf * 16 + kg * 4 + r;
+        if (qrow < Sq)
+          LSE[((long)b * H + h) * Sq + qrow] =
+              m_run[rf][r] + __logf(fmaxf(l_run[rf][r], 1e-30f));
+      }
     }
   }
 }
@@ -620,23 +658,28 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                 kv_pad.size(0) == B && kv_pad.size(1) == Sk);
     pad = kv_pad.data_ptr<unsigned char>();
   }
-  dim3 grid(B * H, cdiv(Sq, WAVES * QW));
+  // RF=2 (32 q-rows/wave) halves K/V re-staging; it costs registers
+  // (~188 vs ~130 VGPR -> 2 vs 3+ waves/SIMD), so it dispatches only for
+  // long sequences where the staging traffic dominates.
+  const bool rf2 = Sk >= 2048;
+  dim3 grid(B * H, cdiv(Sq, WAVES * (rf2 ? 32 : 16)));
   auto stream = at::hip::getCurrentHIPStream();
   DISPATCH_DH(DH, {
-    if (trv)
-      attn_fwd_kernel<DHC, true><<<grid, 256, 0, stream>>>(
-          (const short*)q.data_ptr(), (const short*)k.data_ptr(),
-          (const short*)v.data_ptr(), pad, (short*)o.data_ptr(),
-          lse.data_ptr<float>(), B, H, Sq, Sk, causal ? 1 : 0,
-          (float)scale, q.stride(1), k.stride(1), (long)H * DH,
-          q.stride(0), k.stride(0));
-    else
-      attn_fwd_kernel<DHC, false><<<grid, 256, 0, stream>>>(
-                  (const short*)q.data_ptr(), (const short*)k.data_ptr(),
-                  (const short*)v.data_ptr(), pad, (short*)o.data_ptr(),
-                  lse.data_ptr<float>(), B, H, Sq, Sk, causal ? 1 : 0,
-                  (float)scale, q.stride(1), k.stride(1), (long)H * DH,
-                  q.stride(0), k.stride(0));
+    auto launch = [&](auto trv_c, auto rf_c) {
+      attn_fwd_kernel<DHC, decltype(trv_c)::value, decltype(rf_c)::value>
+          <<<grid, 256, 0, stream>>>(
+              (const short*)q.data_ptr(), (const short*)k.data_ptr(),
+              (const short*)v.data_ptr(), pad, (short*)o.data_ptr(),
+              lse.data_ptr<float>(), B, H, Sq, Sk, causal ? 1 : 0,
+              (float)scale, q.stride(1), k.stride(1), (long)H * DH,
+              q.stride(0), k.stride(0));
+    };
+    using TT = std::true_type;
+    using FF = std::false_type;
+    using R1 = std::integral_constant<int, 1>;
+    using R2 = std::integral_constant<int, 2>;
+    if (trv) { if (rf2) launch(TT{}, R2{}); else launch(TT{}, R1{}); }
+    else     { if (rf2) launch(FF{}, R2{}); else launch(FF{}, R1{}); }
   });
   return {o, lse};
 }
