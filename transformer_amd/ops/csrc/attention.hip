@@ -10,10 +10,11 @@
 // (pad), decoder self (causal+pad), cross (pad, Sq != Sk).
 //
 // Forward: online softmax (running m, l per row), O(Sq·dh) memory, fp32
-// MFMA accumulate, saves LSE for the backward.  Work unit: 4-wave block =
-// 64 q-rows; K/V tiles of 32 staged through swizzled LDS; QK^T and PV on
-// mfma_f32_16x16x32_bf16; P crosses C-layout -> A-layout through a small
-// per-wave LDS buffer.
+// MFMA accumulate, saves LSE for the backward.  Work unit: 4-wave block of
+// 64 q-rows (128 for seq>=2048 via the RF=2 variant); 64-key K/V tiles
+// staged through swizzled LDS; QK^T and PV on mfma_f32_16x16x32_bf16; P
+// crosses C-layout -> A-layout through a small per-wave LDS buffer; the PV
+// B-operand is read from the natural V image with ds_read_b64_tr_b16.
 //
 // Backward: FA2-style two-kernel split (no atomics, deterministic):
 //   attn_bwd_kv: grid over kv-tiles, accumulates dK,dV (recomputing P from
@@ -27,7 +28,7 @@
 
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 
-#define KVT 32        // kv-tile (= MFMA K for PV)
+#define KVT 32        // backward kv/q tile (fwd uses FKVT=64)
 #define QW 16         // q-rows per wave
 #define WAVES 4
 #define NEG_BIG (-1e9f)
