@@ -61,3 +61,29 @@ def test_predict_greedy(tmp_path, toy_corpus):
     assert out.dim() == 1
     assert out[0].item() == tr.tgt_tokenizer.vocab_size  # tgt start token (Q6)
     assert out.shape[0] <= 1 + tr.max_decode_len + 1
+
+
+def test_sigterm_checkpoints_and_stops(tmp_path, toy_corpus):
+    """SURVEY §5 failure handling: a stop request mid-epoch checkpoints and
+    exits the loop cleanly; a fresh Train resumes from that step."""
+    import signal
+
+    tr, train_ds, test_ds = _mk_train(tmp_path, toy_corpus, epochs=50)
+    tr.install_signal_handler()
+    try:
+        os.kill(os.getpid(), signal.SIGTERM)  # handler sets the stop flag
+        tr.training_loop(train_ds, test_ds)
+        assert tr.ckpt_manager.latest_checkpoint is not None
+        tr2, *_ = _mk_train(tmp_path, toy_corpus, epochs=50)
+        meta = tr2.load_ckpt()
+        assert meta is not None and meta.get("step", 0) >= 1
+    finally:
+        signal.signal(signal.SIGTERM, signal.SIG_DFL)
+        signal.signal(signal.SIGINT, signal.default_int_handler)
+
+
+def test_trace_steps_writes_chrome_trace(tmp_path, toy_corpus):
+    """SURVEY §5 tracing: torch.profiler trace of a few steps."""
+    tr, train_ds, _ = _mk_train(tmp_path, toy_corpus)
+    path = tr.trace_steps(train_ds, 2, str(tmp_path / "traces"))
+    assert os.path.exists(path) and os.path.getsize(path) > 100

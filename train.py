@@ -54,7 +54,7 @@ def main(epochs, enable_function, buffer_size, batch_size, sequence_length,
          warmup_steps=60000, label_smoothing=0.0, device=None, dtype=None,
          seed=1234, log_interval=100, eval_steps=50, synthetic_data=False,
          synthetic_vocab=32768, steps_per_epoch=100, max_decode_len=10,
-         **_ignored):
+         trace_dir=None, **_ignored):
     torch.manual_seed(seed)
     device, dtype = pick_device_dtype(device, dtype)
 
@@ -91,6 +91,9 @@ def main(epochs, enable_function, buffer_size, batch_size, sequence_length,
                   log_interval=log_interval, eval_steps=eval_steps,
                   max_decode_len=max_decode_len)
     train.load_ckpt()
+    train.install_signal_handler()  # SIGTERM/SIGINT -> checkpoint + exit
+    if trace_dir:
+        train.trace_steps(train_ds, 5, trace_dir)
     train.training_loop(train_ds, test_ds)
     print(train.predict("he go to school"))
     export_model(transformer, "model", {

@@ -58,6 +58,9 @@ def transformer_flags(parser: argparse.ArgumentParser | None = None) -> argparse
                         help="Output dimesion of all sublayers including Embedding layer")
     parser.add_argument("--dff", type=int, default=1024, help="Dimetionality of inner layer")
     parser.add_argument("--num_heads", type=int, default=4, help="Number of Attention Head")
+    parser.add_argument("--trace_dir", type=str, default=None,
+                        help="If set, profile 5 training steps with torch.profiler "
+                             "and write a chrome trace here before training")
     _add_bool_flag(parser, "enable_function", True,
                    "Enable Function (compile/capture the train step)")
     parser.add_argument("--max_ckpt_keep", type=int, default=5,

@@ -163,10 +163,49 @@ class Train:
         if self.is_rank0 and ((epoch + 1) % 5 == 0 or (epoch + 1) == self.epochs):
             self.ckpt_manager.save(step, epoch)
 
+    def install_signal_handler(self):
+        """Failure handling: SIGTERM/SIGINT requests a clean
+        stop — the loop finishes the current step, checkpoints, and
+        returns, so training resumes from the interruption point."""
+        import signal
+
+        def _handler(signum, _frame):
+            self._stop_requested = True
+            if self.is_rank0:
+                print(f"signal {signum}: will checkpoint and stop "
+                      "after the current step")
+
+        signal.signal(signal.SIGTERM, _handler)
+        signal.signal(signal.SIGINT, _handler)
+
+    def trace_steps(self, train_dataset, n_steps: int, trace_dir: str):
+        """Tracing/profiling: run n_steps under
+        torch.profiler (Kineto-ROCm) and write a chrome trace to
+        trace_dir.  Separate from the timed loops."""
+        from torch.profiler import profile, ProfilerActivity
+        import os
+        os.makedirs(trace_dir, exist_ok=True)
+        it = iter(train_dataset)
+        acts = [ProfilerActivity.CPU]
+        if self.device.type == "cuda":
+            acts.append(ProfilerActivity.CUDA)
+        with profile(activities=acts) as prof:
+            for _ in range(n_steps):
+                try:
+                    self.train_step(next(it))
+                except StopIteration:
+                    break
+        path = os.path.join(trace_dir, "train_steps.json")
+        prof.export_chrome_trace(path)
+        if self.is_rank0:
+            print(f"trace written: {path}")
+        return path
+
     def training_loop(self, train_dataset, test_dataset):
         template = ("Epoch {}  Loss {:.4f} Accuracy {:.4f}, "
                     "Test Loss {:.4f}, Test Accuracy {:.4f}")
         step = self.optimizer.step_count
+        self._stop_requested = getattr(self, "_stop_requested", False)
         for epoch in range(self.epochs):
             start = time.time()
             for m in (self.train_loss, self.train_accuracy,
@@ -178,6 +217,11 @@ class Train:
             for inputs in train_dataset:
                 self.train_step(inputs)
                 step += 1
+                if self._stop_requested:
+                    if self.is_rank0:
+                        self.ckpt_manager.save(step, epoch)
+                        print("checkpointed on stop request; exiting loop")
+                    return
                 if step % self.log_interval == 0:
                     self._run_eval(test_dataset)
                     if self.is_rank0:
