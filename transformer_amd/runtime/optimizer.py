@@ -192,6 +192,8 @@ class NoamAdam:
 
     def load_state_dict(self, d):
         self.step_count = d["step"]
+        if hasattr(self, "_step_t"):  # keep captured-graph schedule in sync
+            self._step_t.fill_(self.step_count)
         if self.flat is not None:
             self.master.copy_(d["master"].to(self.master.device))
             self.m.copy_(d["m"].to(self.m.device))
