@@ -66,7 +66,9 @@ def main():
     print("== 256-template vs 128-kernel vs torch.matmul (hipBLASLt) ==")
     for (m, n, k, tag) in FWD + [(16384, 512, 2048, "FFN2 fwd"),
                                  (1536, 512, 16384, "QKV dW"),
-                                 (32770, 512, 16320, "logits dW")]:
+                                 (32770, 512, 16320, "logits dW"),
+                                 (4096, 4096, 4096, "4096^3 cal"),
+                                 (8192, 8192, 8192, "8192^3 cal")]:
         a = torch.randn(m, k, device="cuda", dtype=torch.bfloat16)
         w = torch.randn(n, k, device="cuda", dtype=torch.bfloat16)
         nob = torch.Tensor()
