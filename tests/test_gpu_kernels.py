@@ -49,7 +49,7 @@ def test_gemm_nt(m, n, k):
 
 
 @pytest.mark.parametrize("m,n,k,epi", [
-    (512, 512, 96, 0),        # minimum-depth pipeline (3 K-tiles)
+    (512, 512, 128, 0),       # minimum-depth pipeline (2 K-tiles)
     (777, 300, 512, 0),       # ragged M/N edge clamping
     (2048, 1536, 512, 1),     # QKV-like + ReLU epilogue
     (300, 1000, 1024, 0),     # partial tail tiles both dims

@@ -1,37 +1,33 @@
-// 256x256 8-wave deep-pipelined NT GEMM for CDNA4 (gfx950).
+// 256x256 8-wave deep-pipelined NT GEMM for CDNA4 (gfx950) — the guide's
+// "8-phase 256^2 template" reconstructed.
 //
-// The 128x128 two-barrier kernel in gemm.hip tops out near the "step-3
-// structure" ceiling (~500-700 TF measured at model shapes): every
-// __syncthreads() drains the global_load_lds queue (implicit vmcnt(0)), so
-// staging latency is exposed once per K-step.  This kernel follows the
-// guide's 8-phase 256^2 template instead: raw s_barrier + counted
-// s_waitcnt vmcnt(N) (never 0 in the main loop) keep prefetch loads in
-// flight across barriers.
+// Geometry: tile 256x256, BK=64, 512 threads = 8 waves in a 2(M) x 4(N)
+// grid; per-wave output 128x64 = acc[8][4] 16x16 fragments
+// (mfma_f32_16x16x32_bf16, fp32 accumulate).  LDS: TWO K-tile slots x
+// (A 32 KiB + B 32 KiB) = 128 KiB.
 //
-// Geometry:
-//   tile 256x256, BK=32, 512 threads = 8 waves in a 2(M) x 4(N) grid;
-//   per-wave output 128x64 = acc[8][4] 16x16 fragments
-//     (mfma_f32_16x16x32_bf16, fp32 accumulate).
-//   LDS: ring of 4 K-tile slots x (A 16 KiB + B 16 KiB) = 128 KiB.
-//   Per K-tile: 2 phases (one 64-row half of the wave's C each, 16 MFMAs);
-//   each phase: 8-or-4 ds_read_b128 fragment loads, 2 global_load_lds
-//   staging half-tiles of K-tile t+3 into slot (t+3)&3, s_barrier,
-//   s_waitcnt lgkmcnt(0), s_setprio(1), 16 x MFMA, s_setprio(0), s_barrier;
-//   at each tile boundary one s_waitcnt vmcnt(8) (loads from tiles t-1,t
-//   may stay in flight; everything older - i.e. tile t+1's data - landed).
+// Per K-tile: 4 quadrant phases.  Each phase hoists its NEW operand
+// fragments from LDS into registers (12/4/8/0 ds_read_b128 — A-halves and
+// B-quadrant-halves are REUSED across phases from registers), prefetches
+// half-tiles of K-tile t+1 into the other slot with global_load_lds, then
+// raw s_barrier + s_waitcnt lgkmcnt(0) + 16 MFMAs + s_barrier:
+//   q0: read A-half0(8) + B-ch0(4); stage A0',A1'  ; mfma quadrant (0,0)
+//   q1: read B-ch1(4)             ; stage B0',B1'  ; mfma quadrant (0,1)
+//   q2: read A-half1(8)           ;                ; mfma quadrant (1,0)
+//   q3: (all in registers)        ; vmcnt(0)       ; mfma quadrant (1,1)
+// The other slot is idle for the whole group (its tile was consumed last
+// group), so all four prefetch halves are issued in the first two phases
+// and have >=2.5 phases (~1400 cycles) to land — the boundary vmcnt(0)
+// drains an (almost always) empty queue, unlike a __syncthreads() pipeline
+// which exposes the full staging latency every K-step.
 //
-// LDS swizzle (st_16x32): phys = off ^ (((off>>9)&1)<<5) applied to the
-// linear offset within each operand tile ([256 rows][32 k] bf16, 64-B
-// rows).  16-lane ds_read_b128 column groups then hit 8 distinct 16-B
-// slots per 256-B bank row (2-way conflict) instead of 4-way linear.
-// Staged via global_load_lds with the inverse swizzle applied to the
-// SOURCE address (LDS destination stays lane-linear).
+// LDS swizzle (st_16x32 on 128-B rows): phys = off ^ (((off>>9)&1)<<5) —
+// 16-lane b128 column groups land on 4 distinct 16-B slots per bank row.
+// Staged via global_load_lds with the inverse swizzle on the SOURCE address.
 //
-// M/N edge handling: out-of-range tile rows are clamped to the last valid
-// row at staging (duplicate data, discarded at the guarded epilogue), so
-// any M,N work; K must be a multiple of 32 (all model shapes qualify
-// except the V=32770 logits-dX contraction, which falls back to the
-// 128x128 kernel).
+// M/N edges: out-of-range rows clamp to the last valid row at staging
+// (duplicates, discarded by the guarded epilogue); K must be a multiple
+// of 64 (all dispatched model shapes qualify; others use the 128 kernel).
 #include "common.h"
 
 #include <torch/extension.h>
@@ -43,44 +39,44 @@ namespace {
 
 #define G_BM 256
 #define G_BN 256
-#define G_BK 32
+#define G_BK 64
 #define G_THREADS 512
-#define G_SLOTS 4
-// shorts per operand tile (256 x 32)
+// shorts per operand tile (256 x 64) and per slot (A + B)
 #define G_TILE_ELEMS (G_BM * G_BK)
-// shorts per slot (A tile + B tile)
 #define G_SLOT_ELEMS (2 * G_TILE_ELEMS)
 
 DEV_INLINE int sw256(int off) { return off ^ (((off >> 9) & 1) << 5); }
 
-// Stage half h (tile rows [h*128, h*128+128)) of one 256x32 operand tile:
-// 8 chunks of 1 KiB, one global_load_lds(16B) per thread.  `g` points at
-// column k0 of the operand (row-major, ldg elems per row); rows are
-// row0+tile_row clamped to maxrow-1.
+// Stage half h (tile rows [h*128, h*128+128)) of one 256x64 operand tile:
+// 16 chunks of 1 KiB, 2 global_load_lds(16B) per thread.  `g` points at
+// column k0 of the operand (row-major, ldg elems/row); rows clamped.
 DEV_INLINE void stage256(const short* __restrict__ g, long ldg, int row0,
                          int maxrow, int h, short* lds) {
   const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  const int chunk = h * 8 + wid;
-  const int d = chunk * 1024 + lane * 16;  // dest byte offset in tile
-  const int lg = sw256(d);                 // logical byte offset
-  int row = row0 + (lg >> 6);              // 64-B rows
-  if (row > maxrow - 1) row = maxrow - 1;
-  const int colb = lg & 63;
-  __builtin_amdgcn_global_load_lds(
-      (const __attribute__((address_space(1))) void*)(g + (long)row * ldg +
-                                                      (colb >> 1)),
-      (__attribute__((address_space(3))) void*)((char*)lds + d), 16, 0, 0);
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const int chunk = h * 16 + p * 8 + wid;
+    const int d = chunk * 1024 + lane * 16;  // dest byte offset in tile
+    const int lg = sw256(d);                 // logical byte offset
+    int row = row0 + (lg >> 7);              // 128-B rows
+    if (row > maxrow - 1) row = maxrow - 1;
+    const int colb = lg & 127;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)(g + (long)row * ldg +
+                                                        (colb >> 1)),
+        (__attribute__((address_space(3))) void*)((char*)lds + d), 16, 0, 0);
+  }
 }
 
-// Fragment load: A/B tile row `row` (0..255), k-group kg (0..3 -> 8 elems).
-DEV_INLINE bf16x8g frag256(const short* lds, int row, int kg) {
-  const int off = sw256(row * 64 + kg * 16);
+// Fragment load: tile row `row` (0..255), k-group kg (0..3 -> 8 elems),
+// k-step ks (0/1 -> elems 0..31 / 32..63).
+DEV_INLINE bf16x8g frag256(const short* lds, int row, int kb /*bytes*/) {
+  const int off = sw256(row * 128 + kb);
   return (bf16x8g)*(const s16x8*)((const char*)lds + off);
 }
 
 #define G_BARRIER() __builtin_amdgcn_s_barrier()
 #define G_WAIT_LGKM0() asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory")
-#define G_WAIT_VM8() asm volatile("s_waitcnt vmcnt(8)" ::: "memory")
 #define G_WAIT_VM0() asm volatile("s_waitcnt vmcnt(0)" ::: "memory")
 
 template <int EPILOGUE>
@@ -115,81 +111,96 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const int ntiles = K / G_BK;
-  const short* Acol = A;  // advanced by k0 via explicit offset below
-  const short* Bcol = B;
 
-  // Prologue: stage K-tiles 0..2 into slots 0..2, full drain once.
+  // Prologue: stage K-tile 0 into slot 0, drain once.
+  {
+    short* a0 = smem;
+    short* b0 = smem + G_TILE_ELEMS;
 #pragma unroll
-  for (int t = 0; t < 3; ++t) {
-    if (t < ntiles) {
-      short* a_lds = smem + t * G_SLOT_ELEMS;
-      short* b_lds = a_lds + G_TILE_ELEMS;
-      const long k0 = (long)t * G_BK;
-#pragma unroll
-      for (int h = 0; h < 2; ++h) {
-        stage256(Acol + k0, lda, bm0, M, h, a_lds);
-        stage256(Bcol + k0, ldb, bn0, N, h, b_lds);
-      }
+    for (int h = 0; h < 2; ++h) {
+      stage256(A, lda, bm0, M, h, a0);
+      stage256(B, ldb, bn0, N, h, b0);
     }
   }
   G_WAIT_VM0();
   G_BARRIER();
 
-  bf16x8g af[4], bf_[4];
+  // Per-wave operand fragments in registers:
+  //   afr[i][ks]: A rows (wm + rh*64 + i*16 + fr), one rh half at a time
+  //   bfr[j][ks]: B rows (wn + j*16 + fr), all four col-frags live
+  bf16x8g afr[4][2], bfr[4][2];
+
   for (int t = 0; t < ntiles; ++t) {
-    const short* a_lds = smem + (t & 3) * G_SLOT_ELEMS;
+    const short* a_lds = smem + (t & 1) * G_SLOT_ELEMS;
     const short* b_lds = a_lds + G_TILE_ELEMS;
-    const int pf = t + 3;          // prefetch tile
-    short* pa_lds = smem + (pf & 3) * G_SLOT_ELEMS;
+    short* pa_lds = smem + ((t + 1) & 1) * G_SLOT_ELEMS;
     short* pb_lds = pa_lds + G_TILE_ELEMS;
-    const long pk0 = (long)pf * G_BK;
-    const bool do_pf = pf < ntiles;
+    const long pk0 = (long)(t + 1) * G_BK;
+    const bool do_pf = t + 1 < ntiles;
 
-    // ---- phase 0: rows half rh=0 (frags 0..3), all 4 col frags --------
+#define G_MFMA_QUAD(RH, CH)                                                \
+  _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                         \
+    _Pragma("unroll") for (int i = 0; i < 4; ++i)                          \
+      _Pragma("unroll") for (int j = 0; j < 2; ++j)                        \
+        acc[(RH) * 4 + i][(CH) * 2 + j] =                                  \
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(                       \
+                afr[i][ks], bfr[(CH) * 2 + j][ks],                         \
+                acc[(RH) * 4 + i][(CH) * 2 + j], 0, 0, 0);
+
+    // ---- q0: quadrant (rh=0, ch=0) ------------------------------------
 #pragma unroll
     for (int i = 0; i < 4; ++i)
-      af[i] = frag256(a_lds, wm + i * 16 + fr, kg);
 #pragma unroll
-    for (int j = 0; j < 4; ++j)
-      bf_[j] = frag256(b_lds, wn + j * 16 + fr, kg);
+      for (int ks = 0; ks < 2; ++ks)
+        afr[i][ks] = frag256(a_lds, wm + i * 16 + fr, ks * 64 + kg * 16);
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+        bfr[j][ks] = frag256(b_lds, wn + j * 16 + fr, ks * 64 + kg * 16);
     if (do_pf) {
-      stage256(Acol + pk0, lda, bm0, M, 0, pa_lds);
-      stage256(Acol + pk0, lda, bm0, M, 1, pa_lds);
+      stage256(A + pk0, lda, bm0, M, 0, pa_lds);
+      stage256(A + pk0, lda, bm0, M, 1, pa_lds);
     }
     G_BARRIER();
     G_WAIT_LGKM0();
-    __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int i = 0; i < 4; ++i)
-#pragma unroll
-      for (int j = 0; j < 4; ++j)
-        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[i], bf_[j], acc[i][j], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
+    G_MFMA_QUAD(0, 0)
     G_BARRIER();
 
-    // ---- phase 1: rows half rh=1 (frags 4..7), B frags reused ---------
+    // ---- q1: quadrant (0, 1) ------------------------------------------
 #pragma unroll
-    for (int i = 0; i < 4; ++i)
-      af[i] = frag256(a_lds, wm + 64 + i * 16 + fr, kg);
+    for (int j = 2; j < 4; ++j)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+        bfr[j][ks] = frag256(b_lds, wn + j * 16 + fr, ks * 64 + kg * 16);
     if (do_pf) {
-      stage256(Bcol + pk0, ldb, bn0, N, 0, pb_lds);
-      stage256(Bcol + pk0, ldb, bn0, N, 1, pb_lds);
+      stage256(B + pk0, ldb, bn0, N, 0, pb_lds);
+      stage256(B + pk0, ldb, bn0, N, 1, pb_lds);
     }
     G_BARRIER();
     G_WAIT_LGKM0();
-    __builtin_amdgcn_s_setprio(1);
+    G_MFMA_QUAD(0, 1)
+    G_BARRIER();
+
+    // ---- q2: quadrant (1, 0) ------------------------------------------
 #pragma unroll
     for (int i = 0; i < 4; ++i)
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
-        acc[4 + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[i], bf_[j], acc[4 + i][j], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
-    // Tile boundary: everything staged before tiles t-1,t (8 loads/wave)
-    // has landed -> tile t+1's slot is complete; never vmcnt(0).
-    G_WAIT_VM8();
+      for (int ks = 0; ks < 2; ++ks)
+        afr[i][ks] =
+            frag256(a_lds, wm + 64 + i * 16 + fr, ks * 64 + kg * 16);
     G_BARRIER();
+    G_WAIT_LGKM0();
+    G_MFMA_QUAD(1, 0)
+    G_BARRIER();
+
+    // ---- q3: quadrant (1, 1), all operands already in registers -------
+    G_MFMA_QUAD(1, 1)
+    // Boundary: tile t+1's stages were issued >=2.5 phases ago; this
+    // drain is (almost always) a no-op, unlike a per-K-step vmcnt(0).
+    G_WAIT_VM0();
+    G_BARRIER();
+#undef G_MFMA_QUAD
   }
 
   // Epilogue: C/D lane map col = lane&15, row = (lane>>4)*4 + r.
@@ -217,12 +228,12 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
 
 // Host-side eligibility check shared with the gemm.hip dispatcher.
 bool gemm256_viable(int M, int N, int K, int lda, int ldb) {
-  if (K % G_BK != 0 || K < 3 * G_BK) return false;
+  if (K % G_BK != 0 || K < 2 * G_BK) return false;
   if (lda % 8 != 0 || ldb % 8 != 0) return false;
   long nwg = (long)cdiv(M, G_BM) * cdiv(N, G_BN);
-  // Measured rule (tools/gemm_bench.py on MI355X): the pipelined kernel
-  // needs the chip full (>=224 WGs at 1 WG/CU) and either deep K or a
-  // grid big enough to amortize the 3-tile prologue per WG.
+  // Measured rule (tools/gemm_bench.py on MI355X): needs the chip full
+  // (1 WG/CU at 128 KiB LDS) and either deep K or a grid big enough to
+  // amortize the prologue.
   return nwg >= 224 && (K >= 1024 || nwg >= 512);
 }
 
@@ -234,8 +245,8 @@ torch::Tensor gemm256_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
               w.is_contiguous(), "gemm256_nt: w must be contiguous bf16 2-D");
   const int M = a.size(0), K = a.size(1), N = w.size(0);
   TORCH_CHECK(w.size(1) == K, "gemm256_nt: K mismatch");
-  TORCH_CHECK(K % G_BK == 0 && K >= 3 * G_BK,
-              "gemm256_nt: K must be a multiple of 32 (>=96)");
+  TORCH_CHECK(K % G_BK == 0 && K >= 2 * G_BK,
+              "gemm256_nt: K must be a multiple of 64 (>=128)");
   const bool has_bias = bias.defined() && bias.numel() > 0;
   torch::Tensor c;
   if (out.has_value()) {
@@ -247,7 +258,7 @@ torch::Tensor gemm256_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
     c = torch::empty({M, N}, a.options());
   }
   const int nbm = cdiv(M, G_BM), nbn = cdiv(N, G_BN);
-  const size_t smem = G_SLOTS * G_SLOT_ELEMS * sizeof(short);  // 128 KiB
+  const size_t smem = 2 * G_SLOT_ELEMS * sizeof(short);  // 128 KiB
   auto stream = at::hip::getCurrentHIPStream();
   static bool attr_set[2] = {false, false};
   auto launch = [&](auto epi) {
