@@ -234,7 +234,7 @@ bool gemm256_viable(int M, int N, int K, int lda, int ldb) {
   // Measured rule (tools/gemm_bench.py on MI355X): needs the chip full
   // (1 WG/CU at 128 KiB LDS) and either deep K or a grid big enough to
   // amortize the prologue.
-  return nwg >= 224 && (K >= 1024 || nwg >= 512);
+  return nwg >= 224 && (K >= 1024 || nwg >= 384);
 }
 
 torch::Tensor gemm256_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
