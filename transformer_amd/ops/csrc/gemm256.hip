@@ -38,24 +38,24 @@ typedef __bf16 bf16x8g __attribute__((ext_vector_type(8)));
 namespace {
 
 #define G_BM 256
-#define G_BN 256
 #define G_BK 64
 #define G_THREADS 512
-// shorts per operand tile (256 x 64) and per slot (A + B)
+// shorts per A tile (256 x 64); B tile is BN_ x 64 (BN_ = 256 or 128)
 #define G_TILE_ELEMS (G_BM * G_BK)
-#define G_SLOT_ELEMS (2 * G_TILE_ELEMS)
 
 DEV_INLINE int sw256(int off) { return off ^ (((off >> 9) & 1) << 5); }
 
-// Stage half h (tile rows [h*128, h*128+128)) of one 256x64 operand tile:
-// 16 chunks of 1 KiB, 2 global_load_lds(16B) per thread.  `g` points at
-// column k0 of the operand (row-major, ldg elems/row); rows clamped.
+// Stage half h (tile rows [h*ROWS/2, ...)) of one ROWSx64 operand tile:
+// ROWS/8 chunks of 1 KiB, ROWS/128 global_load_lds(16B) per thread.
+// `g` points at column k0 of the operand (row-major, ldg elems/row);
+// rows clamped.
+template <int ROWS = 256>
 DEV_INLINE void stage256(const short* __restrict__ g, long ldg, int row0,
                          int maxrow, int h, short* lds) {
   const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
 #pragma unroll
-  for (int p = 0; p < 2; ++p) {
-    const int chunk = h * 16 + p * 8 + wid;
+  for (int p = 0; p < ROWS / 128; ++p) {
+    const int chunk = h * (ROWS / 16) + p * 8 + wid;
     const int d = chunk * 1024 + lane * 16;  // dest byte offset in tile
     const int lg = sw256(d);                 // logical byte offset
     int row = row0 + (lg >> 7);              // 128-B rows
@@ -79,12 +79,17 @@ DEV_INLINE bf16x8g frag256(const short* lds, int row, int kb /*bytes*/) {
 #define G_WAIT_LGKM0() asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory")
 #define G_WAIT_VM0() asm volatile("s_waitcnt vmcnt(0)" ::: "memory")
 
-template <int EPILOGUE>
+// BN_ = 256 (per-wave 128x64, acc[8][4]) or 128 (per-wave 128x32,
+// acc[8][2]) — the narrow variant keeps the chip full on N=512 shapes.
+template <int EPILOGUE, int BN_>
 __global__ __launch_bounds__(G_THREADS, 1)
 void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
                     const short* __restrict__ bias, short* __restrict__ C,
                     int M, int N, int K, int lda, int ldb, int has_bias,
                     int nbm, int nbn) {
+  constexpr int NF = BN_ / 64;          // B col-frags per wave (4 or 2)
+  constexpr int B_ELEMS = BN_ * G_BK;   // shorts per B tile
+  constexpr int SLOT = G_TILE_ELEMS + B_ELEMS;
   extern __shared__ short smem[];
   const int lane = threadIdx.x & 63;
 
@@ -96,19 +101,19 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
     wg = (x < r ? x * (q + 1) : r * (q + 1) + (x - r) * q) + o;
   }
   const int bm0 = (wg / nbn) * G_BM;
-  const int bn0 = (wg % nbn) * G_BN;
+  const int bn0 = (wg % nbn) * BN_;
 
   const int wid = threadIdx.x >> 6;
-  const int wm = (wid >> 2) * 128;  // wave rows [wm, wm+128) of the C tile
-  const int wn = (wid & 3) * 64;    // wave cols [wn, wn+64)
+  const int wm = (wid >> 2) * 128;        // wave rows [wm, wm+128)
+  const int wn = (wid & 3) * (BN_ / 4);   // wave cols
   const int fr = lane & 15;         // fragment lane row/col
   const int kg = lane >> 4;         // k-group 0..3
 
-  f32x4 acc[8][4];
+  f32x4 acc[8][NF];
 #pragma unroll
   for (int i = 0; i < 8; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const int ntiles = K / G_BK;
 
@@ -118,8 +123,8 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
     short* b0 = smem + G_TILE_ELEMS;
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
-      stage256(A, lda, bm0, M, h, a0);
-      stage256(B, ldb, bn0, N, h, b0);
+      stage256<256>(A, lda, bm0, M, h, a0);
+      stage256<BN_>(B, ldb, bn0, N, h, b0);
     }
   }
   G_WAIT_VM0();
@@ -128,12 +133,12 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
   // Per-wave operand fragments in registers:
   //   afr[i][ks]: A rows (wm + rh*64 + i*16 + fr), one rh half at a time
   //   bfr[j][ks]: B rows (wn + j*16 + fr), all four col-frags live
-  bf16x8g afr[4][2], bfr[4][2];
+  bf16x8g afr[4][2], bfr[NF][2];
 
   for (int t = 0; t < ntiles; ++t) {
-    const short* a_lds = smem + (t & 1) * G_SLOT_ELEMS;
+    const short* a_lds = smem + (t & 1) * SLOT;
     const short* b_lds = a_lds + G_TILE_ELEMS;
-    short* pa_lds = smem + ((t + 1) & 1) * G_SLOT_ELEMS;
+    short* pa_lds = smem + ((t + 1) & 1) * SLOT;
     short* pb_lds = pa_lds + G_TILE_ELEMS;
     const long pk0 = (long)(t + 1) * G_BK;
     const bool do_pf = t + 1 < ntiles;
@@ -141,11 +146,11 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
 #define G_MFMA_QUAD(RH, CH)                                                \
   _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                         \
     _Pragma("unroll") for (int i = 0; i < 4; ++i)                          \
-      _Pragma("unroll") for (int j = 0; j < 2; ++j)                        \
-        acc[(RH) * 4 + i][(CH) * 2 + j] =                                  \
+      _Pragma("unroll") for (int j = 0; j < NF / 2; ++j)                   \
+        acc[(RH) * 4 + i][(CH) * (NF / 2) + j] =                           \
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(                       \
-                afr[i][ks], bfr[(CH) * 2 + j][ks],                         \
-                acc[(RH) * 4 + i][(CH) * 2 + j], 0, 0, 0);
+                afr[i][ks], bfr[(CH) * (NF / 2) + j][ks],                  \
+                acc[(RH) * 4 + i][(CH) * (NF / 2) + j], 0, 0, 0);
 
     // ---- q0: quadrant (rh=0, ch=0) ------------------------------------
 #pragma unroll
@@ -154,13 +159,13 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
       for (int ks = 0; ks < 2; ++ks)
         afr[i][ks] = frag256(a_lds, wm + i * 16 + fr, ks * 64 + kg * 16);
 #pragma unroll
-    for (int j = 0; j < 2; ++j)
+    for (int j = 0; j < NF / 2; ++j)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks)
         bfr[j][ks] = frag256(b_lds, wn + j * 16 + fr, ks * 64 + kg * 16);
     if (do_pf) {
-      stage256(A + pk0, lda, bm0, M, 0, pa_lds);
-      stage256(A + pk0, lda, bm0, M, 1, pa_lds);
+      stage256<256>(A + pk0, lda, bm0, M, 0, pa_lds);
+      stage256<256>(A + pk0, lda, bm0, M, 1, pa_lds);
     }
     G_BARRIER();
     G_WAIT_LGKM0();
@@ -169,13 +174,13 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
 
     // ---- q1: quadrant (0, 1) ------------------------------------------
 #pragma unroll
-    for (int j = 2; j < 4; ++j)
+    for (int j = NF / 2; j < NF; ++j)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks)
         bfr[j][ks] = frag256(b_lds, wn + j * 16 + fr, ks * 64 + kg * 16);
     if (do_pf) {
-      stage256(B + pk0, ldb, bn0, N, 0, pb_lds);
-      stage256(B + pk0, ldb, bn0, N, 1, pb_lds);
+      stage256<BN_>(B + pk0, ldb, bn0, N, 0, pb_lds);
+      stage256<BN_>(B + pk0, ldb, bn0, N, 1, pb_lds);
     }
     G_BARRIER();
     G_WAIT_LGKM0();
@@ -208,7 +213,7 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
   for (int i = 0; i < 8; ++i) {
     const int grow_base = bm0 + wm + i * 16 + kg * 4;
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+    for (int j = 0; j < NF; ++j) {
       const int gcol = bn0 + wn + j * 16 + fr;
       if (gcol >= N) continue;
       const float bv = (has_bias && bias) ? bfbits2f(bias[gcol]) : 0.0f;
@@ -226,13 +231,16 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
 
 }  // namespace
 
+// The narrow-B (BN=128) instance keeps the chip full on N<=512 shapes.
+static int gemm256_bn(int N) { return N <= 512 ? 128 : 256; }
+
 // Host-side eligibility check shared with the gemm.hip dispatcher.
 bool gemm256_viable(int M, int N, int K, int lda, int ldb) {
   if (K % G_BK != 0 || K < 2 * G_BK) return false;
   if (lda % 8 != 0 || ldb % 8 != 0) return false;
-  long nwg = (long)cdiv(M, G_BM) * cdiv(N, G_BN);
+  long nwg = (long)cdiv(M, G_BM) * cdiv(N, gemm256_bn(N));
   // Measured rule (tools/gemm_bench.py on MI355X): needs the chip full
-  // (1 WG/CU at 128 KiB LDS) and either deep K or a grid big enough to
+  // (1 WG/CU at >=96 KiB LDS) and either deep K or a grid big enough to
   // amortize the prologue.
   return nwg >= 224 && (K >= 1024 || nwg >= 384);
 }
@@ -257,26 +265,35 @@ torch::Tensor gemm256_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
   } else {
     c = torch::empty({M, N}, a.options());
   }
-  const int nbm = cdiv(M, G_BM), nbn = cdiv(N, G_BN);
-  const size_t smem = 2 * G_SLOT_ELEMS * sizeof(short);  // 128 KiB
+  const int BNv = gemm256_bn(N);
+  const int nbm = cdiv(M, G_BM), nbn = cdiv(N, BNv);
+  const size_t smem = 2 * (G_TILE_ELEMS + (size_t)BNv * G_BK) *
+                      sizeof(short);  // 128 or 96 KiB
   auto stream = at::hip::getCurrentHIPStream();
-  static bool attr_set[2] = {false, false};
-  auto launch = [&](auto epi) {
+  static bool attr_set[2][2] = {{false, false}, {false, false}};
+  auto launch = [&](auto epi, auto bnc) {
     constexpr int E = decltype(epi)::value;
-    if (!attr_set[E]) {
-      (void)hipFuncSetAttribute((const void*)gemm256_kernel<E>,
+    constexpr int BN_ = decltype(bnc)::value;
+    constexpr int bi = BN_ == 128 ? 1 : 0;
+    if (!attr_set[E][bi]) {
+      (void)hipFuncSetAttribute((const void*)gemm256_kernel<E, BN_>,
                                 hipFuncAttributeMaxDynamicSharedMemorySize,
                                 (int)smem);
-      attr_set[E] = true;
+      attr_set[E][bi] = true;
     }
-    gemm256_kernel<E><<<nbm * nbn, G_THREADS, smem, stream>>>(
+    gemm256_kernel<E, BN_><<<nbm * nbn, G_THREADS, smem, stream>>>(
         (const short*)a.data_ptr(), (const short*)w.data_ptr(),
         has_bias ? (const short*)bias.data_ptr() : nullptr,
         (short*)c.data_ptr(), M, N, K, K, K, has_bias, nbm, nbn);
   };
-  if (epilogue == 1)
-    launch(std::integral_constant<int, 1>{});
-  else
-    launch(std::integral_constant<int, 0>{});
+  using E0 = std::integral_constant<int, 0>;
+  using E1 = std::integral_constant<int, 1>;
+  using B256 = std::integral_constant<int, 256>;
+  using B128 = std::integral_constant<int, 128>;
+  if (epilogue == 1) {
+    if (BNv == 128) launch(E1{}, B128{}); else launch(E1{}, B256{});
+  } else {
+    if (BNv == 128) launch(E0{}, B128{}); else launch(E0{}, B256{});
+  }
   return c;
 }
