@@ -231,14 +231,19 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
 
 }  // namespace
 
-// The narrow-B (BN=128) instance keeps the chip full on N<=512 shapes.
-static int gemm256_bn(int N) { return N <= 512 ? 128 : 256; }
+// The narrow-B (BN=128) instance is used only when the BN=256 grid would
+// underfill the 256-CU chip (measured: it wins FFN2-class shapes +8%, but
+// loses on deep-K tall-M shapes like the logits dW where BN=256 already
+// fills — fewer B re-reads).
+static int gemm256_bn(int M, int N) {
+  return ((long)cdiv(M, G_BM) * cdiv(N, 256) < 224 && N <= 512) ? 128 : 256;
+}
 
 // Host-side eligibility check shared with the gemm.hip dispatcher.
 bool gemm256_viable(int M, int N, int K, int lda, int ldb) {
   if (K % G_BK != 0 || K < 2 * G_BK) return false;
   if (lda % 8 != 0 || ldb % 8 != 0) return false;
-  long nwg = (long)cdiv(M, G_BM) * cdiv(N, gemm256_bn(N));
+  long nwg = (long)cdiv(M, G_BM) * cdiv(N, gemm256_bn(M, N));
   // Measured rule (tools/gemm_bench.py on MI355X): needs the chip full
   // (1 WG/CU at >=96 KiB LDS) and either deep K or a grid big enough to
   // amortize the prologue.
@@ -265,7 +270,7 @@ torch::Tensor gemm256_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
   } else {
     c = torch::empty({M, N}, a.options());
   }
-  const int BNv = gemm256_bn(N);
+  const int BNv = gemm256_bn(M, N);
   const int nbm = cdiv(M, G_BM), nbn = cdiv(N, BNv);
   const size_t smem = 2 * (G_TILE_ELEMS + (size_t)BNv * G_BK) *
                       sizeof(short);  // 128 or 96 KiB
