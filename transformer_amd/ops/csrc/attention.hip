@@ -659,10 +659,12 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                 kv_pad.size(0) == B && kv_pad.size(1) == Sk);
     pad = kv_pad.data_ptr<unsigned char>();
   }
-  // RF=2 (32 q-rows/wave) halves K/V re-staging; it costs registers
-  // (~188 vs ~130 VGPR -> 2 vs 3+ waves/SIMD), so it dispatches only for
-  // long sequences where the staging traffic dominates.
-  const bool rf2 = Sk >= 2048;
+  // RF=2 (32 q-rows/wave) halves K/V re-staging but costs registers
+  // (~188 vs ~130 VGPR -> 2 vs 3+ waves/SIMD).  MEASURED: end-to-end
+  // seq-4096 training is ~6% SLOWER with it (L2 already absorbs much of
+  // the re-read; occupancy is the binding constraint), so it stays
+  // disabled; the RF template and its numerics test remain.
+  const bool rf2 = false;
   dim3 grid(B * H, cdiv(Sq, WAVES * (rf2 ? 32 : 16)));
   auto stream = at::hip::getCurrentHIPStream();
   DISPATCH_DH(DH, {
