@@ -135,8 +135,12 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
                     short* __restrict__ C, float* __restrict__ CW,
                     float* __restrict__ DBW, int M,
                     int N, int K, long m_per_slice, int nbk) {
-  __shared__ short a_img[DW_BM * DW_BN];  // dY block
-  __shared__ short b_img[DW_BM * DW_BK];  // X block
+  // Two-slot ring (PMC: this kernel fetches each operand from HBM once —
+  // L2 absorbs all tile re-reads — so it is pipeline-stall bound, not
+  // traffic bound; the old single-buffer __syncthreads() drain exposed
+  // the full staging latency every 64-token block).
+  __shared__ short a_img[2][DW_BM * DW_BN];  // dY block
+  __shared__ short b_img[2][DW_BM * DW_BK];  // X block
 
   // XCD-aware bijective remap: consecutive logical tiles (same n-block,
   // varying k) land on the SAME XCD, so a dY slice is read into one XCD's
@@ -176,6 +180,20 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
   const int db_c4 = (threadIdx.x & 31) << 2;
   float db_acc[4] = {0.f, 0.f, 0.f, 0.f};
 
+  // Prologue: stage block 0 into slot 0, drain once.
+  if (m_lo < m_hi) {
+    if (m_lo + DW_BM <= m_hi && a_full)
+      dw_stage_glds(dY, N, m_lo, bn0, a_img[0]);
+    else
+      dw_stage(dY, N, m_lo, m_hi, bn0, N, a_img[0]);
+    if (m_lo + DW_BM <= m_hi && b_full)
+      dw_stage_glds(X, K, m_lo, bk0, b_img[0]);
+    else
+      dw_stage(X, K, m_lo, m_hi, bk0, K, b_img[0]);
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
   for (long m0 = m_lo; m0 < m_hi; m0 += DW_BM) {
     if (do_db) {
 #pragma unroll
@@ -194,34 +212,62 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
       }
     }
     const bool m_full = m0 + DW_BM <= m_hi;
-    if (m_full && a_full)
-      dw_stage_glds(dY, N, m0, bn0, a_img);
-    else
-      dw_stage(dY, N, m0, m_hi, bn0, N, a_img);
-    if (m_full && b_full)
-      dw_stage_glds(X, K, m0, bk0, b_img);
-    else
-      dw_stage(X, K, m0, m_hi, bk0, K, b_img);
-    __syncthreads();  // drains the glds queue (vmcnt 0) + ds_writes
-    // 2 MFMA k-steps of 32 tokens each
+    const int cur = (int)((m0 - m_lo) / DW_BM);
+    const short* a_lds = a_img[cur & 1];
+    const short* b_lds = b_img[cur & 1];
+    // Prologue staged block 0; here prefetch block cur+1 into the other
+    // slot (its previous tenant was consumed last block) — all 8 glds
+    // issued in phase 0, ~1.5 compute phases to land before the boundary
+    // vmcnt(0) (template schedule, see gemm256.hip).
+    const long mN = m0 + DW_BM;
+    const bool pf = mN < m_hi;
+    bf16x8d af[4], bf_[4];
+    // ---- phase 0: k-step 0 -------------------------------------------
 #pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      const int ms = ks * 32 + kg * 8;
-      bf16x8d af[4], bf_[4];
+    for (int i = 0; i < 4; ++i)
+      af[i] = dw_frag(a_lds, kg * 8, wn + i * 16);
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
-        af[i] = dw_frag(a_img, ms, wn + i * 16);
+    for (int j = 0; j < 4; ++j)
+      bf_[j] = dw_frag(b_lds, kg * 8, wk + j * 16);
+    if (pf) {
+      short* pa = a_img[(cur + 1) & 1];
+      short* pb = b_img[(cur + 1) & 1];
+      if (mN + DW_BM <= m_hi && a_full)
+        dw_stage_glds(dY, N, mN, bn0, pa);
+      else
+        dw_stage(dY, N, mN, m_hi, bn0, N, pa);
+      if (mN + DW_BM <= m_hi && b_full)
+        dw_stage_glds(X, K, mN, bk0, pb);
+      else
+        dw_stage(X, K, mN, m_hi, bk0, K, pb);
+    }
+    __builtin_amdgcn_s_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
 #pragma unroll
       for (int j = 0; j < 4; ++j)
-        bf_[j] = dw_frag(b_img, ms, wk + j * 16);
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[i], bf_[j], acc[i][j], 0, 0, 0);
+    __builtin_amdgcn_s_barrier();
+    // ---- phase 1: k-step 1 -------------------------------------------
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
+    for (int i = 0; i < 4; ++i)
+      af[i] = dw_frag(a_lds, 32 + kg * 8, wn + i * 16);
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[i], bf_[j], acc[i][j], 0, 0, 0);
-    }
-    __syncthreads();
+    for (int j = 0; j < 4; ++j)
+      bf_[j] = dw_frag(b_lds, 32 + kg * 8, wk + j * 16);
+    __builtin_amdgcn_s_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[i], bf_[j], acc[i][j], 0, 0, 0);
+    // boundary: prefetched block cur+1 must have landed
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
   }
 
   if (do_db) {
