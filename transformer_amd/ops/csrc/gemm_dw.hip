@@ -343,8 +343,15 @@ torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x,
   // savings, so the 128-tile kernel is the only dispatch target.
   const int nbn = cdiv(N, DW_BN), nbk = cdiv(K, DW_BK);
   const int ntiles = nbn * nbk;
-  // slices: fill >=512 WGs, 64-token quanta
-  int nslices = (int)min((M + DW_BM - 1) / DW_BM, (long)cdiv(512, ntiles));
+  // slices: fill >=DW_TARGET_WGS workgroups, 64-token quanta.  More slices
+  // = more parallelism but more fp32 atomic traffic + per-WG prologue;
+  // 512 measured best (sweepable via TFMX_DW_WGS for tools/gemm_bench.py).
+  static int target_wgs = [] {
+    const char* e = getenv("TFMX_DW_WGS");
+    return e ? atoi(e) : 512;
+  }();
+  int nslices = (int)min((M + DW_BM - 1) / DW_BM,
+                         (long)cdiv(target_wgs, ntiles));
   long m_per_slice = (M + nslices - 1) / nslices;
   m_per_slice = (m_per_slice + DW_BM - 1) / DW_BM * DW_BM;
   nslices = (int)((M + m_per_slice - 1) / m_per_slice);
