@@ -129,3 +129,67 @@ def test_flat_grad_aliasing_survives_backward():
         loss.backward()
     opt.flat.check()
     assert opt.flat.flat_g.abs().sum() > 0
+
+
+def _cb_worker(rank, world, tmpdir, q):
+    """Drives the OPS-LAYER readiness callback path (the one the fused GPU
+    kernels fire via _grad_ready) instead of autograd hooks: fill flat_g
+    directly, fire callbacks in reverse order for MOST params, leave a
+    straggler bucket for finalize()."""
+    dist.init_process_group("gloo", init_method=f"file://{tmpdir}/pg2",
+                            rank=rank, world_size=world)
+    try:
+        from transformer_amd.ops import functional as F
+
+        model = _model()
+        opt = NoamAdam(model, 16, warmup_steps=10, use_flat=True)
+        ddp = BucketedDataParallel(opt.flat, bucket_mb=0.01)
+        ddp.broadcast_parameters()
+        # fill only the PARAM regions: alignment padding between params
+        # is deliberately outside every bucket (its gradient is always
+        # zero in real training, so it needs no all-reduce)
+        torch.manual_seed(1000 + rank)
+        full = torch.randn_like(opt.flat.flat_g)
+        opt.flat.flat_g.zero_()
+        for pr, off in zip(opt.flat.params, opt.flat.offsets):
+            opt.flat.flat_g[off:off + pr.numel()] = full[off:off + pr.numel()]
+        mine = opt.flat.flat_g.detach().clone()
+        # fire callbacks for all params EXCEPT the ones in the last bucket
+        skip = {id(p) for p in ddp.buckets[-1]["params"]}
+        for p in reversed(opt.flat.params):
+            if id(p) not in skip:
+                F._grad_ready(p)
+        ddp.finalize()  # must launch the straggler bucket and wait all
+        if rank == 0:
+            q.put((mine, opt.flat.flat_g.detach().clone()))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_callback_readiness_path_and_stragglers(tmp_path):
+    """The kernel-enqueue readiness callbacks (used by the flat GPU path)
+    launch buckets as they complete, and finalize() handles buckets whose
+    params never fired — the all-reduced flat gradient must equal the sum
+    of both ranks' local gradients in every bucket."""
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    ps = [ctx.Process(target=_cb_worker, args=(r, 2, str(tmp_path), q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    mine0, reduced = q.get()
+    for p in ps:
+        p.join(120)
+        assert p.exitcode == 0
+    torch.manual_seed(1001)
+    other_full = torch.randn_like(mine0)
+    other = torch.zeros_like(mine0)
+    # rank 1 masked the same param regions
+    from transformer_amd.runtime.optimizer import FlatParams
+    m = _model()
+    fp = FlatParams(m)
+    for pr, off in zip(fp.params, fp.offsets):
+        other[off:off + pr.numel()] = other_full[off:off + pr.numel()]
+    assert torch.allclose(reduced, mine0 + other, atol=1e-5), \
+        (reduced - mine0 - other).abs().max()
