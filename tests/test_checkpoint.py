@@ -94,3 +94,40 @@ def test_export_and_reload_forward_parity(tmp_path):
         a, _ = m((src, tar), training=False)
         b, _ = m2((src, tar), training=False)
     assert torch.equal(a, b)
+
+
+def test_serve_endpoints(tmp_path, toy_corpus):
+    """serve.py: export a tiny trained-ish model + vocabs, then exercise
+    /health and /translate through FastAPI's in-process test client."""
+    import torch
+    from fastapi.testclient import TestClient
+
+    from transformer_amd.data.dataset import load_dataset
+    from transformer_amd.models import Transformer
+    from transformer_amd.runtime import export_model
+    import serve
+
+    _, _, src_tok, tgt_tok = load_dataset(
+        toy_corpus, str(tmp_path / "sv"), str(tmp_path / "tv"),
+        sequence_length=50, batch_size=4, seed=1)
+    torch.manual_seed(0)
+    m = Transformer(num_layers=1, d_model=16, num_heads=2, dff=32,
+                    input_vocab_size=src_tok.vocab_size + 2,
+                    target_vocab_size=tgt_tok.vocab_size + 2,
+                    rate=0.0, max_position=64)
+    cfg = dict(num_layers=1, d_model=16, num_heads=2, dff=32,
+               input_vocab_size=src_tok.vocab_size + 2,
+               target_vocab_size=tgt_tok.vocab_size + 2,
+               dropout_rate=0.0, max_position=64)
+    export_model(m, str(tmp_path / "model"), cfg)
+
+    app = serve.build_app(str(tmp_path / "model"), str(tmp_path / "sv"),
+                          str(tmp_path / "tv"))
+    client = TestClient(app)
+    r = client.get("/health")
+    assert r.status_code == 200 and r.json()["status"] == "ok"
+    r = client.post("/translate", json={"text": "one two", "max_len": 4})
+    assert r.status_code == 200
+    body = r.json()
+    assert isinstance(body["tokens"], list) and len(body["tokens"]) >= 1
+    assert "text" in body
