@@ -17,9 +17,10 @@
 //
 // Tiling: output 128(n) x 128(k) per 256-thread WG (4 waves, 64x64 each,
 // acc[4][4] 16x16 frags); contraction in 64-token blocks (2 MFMA k-steps);
-// the M axis is split across blockIdx.y slices with fp32 atomic
-// accumulation + a finalize pass so the 256-CU chip stays full even when
-// N=K=512 (16 output tiles).
+// the M axis is split across blockIdx.y slices writing DISJOINT fp32
+// partials (an atomicAdd epilogue measured at the chip's fp32-atomic
+// rate), reduced by a chip-filling pass, so the 256-CU chip stays full
+// even when N=K=512 (16 output tiles).
 #include "common.h"
 
 #include <torch/extension.h>
@@ -279,6 +280,11 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
   // Epilogue: D lane map col = lane&15 -> k, row = (lane>>4)*4 + r -> n.
   // The fragment A row role is n (tr-read result lane = n), so output rows
   // follow the standard C/D mapping with rows = n, cols = k.
+  // SPLIT epilogue writes DISJOINT per-slice fp32 partials (plain
+  // stores): an atomicAdd version measured at the chip's fp32-atomic
+  // rate (~8.4M atomics/call ~= the whole kernel time); the partials
+  // are reduced by dw_reduce_kernel over a chip-filling N*K grid.
+  float* cw_slice = SPLIT ? CW + (long)blockIdx.y * N * K : nullptr;
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     const int nrow_base = bn0 + wn + i * 16 + kg * 4;
@@ -291,7 +297,7 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
         const int nrow = nrow_base + r;
         if (nrow >= N) continue;
         if (SPLIT)
-          atomicAdd(&CW[(long)nrow * K + kcol], acc[i][j][r]);
+          cw_slice[(long)nrow * K + kcol] = acc[i][j][r];
         else
           C[(long)nrow * K + kcol] = f2bfbits(acc[i][j][r]);
       }
@@ -299,16 +305,22 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
   }
 }
 
-// CLEAR: re-zero the workspace after reading so a cached workspace is
-// ready for the next same-shape call without a separate fill_ launch.
-template <bool CLEAR>
-__global__ void dw_finalize_kernel(float* __restrict__ cw,
-                                   short* __restrict__ c, long nk) {
+// Reduce the nslices disjoint fp32 partials -> bf16 C.  N*K is large
+// (>=2^18 for every dispatched shape), so this grid fills the chip.
+__global__ void dw_reduce_kernel(const float* __restrict__ cw,
+                                 short* __restrict__ c, long nk,
+                                 int nslices) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < nk) {
-    c[i] = f2bfbits(cw[i]);
-    if (CLEAR) cw[i] = 0.f;
-  }
+  if (i >= nk) return;
+  float s = 0.f;
+  for (int sl = 0; sl < nslices; ++sl) s += cw[(long)sl * nk + i];
+  c[i] = f2bfbits(s);
+}
+
+__global__ void dw_cast_kernel(const float* __restrict__ cw,
+                               short* __restrict__ c, long nk) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < nk) c[i] = f2bfbits(cw[i]);
 }
 
 }  // namespace
@@ -378,29 +390,28 @@ torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x,
   if (nslices == 1) {
     if (has_db) launch(F{}, T{}, nullptr); else launch(F{}, F{}, nullptr);
   } else {
-    // caller may pass a cached ALREADY-ZERO workspace (the finalize pass
-    // below re-zeroes it), killing the per-call fill_ launch+write.
+    // Disjoint per-slice fp32 partials, plain stores, no zeroing needed
+    // (every element of every slice is written); reduced by a
+    // chip-filling pass.  `cw_cached` (an optional caller workspace from
+    // the old atomic scheme) is reused as backing storage when large
+    // enough.
+    long nk = (long)N * K;
     torch::Tensor cw;
-    bool cached = cw_cached.has_value();
-    if (cached) {
+    if (cw_cached.has_value() &&
+        cw_cached->numel() >= (long)nslices * nk &&
+        cw_cached->dtype() == torch::kFloat32) {
       cw = *cw_cached;
-      TORCH_CHECK(cw.dtype() == torch::kFloat32 && cw.is_cuda() &&
-                  cw.numel() >= (long)N * K, "gemm_dw: bad cw workspace");
     } else {
-      cw = torch::zeros({N, K}, dy.options().dtype(torch::kFloat32));
+      cw = torch::empty({(long)nslices * nk},
+                        dy.options().dtype(torch::kFloat32));
     }
     if (has_db) launch(T{}, T{}, cw.data_ptr<float>());
     else launch(T{}, F{}, cw.data_ptr<float>());
-    long nk = (long)N * K;
-    if (cached)
-      dw_finalize_kernel<true><<<(nk + 1023) / 1024, 1024, 0, stream>>>(
-          cw.data_ptr<float>(), (short*)c.data_ptr(), nk);
-    else
-      dw_finalize_kernel<false><<<(nk + 1023) / 1024, 1024, 0, stream>>>(
-          cw.data_ptr<float>(), (short*)c.data_ptr(), nk);
+    dw_reduce_kernel<<<(nk + 255) / 256, 256, 0, stream>>>(
+        cw.data_ptr<float>(), (short*)c.data_ptr(), nk, nslices);
   }
   if (has_db)
-    dw_finalize_kernel<false><<<cdiv(N, 256), 256, 0, stream>>>(
+    dw_cast_kernel<<<cdiv(N, 256), 256, 0, stream>>>(
         dbw.data_ptr<float>(), (short*)db->data_ptr(), N);
   return c;
 }

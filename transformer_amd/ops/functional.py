@@ -53,9 +53,6 @@ def _flat(p):
 # Linear: y = x @ W^T + b, optional fused ReLU epilogue (K1/K7/K8/K12).
 # ---------------------------------------------------------------------------
 
-_DW_WORKSPACES: dict = {}
-
-
 def _dw_gemm(dy, x, out=None):
     """dW[N,K] = dY^T @ X — plain GEMM; backend measured per shape on
     MI355X (tools/gemm_bench.py): hipBLASLt TN wins at the wide logits
@@ -66,16 +63,11 @@ def _dw_gemm(dy, x, out=None):
         if out is not None:
             return torch.matmul(dy.t(), x, out=out)
         return torch.matmul(dy.t(), x)
-    # per-shape fp32 workspace cache: gemm_dw's finalize re-zeroes it, so
-    # only the first call per shape pays the zero-fill (was one fill_
-    # launch per linear per step).
-    key = (dy.shape[1], x.shape[1], dy.device)
-    cw = _DW_WORKSPACES.get(key)
-    if cw is None:
-        cw = torch.zeros(dy.shape[1] * x.shape[1], device=dy.device,
-                         dtype=torch.float32)
-        _DW_WORKSPACES[key] = cw
-    return E.gemm_dw(dy, x, out, None, cw)
+    # gemm_dw writes DISJOINT per-slice fp32 partials (no atomics, no
+    # zeroing — the fp32-atomic epilogue measured at the chip's atomic
+    # rate); it sizes its own torch::empty workspace via the caching
+    # allocator, so no Python-side workspace is needed.
+    return E.gemm_dw(dy, x, out, None, None)
 
 
 def _dw_db_gemm(dy, x, has_bias, w_out=None, b_out=None):
