@@ -363,43 +363,24 @@ void transpose2d_kernel(const short* __restrict__ in, short* __restrict__ out,
 // disjoint-partials scheme (no atomics) was tried and measured SLOWER:
 // its gy-deep serial reduction ran on a 2-8 block grid.
 #define COLSUM_ROWS 128
-// 2 columns per thread (s16x2 loads) x 4-row unroll, 64-thread blocks so
-// small-N shapes still fill the 256-CU chip — the scalar one-col version
-// ran latency-bound at ~1 TB/s (same disease ln_gb had).
-__global__ __launch_bounds__(64)
+__global__ __launch_bounds__(256)
 void colsum_part_kernel(const short* __restrict__ dy, float* __restrict__ acc,
                         int M, int N) {
-  const int c = (blockIdx.x * 64 + threadIdx.x) * 2;
-  if (c >= N) return;
-  const long m0 = (long)blockIdx.y * COLSUM_ROWS;
-  const long m1 = min((long)M, m0 + COLSUM_ROWS);
-  if (c + 1 < N && (N % 2 == 0)) {
-    float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
-    long m = m0;
-    for (; m + 4 <= m1; m += 4) {
-      s16x2 a = *(const s16x2*)(dy + m * N + c);
-      s16x2 b = *(const s16x2*)(dy + (m + 1) * N + c);
-      s16x2 d = *(const s16x2*)(dy + (m + 2) * N + c);
-      s16x2 e = *(const s16x2*)(dy + (m + 3) * N + c);
-      s0 += bfbits2f(a[0]) + bfbits2f(b[0]);
-      s1 += bfbits2f(a[1]) + bfbits2f(b[1]);
-      s2 += bfbits2f(d[0]) + bfbits2f(e[0]);
-      s3 += bfbits2f(d[1]) + bfbits2f(e[1]);
-    }
-    for (; m < m1; ++m) {
-      s16x2 a = *(const s16x2*)(dy + m * N + c);
-      s0 += bfbits2f(a[0]);
-      s1 += bfbits2f(a[1]);
-    }
-    atomicAdd(&acc[c], s0 + s2);
-    atomicAdd(&acc[c + 1], s1 + s3);
-  } else {
-    for (int cc = c; cc < min(c + 2, N); ++cc) {
-      float s = 0.f;
-      for (long m = m0; m < m1; ++m) s += bfbits2f(dy[m * N + cc]);
-      atomicAdd(&acc[cc], s);
-    }
+  int n = blockIdx.x * 256 + threadIdx.x;
+  if (n >= N) return;
+  long m0 = (long)blockIdx.y * COLSUM_ROWS;
+  long m1 = min((long)M, m0 + COLSUM_ROWS);
+  // 4 independent accumulators for memory-level parallelism
+  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+  long m = m0;
+  for (; m + 4 <= m1; m += 4) {
+    s0 += bfbits2f(dy[m * N + n]);
+    s1 += bfbits2f(dy[(m + 1) * N + n]);
+    s2 += bfbits2f(dy[(m + 2) * N + n]);
+    s3 += bfbits2f(dy[(m + 3) * N + n]);
   }
+  for (; m < m1; ++m) s0 += bfbits2f(dy[m * N + n]);
+  atomicAdd(&acc[n], (s0 + s1) + (s2 + s3));
 }
 
 __global__ void cast_colsum_kernel(float* __restrict__ in,
@@ -602,8 +583,8 @@ torch::Tensor colsum(torch::Tensor a, c10::optional<torch::Tensor> out_opt) {
     out = torch::empty({N}, a.options());
   }
   auto stream = at::hip::getCurrentHIPStream();
-  dim3 grid(cdiv(cdiv(N, 2), 64), cdiv(M, COLSUM_ROWS));
-  colsum_part_kernel<<<grid, 64, 0, stream>>>(
+  dim3 grid(cdiv(N, 256), cdiv(M, COLSUM_ROWS));
+  colsum_part_kernel<<<grid, 256, 0, stream>>>(
       (const short*)a.data_ptr(), acc.data_ptr<float>(), M, N);
   cast_colsum_kernel<<<cdiv(N, 256), 256, 0, stream>>>(
       acc.data_ptr<float>(), (short*)out.data_ptr(), N);
