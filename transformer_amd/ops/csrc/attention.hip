@@ -80,6 +80,31 @@ DEV_INLINE bf16x8 lds_read8_tr(const short* base, int kb, int col0) {
   return (bf16x8)v;
 }
 
+
+// Per-wave C-scratch in TRANSPOSED layout [C-col][16 C-rows]: each lane
+// writes its 4 accumulator rows (fixed C-col) as ONE b64 store, and the
+// next stage reads MFMA A-fragments back with ds_read_b64_tr_b16 — no
+// scalar b16 scatter in either direction.
+DEV_INLINE void scrT_write4(short* img, int ccol, int rr0, s16x4 v) {
+  *(s16x4*)&img[ccol * 16 + rr0] = v;
+}
+
+// A-fragment from the transposed scratch: lane l receives rows
+// (inner C-row = l&15) x outer chunk [ob, ob+8).
+DEV_INLINE bf16x8 scrT_read8(const short* img, int ob) {
+  const int mp = threadIdx.x & 15;
+  const char* base = (const char*)img;
+  s16x4t lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (__attribute__((address_space(3))) s16x4t*)(
+          const_cast<char*>(base) + ((ob + (mp >> 2)) * 16 + 4 * (mp & 3)) * 2));
+  s16x4t hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (__attribute__((address_space(3))) s16x4t*)(
+          const_cast<char*>(base) +
+          ((ob + 4 + (mp >> 2)) * 16 + 4 * (mp & 3)) * 2));
+  s16x8 v = {lo[0], lo[1], lo[2], lo[3], hi[0], hi[1], hi[2], hi[3]};
+  return (bf16x8)v;
+}
+
 // Stage a [KVT][DH] bf16 tile from global rows (stride row_stride elems)
 // into swizzled LDS: natural layout into lds_n (if WRITE_N) and/or the
 // transpose [DH][KVT] into lds_t (if WRITE_T).
@@ -270,13 +295,13 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
           l_run[rf][r] = l_run[rf][r] * (half == 0 ? alpha[r] : 1.f) +
                          psum[r];
         }
-#pragma unroll
-        for (int r = 0; r < 4; ++r)
-          lds_write1<FKVT * 2>(p_lds[wid], kg * 4 + r, half * 16 + fr,
-                               f2bfbits(p_raw[half][r]));
+        // one b64 store: P rows q=kg*4.. at transposed-scratch col = key
+        s16x4 pw = {f2bfbits(p_raw[half][0]), f2bfbits(p_raw[half][1]),
+                    f2bfbits(p_raw[half][2]), f2bfbits(p_raw[half][3])};
+        scrT_write4(p_lds[wid], half * 16 + fr, kg * 4, pw);
       }
-      // p_lds is per-wave: a wave barrier orders the scalar P writes
-      // against this wave's own PV reads.
+      // p_lds is per-wave: a wave barrier orders the P writes against
+      // this wave's own PV reads.
       __builtin_amdgcn_wave_barrier();
 
       // ---- O += P · V  (A = P[q][key] from LDS, B = V^T[d][key]) -----
@@ -285,8 +310,7 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
       for (int ks = 0; ks < FKVT / 32; ++ks)
 #pragma unroll
         for (int i = 0; i < D16; ++i) {
-          bf16x8 pa =
-              lds_read8<FKVT * 2>(p_lds[wid], fr, ks * 32 + kg * 8);
+          bf16x8 pa = scrT_read8(p_lds[wid], ks * 32 + kg * 8);
           bf16x8 vb = TRV
               ? lds_read8_tr<DH * 2>(v_lds, ks * 32 + kg * 8, i * 16)
               : lds_read8<FKVT * 2>(v_lds, i * 16 + fr, ks * 32 + kg * 8);
@@ -442,6 +466,7 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
       const bool q_ok = qcol < Sq;
       const float lse_q = q_ok ? lse[min(qcol, Sq - 1)] : 0.f;
       const float d_q = q_ok ? dl[min(qcol, Sq - 1)] : 0.f;
+      s16x4 pw;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int key = k0w + kg * 4 + r;
@@ -450,31 +475,32 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
                       (causal && key > qcol);
         float p = masked ? 0.f : __expf(x - lse_q);
         ds_reg[half][r] = p * (dpt[r] - d_q) * scale;
-        lds_write1<KVT * 2>(x_lds[wid], kg * 4 + r, half * 16 + fr, f2bfbits(p));
+        pw[r] = f2bfbits(p);
       }
+      scrT_write4(x_lds[wid], half * 16 + fr, kg * 4, pw);
     }
     __builtin_amdgcn_wave_barrier();
     // dV += P^T · dO   (A = full P^T[key][q0..31], B = dO^T[d][q])
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int i = 0; i < D16; ++i) {
-      bf16x8 pa = lds_read8<KVT * 2>(x_lds[wid], fr, kg * 8);
+      bf16x8 pa = scrT_read8(x_lds[wid], kg * 8);
       bf16x8 db = lds_read8_tr<DH * 2>(do_lds, kg * 8, i * 16);
       acc_dv[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, db, acc_dv[i], 0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
     // pass 2: both halves of dS^T, then dK += dS^T · Q
 #pragma unroll
-    for (int half = 0; half < 2; ++half)
-#pragma unroll
-      for (int r = 0; r < 4; ++r)
-        lds_write1<KVT * 2>(x_lds[wid], kg * 4 + r, half * 16 + fr,
-                            f2bfbits(ds_reg[half][r]));
+    for (int half = 0; half < 2; ++half) {
+      s16x4 dw = {f2bfbits(ds_reg[half][0]), f2bfbits(ds_reg[half][1]),
+                  f2bfbits(ds_reg[half][2]), f2bfbits(ds_reg[half][3])};
+      scrT_write4(x_lds[wid], half * 16 + fr, kg * 4, dw);
+    }
     __builtin_amdgcn_wave_barrier();
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int i = 0; i < D16; ++i) {
-      bf16x8 sa = lds_read8<KVT * 2>(x_lds[wid], fr, kg * 8);
+      bf16x8 sa = scrT_read8(x_lds[wid], kg * 8);
       bf16x8 qb2 = lds_read8_tr<DH * 2>(q_lds, kg * 8, i * 16);
       acc_dk[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(sa, qb2, acc_dk[i], 0, 0, 0);
     }
@@ -563,6 +589,7 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
 #pragma unroll
     for (int half = 0; half < 2; ++half) {
       f32x4 s = {0, 0, 0, 0}, dp = {0, 0, 0, 0};
+      s16x4 dsw;
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int d = 0; d < D32; ++d) {
@@ -582,17 +609,16 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
         float d_q = row_ok ? dl[min(qrow, Sq - 1)] : 0.f;
         bool masked = !row_ok || col_pad || (causal && kcol > qrow);
         float p = masked ? 0.f : __expf(s[r] * scale - lse_q);
-        float ds = p * (dp[r] - d_q) * scale;
-        lds_write1<KVT * 2>(x_lds[wid], kg * 4 + r, half * 16 + fr,
-                            f2bfbits(ds));
+        dsw[r] = f2bfbits(p * (dp[r] - d_q) * scale);
       }
+      scrT_write4(x_lds[wid], half * 16 + fr, kg * 4, dsw);
     }
     __builtin_amdgcn_wave_barrier();
     // dQ += dS · K   (A = dS[q][key] from LDS, B = K^T[d][key])
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int i = 0; i < D16; ++i) {
-      bf16x8 sa = lds_read8<KVT * 2>(x_lds[wid], fr, kg * 8);
+      bf16x8 sa = scrT_read8(x_lds[wid], kg * 8);
       bf16x8 kb2 = lds_read8_tr<DH * 2>(k_lds, kg * 8, i * 16);
       acc_dq[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(sa, kb2, acc_dq[i], 0, 0, 0);
     }
