@@ -232,24 +232,30 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
       float tile_pmax[4];
 #pragma unroll
       for (int r = 0; r < 4; ++r) tile_pmax[r] = -1e30f;
+      // all NHALF QK^T chains issued back-to-back (independent
+      // accumulators -> MFMA ILP) before any softmax VALU
+      f32x4 s2h[NHALF];
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int half = 0; half < NHALF; ++half) {
-        f32x4 s2 = {0, 0, 0, 0};
-        __builtin_amdgcn_s_setprio(1);
+        s2h[half] = {0, 0, 0, 0};
 #pragma unroll
         for (int d = 0; d < D32; ++d) {
           bf16x8 kf =
               lds_read8<DH * 2>(k_lds, half * 16 + fr, d * 32 + kg * 8);
-          s2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[rf][d], kf, s2,
-                                                       0, 0, 0);
+          s2h[half] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              qf[rf][d], kf, s2h[half], 0, 0, 0);
         }
-        __builtin_amdgcn_s_setprio(0);
+      }
+      __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+      for (int half = 0; half < NHALF; ++half) {
         const int kcol = k0 + half * 16 + fr;   // C col = lane&15
         const bool col_pad = (kcol >= Sk) || (pad && pad[min(kcol, Sk - 1)]);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int qrow = qr0 + kg * 4 + r;    // C row = (lane>>4)*4+r
-          float x = s2[r] * scale;
+          float x = s2h[half][r] * scale;
           if (col_pad) x += NEG_BIG;
           if (causal && kcol > qrow) x += NEG_BIG;
           p_raw[half][r] = x;
