@@ -455,28 +455,19 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
 
     // pass 1: both halves of P^T into x_lds; dS^T kept in registers
     float ds_reg[2][4];
-    // both halves' S^T and dP^T chains issued together: 4 independent
-    // MFMA accumulators before any softmax/VALU work
-    f32x4 st2[2], dpt2[2];
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int half = 0; half < 2; ++half) {
-      st2[half] = {0, 0, 0, 0};
-      dpt2[half] = {0, 0, 0, 0};
+      // S^T[key][q] = K·Q^T ; dP^T[key][q] = V·dO^T   (C row=key, col=q)
+      f32x4 st = {0, 0, 0, 0}, dpt = {0, 0, 0, 0};
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int d = 0; d < D32; ++d) {
         bf16x8 qb_ = lds_read8<DH * 2>(q_lds, half * 16 + fr, d * 32 + kg * 8);
         bf16x8 dob = lds_read8<DH * 2>(do_lds, half * 16 + fr, d * 32 + kg * 8);
-        st2[half] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            kf[d], qb_, st2[half], 0, 0, 0);
-        dpt2[half] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            vf[d], dob, dpt2[half], 0, 0, 0);
+        st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[d], qb_, st, 0, 0, 0);
+        dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vf[d], dob, dpt, 0, 0, 0);
       }
-    }
-    __builtin_amdgcn_s_setprio(0);
-#pragma unroll
-    for (int half = 0; half < 2; ++half) {
-      const f32x4 st = st2[half], dpt = dpt2[half];
+      __builtin_amdgcn_s_setprio(0);
       const int qcol = j0 + half * 16 + fr;
       const bool q_ok = qcol < Sq;
       const float lse_q = q_ok ? lse[min(qcol, Sq - 1)] : 0.f;
@@ -601,27 +592,19 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
                               v_lds, nullptr);
     __syncthreads();
 
-    f32x4 s2b[2], dp2[2];
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int half = 0; half < 2; ++half) {
-      s2b[half] = {0, 0, 0, 0};
-      dp2[half] = {0, 0, 0, 0};
+      f32x4 s = {0, 0, 0, 0}, dp = {0, 0, 0, 0};
+      s16x4 dsw;
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int d = 0; d < D32; ++d) {
         bf16x8 kb_ = lds_read8<DH * 2>(k_lds, half * 16 + fr, d * 32 + kg * 8);
         bf16x8 vb_ = lds_read8<DH * 2>(v_lds, half * 16 + fr, d * 32 + kg * 8);
-        s2b[half] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            qf[d], kb_, s2b[half], 0, 0, 0);
-        dp2[half] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            dof[d], vb_, dp2[half], 0, 0, 0);
+        s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[d], kb_, s, 0, 0, 0);
+        dp = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof[d], vb_, dp, 0, 0, 0);
       }
-    }
-    __builtin_amdgcn_s_setprio(0);
-#pragma unroll
-    for (int half = 0; half < 2; ++half) {
-      const f32x4 s = s2b[half], dp = dp2[half];
-      s16x4 dsw;
+      __builtin_amdgcn_s_setprio(0);
       const int kcol = k0 + half * 16 + fr;
       const bool col_pad = (kcol >= Sk) || (pad && pad[min(kcol, Sk - 1)]);
 #pragma unroll
