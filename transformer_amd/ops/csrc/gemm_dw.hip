@@ -222,19 +222,14 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
     // vmcnt(0) (template schedule, see gemm256.hip).
     const long mN = m0 + DW_BM;
     const bool pf = mN < m_hi;
-    // single phase per 64-token block: both k-steps' fragments hoisted
-    // (32 tr16 reads), prefetch issued before the barrier, then 32
-    // independent MFMAs — one barrier pair per block instead of two.
-    bf16x8d af[2][4], bf_[2][4];
+    bf16x8d af[4], bf_[4];
+    // ---- phase 0: k-step 0 -------------------------------------------
 #pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
+    for (int i = 0; i < 4; ++i)
+      af[i] = dw_frag(a_lds, kg * 8, wn + i * 16);
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
-        af[ks][i] = dw_frag(a_lds, ks * 32 + kg * 8, wn + i * 16);
-#pragma unroll
-      for (int j = 0; j < 4; ++j)
-        bf_[ks][j] = dw_frag(b_lds, ks * 32 + kg * 8, wk + j * 16);
-    }
+    for (int j = 0; j < 4; ++j)
+      bf_[j] = dw_frag(b_lds, kg * 8, wk + j * 16);
     if (pf) {
       short* pa = a_img[(cur + 1) & 1];
       short* pb = b_img[(cur + 1) & 1];
@@ -250,13 +245,27 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 #pragma unroll
-    for (int ks = 0; ks < 2; ++ks)
+    for (int i = 0; i < 4; ++i)
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
+      for (int j = 0; j < 4; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[i], bf_[j], acc[i][j], 0, 0, 0);
+    __builtin_amdgcn_s_barrier();
+    // ---- phase 1: k-step 1 -------------------------------------------
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[ks][i], bf_[ks][j], acc[i][j], 0, 0, 0);
+    for (int i = 0; i < 4; ++i)
+      af[i] = dw_frag(a_lds, 32 + kg * 8, wn + i * 16);
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      bf_[j] = dw_frag(b_lds, 32 + kg * 8, wk + j * 16);
+    __builtin_amdgcn_s_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[i], bf_[j], acc[i][j], 0, 0, 0);
     // boundary: prefetched block cur+1 must have landed
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
