@@ -214,7 +214,11 @@ class Train:
             if hasattr(train_dataset, "set_epoch"):
                 train_dataset.set_epoch(epoch)
 
+            interval_tokens = 0
+            interval_t0 = time.time()
             for inputs in train_dataset:
+                src, tar = inputs
+                interval_tokens += src.numel() + tar.numel()
                 self.train_step(inputs)
                 step += 1
                 if self._stop_requested:
@@ -223,13 +227,27 @@ class Train:
                         print("checkpointed on stop request; exiting loop")
                     return
                 if step % self.log_interval == 0:
+                    # tokens/sec over the interval (the north-star metric;
+                    # SURVEY.md §5 metrics) — whole-job rate in DP
+                    import torch.distributed as dist
+                    world = (dist.get_world_size()
+                             if dist.is_available() and dist.is_initialized()
+                             else 1)
+                    dt = max(time.time() - interval_t0, 1e-9)
+                    tps = interval_tokens * world / dt
+                    interval_tokens = 0
+                    interval_t0 = time.time()
                     self._run_eval(test_dataset)
                     if self.is_rank0:
                         print(template.format(
                             epoch + 1, self.train_loss.result(),
                             self.train_accuracy.result(),
                             self.test_loss.result(),
-                            self.test_accuracy.result()))
+                            self.test_accuracy.result())
+                            + f", {tps:,.0f} tokens/s")
+                    if self.train_summary_writer:
+                        self.train_summary_writer.add_scalar(
+                            "tokens_per_sec", tps, step)
 
             # end-of-epoch eval so the epoch summary always reflects a real
             # test sweep (Q8/Q10: the reference could silently report
