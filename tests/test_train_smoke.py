@@ -87,3 +87,18 @@ def test_trace_steps_writes_chrome_trace(tmp_path, toy_corpus):
     tr, train_ds, _ = _mk_train(tmp_path, toy_corpus)
     path = tr.trace_steps(train_ds, 2, str(tmp_path / "traces"))
     assert os.path.exists(path) and os.path.getsize(path) > 100
+
+
+def test_predict_batch(tmp_path, toy_corpus):
+    """predict() with a list runs one padded batch through the KV-cached
+    decoder; per-row results must match single-sentence decodes."""
+    import torch
+
+    tr, *_ = _mk_train(tmp_path, toy_corpus)
+    sents = ["one two three", "four", "two three four one two"]
+    batch_out = tr.predict(sents)
+    assert batch_out.dim() == 2 and batch_out.shape[0] == 3
+    for i, s in enumerate(sents):
+        single = tr.predict(s)
+        n = min(len(single), batch_out.shape[1])
+        assert torch.equal(batch_out[i, :n], single[:n]), (i, s)

@@ -123,15 +123,23 @@ class Train:
     # -- greedy inference (C19, reference train.py:91-121; Q6 fixes) --------
     @torch.no_grad()
     def predict(self, input_sentence):
-        if isinstance(input_sentence, (list, tuple)):
-            input_sentence = input_sentence[0]
+        """Greedy decode one sentence (str) or a batch (list[str]) — the
+        reference accepted a list but decoded only sequentially; here a
+        list runs as one padded batch through the KV-cached decoder."""
+        batch = (list(input_sentence)
+                 if isinstance(input_sentence, (list, tuple))
+                 else [input_sentence])
         src_start = self.src_tokenizer.vocab_size
         src_end = src_start + 1
         tgt_start = self.tgt_tokenizer.vocab_size
         tgt_end = tgt_start + 1
-        tokens = [src_start] + self.src_tokenizer.encode(input_sentence) + [src_end]
-        encoder_input = torch.tensor([tokens], dtype=torch.int64,
-                                     device=self.device)
+        seqs = [[src_start] + self.src_tokenizer.encode(t) + [src_end]
+                for t in batch]
+        S = max(len(s) for s in seqs)
+        encoder_input = torch.zeros(len(seqs), S, dtype=torch.int64,
+                                    device=self.device)  # pad id 0
+        for i, s in enumerate(seqs):
+            encoder_input[i, :len(s)] = torch.tensor(s, dtype=torch.int64)
         # KV-cached greedy decode: encoder runs once, each step is O(1) in
         # prefix length (SURVEY.md §3.3 — vs the reference's full re-run per
         # token, train.py:109-118).
@@ -139,6 +147,8 @@ class Train:
         output = greedy_decode(self.transformer, encoder_input,
                                tgt_start, tgt_end,
                                max_len=self.max_decode_len)
+        if isinstance(input_sentence, (list, tuple)):
+            return output.cpu()
         return output.squeeze(0).cpu()
 
     def load_ckpt(self):
