@@ -79,17 +79,20 @@ DEV_INLINE bf16x8g frag256(const short* lds, int row, int kb /*bytes*/) {
 #define G_WAIT_LGKM0() asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory")
 #define G_WAIT_VM0() asm volatile("s_waitcnt vmcnt(0)" ::: "memory")
 
-// BN_ = 256 (per-wave 128x64, acc[8][4]) or 128 (per-wave 128x32,
-// acc[8][2]) — the narrow variant keeps the chip full on N=512 shapes.
-template <int EPILOGUE, int BN_>
+// BM_/BN_ in {256, 128}: per-wave output (BM_/2) x (BN_/4); narrower
+// instances keep the chip full on small-M/N shapes with the same
+// pipelined phase structure.
+template <int EPILOGUE, int BM_, int BN_>
 __global__ __launch_bounds__(G_THREADS, 1)
 void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
                     const short* __restrict__ bias, short* __restrict__ C,
                     int M, int N, int K, int lda, int ldb, int has_bias,
                     int nbm, int nbn) {
+  constexpr int MF = BM_ / 32;          // A row-frags per wave (8 or 4)
   constexpr int NF = BN_ / 64;          // B col-frags per wave (4 or 2)
+  constexpr int A_ELEMS = BM_ * G_BK;   // shorts per A tile
   constexpr int B_ELEMS = BN_ * G_BK;   // shorts per B tile
-  constexpr int SLOT = G_TILE_ELEMS + B_ELEMS;
+  constexpr int SLOT = A_ELEMS + B_ELEMS;
   extern __shared__ short smem[];
   const int lane = threadIdx.x & 63;
 
@@ -100,18 +103,18 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
     int q = nwg / 8, r = nwg % 8, x = wg % 8, o = wg / 8;
     wg = (x < r ? x * (q + 1) : r * (q + 1) + (x - r) * q) + o;
   }
-  const int bm0 = (wg / nbn) * G_BM;
+  const int bm0 = (wg / nbn) * BM_;
   const int bn0 = (wg % nbn) * BN_;
 
   const int wid = threadIdx.x >> 6;
-  const int wm = (wid >> 2) * 128;        // wave rows [wm, wm+128)
+  const int wm = (wid >> 2) * (BM_ / 2);  // wave rows
   const int wn = (wid & 3) * (BN_ / 4);   // wave cols
   const int fr = lane & 15;         // fragment lane row/col
   const int kg = lane >> 4;         // k-group 0..3
 
-  f32x4 acc[8][NF];
+  f32x4 acc[MF][NF];
 #pragma unroll
-  for (int i = 0; i < 8; ++i)
+  for (int i = 0; i < MF; ++i)
 #pragma unroll
     for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
@@ -120,10 +123,10 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
   // Prologue: stage K-tile 0 into slot 0, drain once.
   {
     short* a0 = smem;
-    short* b0 = smem + G_TILE_ELEMS;
+    short* b0 = smem + A_ELEMS;
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
-      stage256<256>(A, lda, bm0, M, h, a0);
+      stage256<BM_>(A, lda, bm0, M, h, a0);
       stage256<BN_>(B, ldb, bn0, N, h, b0);
     }
   }
@@ -133,28 +136,28 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
   // Per-wave operand fragments in registers:
   //   afr[i][ks]: A rows (wm + rh*64 + i*16 + fr), one rh half at a time
   //   bfr[j][ks]: B rows (wn + j*16 + fr), all four col-frags live
-  bf16x8g afr[4][2], bfr[NF][2];
+  bf16x8g afr[MF / 2][2], bfr[NF][2];
 
   for (int t = 0; t < ntiles; ++t) {
     const short* a_lds = smem + (t & 1) * SLOT;
-    const short* b_lds = a_lds + G_TILE_ELEMS;
+    const short* b_lds = a_lds + A_ELEMS;
     short* pa_lds = smem + ((t + 1) & 1) * SLOT;
-    short* pb_lds = pa_lds + G_TILE_ELEMS;
+    short* pb_lds = pa_lds + A_ELEMS;
     const long pk0 = (long)(t + 1) * G_BK;
     const bool do_pf = t + 1 < ntiles;
 
 #define G_MFMA_QUAD(RH, CH)                                                \
   _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                         \
-    _Pragma("unroll") for (int i = 0; i < 4; ++i)                          \
+    _Pragma("unroll") for (int i = 0; i < MF / 2; ++i)                     \
       _Pragma("unroll") for (int j = 0; j < NF / 2; ++j)                   \
-        acc[(RH) * 4 + i][(CH) * (NF / 2) + j] =                           \
+        acc[(RH) * (MF / 2) + i][(CH) * (NF / 2) + j] =                    \
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(                       \
                 afr[i][ks], bfr[(CH) * (NF / 2) + j][ks],                  \
-                acc[(RH) * 4 + i][(CH) * (NF / 2) + j], 0, 0, 0);
+                acc[(RH) * (MF / 2) + i][(CH) * (NF / 2) + j], 0, 0, 0);
 
     // ---- q0: quadrant (rh=0, ch=0) ------------------------------------
 #pragma unroll
-    for (int i = 0; i < 4; ++i)
+    for (int i = 0; i < MF / 2; ++i)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks)
         afr[i][ks] = frag256(a_lds, wm + i * 16 + fr, ks * 64 + kg * 16);
@@ -164,8 +167,8 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
       for (int ks = 0; ks < 2; ++ks)
         bfr[j][ks] = frag256(b_lds, wn + j * 16 + fr, ks * 64 + kg * 16);
     if (do_pf) {
-      stage256<256>(A + pk0, lda, bm0, M, 0, pa_lds);
-      stage256<256>(A + pk0, lda, bm0, M, 1, pa_lds);
+      stage256<BM_>(A + pk0, lda, bm0, M, 0, pa_lds);
+      stage256<BM_>(A + pk0, lda, bm0, M, 1, pa_lds);
     }
     G_BARRIER();
     G_WAIT_LGKM0();
@@ -189,11 +192,11 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
 
     // ---- q2: quadrant (1, 0) ------------------------------------------
 #pragma unroll
-    for (int i = 0; i < 4; ++i)
+    for (int i = 0; i < MF / 2; ++i)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks)
-        afr[i][ks] =
-            frag256(a_lds, wm + 64 + i * 16 + fr, ks * 64 + kg * 16);
+        afr[i][ks] = frag256(a_lds, wm + (BM_ / 4) + i * 16 + fr,
+                             ks * 64 + kg * 16);
     G_BARRIER();
     G_WAIT_LGKM0();
     G_MFMA_QUAD(1, 0)
@@ -210,7 +213,7 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
 
   // Epilogue: C/D lane map col = lane&15, row = (lane>>4)*4 + r.
 #pragma unroll
-  for (int i = 0; i < 8; ++i) {
+  for (int i = 0; i < MF; ++i) {
     const int grow_base = bm0 + wm + i * 16 + kg * 4;
 #pragma unroll
     for (int j = 0; j < NF; ++j) {
@@ -231,22 +234,27 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
 
 }  // namespace
 
-// The narrow-B (BN=128) instance is used only when the BN=256 grid would
-// underfill the 256-CU chip (measured: it wins FFN2-class shapes +8%, but
-// loses on deep-K tall-M shapes like the logits dW where BN=256 already
-// fills — fewer B re-reads).
-static int gemm256_bn(int M, int N) {
-  return ((long)cdiv(M, G_BM) * cdiv(N, 256) < 224 && N <= 512) ? 128 : 256;
+// Tile choice: widest instance whose grid still fills the 256-CU chip
+// (narrower tiles re-read operands more, so wider wins when it fits —
+// measured: BN=128 wins FFN2-class shapes +8% but loses on the logits dW
+// where the wide grid already fills).
+static void gemm256_tile(int M, int N, int& BM, int& BN) {
+  const long g22 = (long)cdiv(M, 256) * cdiv(N, 256);
+  const long g21 = (long)cdiv(M, 256) * cdiv(N, 128);
+  if (g22 >= 224) { BM = 256; BN = 256; }
+  else if (g21 >= 224) { BM = 256; BN = 128; }
+  else { BM = 128; BN = 128; }
 }
 
 // Host-side eligibility check shared with the gemm.hip dispatcher.
 bool gemm256_viable(int M, int N, int K, int lda, int ldb) {
   if (K % G_BK != 0 || K < 2 * G_BK) return false;
   if (lda % 8 != 0 || ldb % 8 != 0) return false;
-  long nwg = (long)cdiv(M, G_BM) * cdiv(N, gemm256_bn(M, N));
+  int BM, BN;
+  gemm256_tile(M, N, BM, BN);
+  long nwg = (long)cdiv(M, BM) * cdiv(N, BN);
   // Measured rule (tools/gemm_bench.py on MI355X): needs the chip full
-  // (1 WG/CU at >=96 KiB LDS) and either deep K or a grid big enough to
-  // amortize the prologue.
+  // and either deep K or a grid big enough to amortize the prologue.
   return nwg >= 224 && (K >= 1024 || nwg >= 384);
 }
 
@@ -270,35 +278,37 @@ torch::Tensor gemm256_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
   } else {
     c = torch::empty({M, N}, a.options());
   }
-  const int BNv = gemm256_bn(M, N);
-  const int nbm = cdiv(M, G_BM), nbn = cdiv(N, BNv);
-  const size_t smem = 2 * (G_TILE_ELEMS + (size_t)BNv * G_BK) *
-                      sizeof(short);  // 128 or 96 KiB
+  int BMv, BNv;
+  gemm256_tile(M, N, BMv, BNv);
+  const int nbm = cdiv(M, BMv), nbn = cdiv(N, BNv);
+  const size_t smem = 2 * ((size_t)BMv + BNv) * G_BK * sizeof(short);
   auto stream = at::hip::getCurrentHIPStream();
-  static bool attr_set[2][2] = {{false, false}, {false, false}};
-  auto launch = [&](auto epi, auto bnc) {
+  static bool attr_set[2][3] = {};
+  auto launch = [&](auto epi, auto bmc, auto bnc) {
     constexpr int E = decltype(epi)::value;
+    constexpr int BM_ = decltype(bmc)::value;
     constexpr int BN_ = decltype(bnc)::value;
-    constexpr int bi = BN_ == 128 ? 1 : 0;
-    if (!attr_set[E][bi]) {
-      (void)hipFuncSetAttribute((const void*)gemm256_kernel<E, BN_>,
+    constexpr int ti = BM_ == 128 ? 2 : (BN_ == 128 ? 1 : 0);
+    if (!attr_set[E][ti]) {
+      (void)hipFuncSetAttribute((const void*)gemm256_kernel<E, BM_, BN_>,
                                 hipFuncAttributeMaxDynamicSharedMemorySize,
                                 (int)smem);
-      attr_set[E][bi] = true;
+      attr_set[E][ti] = true;
     }
-    gemm256_kernel<E, BN_><<<nbm * nbn, G_THREADS, smem, stream>>>(
+    gemm256_kernel<E, BM_, BN_><<<nbm * nbn, G_THREADS, smem, stream>>>(
         (const short*)a.data_ptr(), (const short*)w.data_ptr(),
         has_bias ? (const short*)bias.data_ptr() : nullptr,
         (short*)c.data_ptr(), M, N, K, K, K, has_bias, nbm, nbn);
   };
   using E0 = std::integral_constant<int, 0>;
   using E1 = std::integral_constant<int, 1>;
-  using B256 = std::integral_constant<int, 256>;
-  using B128 = std::integral_constant<int, 128>;
-  if (epilogue == 1) {
-    if (BNv == 128) launch(E1{}, B128{}); else launch(E1{}, B256{});
-  } else {
-    if (BNv == 128) launch(E0{}, B128{}); else launch(E0{}, B256{});
-  }
+  using T256 = std::integral_constant<int, 256>;
+  using T128 = std::integral_constant<int, 128>;
+  auto dis = [&](auto epi) {
+    if (BMv == 128) launch(epi, T128{}, T128{});
+    else if (BNv == 128) launch(epi, T256{}, T128{});
+    else launch(epi, T256{}, T256{});
+  };
+  if (epilogue == 1) dis(E1{}); else dis(E0{});
   return c;
 }
