@@ -36,36 +36,64 @@ void ce_fwd_kernel(const short* __restrict__ logits,
   const short* lrow = logits + row * V;
   const int t = threadIdx.x;
 
+  // ONE pass over the 32k-vocab row (the two-pass version reads the
+  // whole row twice — the second read is partly L2-resident, but this
+  // kernel is still logits-read bound).  Branchless per-chunk online
+  // rescale: always `s = s*exp(m_old - m_new) + chunk_sum` (exp of <=0),
+  // then an (m, s) pair merge across lanes and waves.
   float mx = -1e30f, sume = 0.f, sumx = 0.f;
-  for (int c = t * 8; c < V; c += 256 * 8) {
-    if (c + 8 <= V) {
-      s16x8 v = *(const s16x8*)(lrow + c);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) mx = fmaxf(mx, bfbits2f(v[j]));
-    } else {
-      for (int j = 0; c + j < V; ++j) mx = fmaxf(mx, bfbits2f(lrow[c + j]));
-    }
+  for (int c = t * 8; c + 8 <= V; c += 256 * 8) {
+    s16x8 v = *(const s16x8*)(lrow + c);
+    float x0 = bfbits2f(v[0]), x1 = bfbits2f(v[1]);
+    float x2 = bfbits2f(v[2]), x3 = bfbits2f(v[3]);
+    float x4 = bfbits2f(v[4]), x5 = bfbits2f(v[5]);
+    float x6 = bfbits2f(v[6]), x7 = bfbits2f(v[7]);
+    float cm = fmaxf(fmaxf(fmaxf(x0, x1), fmaxf(x2, x3)),
+                     fmaxf(fmaxf(x4, x5), fmaxf(x6, x7)));
+    float mn = fmaxf(mx, cm);
+    float cs = __expf(x0 - mn) + __expf(x1 - mn) + __expf(x2 - mn) +
+               __expf(x3 - mn) + __expf(x4 - mn) + __expf(x5 - mn) +
+               __expf(x6 - mn) + __expf(x7 - mn);
+    sume = sume * __expf(mx - mn) + cs;
+    mx = mn;
+    sumx += ((x0 + x1) + (x2 + x3)) + ((x4 + x5) + (x6 + x7));
   }
-  mx = block_reduce(mx, scratch, 0);
-  for (int c = t * 8; c < V; c += 256 * 8) {
-    if (c + 8 <= V) {
-      s16x8 v = *(const s16x8*)(lrow + c);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float x = bfbits2f(v[j]);
-        sume += __expf(x - mx);
-        sumx += x;
-      }
-    } else {
+  // V tail (V % 2048 partial chunk for this thread)
+  {
+    int c = ((V / 2048) * 2048) + t * 8;  // last sweep start for thread t
+    if (c < V && c + 8 > V) {
       for (int j = 0; c + j < V; ++j) {
         float x = bfbits2f(lrow[c + j]);
-        sume += __expf(x - mx);
+        float mn = fmaxf(mx, x);
+        sume = sume * __expf(mx - mn) + __expf(x - mn);
+        mx = mn;
         sumx += x;
       }
     }
   }
-  sume = block_reduce(sume, scratch, 1);
-  __syncthreads();
+  // merge (mx, sume) across the wave, then across the 4 waves
+  {
+#pragma unroll
+    for (int off = 1; off < 64; off <<= 1) {
+      float m2 = __shfl_xor(mx, off);
+      float s2 = __shfl_xor(sume, off);
+      float mn = fmaxf(mx, m2);
+      sume = sume * __expf(mx - mn) + s2 * __expf(m2 - mn);
+      mx = mn;
+    }
+    __shared__ float sm[4], ss[4];
+    const int wid = threadIdx.x >> 6;
+    if ((threadIdx.x & 63) == 0) {
+      sm[wid] = mx;
+      ss[wid] = sume;
+    }
+    __syncthreads();
+    float M = fmaxf(fmaxf(sm[0], sm[1]), fmaxf(sm[2], sm[3]));
+    sume = ss[0] * __expf(sm[0] - M) + ss[1] * __expf(sm[1] - M) +
+           ss[2] * __expf(sm[2] - M) + ss[3] * __expf(sm[3] - M);
+    mx = M;
+    __syncthreads();
+  }
   sumx = block_reduce(sumx, scratch, 1);
   const float lse = mx + __logf(sume);
   if (t == 0) {
