@@ -40,3 +40,24 @@ def test_save_load(tmp_path):
     assert tok2.vocab_size == tok.vocab_size
     for line in CORPUS:
         assert tok2.encode(line) == tok.encode(line)
+
+
+def test_tokenizer_roundtrip_fuzz():
+    """Byte fallback must round-trip arbitrary text, including unicode,
+    underscores (the end-of-word marker) and OOV symbols."""
+    import random
+
+    from transformer_amd.data.tokenizer import SubwordTokenizer
+
+    tok = SubwordTokenizer.build_from_corpus(
+        ["the quick brown fox", "pack my box with five dozen jugs"] * 5,
+        target_vocab_size=200)
+    rng = random.Random(7)
+    alphabet = "abc ДЖ中文🙂_\\ xyz0189"
+    for _ in range(40):
+        s = "".join(rng.choice(alphabet) for _ in range(rng.randrange(1, 40)))
+        # round-trip is defined up to whitespace normalization
+        want = " ".join(s.split())
+        ids = tok.encode(s)
+        assert all(0 < i < tok.vocab_size for i in ids)
+        assert tok.decode(ids) == want, (s, tok.decode(ids))
