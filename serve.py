@@ -11,7 +11,13 @@ Endpoints:
                                -> {"tokens": [...], "text": "..."}
 
 Greedy KV-cached decode (models/transformer.py greedy_decode); the decode
-runs under torch.no_grad on whatever device the model was loaded to.
+runs under torch.no_grad on whatever device the model was loaded to.  On
+CUDA, requests are padded into (src_len, max_len) shape buckets and served
+by a hipGraph-captured decoder (models/transformer.py GraphedDecoder): one
+graph replay per generated token instead of hundreds of eager launches —
+the eager loop is launch-bound at B=1.  Padding the source is semantics-
+preserving (pad keys are masked in every attention; pad rows never feed a
+real query), so bucketed answers match the unpadded eager decode.
 """
 
 from __future__ import annotations
@@ -27,12 +33,26 @@ def build_app(model_dir: str, src_vocab: str, tgt_vocab: str,
 
     from transformer_amd.runtime.export import load_exported
     from transformer_amd.data.tokenizer import SubwordTokenizer
-    from transformer_amd.models.transformer import greedy_decode
+    from transformer_amd.models.transformer import (GraphedDecoder,
+                                                    greedy_decode)
 
     model, config = load_exported(model_dir, device=device, dtype=dtype)
     model.eval()
     src_tok = SubwordTokenizer.load_from_file(src_vocab)
     tgt_tok = SubwordTokenizer.load_from_file(tgt_vocab)
+
+    use_graph = device.startswith("cuda") and torch.cuda.is_available()
+    graphed: dict = {}  # (S_src_bucket, max_len_bucket) -> GraphedDecoder
+
+    def get_graphed(s_src: int, max_len: int):
+        key = (max(16, (s_src + 15) // 16 * 16),
+               max(16, (max_len + 15) // 16 * 16))
+        if key not in graphed:
+            graphed[key] = GraphedDecoder(
+                model, B=1, S_src=key[0], max_len=key[1],
+                start_id=tgt_tok.vocab_size,
+                device=next(model.parameters()).device)
+        return graphed[key], key[0]
 
     app = FastAPI(title="transformer_amd", version="1.0")
 
@@ -50,8 +70,15 @@ def build_app(model_dir: str, src_vocab: str, tgt_vocab: str,
                               device=next(model.parameters()).device)
         tgt_start = tgt_tok.vocab_size
         with torch.no_grad():
-            out = greedy_decode(model, enc_in, tgt_start, tgt_start + 1,
-                                max_len=max_len)
+            if use_graph:
+                dec, s_bucket = get_graphed(len(tokens), max_len)
+                padded = torch.zeros(1, s_bucket, dtype=torch.int64,
+                                     device=enc_in.device)
+                padded[0, :len(tokens)] = enc_in[0]
+                out = dec(padded, tgt_start + 1, max_len=max_len)
+            else:
+                out = greedy_decode(model, enc_in, tgt_start, tgt_start + 1,
+                                    max_len=max_len)
         ids = out.squeeze(0).tolist()
         body = [t for t in ids if 0 < t < tgt_tok.vocab_size]
         return {"tokens": ids, "text": tgt_tok.decode(body)}

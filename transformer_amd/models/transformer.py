@@ -368,3 +368,122 @@ def greedy_decode(model, inp, start_id, end_id, max_len=10):
         if bool(finished.all()):
             break
     return out
+
+
+# ---------------------------------------------------------------------------
+# hipGraph-captured decoding (serving): one replay per token.  The eager
+# decode_step launches hundreds of tiny kernels per token (launch-bound at
+# small batch, ~2.3 ms/token at B=1); here the whole step is captured once
+# with a DEVICE position counter and replayed.  Fixed-size zero-initialized
+# KV caches are attended with a pad mask that the captured step itself
+# unmasks one slot per replay (index_fill_ on the device position), so no
+# host work happens between tokens beyond the replay call.
+# ---------------------------------------------------------------------------
+
+class GraphedDecoder:
+    """Captures one decoder step for fixed (B, src) shapes; __call__ decodes
+    max_len tokens with one graph replay per token."""
+
+    @torch.no_grad()
+    def __init__(self, model: "Transformer", B: int, S_src: int, max_len: int,
+                 start_id: int, device):
+        assert max_len >= 4, "warmup+capture advance the position past 3"
+        self.model = model
+        self.B, self.S_src, self.max_len = B, S_src, max_len
+        self.start_id = start_id
+        dec = model.decoder
+        layer0 = dec.layers[0]
+        H, dh = layer0.mha1.num_heads, layer0.mha1.depth
+        dt = next(model.parameters()).dtype
+        self.inp = torch.zeros(B, S_src, dtype=torch.int64, device=device)
+        self.enc_output = torch.zeros(B, S_src, dec.d_model, device=device,
+                                      dtype=dt)
+        self.src_pad = torch.zeros(B, S_src, dtype=torch.uint8, device=device)
+        self.self_kv = [torch.zeros(B, max_len, 2, H, dh, device=device,
+                                    dtype=dt) for _ in dec.layers]
+        self.cross_kv = [torch.zeros(B, S_src, 2, H, dh, device=device,
+                                     dtype=dt) for _ in dec.layers]
+        self.kv_pad = torch.ones(B, max_len, dtype=torch.uint8, device=device)
+        self.tok = torch.zeros(B, max_len + 1, dtype=torch.int64,
+                               device=device)
+        self.pos = torch.zeros(1, dtype=torch.int64, device=device)
+
+        def step():
+            from .. import ops as O
+            pos = self.pos
+            # this replay's cache slot becomes visible
+            self.kv_pad.index_fill_(1, pos, 0)
+            cur = self.tok.index_select(1, pos)            # (B,1)
+            pe_row = dec.pe.index_select(0, pos)           # (1,d) model dtype
+            # same kernel as the eager decode path -> bit-identical tokens
+            x = O.ext().embed_pe_fwd(cur, dec.embedding, pe_row)
+            for li, layer in enumerate(dec.layers):
+                qkv = O.linear(x, layer.mha1.w_qkv, layer.mha1.b_qkv)
+                qkv = qkv.view(B, 1, 3, H, dh)
+                self.self_kv[li].index_copy_(1, pos, qkv[:, :, 1:3])
+                q = qkv[:, :, 0].contiguous()
+                k, v = self.self_kv[li].unbind(dim=2)
+                attn1 = O.fused_attention(q, k, v, kv_pad=self.kv_pad)
+                attn1 = O.linear(attn1.reshape(B, 1, H * dh),
+                                 layer.mha1.w_o, layer.mha1.b_o)
+                out1 = O.residual_layernorm(attn1, x, layer.ln1.gamma,
+                                            layer.ln1.beta, layer.ln1.eps)
+                q2 = O.linear(out1, layer.mha2.w_q, layer.mha2.b_q)
+                q2 = q2.view(B, 1, H, dh)
+                k2, v2 = self.cross_kv[li].unbind(dim=2)
+                attn2 = O.fused_attention(q2, k2, v2, kv_pad=self.src_pad)
+                attn2 = O.linear(attn2.reshape(B, 1, H * dh),
+                                 layer.mha2.w_o, layer.mha2.b_o)
+                out2 = O.residual_layernorm(attn2, out1, layer.ln2.gamma,
+                                            layer.ln2.beta, layer.ln2.eps)
+                ffn = layer.ffn(out2)
+                x = O.residual_layernorm(ffn, out2, layer.ln3.gamma,
+                                         layer.ln3.beta, layer.ln3.eps)
+            logits = O.linear(x, self.model.w_final, self.model.b_final)
+            nxt = O.argmax_lastdim(logits.float()).view(B, 1)
+            self.tok.index_copy_(1, pos + 1, nxt)
+            self.pos.add_(1)
+
+        # warmup on a side stream, then capture
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            step()
+            step()
+        torch.cuda.current_stream().wait_stream(s)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            step()
+
+    @torch.no_grad()
+    def __call__(self, inp: torch.Tensor, end_id: int,
+                 max_len: int | None = None) -> torch.Tensor:
+        """inp (B, S_src) int64 -> (B, n+1) ids starting with start_id;
+        tokens after the first end_id per row are zeroed."""
+        n = min(max_len or self.max_len, self.max_len)
+        assert inp.shape == (self.B, self.S_src), (inp.shape, self.B)
+        self.inp.copy_(inp)
+        enc, src_pad = encode(self.model, self.inp)
+        self.enc_output.copy_(enc)
+        self.src_pad.copy_(src_pad.to(torch.uint8))
+        dec = self.model.decoder
+        for li, layer in enumerate(dec.layers):
+            kv = ops.linear(self.enc_output, layer.mha2.w_kv, layer.mha2.b_kv)
+            self.cross_kv[li].copy_(
+                kv.view(self.B, self.S_src, 2, layer.mha2.num_heads,
+                        layer.mha2.depth))
+        self.kv_pad.fill_(1)
+        self.self_kv and [t.zero_() for t in self.self_kv]
+        self.tok.zero_()
+        self.tok[:, 0] = self.start_id
+        self.pos.zero_()
+        for _ in range(n):
+            self.graph.replay()
+        out = self.tok[:, :n + 1].clone()
+        # zero everything after the first end_id per row (parity with the
+        # early-stopping eager decode)
+        hit = (out == end_id).cumsum(dim=1) > 0
+        mask = hit.roll(1, dims=1)
+        mask[:, 0] = False
+        out[mask] = 0
+        return out
