@@ -8,6 +8,8 @@
 // recomputes softmax from logits+LSE (no S×V weight materialization).
 #include "common.h"
 
+#include <map>
+
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
@@ -28,7 +30,8 @@ DEV_INLINE float block_reduce(float v, float* scratch, int op /*0=max 1=sum*/) {
 __global__ __launch_bounds__(256)
 void ce_fwd_kernel(const short* __restrict__ logits,
                    const long* __restrict__ targets,
-                   float* __restrict__ lse_out, float* __restrict__ loss_out,
+                   float* __restrict__ lse_out,
+                   float* __restrict__ loss_banks,
                    long R, int V, float inv_batch, float eps_ls) {
   __shared__ float scratch[4];
   const long row = blockIdx.x;
@@ -103,7 +106,9 @@ void ce_fwd_kernel(const short* __restrict__ logits,
       float xt = bfbits2f(lrow[tgt]);
       // (1-eps)*(lse - x_t) + eps*(lse - mean_j x_j)
       float loss = lse - (1.f - eps_ls) * xt - eps_ls * (sumx / V);
-      atomicAdd(loss_out, loss * inv_batch);
+      // 256 banks: one atomic per row to a single scalar serialized all
+      // 16k row-blocks (~60 us); banked contention is R/256-way.
+      atomicAdd(&loss_banks[row & 255], loss * inv_batch);
     }
   }
 }
@@ -157,6 +162,20 @@ void ce_bwd_kernel(const short* __restrict__ logits,
   }
 }
 
+// one wave: sum the 256 banks into the loss scalar and re-zero them for
+// the next call (the bank workspace is cached per device).
+__global__ void ce_loss_reduce_kernel(float* __restrict__ banks,
+                                      float* __restrict__ loss_out) {
+  float v = banks[threadIdx.x] + banks[threadIdx.x + 64] +
+            banks[threadIdx.x + 128] + banks[threadIdx.x + 192];
+  banks[threadIdx.x] = 0.f;
+  banks[threadIdx.x + 64] = 0.f;
+  banks[threadIdx.x + 128] = 0.f;
+  banks[threadIdx.x + 192] = 0.f;
+  v = wave_sum(v);
+  if (threadIdx.x == 0) *loss_out = v;
+}
+
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets,
                                   double batch_size, double label_smoothing) {
   TORCH_CHECK(logits.is_cuda() && logits.dtype() == torch::kBFloat16 &&
@@ -165,12 +184,20 @@ std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets,
   const long R = logits.size(0);
   const int V = logits.size(1);
   auto lse = torch::empty({R}, logits.options().dtype(torch::kFloat32));
-  auto loss = torch::zeros({}, logits.options().dtype(torch::kFloat32));
+  auto loss = torch::empty({}, logits.options().dtype(torch::kFloat32));
+  static std::map<int, torch::Tensor> bank_cache;
+  auto it = bank_cache.find((int)logits.get_device());
+  if (it == bank_cache.end())
+    it = bank_cache.emplace((int)logits.get_device(), torch::zeros(
+        {256}, logits.options().dtype(torch::kFloat32))).first;
+  auto banks = it->second;
   auto stream = at::hip::getCurrentHIPStream();
   ce_fwd_kernel<<<R, 256, 0, stream>>>(
       (const short*)logits.data_ptr(), targets.data_ptr<long>(),
-      lse.data_ptr<float>(), loss.data_ptr<float>(), R, V,
+      lse.data_ptr<float>(), banks.data_ptr<float>(), R, V,
       1.0f / (float)batch_size, (float)label_smoothing);
+  ce_loss_reduce_kernel<<<1, 64, 0, stream>>>(banks.data_ptr<float>(),
+                                              loss.data_ptr<float>());
   return {loss, lse};
 }
 
