@@ -363,63 +363,9 @@ void transpose2d_kernel(const short* __restrict__ in, short* __restrict__ out,
 // disjoint-partials scheme (no atomics) was tried and measured SLOWER:
 // its gy-deep serial reduction ran on a 2-8 block grid.
 #define COLSUM_ROWS 128
-// v8 layout (measured vs the 1-col-per-thread version, which ran 3.8x off
-// the HBM bound at N=512): each block covers a 512-column span; lane c of
-// every wave owns the same 8 consecutive columns (s16x8 loads, one fully
-// coalesced 1 KiB request per wave per row) and the 4 waves stripe the
-// block's row range with stride 4.  Wave partials combine through LDS so
-// each block issues ONE set of 512 column atomics.
 __global__ __launch_bounds__(256)
 void colsum_part_kernel(const short* __restrict__ dy, float* __restrict__ acc,
-                        int M, int N, int rows_per_block) {
-  const int c0 = blockIdx.x * 512;
-  const long m0 = (long)blockIdx.y * rows_per_block;
-  const long m1 = min((long)M, m0 + rows_per_block);
-  const int w = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  __shared__ float part[4][512];
-  if (c0 + 512 <= N) {
-    const int c = c0 + lane * 8;
-    float s[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-    long m = m0 + w;
-    for (; m + 4 < m1; m += 8) {  // 2 rows in flight per lane
-      s16x8 v0 = *(const s16x8*)(dy + m * N + c);
-      s16x8 v1 = *(const s16x8*)(dy + (m + 4) * N + c);
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        s[j] += bfbits2f(v0[j]) + bfbits2f(v1[j]);
-    }
-    if (m < m1) {
-      s16x8 v0 = *(const s16x8*)(dy + m * N + c);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) s[j] += bfbits2f(v0[j]);
-    }
-#pragma unroll
-    for (int j = 0; j < 8; ++j) part[w][lane * 8 + j] = s[j];
-    __syncthreads();
-    if (w == 0) {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int i = lane * 8 + j;
-        atomicAdd(&acc[c0 + i],
-                  (part[0][i] + part[1][i]) + (part[2][i] + part[3][i]));
-      }
-    }
-  } else {
-    // ragged tail block (N % 512 columns), scalar per column
-    for (int n = c0 + threadIdx.x; n < N; n += 256) {
-      float s0 = 0.f;
-      for (long m = m0; m < m1; ++m) s0 += bfbits2f(dy[m * N + n]);
-      atomicAdd(&acc[n], s0);
-    }
-  }
-}
-
-// scalar column-per-thread variant for N % 512 != 0 (odd-N logits colsum:
-// the s16x8 path would issue 2-byte-aligned 16-byte loads there).  Already
-// within ~25% of the read bound at the 16320x32770 shape.
-__global__ __launch_bounds__(256)
-void colsum_scalar_kernel(const short* __restrict__ dy,
-                          float* __restrict__ acc, int M, int N) {
+                        int M, int N) {
   int n = blockIdx.x * 256 + threadIdx.x;
   if (n >= N) return;
   long m0 = (long)blockIdx.y * COLSUM_ROWS;
@@ -637,20 +583,9 @@ torch::Tensor colsum(torch::Tensor a, c10::optional<torch::Tensor> out_opt) {
     out = torch::empty({N}, a.options());
   }
   auto stream = at::hip::getCurrentHIPStream();
-  if (N % 512 == 0) {
-    // ~512 blocks total: fills the 256-CU chip without exploding the
-    // per-column atomic contention (= grid.y contenders per column)
-    const int gx = N / 512;
-    int gy = std::max(1, std::min(512 / gx, cdiv(M, 8)));
-    const int rows = cdiv(M, gy);
-    gy = cdiv(M, rows);
-    colsum_part_kernel<<<dim3(gx, gy), 256, 0, stream>>>(
-        (const short*)a.data_ptr(), acc.data_ptr<float>(), M, N, rows);
-  } else {
-    dim3 grid(cdiv(N, 256), cdiv(M, COLSUM_ROWS));
-    colsum_scalar_kernel<<<grid, 256, 0, stream>>>(
-        (const short*)a.data_ptr(), acc.data_ptr<float>(), M, N);
-  }
+  dim3 grid(cdiv(N, 256), cdiv(M, COLSUM_ROWS));
+  colsum_part_kernel<<<grid, 256, 0, stream>>>(
+      (const short*)a.data_ptr(), acc.data_ptr<float>(), M, N);
   cast_colsum_kernel<<<cdiv(N, 256), 256, 0, stream>>>(
       acc.data_ptr<float>(), (short*)out.data_ptr(), N);
   return out;

@@ -193,59 +193,53 @@ void ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ s,
 }
 
 // column reduction: dgamma[c] = sum_r dy[r][c]*xhat[r][c]; dbeta[c] = sum dy.
-// v8 layout (the s16x2 two-col-per-thread version measured 4x off the read
-// bound at D=512): each block covers a 512-column span; lane c of every
-// wave owns 8 consecutive columns (s16x8 loads — one coalesced 1 KiB
-// request per wave per row per stream) and the 4 waves stripe the block's
-// rows with stride 4.  Wave partials combine through LDS so each block
-// issues one set of 512 column atomics per array.
+// Thread t of block b owns column b*256+t; every row read is coalesced
+// across the block's 256 consecutive columns.
 #define LNGB_ROWS 128
 __global__ __launch_bounds__(256)
 void ln_gb_kernel(const short* __restrict__ dy, const short* __restrict__ s,
                   const float* __restrict__ mean,
                   const float* __restrict__ rstd, float* __restrict__ acc_g,
-                  float* __restrict__ acc_b, int R, int D,
-                  int rows_per_block) {
-  const int c0 = blockIdx.x * 512;
-  const long r0 = (long)blockIdx.y * rows_per_block;
-  const long r1 = min((long)R, r0 + rows_per_block);
-  const int w = threadIdx.x >> 6, lane = threadIdx.x & 63;
-  __shared__ float part_g[4][512];
-  __shared__ float part_b[4][512];
-  if (c0 + 512 <= D && (D % 8) == 0) {
-    const int c = c0 + lane * 8;
-    float sg[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-    float sb[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-    for (long r = r0 + w; r < r1; r += 4) {
-      s16x8 dv = *(const s16x8*)(dy + r * D + c);
-      s16x8 sv = *(const s16x8*)(s + r * D + c);
-      const float mu = mean[r], rs = rstd[r];
+                  float* __restrict__ acc_b, int R, int D) {
+  // 2 columns per thread (s16x2 loads) x 4-row unroll: the scalar
+  // one-col-per-thread version was latency-bound at ~0.9 TB/s.
+  const int c = (blockIdx.x * 64 + threadIdx.x) * 2;
+  if (c >= D) return;
+  const long r0 = (long)blockIdx.y * LNGB_ROWS;
+  const long r1 = min((long)R, r0 + LNGB_ROWS);
+  float sg0 = 0.f, sg1 = 0.f, sb0 = 0.f, sb1 = 0.f;
+  const bool pair = (c + 1 < D) && (D % 2 == 0);
+  if (pair) {
+    long r = r0;
+    for (; r + 4 <= r1; r += 4) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float d = bfbits2f(dv[j]);
-        sg[j] += d * (bfbits2f(sv[j]) - mu) * rs;
-        sb[j] += d;
+      for (int u = 0; u < 4; ++u) {
+        s16x2 dv = *(const s16x2*)(dy + (r + u) * D + c);
+        s16x2 sv = *(const s16x2*)(s + (r + u) * D + c);
+        float mu = mean[r + u], rs = rstd[r + u];
+        float d0 = bfbits2f(dv[0]), d1 = bfbits2f(dv[1]);
+        sg0 += d0 * (bfbits2f(sv[0]) - mu) * rs;
+        sg1 += d1 * (bfbits2f(sv[1]) - mu) * rs;
+        sb0 += d0;
+        sb1 += d1;
       }
     }
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      part_g[w][lane * 8 + j] = sg[j];
-      part_b[w][lane * 8 + j] = sb[j];
+    for (; r < r1; ++r) {
+      s16x2 dv = *(const s16x2*)(dy + r * D + c);
+      s16x2 sv = *(const s16x2*)(s + r * D + c);
+      float mu = mean[r], rs = rstd[r];
+      float d0 = bfbits2f(dv[0]), d1 = bfbits2f(dv[1]);
+      sg0 += d0 * (bfbits2f(sv[0]) - mu) * rs;
+      sg1 += d1 * (bfbits2f(sv[1]) - mu) * rs;
+      sb0 += d0;
+      sb1 += d1;
     }
-    __syncthreads();
-    if (w < 2) {  // wave 0 -> dgamma, wave 1 -> dbeta
-      float(*part)[512] = w ? part_b : part_g;
-      float* acc = w ? acc_b : acc_g;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int i = lane * 8 + j;
-        atomicAdd(&acc[c0 + i],
-                  (part[0][i] + part[1][i]) + (part[2][i] + part[3][i]));
-      }
-    }
+    atomicAdd(&acc_g[c], sg0);
+    atomicAdd(&acc_b[c], sb0);
+    atomicAdd(&acc_g[c + 1], sg1);
+    atomicAdd(&acc_b[c + 1], sb1);
   } else {
-    // ragged span (D % 512 columns, or D < 512), scalar per column
-    for (int cc = c0 + threadIdx.x; cc < D; cc += 256) {
+    for (int cc = c; cc < min(c + 2, D); ++cc) {
       float sg = 0.f, sb = 0.f;
       for (long r = r0; r < r1; ++r) {
         float dyv = bfbits2f(dy[r * D + cc]);
@@ -360,16 +354,11 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor s,
   auto acc_b = wit->second.narrow(0, D, D);
   // 64-thread blocks: at d_model=512 a 256-thread block grid is only 128
   // workgroups — half the 256-CU chip idle.
-  // ~512 blocks total (see kernel comment); 256 threads, 4 row-striped
-  // waves per block, LDS-combined partials.
-  const int gx = cdiv(D, 512);
-  int gy = std::max(1, std::min(512 / gx, cdiv(R, 8)));
-  const int rows = cdiv(R, gy);
-  gy = cdiv(R, rows);
-  ln_gb_kernel<<<dim3(gx, gy), 256, 0, stream>>>(
+  dim3 gbgrid(cdiv(cdiv(D, 2), 64), cdiv(R, LNGB_ROWS));
+  ln_gb_kernel<<<gbgrid, 64, 0, stream>>>(
       (const short*)dy.data_ptr(), (const short*)s.data_ptr(),
       mean.data_ptr<float>(), rstd.data_ptr<float>(),
-      acc_g.data_ptr<float>(), acc_b.data_ptr<float>(), R, D, rows);
+      acc_g.data_ptr<float>(), acc_b.data_ptr<float>(), R, D);
   ln_gb_cast_kernel<<<cdiv(D, 256), 256, 0, stream>>>(
       acc_g.data_ptr<float>(), acc_b.data_ptr<float>(),
       (short*)dgamma.data_ptr(), (short*)dbeta.data_ptr(), D);
