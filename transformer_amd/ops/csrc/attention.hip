@@ -282,25 +282,18 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
 #pragma unroll
         for (int r = 0; r < 4; ++r) acc[rf][i][r] *= alpha[r];
 
-      // P = exp(S - m), row-sum into l, park bf16 P in LDS (A-layout)
+      // P = exp(S - m), park bf16 P in LDS (A-layout).  The row-sum for l
+      // is NOT shuffle-reduced here: it comes out of the PV phase below as
+      // one extra MFMA per 32-key chunk against a ones B-operand (the
+      // per-half 16-lane shuffle chains were ~160 issue slots per tile in
+      // an issue-bound kernel, ~2 MFMAs are ~34 cycles).  l then also
+      // normalizes by the sum of the SAME bf16-rounded P the O
+      // accumulation uses.
 #pragma unroll
       for (int half = 0; half < NHALF; ++half) {
-        float psum[4];
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          float pv = __expf(p_raw[half][r] - m_run[rf][r]);
-          p_raw[half][r] = pv;
-          psum[r] = pv;
-        }
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-#pragma unroll
-          for (int off = 1; off < 16; off <<= 1)
-            psum[r] += __shfl_xor(psum[r], off);
-          // first half applies alpha to l; later halves add directly
-          l_run[rf][r] = l_run[rf][r] * (half == 0 ? alpha[r] : 1.f) +
-                         psum[r];
-        }
+        for (int r = 0; r < 4; ++r)
+          p_raw[half][r] = __expf(p_raw[half][r] - m_run[rf][r]);
         // one b64 store: P rows q=kg*4.. at transposed-scratch col = key
         s16x4 pw = {f2bfbits(p_raw[half][0]), f2bfbits(p_raw[half][1]),
                     f2bfbits(p_raw[half][2]), f2bfbits(p_raw[half][3])};
@@ -311,19 +304,30 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
       __builtin_amdgcn_wave_barrier();
 
       // ---- O += P · V  (A = P[q][key] from LDS, B = V^T[d][key]) -----
+      const short one_bits = (short)0x3F80;  // bf16 1.0
+      const bf16x8 onesv = (bf16x8)(s16x8){one_bits, one_bits, one_bits,
+                                           one_bits, one_bits, one_bits,
+                                           one_bits, one_bits};
+      f32x4 accl = {0.f, 0.f, 0.f, 0.f};
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int ks = 0; ks < FKVT / 32; ++ks)
+      for (int ks = 0; ks < FKVT / 32; ++ks) {
+        bf16x8 pa = scrT_read8(p_lds[wid], ks * 32 + kg * 8);
+        accl = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, onesv, accl,
+                                                       0, 0, 0);
 #pragma unroll
         for (int i = 0; i < D16; ++i) {
-          bf16x8 pa = scrT_read8(p_lds[wid], ks * 32 + kg * 8);
           bf16x8 vb = TRV
               ? lds_read8_tr<DH * 2>(v_lds, ks * 32 + kg * 8, i * 16)
               : lds_read8<FKVT * 2>(v_lds, i * 16 + fr, ks * 32 + kg * 8);
           acc[rf][i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               pa, vb, acc[rf][i], 0, 0, 0);
         }
+      }
       __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        l_run[rf][r] = l_run[rf][r] * alpha[r] + accl[r];
       __builtin_amdgcn_wave_barrier();  // p_lds reused by next rf
     }
     __syncthreads();
