@@ -11,6 +11,8 @@ from __future__ import annotations
 
 import math
 
+import os
+
 import torch
 
 from . import reference as R
@@ -53,6 +55,24 @@ def _flat(p):
 # Linear: y = x @ W^T + b, optional fused ReLU epilogue (K1/K7/K8/K12).
 # ---------------------------------------------------------------------------
 
+
+# Forward GEMM backend (measured, tools/head_gemm_bench.py on MI355X):
+# hipBLASLt's fused-bias/ReLU epilogues (F.linear / torch._addmm_activation)
+# beat the hand-written NT path at every model forward shape (qkv 515->617
+# TF, ffn2 733->1034, logits 611->716; the original fwd A/B had only
+# compared matmul + a separate bias pass, which loses).  TFMX_FWD_GEMM=hip
+# forces the hand-written kernel (A/B + fallback).
+_FWD_HIP = os.environ.get("TFMX_FWD_GEMM", "") == "hip"
+
+
+def _fwd_gemm(E, x, w, b, activation):
+    if _FWD_HIP or b is None:
+        return E.gemm_nt(x, w, b if b is not None else torch.Tensor(),
+                         1 if activation == "relu" else 0)
+    if activation == "relu":
+        return torch._addmm_activation(b, x, w.t())
+    return torch.nn.functional.linear(x, w, b)
+
 def _dw_gemm(dy, x, out=None):
     """dW[N,K] = dY^T @ X — plain GEMM; backend measured per shape on
     MI355X (tools/gemm_bench.py): hipBLASLt TN wins at the wide logits
@@ -86,8 +106,7 @@ class _LinearFn(torch.autograd.Function):
     def forward(ctx, x, w, b, activation):
         # x: (M, K) bf16; w: (N, K) bf16; b: (N,) bf16 or None
         E = ext()
-        y = E.gemm_nt(x, w, b if b is not None else torch.Tensor(),
-                      1 if activation == "relu" else 0)
+        y = _fwd_gemm(E, x, w, b, activation)
         ctx.activation = activation
         ctx.has_bias = b is not None
         ctx.save_for_backward(x, w, y if activation == "relu" else torch.Tensor())
@@ -120,8 +139,7 @@ class _LinearFlatFn(torch.autograd.Function):
     def forward(ctx, x, wb, activation):
         E = ext()
         w, b = wb
-        y = E.gemm_nt(x, w, b if b is not None else torch.Tensor(),
-                      1 if activation == "relu" else 0)
+        y = _fwd_gemm(E, x, w, b, activation)
         ctx.activation = activation
         ctx.wb = wb
         ctx.save_for_backward(x, y if activation == "relu" else torch.Tensor())
