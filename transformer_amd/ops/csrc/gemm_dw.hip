@@ -130,7 +130,7 @@ DEV_INLINE void dw_stage_glds(const short* __restrict__ g, long ldg, long m0,
 // second launch_bounds arg pins >=4 waves/SIMD (VGPR cap 128): the DB
 // variant otherwise allocates 134 VGPRs and drops a whole workgroup of
 // block-level overlap per CU (measured -20% end-to-end).
-template <bool SPLIT, bool DB>
+template <bool SPLIT, bool DB, bool NOBAR = true>
 __global__ __launch_bounds__(DW_THREADS)
 void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
                     short* __restrict__ C, float* __restrict__ CW,
@@ -242,24 +242,35 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
       else
         dw_stage(X, K, mN, m_hi, bk0, K, pb);
     }
-    __builtin_amdgcn_s_barrier();
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    // NOBAR: no intra-block barriers — both k-step phases read the SAME
+    // LDS slot (synced at the block boundary), fragments are wave-private
+    // registers and the glds prefetch lands in the OTHER slot under
+    // vmcnt; the compiler tracks the LDS-read -> MFMA dependencies and
+    // software-pipelines the phase-1 reads under the phase-0 MFMA burst.
+    // else: 3 full s_barrier + lgkmcnt(0) drains per block keep the 4
+    // waves in lockstep (A/B via TFMX_DW_BAR=1).
+    if (!NOBAR) {
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    }
 #pragma unroll
     for (int i = 0; i < 4; ++i)
 #pragma unroll
       for (int j = 0; j < 4; ++j)
         acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[i], bf_[j], acc[i][j], 0, 0, 0);
-    __builtin_amdgcn_s_barrier();
     // ---- phase 1: k-step 1 -------------------------------------------
+    if (!NOBAR) __builtin_amdgcn_s_barrier();
 #pragma unroll
     for (int i = 0; i < 4; ++i)
       af[i] = dw_frag(a_lds, 32 + kg * 8, wn + i * 16);
 #pragma unroll
     for (int j = 0; j < 4; ++j)
       bf_[j] = dw_frag(b_lds, 32 + kg * 8, wk + j * 16);
-    __builtin_amdgcn_s_barrier();
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    if (!NOBAR) {
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    }
 #pragma unroll
     for (int i = 0; i < 4; ++i)
 #pragma unroll
@@ -378,9 +389,15 @@ torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x,
     dbw = torch::zeros({N}, dy.options().dtype(torch::kFloat32));
     dbw_p = dbw.data_ptr<float>();
   }
+  static const bool use_bar = [] {
+    const char* e = getenv("TFMX_DW_BAR");
+    return e && atoi(e) != 0;
+  }();
   auto launch = [&](auto split, auto dbc, float* cwp) {
-    gemm_dw_kernel<decltype(split)::value, decltype(dbc)::value>
-        <<<grid, DW_THREADS, 0, stream>>>(
+    auto kfn = use_bar
+        ? gemm_dw_kernel<decltype(split)::value, decltype(dbc)::value, false>
+        : gemm_dw_kernel<decltype(split)::value, decltype(dbc)::value, true>;
+    kfn<<<grid, DW_THREADS, 0, stream>>>(
             (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
             (short*)c.data_ptr(), cwp, dbw_p, (int)M, N, K, m_per_slice,
             nbk);
