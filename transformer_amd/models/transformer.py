@@ -444,16 +444,22 @@ class GraphedDecoder:
             self.tok.index_copy_(1, pos + 1, nxt)
             self.pos.add_(1)
 
-        # warmup on a side stream, then capture
-        s = torch.cuda.Stream()
-        s.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(s):
-            step()
-            step()
-        torch.cuda.current_stream().wait_stream(s)
-        self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph):
-            step()
+        # warmup on a side stream, then capture.  capture hint: library
+        # GEMM kernels inside the graph (launch cost is replayed away).
+        from ..ops.functional import set_capture_hint
+        set_capture_hint(True)
+        try:
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                step()
+                step()
+            torch.cuda.current_stream().wait_stream(s)
+            self.graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph):
+                step()
+        finally:
+            set_capture_hint(False)
 
     @torch.no_grad()
     def __call__(self, inp: torch.Tensor, end_id: int,

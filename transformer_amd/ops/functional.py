@@ -65,11 +65,23 @@ def _flat(p):
 _FWD_HIP = os.environ.get("TFMX_FWD_GEMM", "") == "hip"
 
 
+_CAPTURE_HINT = False
+
+
+def set_capture_hint(on: bool):
+    """Inside a hipGraph (GraphedDecoder) launch overhead is replayed
+    away and the library kernels win even at M=B rows (0.705 vs 0.788
+    ms/token); eager small-M calls keep the cheap-launch hand-written
+    kernel.  Set around warmup+capture so both trace the same path."""
+    global _CAPTURE_HINT
+    _CAPTURE_HINT = on
+
+
 def _fwd_gemm(E, x, w, b, activation):
     # small-M (decode/serving) rows: the hand-written kernel's launch is
     # cheaper than hipBLASLt's (eager B=1 decode measured 1.8 vs 2.4
     # ms/token through F.linear); training shapes (M ~ 16k) go library.
-    if _FWD_HIP or b is None or x.shape[0] <= 1024:
+    if _FWD_HIP or b is None or (x.shape[0] <= 1024 and not _CAPTURE_HINT):
         return E.gemm_nt(x, w, b if b is not None else torch.Tensor(),
                          1 if activation == "relu" else 0)
     if activation == "relu":
