@@ -226,3 +226,44 @@ def test_graphed_decode_matches_eager():
     # replays advance purely on device: a second call reproduces itself
     out2 = dec(inp, end, max_len=max_len)
     assert torch.equal(out, out2)
+
+
+def test_serve_graphed_path_gpu(tmp_path, toy_corpus):
+    """serve.py on CUDA: /translate must route through the hipGraph-captured
+    GraphedDecoder shape buckets (export -> tokenizers -> padded bucket ->
+    graph replay) and answer consistently across repeat calls."""
+    from fastapi.testclient import TestClient
+
+    from transformer_amd.data.dataset import load_dataset
+    from transformer_amd.models import Transformer
+    from transformer_amd.runtime import export_model
+    import serve
+
+    _, _, src_tok, tgt_tok = load_dataset(
+        toy_corpus, str(tmp_path / "sv"), str(tmp_path / "tv"),
+        sequence_length=50, batch_size=4, seed=1)
+    torch.manual_seed(0)
+    cfg = dict(num_layers=1, d_model=128, num_heads=2, dff=256,
+               input_vocab_size=src_tok.vocab_size + 2,
+               target_vocab_size=tgt_tok.vocab_size + 2,
+               dropout_rate=0.0, max_position=128)
+    m = Transformer(num_layers=cfg["num_layers"], d_model=128, num_heads=2,
+                    dff=256, input_vocab_size=cfg["input_vocab_size"],
+                    target_vocab_size=cfg["target_vocab_size"],
+                    rate=0.0, max_position=128)
+    export_model(m, str(tmp_path / "model"), cfg)
+
+    app = serve.build_app(str(tmp_path / "model"), str(tmp_path / "sv"),
+                          str(tmp_path / "tv"), device="cuda",
+                          dtype=torch.bfloat16)
+    client = TestClient(app)
+    r = client.post("/translate", json={"text": "one two", "max_len": 6})
+    assert r.status_code == 200
+    toks1 = r.json()["tokens"]
+    # repeat: graph replay must reproduce itself
+    r = client.post("/translate", json={"text": "one two", "max_len": 6})
+    assert r.json()["tokens"] == toks1
+    # a longer max_len lands in a second bucket (fresh capture)
+    r = client.post("/translate", json={"text": "one two three",
+                                        "max_len": 20})
+    assert r.status_code == 200 and len(r.json()["tokens"]) >= 1
