@@ -130,7 +130,14 @@ DEV_INLINE void dw_stage_glds(const short* __restrict__ g, long ldg, long m0,
 // second launch_bounds arg pins >=4 waves/SIMD (VGPR cap 128): the DB
 // variant otherwise allocates 134 VGPRs and drops a whole workgroup of
 // block-level overlap per CU (measured -20% end-to-end).
-template <bool SPLIT, bool DB, bool NOBAR = true>
+typedef __bf16 bf16x4d __attribute__((ext_vector_type(4)));
+
+// X16: use v_mfma_f32_16x16x16_bf16 (4-deep, 2-VGPR operands) so each
+// ds_read_b64_tr_b16 result feeds an MFMA DIRECTLY — the 16x16x32 path
+// pays ~65 v_mov per 64-m block pairing two v4i16 tr-reads into each
+// 4-VGPR operand (seen in the compiled loop; ~24% of the MFMA issue
+// time).  Same LDS traffic, twice the (half-size) MFMA instructions.
+template <bool SPLIT, bool DB, bool NOBAR = true, bool X16 = false>
 __global__ __launch_bounds__(DW_THREADS)
 void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
                     short* __restrict__ C, float* __restrict__ CW,
@@ -223,13 +230,27 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
     const long mN = m0 + DW_BM;
     const bool pf = mN < m_hi;
     bf16x8d af[4], bf_[4];
+    s16x4d al[4], ah[4], bl[4], bh[4];
     // ---- phase 0: k-step 0 -------------------------------------------
+    if (X16) {
 #pragma unroll
-    for (int i = 0; i < 4; ++i)
-      af[i] = dw_frag(a_lds, kg * 8, wn + i * 16);
+      for (int i = 0; i < 4; ++i) {
+        al[i] = dw_tr4(a_lds, kg * 8, wn + i * 16);
+        ah[i] = dw_tr4(a_lds, kg * 8 + 4, wn + i * 16);
+      }
 #pragma unroll
-    for (int j = 0; j < 4; ++j)
-      bf_[j] = dw_frag(b_lds, kg * 8, wk + j * 16);
+      for (int j = 0; j < 4; ++j) {
+        bl[j] = dw_tr4(b_lds, kg * 8, wk + j * 16);
+        bh[j] = dw_tr4(b_lds, kg * 8 + 4, wk + j * 16);
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        af[i] = dw_frag(a_lds, kg * 8, wn + i * 16);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        bf_[j] = dw_frag(b_lds, kg * 8, wk + j * 16);
+    }
     if (pf) {
       short* pa = a_img[(cur + 1) & 1];
       short* pb = b_img[(cur + 1) & 1];
@@ -256,17 +277,38 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
 #pragma unroll
     for (int i = 0; i < 4; ++i)
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
-        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[i], bf_[j], acc[i][j], 0, 0, 0);
+      for (int j = 0; j < 4; ++j) {
+        if (X16) {
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x16bf16_1k(
+              (bf16x4d)al[i], (bf16x4d)bl[j], acc[i][j], 0, 0, 0);
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x16bf16_1k(
+              (bf16x4d)ah[i], (bf16x4d)bh[j], acc[i][j], 0, 0, 0);
+        } else {
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf_[j], acc[i][j], 0, 0, 0);
+        }
+      }
     // ---- phase 1: k-step 1 -------------------------------------------
     if (!NOBAR) __builtin_amdgcn_s_barrier();
+    if (X16) {
 #pragma unroll
-    for (int i = 0; i < 4; ++i)
-      af[i] = dw_frag(a_lds, 32 + kg * 8, wn + i * 16);
+      for (int i = 0; i < 4; ++i) {
+        al[i] = dw_tr4(a_lds, 32 + kg * 8, wn + i * 16);
+        ah[i] = dw_tr4(a_lds, 32 + kg * 8 + 4, wn + i * 16);
+      }
 #pragma unroll
-    for (int j = 0; j < 4; ++j)
-      bf_[j] = dw_frag(b_lds, 32 + kg * 8, wk + j * 16);
+      for (int j = 0; j < 4; ++j) {
+        bl[j] = dw_tr4(b_lds, 32 + kg * 8, wk + j * 16);
+        bh[j] = dw_tr4(b_lds, 32 + kg * 8 + 4, wk + j * 16);
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        af[i] = dw_frag(a_lds, 32 + kg * 8, wn + i * 16);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        bf_[j] = dw_frag(b_lds, 32 + kg * 8, wk + j * 16);
+    }
     if (!NOBAR) {
       __builtin_amdgcn_s_barrier();
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -274,9 +316,17 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
 #pragma unroll
     for (int i = 0; i < 4; ++i)
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
-        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[i], bf_[j], acc[i][j], 0, 0, 0);
+      for (int j = 0; j < 4; ++j) {
+        if (X16) {
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x16bf16_1k(
+              (bf16x4d)al[i], (bf16x4d)bl[j], acc[i][j], 0, 0, 0);
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x16bf16_1k(
+              (bf16x4d)ah[i], (bf16x4d)bh[j], acc[i][j], 0, 0, 0);
+        } else {
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf_[j], acc[i][j], 0, 0, 0);
+        }
+      }
     // boundary: prefetched block cur+1 must have landed
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
@@ -393,10 +443,17 @@ torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x,
     const char* e = getenv("TFMX_DW_BAR");
     return e && atoi(e) != 0;
   }();
+  static const bool use_x16 = [] {
+    const char* e = getenv("TFMX_DW_X16");
+    return e && atoi(e) != 0;
+  }();
   auto launch = [&](auto split, auto dbc, float* cwp) {
-    auto kfn = use_bar
-        ? gemm_dw_kernel<decltype(split)::value, decltype(dbc)::value, false>
-        : gemm_dw_kernel<decltype(split)::value, decltype(dbc)::value, true>;
+    constexpr bool SP = decltype(split)::value, DBV = decltype(dbc)::value;
+    auto kfn = use_x16
+        ? (use_bar ? gemm_dw_kernel<SP, DBV, false, true>
+                   : gemm_dw_kernel<SP, DBV, true, true>)
+        : (use_bar ? gemm_dw_kernel<SP, DBV, false, false>
+                   : gemm_dw_kernel<SP, DBV, true, false>);
     kfn<<<grid, DW_THREADS, 0, stream>>>(
             (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
             (short*)c.data_ptr(), cwp, dbw_p, (int)M, N, K, m_per_slice,
