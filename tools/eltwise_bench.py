@@ -36,7 +36,7 @@ for N in (512, 2048, 32770):
     ref = dy.float().sum(0)
     got = E.colsum(dy).float()
     err = (ref - got).abs().max().item() / ref.abs().max().item()
-    print(f"  colsum {M}x{N}: {us:8.1f} us  ({gb/us*1e3:7.0f} GB/s, "
+    print(f"  colsum {M}x{N}: {us:8.1f} us  ({gb/us*1e3:7.0f} TB/s, "
           f"bound {gb/6.3*1e3:6.1f} us)  relerr {err:.1e}")
 
 print("== ce fwd/bwd: logits 16320x32770 ==")
@@ -53,17 +53,17 @@ ref_loss = per[real].sum() / 64.0
 print(f"  ce loss relerr {abs(loss_lse[0].item()-ref_loss.item())/abs(ref_loss.item()):.1e}")
 us = t(lambda: E.ce_fwd(logits, tgt, 64.0, 0.1))
 gb = M * V * 2 / 1e9
-print(f"  ce_fwd : {us:8.1f} us  ({gb/us*1e3:7.0f} GB/s, bound {gb/6.3*1e3:6.1f} us)")
+print(f"  ce_fwd : {us:8.1f} us  ({gb/us*1e3:7.0f} TB/s, bound {gb/6.3*1e3:6.1f} us)")
 dloss = torch.ones((), device="cuda")
 us = t(lambda: E.ce_bwd(logits, tgt, loss_lse[1], dloss, 64.0, 0.1))
-print(f"  ce_bwd : {us:8.1f} us  ({2*gb/us*1e3:7.0f} GB/s r+w, bound {2*gb/6.3*1e3:6.1f} us)")
+print(f"  ce_bwd : {us:8.1f} us  ({2*gb/us*1e3:7.0f} TB/s r+w, bound {2*gb/6.3*1e3:6.1f} us)")
 
 print("== relu_bwd 16320x2048 ==")
 h = torch.randn(M, 2048, device="cuda").bfloat16().contiguous()
 dyh = torch.randn(M, 2048, device="cuda").bfloat16().contiguous()
 us = t(lambda: E.relu_bwd(dyh, h))
 gb = 3 * M * 2048 * 2 / 1e9
-print(f"  relu_bwd: {us:7.1f} us  ({gb/us*1e3:7.0f} GB/s rrw, bound {gb/6.3*1e3:6.1f} us)")
+print(f"  relu_bwd: {us:7.1f} us  ({gb/us*1e3:7.0f} TB/s rrw, bound {gb/6.3*1e3:6.1f} us)")
 
 print("== ln_gb shapes (dgamma/dbeta over rows, D=512) ==")
 s = torch.randn(M, 512, device="cuda").bfloat16().contiguous()
@@ -79,7 +79,7 @@ rg = (dy5.float() * xh).sum(0)
 rb = dy5.float().sum(0)
 eg = (rg - dgm.float()).abs().max().item() / rg.abs().max().item()
 eb = (rb - dbt.float()).abs().max().item() / rb.abs().max().item()
-print(f"  ln_bwd+gb: {us:6.1f} us  ({gb/us*1e3:7.0f} GB/s, bound {gb/6.3*1e3:6.1f} us)  relerr g {eg:.1e} b {eb:.1e}")
+print(f"  ln_bwd+gb: {us:6.1f} us  ({gb/us*1e3:7.0f} TB/s, bound {gb/6.3*1e3:6.1f} us)  relerr g {eg:.1e} b {eb:.1e}")
 
 print("== dh GEMM: dY2[16320,512] @ W2[512,2048] (NN) ==")
 a = torch.randn(M, 512, device="cuda").bfloat16().contiguous()
