@@ -41,10 +41,11 @@ namespace {
 // Within a subtile the m-row is stored permuted (pr = low-3-bits<<2 | top-2)
 // so the four 16-lane tr-read groups (m windows 8 apart) interleave across
 // the 512-B subtile instead of stacking on the same 256-B bank row.
+template <int ROWS = 64>
 DEV_INLINE int dw_img(int m, int col) {
   const int r = m & 31;
   const int pr = ((r & 7) << 2) | (r >> 3);
-  return ((col >> 4) * 2 + (m >> 5)) * 512 + pr * 16 + (col & 15);
+  return ((col >> 4) * (ROWS / 32) + (m >> 5)) * 512 + pr * 16 + (col & 15);
 }
 
 // Stage a [64 m][128 col] block from global (row-major, ldg elems/row) into
@@ -52,14 +53,15 @@ DEV_INLINE int dw_img(int m, int col) {
 // along col) + 4 ds_writes.  Out-of-range token rows and columns are
 // ZERO-filled — rows are contraction terms here, so clamping (as the NT
 // kernels do for discarded output rows) would add duplicate contributions.
+template <int COLS = 128, int ROWS = 64>
 DEV_INLINE void dw_stage(const short* __restrict__ g, long ldg, long m0,
                          long mmax, int col0, int ncols, short* lds) {
   const int t = threadIdx.x;
 #pragma unroll
-  for (int p = 0; p < 4; ++p) {
-    int idx = p * DW_THREADS + t;      // 1024 chunks of 8 shorts
-    int m = idx >> 4;                  // 16 chunks per 128-col row
-    int c8 = (idx & 15) << 3;
+  for (int p = 0; p < ROWS * COLS / 2048; ++p) {
+    int idx = p * DW_THREADS + t;      // ROWS*COLS/8 chunks of 8 shorts
+    int m = idx / (COLS / 8);
+    int c8 = (idx % (COLS / 8)) << 3;
     long gm = m0 + m;
     s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
     if (gm < mmax) {
@@ -72,26 +74,28 @@ DEV_INLINE void dw_stage(const short* __restrict__ g, long ldg, long m0,
       }
     }
     // image dest: subtile row-half [m][c8..c8+8) — contiguous 16 B
-    *(s16x4*)&lds[dw_img(m, c8)] = {v[0], v[1], v[2], v[3]};
-    *(s16x4*)&lds[dw_img(m, c8 + 4)] = {v[4], v[5], v[6], v[7]};
+    *(s16x4*)&lds[dw_img<ROWS>(m, c8)] = {v[0], v[1], v[2], v[3]};
+    *(s16x4*)&lds[dw_img<ROWS>(m, c8 + 4)] = {v[4], v[5], v[6], v[7]};
   }
 }
 
 // Transpose-read one 4-deep fragment slice: returns T[mb..mb+4)[colb + n]
 // for this lane's n-role (n = lane & 15).  mb must be a multiple of 4
 // within one 32-m subtile.
+template <int ROWS = 64>
 DEV_INLINE s16x4d dw_tr4(const short* lds, int mb, int colb) {
   const int mp = threadIdx.x & 15;   // this lane's address role
-  const int off = dw_img(mb + (mp >> 2), colb + 4 * (mp & 3));
+  const int off = dw_img<ROWS>(mb + (mp >> 2), colb + 4 * (mp & 3));
   return __builtin_amdgcn_ds_read_tr16_b64_v4i16(
       (__attribute__((address_space(3))) s16x4d*)(
           const_cast<short*>(&lds[off])));
 }
 
 // Full 8-deep bf16x8 fragment: contraction ms..ms+8, column colb + (l&15).
+template <int ROWS = 64>
 DEV_INLINE bf16x8d dw_frag(const short* lds, int ms, int colb) {
-  s16x4d lo = dw_tr4(lds, ms, colb);
-  s16x4d hi = dw_tr4(lds, ms + 4, colb);
+  s16x4d lo = dw_tr4<ROWS>(lds, ms, colb);
+  s16x4d hi = dw_tr4<ROWS>(lds, ms + 4, colb);
   s16x8 v = {lo[0], lo[1], lo[2], lo[3], hi[0], hi[1], hi[2], hi[3]};
   return (bf16x8d)v;
 }
@@ -101,11 +105,12 @@ DEV_INLINE bf16x8d dw_frag(const short* lds, int ms, int colb) {
 // inverse row permutation is applied to the per-lane SOURCE address — no
 // register bounce, no ds_write pass (guide §5 rule 1).  Each wave stages 4
 // of the 16 1-KiB chunks.
+template <int COLS = 128, int ROWS = 64>
 DEV_INLINE void dw_stage_glds(const short* __restrict__ g, long ldg, long m0,
                               int col0, short* lds) {
   const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
 #pragma unroll
-  for (int p = 0; p < 4; ++p) {
+  for (int p = 0; p < ROWS * COLS / 2048; ++p) {
     const int chunk = p * 4 + wid;
     const int d = chunk * 1024 + lane * 16;  // dest byte in image
     const int e = d >> 1;                    // dest element
@@ -113,8 +118,8 @@ DEV_INLINE void dw_stage_glds(const short* __restrict__ g, long ldg, long m0,
     const int we = e & 511;
     const int pr = we >> 4;
     const int r = ((pr & 3) << 3) | (pr >> 2);     // inverse permutation
-    const int m = (sub & 1) * 32 + r;
-    const int col = (sub >> 1) * 16 + (we & 15);
+    const int m = (sub % (ROWS / 32)) * 32 + r;
+    const int col = (sub / (ROWS / 32)) * 16 + (we & 15);
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)(g + (m0 + m) * ldg +
                                                         col0 + col),
@@ -137,7 +142,8 @@ typedef __bf16 bf16x4d __attribute__((ext_vector_type(4)));
 // pays ~65 v_mov per 64-m block pairing two v4i16 tr-reads into each
 // 4-VGPR operand (seen in the compiled loop; ~24% of the MFMA issue
 // time).  Same LDS traffic, twice the (half-size) MFMA instructions.
-template <bool SPLIT, bool DB, bool NOBAR = true, bool X16 = false>
+template <bool SPLIT, bool DB, bool NOBAR = true, bool X16 = false,
+          int BM = DW_BM>
 __global__ __launch_bounds__(DW_THREADS)
 void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
                     short* __restrict__ C, float* __restrict__ CW,
@@ -147,8 +153,8 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
   // L2 absorbs all tile re-reads — so it is pipeline-stall bound, not
   // traffic bound; the old single-buffer __syncthreads() drain exposed
   // the full staging latency every 64-token block).
-  __shared__ short a_img[2][DW_BM * DW_BN];  // dY block
-  __shared__ short b_img[2][DW_BM * DW_BK];  // X block
+  __shared__ short a_img[2][BM * DW_BN];  // dY block
+  __shared__ short b_img[2][BM * DW_BK];  // X block
 
   // XCD-aware bijective remap: consecutive logical tiles (same n-block,
   // varying k) land on the SAME XCD, so a dY slice is read into one XCD's
@@ -190,22 +196,22 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
 
   // Prologue: stage block 0 into slot 0, drain once.
   if (m_lo < m_hi) {
-    if (m_lo + DW_BM <= m_hi && a_full)
-      dw_stage_glds(dY, N, m_lo, bn0, a_img[0]);
+    if (m_lo + BM <= m_hi && a_full)
+      dw_stage_glds<128, BM>(dY, N, m_lo, bn0, a_img[0]);
     else
-      dw_stage(dY, N, m_lo, m_hi, bn0, N, a_img[0]);
-    if (m_lo + DW_BM <= m_hi && b_full)
-      dw_stage_glds(X, K, m_lo, bk0, b_img[0]);
+      dw_stage<128, BM>(dY, N, m_lo, m_hi, bn0, N, a_img[0]);
+    if (m_lo + BM <= m_hi && b_full)
+      dw_stage_glds<128, BM>(X, K, m_lo, bk0, b_img[0]);
     else
-      dw_stage(X, K, m_lo, m_hi, bk0, K, b_img[0]);
+      dw_stage<128, BM>(X, K, m_lo, m_hi, bk0, K, b_img[0]);
   }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
-  for (long m0 = m_lo; m0 < m_hi; m0 += DW_BM) {
+  for (long m0 = m_lo; m0 < m_hi; m0 += BM) {
     if (do_db) {
 #pragma unroll
-      for (int p = 0; p < 8; ++p) {
+      for (int p = 0; p < BM / 8; ++p) {
         const long gm = m0 + p * 8 + (threadIdx.x >> 5);
         if (gm < m_hi && bn0 + db_c4 + 4 <= N) {
           s16x4 v = *(const s16x4*)(dY + gm * N + bn0 + db_c4);
@@ -219,114 +225,78 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
         }
       }
     }
-    const bool m_full = m0 + DW_BM <= m_hi;
-    const int cur = (int)((m0 - m_lo) / DW_BM);
+    const int cur = (int)((m0 - m_lo) / BM);
     const short* a_lds = a_img[cur & 1];
     const short* b_lds = b_img[cur & 1];
     // Prologue staged block 0; here prefetch block cur+1 into the other
-    // slot (its previous tenant was consumed last block) — all 8 glds
-    // issued in phase 0, ~1.5 compute phases to land before the boundary
-    // vmcnt(0) (template schedule, see gemm256.hip).
-    const long mN = m0 + DW_BM;
+    // slot (its previous tenant was consumed last block) — all glds
+    // issued in phase 0, the rest of the block to land before the
+    // boundary vmcnt(0) (template schedule, see gemm256.hip).
+    const long mN = m0 + BM;
     const bool pf = mN < m_hi;
     bf16x8d af[4], bf_[4];
     s16x4d al[4], ah[4], bl[4], bh[4];
-    // ---- phase 0: k-step 0 -------------------------------------------
-    if (X16) {
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        al[i] = dw_tr4(a_lds, kg * 8, wn + i * 16);
-        ah[i] = dw_tr4(a_lds, kg * 8 + 4, wn + i * 16);
-      }
+    for (int ph = 0; ph < BM / 32; ++ph) {
+      const int ko = ph * 32 + kg * 8;
+      if (!NOBAR && ph) __builtin_amdgcn_s_barrier();
+      if (X16) {
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        bl[j] = dw_tr4(b_lds, kg * 8, wk + j * 16);
-        bh[j] = dw_tr4(b_lds, kg * 8 + 4, wk + j * 16);
+        for (int i = 0; i < 4; ++i) {
+          al[i] = dw_tr4<BM>(a_lds, ko, wn + i * 16);
+          ah[i] = dw_tr4<BM>(a_lds, ko + 4, wn + i * 16);
+        }
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          bl[j] = dw_tr4<BM>(b_lds, ko, wk + j * 16);
+          bh[j] = dw_tr4<BM>(b_lds, ko + 4, wk + j * 16);
+        }
+      } else {
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+          af[i] = dw_frag<BM>(a_lds, ko, wn + i * 16);
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          bf_[j] = dw_frag<BM>(b_lds, ko, wk + j * 16);
       }
-    } else {
+      if (ph == 0 && pf) {
+        short* pa = a_img[(cur + 1) & 1];
+        short* pb = b_img[(cur + 1) & 1];
+        if (mN + BM <= m_hi && a_full)
+          dw_stage_glds<128, BM>(dY, N, mN, bn0, pa);
+        else
+          dw_stage<128, BM>(dY, N, mN, m_hi, bn0, N, pa);
+        if (mN + BM <= m_hi && b_full)
+          dw_stage_glds<128, BM>(X, K, mN, bk0, pb);
+        else
+          dw_stage<128, BM>(X, K, mN, m_hi, bk0, K, pb);
+      }
+      // NOBAR: no intra-block barriers — all phases read the SAME LDS
+      // slot (synced at the block boundary), fragments are wave-private
+      // registers and the glds prefetch lands in the OTHER slot under
+      // vmcnt; the compiler tracks the LDS-read -> MFMA dependencies and
+      // software-pipelines the next phase's reads under this phase's
+      // MFMA burst.  else: full s_barrier + lgkmcnt(0) drains per phase
+      // keep the 4 waves in lockstep (A/B via TFMX_DW_BAR=1).
+      if (!NOBAR) {
+        __builtin_amdgcn_s_barrier();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      }
 #pragma unroll
       for (int i = 0; i < 4; ++i)
-        af[i] = dw_frag(a_lds, kg * 8, wn + i * 16);
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
-        bf_[j] = dw_frag(b_lds, kg * 8, wk + j * 16);
-    }
-    if (pf) {
-      short* pa = a_img[(cur + 1) & 1];
-      short* pb = b_img[(cur + 1) & 1];
-      if (mN + DW_BM <= m_hi && a_full)
-        dw_stage_glds(dY, N, mN, bn0, pa);
-      else
-        dw_stage(dY, N, mN, m_hi, bn0, N, pa);
-      if (mN + DW_BM <= m_hi && b_full)
-        dw_stage_glds(X, K, mN, bk0, pb);
-      else
-        dw_stage(X, K, mN, m_hi, bk0, K, pb);
-    }
-    // NOBAR: no intra-block barriers — both k-step phases read the SAME
-    // LDS slot (synced at the block boundary), fragments are wave-private
-    // registers and the glds prefetch lands in the OTHER slot under
-    // vmcnt; the compiler tracks the LDS-read -> MFMA dependencies and
-    // software-pipelines the phase-1 reads under the phase-0 MFMA burst.
-    // else: 3 full s_barrier + lgkmcnt(0) drains per block keep the 4
-    // waves in lockstep (A/B via TFMX_DW_BAR=1).
-    if (!NOBAR) {
-      __builtin_amdgcn_s_barrier();
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    }
-#pragma unroll
-    for (int i = 0; i < 4; ++i)
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        if (X16) {
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x16bf16_1k(
-              (bf16x4d)al[i], (bf16x4d)bl[j], acc[i][j], 0, 0, 0);
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x16bf16_1k(
-              (bf16x4d)ah[i], (bf16x4d)bh[j], acc[i][j], 0, 0, 0);
-        } else {
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[i], bf_[j], acc[i][j], 0, 0, 0);
+        for (int j = 0; j < 4; ++j) {
+          if (X16) {
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x16bf16_1k(
+                (bf16x4d)al[i], (bf16x4d)bl[j], acc[i][j], 0, 0, 0);
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x16bf16_1k(
+                (bf16x4d)ah[i], (bf16x4d)bh[j], acc[i][j], 0, 0, 0);
+          } else {
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[i], bf_[j], acc[i][j], 0, 0, 0);
+          }
         }
-      }
-    // ---- phase 1: k-step 1 -------------------------------------------
-    if (!NOBAR) __builtin_amdgcn_s_barrier();
-    if (X16) {
-#pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        al[i] = dw_tr4(a_lds, 32 + kg * 8, wn + i * 16);
-        ah[i] = dw_tr4(a_lds, 32 + kg * 8 + 4, wn + i * 16);
-      }
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        bl[j] = dw_tr4(b_lds, 32 + kg * 8, wk + j * 16);
-        bh[j] = dw_tr4(b_lds, 32 + kg * 8 + 4, wk + j * 16);
-      }
-    } else {
-#pragma unroll
-      for (int i = 0; i < 4; ++i)
-        af[i] = dw_frag(a_lds, 32 + kg * 8, wn + i * 16);
-#pragma unroll
-      for (int j = 0; j < 4; ++j)
-        bf_[j] = dw_frag(b_lds, 32 + kg * 8, wk + j * 16);
     }
-    if (!NOBAR) {
-      __builtin_amdgcn_s_barrier();
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    }
-#pragma unroll
-    for (int i = 0; i < 4; ++i)
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        if (X16) {
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x16bf16_1k(
-              (bf16x4d)al[i], (bf16x4d)bl[j], acc[i][j], 0, 0, 0);
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x16bf16_1k(
-              (bf16x4d)ah[i], (bf16x4d)bh[j], acc[i][j], 0, 0, 0);
-        } else {
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[i], bf_[j], acc[i][j], 0, 0, 0);
-        }
-      }
     // boundary: prefetched block cur+1 must have landed
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
@@ -423,10 +393,15 @@ torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x,
     const char* e = getenv("TFMX_DW_WGS");
     return e ? atoi(e) : 512;
   }();
-  int nslices = (int)min((M + DW_BM - 1) / DW_BM,
+  static const bool use_bm32 = [] {
+    const char* e = getenv("TFMX_DW_BM");
+    return e && atoi(e) == 32;
+  }();
+  const int bm = use_bm32 ? 32 : DW_BM;
+  int nslices = (int)min((M + bm - 1) / bm,
                          (long)cdiv(target_wgs, ntiles));
   long m_per_slice = (M + nslices - 1) / nslices;
-  m_per_slice = (m_per_slice + DW_BM - 1) / DW_BM * DW_BM;
+  m_per_slice = (m_per_slice + bm - 1) / bm * bm;
   nslices = (int)((M + m_per_slice - 1) / m_per_slice);
   dim3 grid(ntiles, nslices);
   auto stream = at::hip::getCurrentHIPStream();
@@ -449,11 +424,13 @@ torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x,
   }();
   auto launch = [&](auto split, auto dbc, float* cwp) {
     constexpr bool SP = decltype(split)::value, DBV = decltype(dbc)::value;
-    auto kfn = use_x16
-        ? (use_bar ? gemm_dw_kernel<SP, DBV, false, true>
-                   : gemm_dw_kernel<SP, DBV, true, true>)
-        : (use_bar ? gemm_dw_kernel<SP, DBV, false, false>
-                   : gemm_dw_kernel<SP, DBV, true, false>);
+    auto kfn = use_bm32
+        ? gemm_dw_kernel<SP, DBV, true, false, 32>
+        : (use_x16
+               ? (use_bar ? gemm_dw_kernel<SP, DBV, false, true>
+                          : gemm_dw_kernel<SP, DBV, true, true>)
+               : (use_bar ? gemm_dw_kernel<SP, DBV, false, false>
+                          : gemm_dw_kernel<SP, DBV, true, false>));
     kfn<<<grid, DW_THREADS, 0, stream>>>(
             (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
             (short*)c.data_ptr(), cwp, dbw_p, (int)M, N, K, m_per_slice,
