@@ -43,6 +43,10 @@ def build_app(model_dir: str, src_vocab: str, tgt_vocab: str,
 
     use_graph = device.startswith("cuda") and torch.cuda.is_available()
     graphed: dict = {}  # (S_src_bucket, max_len_bucket) -> GraphedDecoder
+    # FastAPI runs sync endpoints in a threadpool; GraphedDecoder replays
+    # mutate shared capture buffers, so decodes are serialized.
+    import threading
+    graph_lock = threading.Lock()
 
     def get_graphed(s_src: int, max_len: int):
         key = (max(16, (s_src + 15) // 16 * 16),
@@ -71,11 +75,12 @@ def build_app(model_dir: str, src_vocab: str, tgt_vocab: str,
         tgt_start = tgt_tok.vocab_size
         with torch.no_grad():
             if use_graph:
-                dec, s_bucket = get_graphed(len(tokens), max_len)
-                padded = torch.zeros(1, s_bucket, dtype=torch.int64,
-                                     device=enc_in.device)
-                padded[0, :len(tokens)] = enc_in[0]
-                out = dec(padded, tgt_start + 1, max_len=max_len)
+                with graph_lock:
+                    dec, s_bucket = get_graphed(len(tokens), max_len)
+                    padded = torch.zeros(1, s_bucket, dtype=torch.int64,
+                                         device=enc_in.device)
+                    padded[0, :len(tokens)] = enc_in[0]
+                    out = dec(padded, tgt_start + 1, max_len=max_len)
             else:
                 out = greedy_decode(model, enc_in, tgt_start, tgt_start + 1,
                                     max_len=max_len)
