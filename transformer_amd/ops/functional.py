@@ -85,7 +85,9 @@ def _fwd_gemm(E, x, w, b, activation):
         return E.gemm_nt(x, w, b if b is not None else torch.Tensor(),
                          1 if activation == "relu" else 0)
     if activation == "relu":
-        return torch._addmm_activation(b, x, w.t())
+        if hasattr(torch, "_addmm_activation"):
+            return torch._addmm_activation(b, x, w.t())
+        return torch.relu(torch.nn.functional.linear(x, w, b))
     return torch.nn.functional.linear(x, w, b)
 
 def _dw_gemm(dy, x, out=None):
