@@ -133,9 +133,11 @@ def main():
         dist.barrier()
 
     elapsed = t1 - t0
-    # MAX over ranks
+    # MAX over ranks (tensor must live on the backend's device: RCCL
+    # rejects CPU tensors)
     if world > 1:
-        t = torch.tensor([elapsed], dtype=torch.float64)
+        red_dev = device if dist.get_backend() == "nccl" else "cpu"
+        t = torch.tensor([elapsed], dtype=torch.float64, device=red_dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
 

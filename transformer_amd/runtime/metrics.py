@@ -27,6 +27,11 @@ class Mean:
 
     def sync(self):
         if dist.is_available() and dist.is_initialized():
-            t = torch.tensor([self.total, self.count], dtype=torch.float64)
+            # RCCL ("nccl") only reduces device tensors
+            dev = (torch.device("cuda", torch.cuda.current_device())
+                   if dist.get_backend() == "nccl" and torch.cuda.is_available()
+                   else torch.device("cpu"))
+            t = torch.tensor([self.total, self.count], dtype=torch.float64,
+                             device=dev)
             dist.all_reduce(t)
             self.total, self.count = t[0].item(), t[1].item()
