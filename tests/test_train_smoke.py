@@ -102,3 +102,29 @@ def test_predict_batch(tmp_path, toy_corpus):
         single = tr.predict(s)
         n = min(len(single), batch_out.shape[1])
         assert torch.equal(batch_out[i, :n], single[:n]), (i, s)
+
+
+def test_train_entry_point_subprocess(tmp_path, toy_corpus):
+    """C25: `python train.py` end-to-end — flags, dataset+tokenizer build,
+    training loop, checkpoint write, predict smoke, export — as the user
+    would run it."""
+    import subprocess
+    import sys
+    import os
+
+    out = subprocess.run(
+        [sys.executable, "train.py",
+         "--dataset_path", str(toy_corpus),
+         "--src_vocab_file", str(tmp_path / "sv"),
+         "--tgt_vocab_file", str(tmp_path / "tv"),
+         "--ckpt_path", str(tmp_path / "ckpt"),
+         "--epochs", "1", "--num_layers", "1", "--d_model", "32",
+         "--dff", "64", "--num_heads", "2", "--batch_size", "8",
+         "--sequence_length", "40", "--noenable_function",
+         "--steps_per_epoch", "3"],
+        capture_output=True, text=True, timeout=600,
+        env={**os.environ, "TFMX_EXPORT_DIR": str(tmp_path / "model")})
+    assert out.returncode == 0, out.stderr[-3000:]
+    assert os.path.isdir(tmp_path / "ckpt"), "no checkpoint directory"
+    assert "Epoch 1" in out.stdout or "epoch" in out.stdout.lower(), \
+        out.stdout[-1500:]
