@@ -259,3 +259,25 @@ def test_distributed_train_loop_gloo(tmp_path):
         assert p.exitcode == 0
     assert same, "ranks diverged after DistributedTrain loop"
     assert ckpt_ok, "rank-0 checkpoint directory missing"
+
+
+@pytest.mark.timeout(600)
+def test_distributed_train_entry_point_subprocess(tmp_path, toy_corpus):
+    """C25: `python distributed_train.py --num_gpu 2` end-to-end on CPU —
+    the self-exec under torchrun, gloo DP=2, shared dataset build, rank-0
+    checkpointing."""
+    import subprocess
+    import sys
+
+    out = subprocess.run(
+        [sys.executable, "distributed_train.py", "--num_gpu", "2",
+         "--dataset_path", str(toy_corpus),
+         "--src_vocab_file", str(tmp_path / "sv"),
+         "--tgt_vocab_file", str(tmp_path / "tv"),
+         "--ckpt_path", str(tmp_path / "ckpt"),
+         "--epochs", "1", "--num_layers", "1", "--d_model", "32",
+         "--dff", "64", "--num_heads", "2", "--batch_size", "8",
+         "--sequence_length", "40", "--noenable_function"],
+        capture_output=True, text=True, timeout=540)
+    assert out.returncode == 0, (out.stderr[-3000:], out.stdout[-1000:])
+    assert os.path.isdir(tmp_path / "ckpt"), "no rank-0 checkpoint directory"

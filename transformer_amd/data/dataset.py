@@ -136,7 +136,7 @@ def _encode_corpus(pairs, src_tok, tgt_tok, cache_path=None):
     enc = [([ss] + src_tok.encode(s) + [se], [ts] + tgt_tok.encode(t) + [te])
            for s, t in pairs]
     if cache_path:
-        tmp = cache_file + ".tmp"
+        tmp = f"{cache_file}.{os.getpid()}.tmp"  # pid-unique: no rank races
         torch.save(enc, tmp)
         os.replace(tmp, cache_file)
     return enc
@@ -159,11 +159,22 @@ def load_dataset(dataset_path: str, src_vocab_file: str, tgt_vocab_file: str,
         k = max(batch_size, len(train_pairs) // 50)
         train_pairs, test_pairs = train_pairs[:-k], train_pairs[-k:]
 
+    # In DP, rank 0 builds the vocab + encoded caches alone and the other
+    # ranks wait at a barrier, then hit the persisted files — concurrent
+    # builders raced on the shared cache/vocab writes (os.replace of a
+    # just-replaced tmp) and wasted (world-1) duplicate tokenizations.
+    import torch.distributed as dist
+    is_dist = (dist.is_available() and dist.is_initialized()
+               and dist.get_world_size() > 1)
+    if is_dist and dist.get_rank() != 0:
+        dist.barrier()  # rank 0 building
     src_tok, tgt_tok = load_or_create_tokenizer(train_pairs, src_vocab_file,
                                                 tgt_vocab_file)
     cache = os.path.join(dataset_path, "encoded_cache")
     enc_train = _encode_corpus(train_pairs, src_tok, tgt_tok, cache + "_train")
     enc_test = _encode_corpus(test_pairs, src_tok, tgt_tok, cache + "_test")
+    if is_dist and dist.get_rank() == 0:
+        dist.barrier()  # release the waiting ranks
 
     # length filter (train only, reference utils.py:145-153)
     enc_train = [(s, t) for s, t in enc_train

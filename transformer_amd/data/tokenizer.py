@@ -96,9 +96,15 @@ class SubwordTokenizer:
 
     # -- persistence (the reference's `.subwords` files, utils.py:92-97) ----
     def save_to_file(self, prefix: str):
-        with open(prefix + ".subwords", "w", encoding="utf-8") as f:
+        # atomic (tmp + rename, pid-unique): concurrent builders (e.g. DP
+        # ranks racing before the rank-0 gate existed) can never interleave
+        # writes into a corrupt vocab file.
+        import os
+        tmp = f"{prefix}.subwords.{os.getpid()}.tmp"
+        with open(tmp, "w", encoding="utf-8") as f:
             for s in self._subwords:
                 f.write("'" + s.replace("\\", "\\\\").replace("'", "\\'") + "'\n")
+        os.replace(tmp, prefix + ".subwords")
 
     @classmethod
     def load_from_file(cls, prefix: str) -> "SubwordTokenizer":
