@@ -145,3 +145,46 @@ def test_kv_cached_decode_matches_full_rerun():
     assert out_cached.shape[1] <= output.shape[1]
     T = out_cached.shape[1]
     assert torch.equal(out_cached, output[:, :T]), (out_cached, output)
+
+
+def test_source_pad_extension_invariance():
+    """Padding the source with trailing pad tokens must not change the
+    logits for the same target positions — the property serve.py's
+    shape-bucket padding relies on (pad keys are masked in encoder self-
+    and cross-attention; pad rows never feed a real query)."""
+    import torch
+    from transformer_amd.models import Transformer
+
+    torch.manual_seed(11)
+    m = Transformer(num_layers=2, d_model=32, num_heads=2, dff=64,
+                    input_vocab_size=60, target_vocab_size=60, rate=0.0,
+                    max_position=64)
+    m.eval()
+    inp = torch.randint(2, 58, (2, 7))
+    tar = torch.randint(2, 58, (2, 5))
+    logits1, _ = m((inp, tar), training=False)
+    padded = torch.zeros(2, 13, dtype=torch.int64)
+    padded[:, :7] = inp
+    logits2, _ = m((padded, tar), training=False)
+    assert torch.allclose(logits1, logits2, atol=1e-5), \
+        (logits1 - logits2).abs().max()
+
+
+def test_greedy_decode_pad_extension_invariance():
+    """Same property end-to-end through the KV-cached greedy decode."""
+    import torch
+    from transformer_amd.models import Transformer
+    from transformer_amd.models.transformer import greedy_decode
+
+    torch.manual_seed(12)
+    m = Transformer(num_layers=1, d_model=32, num_heads=2, dff=64,
+                    input_vocab_size=60, target_vocab_size=60, rate=0.0,
+                    max_position=64)
+    m.eval()
+    inp = torch.randint(2, 58, (2, 6))
+    out1 = greedy_decode(m, inp, 58, 59, max_len=8)
+    padded = torch.zeros(2, 11, dtype=torch.int64)
+    padded[:, :6] = inp
+    out2 = greedy_decode(m, padded, 58, 59, max_len=8)
+    n = min(out1.shape[1], out2.shape[1])
+    assert torch.equal(out1[:, :n], out2[:, :n]), (out1, out2)
