@@ -35,7 +35,10 @@ def main(epochs, enable_function, buffer_size, batch_size, sequence_length,
          warmup_steps=60000, label_smoothing=0.0, device=None, dtype=None,
          seed=1234, log_interval=100, eval_steps=50, synthetic_data=False,
          synthetic_vocab=32768, steps_per_epoch=100, max_decode_len=10,
-         **_ignored):
+         debug_sync=False, **_ignored):
+    if debug_sync:  # race/fault localization mode
+        os.environ["AMD_SERIALIZE_KERNEL"] = "3"
+        os.environ["HIP_LAUNCH_BLOCKING"] = "1"
     rank, local_rank, world_size = init_distributed()
     torch.manual_seed(seed)  # same init on every rank (then X1 broadcast)
     if device is None and torch.cuda.is_available():

@@ -54,7 +54,13 @@ def main(epochs, enable_function, buffer_size, batch_size, sequence_length,
          warmup_steps=60000, label_smoothing=0.0, device=None, dtype=None,
          seed=1234, log_interval=100, eval_steps=50, synthetic_data=False,
          synthetic_vocab=32768, steps_per_epoch=100, max_decode_len=10,
-         trace_dir=None, **_ignored):
+         trace_dir=None, debug_sync=False, **_ignored):
+    if debug_sync:
+        # race/fault localization: every kernel launch is
+        # serialized and synchronous so a fault is attributed to the
+        # launching line, and reductions run in a deterministic order.
+        os.environ["AMD_SERIALIZE_KERNEL"] = "3"
+        os.environ["HIP_LAUNCH_BLOCKING"] = "1"
     torch.manual_seed(seed)
     device, dtype = pick_device_dtype(device, dtype)
 

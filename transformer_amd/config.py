@@ -88,6 +88,10 @@ def transformer_flags(parser: argparse.ArgumentParser | None = None) -> argparse
                         help="token vocab for --synthetic_data (model vocab = +2)")
     parser.add_argument("--steps_per_epoch", type=int, default=100,
                         help="steps per epoch when --synthetic_data")
+    parser.add_argument("--debug_sync", action="store_true",
+                        help="serialize+block every kernel launch "
+                             "(AMD_SERIALIZE_KERNEL=3, HIP_LAUNCH_BLOCKING=1)"
+                             " — race/fault localization mode, SURVEY.md §5")
     parser.add_argument("--max_decode_len", type=int, default=10,
                         help="greedy decode steps for predict (reference train.py:109 uses 10)")
     return parser
@@ -125,6 +129,7 @@ def flags_dict(args: argparse.Namespace) -> dict:
         "synthetic_vocab": args.synthetic_vocab,
         "steps_per_epoch": args.steps_per_epoch,
         "max_decode_len": args.max_decode_len,
+        "debug_sync": args.debug_sync,
     }
 
 
