@@ -362,7 +362,7 @@ def greedy_decode(model, inp, start_id, end_id, max_len=10):
     finished = torch.zeros(B, dtype=torch.bool, device=inp.device)
     for _ in range(max_len):
         logits = decode_step(model, out[:, -1:], cache)
-        nxt = ops.argmax_lastdim(logits.float()).view(B, 1)
+        nxt = ops.argmax_lastdim(logits).view(B, 1)
         out = torch.cat([out, nxt], dim=-1)
         finished |= (nxt.squeeze(1) == end_id)
         if bool(finished.all()):
@@ -440,7 +440,7 @@ class GraphedDecoder:
                 x = O.residual_layernorm(ffn, out2, layer.ln3.gamma,
                                          layer.ln3.beta, layer.ln3.eps)
             logits = O.linear(x, self.model.w_final, self.model.b_final)
-            nxt = O.argmax_lastdim(logits.float()).view(B, 1)
+            nxt = O.argmax_lastdim(logits).view(B, 1)
             self.tok.index_copy_(1, pos + 1, nxt)
             self.pos.add_(1)
 
