@@ -128,3 +128,32 @@ def test_train_entry_point_subprocess(tmp_path, toy_corpus):
     assert os.path.isdir(tmp_path / "ckpt"), "no checkpoint directory"
     assert "Epoch 1" in out.stdout or "epoch" in out.stdout.lower(), \
         out.stdout[-1500:]
+
+
+def test_checkpoint_cadence_q7_intent(tmp_path):
+    """the reference's precedence bug saved every epoch
+    EXCEPT multiples of 5; the implemented intent is save at epochs
+    5, 10, ... and at the last epoch."""
+    from transformer_amd.models import Transformer
+    from transformer_amd.runtime import Train
+
+    class Tok:
+        vocab_size = 30
+        def encode(self, s):
+            return [1]
+        def decode(self, ids):
+            return "x"
+
+    saved = []
+    torch.manual_seed(0)
+    m = Transformer(num_layers=1, d_model=16, num_heads=2, dff=32,
+                    input_vocab_size=32, target_vocab_size=32, rate=0.0,
+                    max_position=16)
+    tr = Train(epochs=7, enable_function=False, transformer=m,
+               src_tokenizer=Tok(), tgt_tokenizer=Tok(), batch_size=2,
+               train_log_dir=None, test_log_dir=None, max_ckpt_keep=3,
+               ckpt_path=str(tmp_path), d_model=16)
+    tr.ckpt_manager.save = lambda step, epoch=None: saved.append(epoch)
+    for epoch in range(7):
+        tr._save_if_due(epoch, step=epoch + 1)
+    assert saved == [4, 6], saved  # epochs 5 and 7 (0-indexed 4, 6)
