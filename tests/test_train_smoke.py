@@ -157,3 +157,41 @@ def test_checkpoint_cadence_q7_intent(tmp_path):
     for epoch in range(7):
         tr._save_if_due(epoch, step=epoch + 1)
     assert saved == [4, 6], saved  # epochs 5 and 7 (0-indexed 4, 6)
+
+
+def test_eval_runs_fixed_batch_count_q8(tmp_path):
+    """the reference's distributed eval effectively ran ONE
+    test step (never-reset modulo counter); here eval consumes exactly
+    min(eval_steps, len(dataset)) batches every time it runs."""
+    from transformer_amd.models import Transformer
+    from transformer_amd.runtime import Train
+
+    class Tok:
+        vocab_size = 30
+        def encode(self, s):
+            return [1]
+        def decode(self, ids):
+            return "x"
+
+    torch.manual_seed(0)
+    m = Transformer(num_layers=1, d_model=16, num_heads=2, dff=32,
+                    input_vocab_size=32, target_vocab_size=32, rate=0.0,
+                    max_position=16)
+    tr = Train(epochs=1, enable_function=False, transformer=m,
+               src_tokenizer=Tok(), tgt_tokenizer=Tok(), batch_size=2,
+               train_log_dir=None, test_log_dir=None, max_ckpt_keep=1,
+               ckpt_path=str(tmp_path), d_model=16, eval_steps=3)
+    batch = (torch.randint(1, 30, (2, 6)), torch.randint(1, 30, (2, 6)))
+    consumed = []
+
+    def gen(n):
+        for i in range(n):
+            consumed.append(i)
+            yield batch
+
+    tr._run_eval(gen(10))
+    assert len(consumed) == 3, consumed        # capped at eval_steps
+    consumed.clear()
+    tr._run_eval(gen(2))
+    assert len(consumed) == 2, consumed        # short dataset: all of it
+    assert tr.test_loss.count == 2             # metrics reset per eval
