@@ -1,10 +1,16 @@
 """Autograd-integrated op layer with device dispatch.
 
-GPU (CUDA/ROCm) tensors run the hand-written CDNA4 HIP kernels via
-`torch.autograd.Function` wrappers with hand-written backward kernels
-(SURVEY.md §2.3); CPU tensors run the fp32 reference math in reference.py
-through plain differentiable torch ops.  The GPU path never falls back:
-if the extension is missing it raises (see ops/__init__.py).
+GPU (CUDA/ROCm) tensors run via `torch.autograd.Function` wrappers
+(SURVEY.md §2.3): the fused/custom ops (attention fwd+bwd, dW-transpose
+GEMM, residual-LN(+dropout), CE, Adam, embedding+PE, dropout, argmax/
+accuracy) are hand-written CDNA4 HIP kernels; PLAIN GEMMs go to the
+backend that measured fastest per shape (hipBLASLt fused bias/ReLU
+epilogues for training-shape forwards and for dX / the wide logits dW;
+the hand-written NT/dW kernels for small-M forwards and the d_model dW
+shapes — see docs/PERF.md "GEMM backends").  CPU tensors run the fp32
+reference math in reference.py through plain differentiable torch ops.
+The GPU path never falls back silently: if the extension is missing it
+raises (see ops/__init__.py).
 """
 
 from __future__ import annotations
