@@ -61,3 +61,25 @@ def test_tokenizer_roundtrip_fuzz():
         ids = tok.encode(s)
         assert all(0 < i < tok.vocab_size for i in ids)
         assert tok.decode(ids) == want, (s, tok.decode(ids))
+
+
+def test_tokenizer_roundtrip_hypothesis():
+    """Property: decode(encode(text)) recovers text modulo whitespace
+    normalization, for arbitrary unicode (byte fallback covers OOV)."""
+    from hypothesis import given, settings, strategies as st
+
+    from transformer_amd.data.tokenizer import SubwordTokenizer
+
+    tok = SubwordTokenizer.build_from_corpus(
+        ["the quick brown fox", "pack my box with five dozen jugs"], 300)
+
+    @settings(max_examples=200, deadline=None)
+    @given(st.text(max_size=40))
+    def check(text):
+        ids = tok.encode(text)
+        assert all(0 < i < tok.vocab_size for i in ids)
+        # whitespace runs collapse to single spaces by construction
+        expect = " ".join(text.split())
+        assert tok.decode(ids) == expect, (text, ids)
+
+    check()

@@ -28,9 +28,31 @@ _MAX_SUBWORD_LEN = 20
 
 def _words_with_markers(line: str):
     """Split on whitespace; append '_' end-of-word marker (decode joins on
-    it).  '_' inside the text is escaped to '\\u'."""
+    it).  Escapes: backslash -> '\\\\' first, then '_' -> '\\u', so a
+    literal backslash-u in the input cannot collide with the underscore
+    escape (caught by the hypothesis round-trip fuzz)."""
     for w in _WORD_RE.findall(line):
-        yield w.replace("_", "\\u") + "_"
+        yield w.replace("\\", "\\\\").replace("_", "\\u") + "_"
+
+
+def _unescape(text: str) -> str:
+    out = []
+    i, n = 0, len(text)
+    while i < n:
+        c = text[i]
+        if c == "\\" and i + 1 < n:
+            nxt = text[i + 1]
+            if nxt == "\\":
+                out.append("\\")
+                i += 2
+                continue
+            if nxt == "u":
+                out.append("_")
+                i += 2
+                continue
+        out.append(c)
+        i += 1
+    return "".join(out)
 
 
 class SubwordTokenizer:
@@ -92,7 +114,7 @@ class SubwordTokenizer:
                 flush()
                 parts.append(self._subwords[t - 1])
         flush()
-        return "".join(parts).replace("_", " ").replace("\\u", "_").rstrip()
+        return _unescape("".join(parts).replace("_", " ")).rstrip()
 
     # -- persistence (the reference's `.subwords` files, utils.py:92-97) ----
     def save_to_file(self, prefix: str):
