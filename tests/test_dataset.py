@@ -61,3 +61,32 @@ def test_synthetic_dataset_shapes():
     assert src.shape == (4, 16) and tgt.shape == (4, 16)
     assert (src[:, 0] == 1000).all() and (src[:, -1] == 1001).all()
     assert (src != 0).all()  # no padding -> fixed work per step
+
+
+def test_dp_shards_partition_global_batch():
+    """Property: for any world size dividing the batch, the per-rank
+    shards of every batch are disjoint and their union is exactly the
+    global batch."""
+    from transformer_amd.data.dataset import BatchedDataset
+
+    pairs = [([i, i + 1], [i + 2]) for i in range(2, 53)]  # 51 pairs, ragged tail
+    for world in (1, 2, 4):
+        per_rank = [list(BatchedDataset(pairs, 8, shuffle=True, seed=3,
+                                        rank=r, world_size=world))
+                    for r in range(world)]
+        # same number of batches per rank (lockstep: no collective desync)
+        assert len({len(b) for b in per_rank}) == 1
+        full = list(BatchedDataset(pairs, 8, shuffle=True, seed=3))
+        # a ragged tail smaller than the world is dropped UNIFORMLY
+        n_common = len(per_rank[0])
+        assert n_common in (len(full), len(full) - 1)
+        for bi in range(n_common):
+            gsrc = full[bi][0]
+            rows = [tuple(row.tolist())
+                    for r in range(world)
+                    for row in per_rank[r][bi][0]]
+            grows = [tuple(row.tolist()) for row in gsrc]
+            def strip(t):
+                return tuple(x for x in t if x != 0)  # padding widths differ
+            assert sorted(strip(t) for t in rows) == \
+                sorted(strip(t) for t in grows), (world, bi)

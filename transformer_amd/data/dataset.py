@@ -100,11 +100,17 @@ class BatchedDataset:
             idx = order[i:i + gb]
             if len(idx) < gb and self.drop_last:
                 break
+            # A ragged tail smaller than the world would leave some ranks
+            # with an empty shard — those ranks would skip the batch while
+            # the others step, desynchronizing the collectives (deadlock).
+            # The decision must be UNIFORM across ranks: every rank drops
+            # such a tail.  For len(idx) >= world the floor-split below
+            # gives every rank at least one item.
+            if self.world_size > 1 and len(idx) < self.world_size:
+                continue
             # per-replica shard of the global batch
             shard = idx[self.rank * len(idx) // self.world_size:
                         (self.rank + 1) * len(idx) // self.world_size]
-            if not shard:
-                continue
             src = [self.pairs[j][0] for j in shard]
             tgt = [self.pairs[j][1] for j in shard]
             yield _pad_batch(src), _pad_batch(tgt)
