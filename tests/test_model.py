@@ -188,3 +188,26 @@ def test_greedy_decode_pad_extension_invariance():
     out2 = greedy_decode(m, padded, 58, 59, max_len=8)
     n = min(out1.shape[1], out2.shape[1])
     assert torch.equal(out1[:, :n], out2[:, :n]), (out1, out2)
+
+
+def test_decoder_causality_property():
+    """Changing target tokens at positions >= t must not change logits at
+    positions < t (the causal mask contract the KV-cached and graphed
+    decoders rely on)."""
+    import torch
+    from transformer_amd.models import Transformer
+
+    torch.manual_seed(13)
+    m = Transformer(num_layers=2, d_model=32, num_heads=2, dff=64,
+                    input_vocab_size=60, target_vocab_size=60, rate=0.0,
+                    max_position=32)
+    m.eval()
+    inp = torch.randint(2, 58, (2, 6))
+    tar = torch.randint(2, 58, (2, 8))
+    logits1, _ = m((inp, tar), training=False)
+    tar2 = tar.clone()
+    tar2[:, 5:] = torch.randint(2, 58, (2, 3))
+    logits2, _ = m((inp, tar2), training=False)
+    assert torch.allclose(logits1[:, :5], logits2[:, :5], atol=1e-5), \
+        (logits1[:, :5] - logits2[:, :5]).abs().max()
+    assert not torch.allclose(logits1[:, 5:], logits2[:, 5:])
