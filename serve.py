@@ -42,7 +42,14 @@ def build_app(model_dir: str, src_vocab: str, tgt_vocab: str,
     tgt_tok = SubwordTokenizer.load_from_file(tgt_vocab)
 
     use_graph = device.startswith("cuda") and torch.cuda.is_available()
-    graphed: dict = {}  # (S_src_bucket, max_len_bucket) -> GraphedDecoder
+    # Bucket cache is LRU-bounded and keys are clamped: each entry holds KV
+    # caches + a captured graph, so request-controlled (src_len, max_len)
+    # must not grow GPU memory without limit.
+    from collections import OrderedDict
+    MAX_DECODE_LEN = 256          # server ceiling on requested max_len
+    MAX_SRC_LEN = 512             # requests longer than this are rejected
+    MAX_GRAPH_BUCKETS = 8
+    graphed = OrderedDict()  # (S_src_bucket, max_len_bucket) -> decoder
     # FastAPI runs sync endpoints in a threadpool; GraphedDecoder replays
     # mutate shared capture buffers, so decodes are serialized.
     import threading
@@ -51,7 +58,11 @@ def build_app(model_dir: str, src_vocab: str, tgt_vocab: str,
     def get_graphed(s_src: int, max_len: int):
         key = (max(16, (s_src + 15) // 16 * 16),
                max(16, (max_len + 15) // 16 * 16))
-        if key not in graphed:
+        if key in graphed:
+            graphed.move_to_end(key)
+        else:
+            while len(graphed) >= MAX_GRAPH_BUCKETS:
+                graphed.popitem(last=False)
             graphed[key] = GraphedDecoder(
                 model, B=1, S_src=key[0], max_len=key[1],
                 start_id=tgt_tok.vocab_size,
@@ -67,9 +78,12 @@ def build_app(model_dir: str, src_vocab: str, tgt_vocab: str,
     @app.post("/translate")
     def translate(payload: dict = Body(...)):
         text = str(payload.get("text", ""))
-        max_len = int(payload.get("max_len", 10))
+        max_len = min(max(int(payload.get("max_len", 10)), 1),
+                      MAX_DECODE_LEN)
         src_start = src_tok.vocab_size
         tokens = [src_start] + src_tok.encode(text) + [src_start + 1]
+        if len(tokens) > MAX_SRC_LEN:
+            return {"error": f"source too long (> {MAX_SRC_LEN} tokens)"}
         enc_in = torch.tensor([tokens], dtype=torch.int64,
                               device=next(model.parameters()).device)
         tgt_start = tgt_tok.vocab_size

@@ -267,3 +267,42 @@ def test_serve_graphed_path_gpu(tmp_path, toy_corpus):
     r = client.post("/translate", json={"text": "one two three",
                                         "max_len": 20})
     assert r.status_code == 200 and len(r.json()["tokens"]) >= 1
+
+
+def test_recapture_is_state_neutral():
+    """a mid-training graph recapture must not move
+    optimizer state — warmup runs real fwd+bwd+Adam iterations on the
+    static buffers, which must be snapshot/restored."""
+    import torch
+    from transformer_amd.models import Transformer
+    from transformer_amd.runtime import NoamAdam
+    from transformer_amd.runtime.graph import CapturedTrainStep
+    from transformer_amd import ops
+
+    torch.manual_seed(11)
+    model = Transformer(num_layers=1, d_model=64, num_heads=2, dff=128,
+                        input_vocab_size=90, target_vocab_size=90,
+                        rate=0.1, max_position=64).cuda().bfloat16()
+    opt = NoamAdam(model, 64, warmup_steps=100, use_flat=True)
+    B = 4
+    loss_fn = lambda real, pred: ops.masked_cross_entropy(pred, real, B, 0.0)
+    cap = CapturedTrainStep(model, opt, loss_fn, (B, 8), (B, 8),
+                            torch.device("cuda"))
+    src = torch.randint(1, 80, (B, 8), device="cuda")
+    tar = torch.randint(1, 80, (B, 8), device="cuda")
+    for _ in range(3):
+        cap(src, tar)
+    torch.cuda.synchronize()
+    snap = (opt.master.clone(), opt.m.clone(), opt.v.clone(),
+            opt.flat.flat_w.clone(), opt.step_count)
+    # recapture at a bigger shape (what a longer batch triggers)
+    CapturedTrainStep(model, opt, loss_fn, (B, 16), (B, 16),
+                      torch.device("cuda"))
+    torch.cuda.synchronize()
+    assert torch.equal(opt.master, snap[0]), "master moved in recapture"
+    assert torch.equal(opt.m, snap[1]), "m moved in recapture"
+    assert torch.equal(opt.v, snap[2]), "v moved in recapture"
+    assert torch.equal(opt.flat.flat_w, snap[3]), "weights moved"
+    assert opt.step_count == snap[4]
+    step_t, _ = opt.graph_state()
+    assert int(step_t.item()) == opt.step_count

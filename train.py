@@ -30,7 +30,11 @@ class _SyntheticTok:
         self.vocab_size = vocab_size
 
     def encode(self, text):
-        return [2 + (hash(w) % (self.vocab_size - 2)) for w in text.split()]
+        # stable across processes/runs (Python hash() is PYTHONHASHSEED-
+        # dependent, which would desync DP ranks on predict)
+        import zlib
+        return [2 + (zlib.crc32(w.encode()) % (self.vocab_size - 2))
+                for w in text.split()]
 
     def decode(self, ids):
         return " ".join(str(int(i)) for i in ids)

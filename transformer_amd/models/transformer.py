@@ -367,6 +367,18 @@ def greedy_decode(model, inp, start_id, end_id, max_len=10):
         finished |= (nxt.squeeze(1) == end_id)
         if bool(finished.all()):
             break
+    # Rows that finished early kept receiving argmax tokens while other
+    # rows decoded — mask them so batched output matches single-sentence
+    # and GraphedDecoder output for the same sentence.
+    return mask_after_end(out, end_id)
+
+
+def mask_after_end(out: torch.Tensor, end_id: int) -> torch.Tensor:
+    """Zero every token after the first end_id per row (in place)."""
+    hit = (out == end_id).cumsum(dim=1) > 0
+    mask = hit.roll(1, dims=1)
+    mask[:, 0] = False
+    out[mask] = 0
     return out
 
 
@@ -488,8 +500,4 @@ class GraphedDecoder:
         out = self.tok[:, :n + 1].clone()
         # zero everything after the first end_id per row (parity with the
         # early-stopping eager decode)
-        hit = (out == end_id).cumsum(dim=1) > 0
-        mask = hit.roll(1, dims=1)
-        mask[:, 0] = False
-        out[mask] = 0
-        return out
+        return mask_after_end(out, end_id)

@@ -59,5 +59,6 @@ from .functional import (  # noqa: E402
     dropout,
     masked_cross_entropy,
     masked_accuracy,
+    masked_accuracy_counts,
     argmax_lastdim,
 )

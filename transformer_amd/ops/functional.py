@@ -35,22 +35,27 @@ from . import ext
 # subscribes here for gradient-readiness instead of post-accumulate hooks.
 # ---------------------------------------------------------------------------
 
-_GRAD_READY_CB = None
-
-
-def set_grad_ready_callback(fn):
+def set_grad_ready_callback(params, fn):
     """fn(param) is invoked (backward order) right after a parameter's
     gradient kernels are enqueued on the compute stream.  Used by
-    parallel/ddp.py to launch bucket all-reduces."""
-    global _GRAD_READY_CB
-    _GRAD_READY_CB = fn
+    parallel/ddp.py to launch bucket all-reduces.  Registration is scoped
+    to the given parameters (not process-global): a second
+    BucketedDataParallel over a different model cannot hijack the first,
+    and fn=None unregisters."""
+    for p in params:
+        if fn is None:
+            if hasattr(p, "_grad_ready_cb"):
+                del p._grad_ready_cb
+        else:
+            p._grad_ready_cb = fn
 
 
 def _grad_ready(*params):
-    if _GRAD_READY_CB is not None:
-        for p in params:
-            if p is not None:
-                _GRAD_READY_CB(p)
+    for p in params:
+        if p is not None:
+            cb = getattr(p, "_grad_ready_cb", None)
+            if cb is not None:
+                cb(p)
 
 
 def _flat(p):
@@ -635,9 +640,21 @@ def argmax_lastdim(logits):
 
 
 def masked_accuracy(logits, targets):
+    correct, total = masked_accuracy_counts(logits, targets)
+    return correct / max(total, 1)
+
+
+def masked_accuracy_counts(logits, targets):
+    """(correct, total) over non-pad target positions — lets callers
+    accumulate token-weighted accuracy (the reference's streaming
+    SparseCategoricalAccuracy semantics, train.py:72-73) instead of a
+    mean of batch-means."""
     if logits.is_cuda:
-        B, T, V = logits.shape
+        B = logits.shape
+        V = logits.shape[-1]
         correct, total = ext().accuracy(logits.reshape(-1, V).contiguous(),
                                         targets.reshape(-1).contiguous())
-        return correct / max(total, 1)
-    return float(R.masked_accuracy(logits, targets))
+        return float(correct), float(total)
+    pred = logits.argmax(dim=-1)
+    mask = targets != 0
+    return (float(((pred == targets) & mask).sum()), float(mask.sum()))

@@ -92,8 +92,17 @@ class BucketedDataParallel:
             for p in flat.params
         ]
         from ..ops import functional as _F
-        _F.set_grad_ready_callback(self._on_param_ready)
+        _F.set_grad_ready_callback(flat.params, self._on_param_ready)
         self._reset_step()
+
+    def detach(self):
+        """Unregister hooks/callbacks (a discarded instance must not keep
+        firing into dead bucket state)."""
+        from ..ops import functional as _F
+        _F.set_grad_ready_callback(self.flat.params, None)
+        for h in self._hooks:
+            h.remove()
+        self._hooks = []
 
     def _on_param_ready(self, p):
         if self.world <= 1:

@@ -59,8 +59,13 @@ class CheckpointManager:
                 os.remove(os.path.join(self.directory, old))
             except FileNotFoundError:
                 pass
-        with open(self._index_path, "w") as f:
+        # atomic index write: a crash mid-write must not orphan the window
+        itmp = self._index_path + ".tmp"
+        with open(itmp, "w") as f:
             json.dump({"checkpoints": ckpts}, f)
+            f.flush()
+            os.fsync(f.fileno())
+        os.replace(itmp, self._index_path)
         return path
 
     def restore(self, path: str | None = None) -> dict | None:

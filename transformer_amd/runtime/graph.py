@@ -46,8 +46,15 @@ class CapturedTrainStep:
             optimizer.step_captured()
             return loss, logits, tar_real
 
-        # warmup on a side stream (cuDNN-style), then capture
+        # warmup on a side stream (cuDNN-style), then capture.  The warmup
+        # iterations execute REAL fwd+bwd+Adam steps on the zero-filled
+        # static buffers — harmless at first capture (m=v=0) but a
+        # mid-training recapture (batch exceeding the captured shape) would
+        # apply zero-gradient Adam updates to live state.  Snapshot the fp32
+        # master + m/v around the warmup so recapture is state-neutral.
         ops.functional.set_graph_rng(step_t)
+        snap = (optimizer.master.clone(), optimizer.m.clone(),
+                optimizer.v.clone())
         try:
             s = torch.cuda.Stream()
             s.wait_stream(torch.cuda.current_stream())
@@ -55,6 +62,12 @@ class CapturedTrainStep:
                 for _ in range(warmup_iters):
                     run()
             torch.cuda.current_stream().wait_stream(s)
+            optimizer.master.copy_(snap[0])
+            optimizer.m.copy_(snap[1])
+            optimizer.v.copy_(snap[2])
+            optimizer.flat.flat_w.copy_(
+                optimizer.master.to(optimizer.flat.flat_w.dtype))
+            del snap
 
             self.graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(self.graph):

@@ -85,8 +85,8 @@ class Train:
             self._captured = cap
         loss = cap(src, tar)
         self.train_loss.update(loss.item())
-        self.train_accuracy.update(ops.masked_accuracy(cap.logits,
-                                                       cap.tar_real))
+        c, t = ops.masked_accuracy_counts(cap.logits, cap.tar_real)
+        self.train_accuracy.update(c / max(t, 1), weight=t)
         return loss
 
     def train_step(self, inputs):
@@ -105,8 +105,10 @@ class Train:
         self.optimizer.step()
 
         self.train_loss.update(loss.detach().item())
-        self.train_accuracy.update(ops.masked_accuracy(predictions.detach(),
-                                                       tar_real))
+        # token-weighted accumulation: batches contribute by real-token
+        # count, matching the reference's streaming accuracy semantics
+        c, t = ops.masked_accuracy_counts(predictions.detach(), tar_real)
+        self.train_accuracy.update(c / max(t, 1), weight=t)
         return loss
 
     @torch.no_grad()
@@ -118,7 +120,8 @@ class Train:
         predictions, _ = self.transformer((src, tar_inp), training=False)
         t_loss = self.loss_function(tar_real, predictions)
         self.test_loss.update(t_loss.item())
-        self.test_accuracy.update(ops.masked_accuracy(predictions, tar_real))
+        c, t = ops.masked_accuracy_counts(predictions, tar_real)
+        self.test_accuracy.update(c / max(t, 1), weight=t)
 
     # -- greedy inference (C19, reference train.py:91-121; Q6 fixes) --------
     @torch.no_grad()
