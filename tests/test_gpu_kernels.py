@@ -538,3 +538,123 @@ def test_fwd_gemm_backends_agree():
         y_lib = torch._addmm_activation(b, x, w.t()).float()
         err = (y_hip - y_lib).abs().max() / y_lib.abs().max().clamp(min=1)
         assert err < 1e-2, f"relu epilogue disagreement {err} at M={M}"
+
+
+# ---------------------------------------------------------------------------
+# gemm_uni: deep-pipelined counted-vmcnt kernels (round 2)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("m,n,k,epi", [
+    (16384, 512, 512, 0),     # out-proj fwd shape (BN=128 grid)
+    (16384, 1536, 512, 0),    # packed QKV fwd
+    (16384, 2048, 512, 1),    # FFN1 + ReLU epilogue
+    (16384, 512, 2048, 0),    # FFN2
+    (16320, 1000, 512, 0),    # ragged M + ragged N (NT clamps both)
+    (2048, 32770, 512, 0),    # logits-head slice (big-N edge)
+])
+def test_gemm_uni_nt(m, n, k, epi):
+    torch.manual_seed(3)
+    a = torch.randn(m, k, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(n, k, device="cuda", dtype=torch.bfloat16) * 0.1
+    b = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    assert _ext().gemm_uni_viable(m, n, k), "dispatch shape must be viable"
+    c = _ext().gemm_uni_nt(a, w, b, epi)
+    ref = a.float() @ w.float().T + b.float()
+    if epi == 1:
+        ref = torch.relu(ref)
+    assert_close(c, ref, 0.03, f"gemm_uni_nt {m}x{n}x{k}")
+
+
+@pytest.mark.parametrize("m,n,k", [
+    (16384, 512, 1536),   # dX of packed QKV
+    (16384, 2048, 512),   # dX of FFN2's weight... (dY @ W with W (512,2048))
+    (16384, 512, 2048),   # dX of FFN1
+    (16320, 512, 512),    # ragged-M dX of out-proj
+])
+def test_gemm_uni_nn(m, n, k):
+    """dX: C[M,N] = A[M,K] @ B[K,N], B (the weight) read red-major via
+    tr16 — asymmetric operands catch transposed fragment maps."""
+    torch.manual_seed(4)
+    a = torch.randn(m, k, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(k, n, device="cuda", dtype=torch.bfloat16) * 0.1
+    c = _ext().gemm_uni_nn(a, w)
+    ref = a.float() @ w.float()
+    assert_close(c, ref, 0.03, f"gemm_uni_nn {m}x{n}x{k}")
+
+
+@pytest.mark.parametrize("mt,n,k,splitr", [
+    (16320, 32768, 512, 1),   # wide (logits-like) dW, aligned vocab
+    (16320, 2048, 512, 1),    # mid dW
+    (16384, 512, 512, 8),     # narrow deep dW, split contraction
+    (16384, 1536, 512, 4),    # QKV dW, split contraction
+])
+def test_gemm_uni_tn(mt, n, k, splitr):
+    """dW: C[N,K] = dY[Mt,N]^T @ X[Mt,K], both operands tr16-read."""
+    torch.manual_seed(5)
+    dy = torch.randn(mt, n, device="cuda", dtype=torch.bfloat16) * 0.1
+    x = torch.randn(mt, k, device="cuda", dtype=torch.bfloat16) * 0.1
+    c = _ext().gemm_uni_tn(dy, x, None, splitr)
+    ref = dy.float().T @ x.float()
+    assert_close(c, ref, 0.03, f"gemm_uni_tn {mt}x{n}x{k} s{splitr}")
+
+
+def test_gemm_uni_tn_padded_vocab():
+    """The ragged-vocab dW path: dY is a (M, V) view of a 256-padded
+    buffer with zero pad columns (what ce_bwd emits)."""
+    torch.manual_seed(6)
+    mt, v, k = 4096, 32770, 512
+    vp = (v + 255) // 256 * 256
+    full = torch.zeros(mt, vp, device="cuda", dtype=torch.bfloat16)
+    full[:, :v].normal_().mul_(0.1)
+    dy = full[:, :v]
+    x = torch.randn(mt, k, device="cuda", dtype=torch.bfloat16) * 0.1
+    c = _ext().gemm_uni_tn(dy, x, None, 1)
+    ref = dy.float().T @ x.float()
+    assert_close(c, ref, 0.03, "gemm_uni_tn padded vocab")
+
+
+def test_ce_bwd_pad_columns_zero():
+    """ce_bwd returns (R, Vp) with Vp = roundup(V, 256); pad columns must
+    be exactly zero (downstream GEMMs contract over Vp), and the [:, :V]
+    grad must match the fp32 reference."""
+    torch.manual_seed(7)
+    R, V = 64, 1000
+    vp = (V + 255) // 256 * 256
+    logits = torch.randn(R, V, device="cuda", dtype=torch.bfloat16)
+    tgt = torch.randint(1, V, (R,), device="cuda")
+    tgt[::7] = 0  # some padded positions
+    loss, lse = _ext().ce_fwd(logits, tgt, 4.0, 0.1)
+    dfull = _ext().ce_bwd(logits, tgt, lse,
+                          torch.ones(1, device="cuda"), 4.0, 0.1)
+    assert dfull.shape == (R, vp)
+    assert (dfull[:, V:].float() == 0).all(), "pad columns not zero"
+    # fp32 reference grad
+    lf = logits.detach().float().requires_grad_(True)
+    from transformer_amd.ops import reference as Rf
+    ref_loss = Rf.masked_cross_entropy(lf.view(1, R, V), tgt.view(1, R),
+                                       4, 0.1)
+    ref_loss.backward()
+    assert_close(dfull[:, :V], lf.grad, 0.05, "ce pad grad")
+
+
+def test_dx_gemm_padded_logits_path():
+    """_dx_gemm on a padded-strided dY must match the dense reference
+    (exercises _wt_padded + the NT padded contraction)."""
+    from transformer_amd.ops import functional as F
+    torch.manual_seed(8)
+    M, V, d = 4096, 32770, 512
+    vp = (V + 255) // 256 * 256
+    full = torch.zeros(M, vp, device="cuda", dtype=torch.bfloat16)
+    full[:, :V].normal_().mul_(0.05)
+    dy = full[:, :V]
+    w = torch.randn(V, d, device="cuda", dtype=torch.bfloat16) * 0.05
+    F.bump_weight_version()
+    dx = F._dx_gemm(_ext(), dy, w)
+    ref = dy.float() @ w.float()
+    assert_close(dx, ref, 0.05, "padded dx")
+    # cache refresh: after a weight update + version bump, result follows
+    with torch.no_grad():
+        w.mul_(2.0)
+    F.bump_weight_version()
+    dx2 = F._dx_gemm(_ext(), dy, w)
+    assert_close(dx2, ref * 2.0, 0.05, "padded dx after update")

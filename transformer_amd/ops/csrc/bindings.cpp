@@ -21,7 +21,16 @@ torch::Tensor gemm_nn(torch::Tensor a, torch::Tensor b,
 torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b,
                       c10::optional<torch::Tensor> out);
 torch::Tensor transpose2d(torch::Tensor a);
+void transpose2d_into(torch::Tensor a, torch::Tensor out);
 torch::Tensor colsum(torch::Tensor a, c10::optional<torch::Tensor> out);
+torch::Tensor gemm_uni_nt(torch::Tensor a, torch::Tensor w,
+                          c10::optional<torch::Tensor> bias, int64_t epilogue,
+                          c10::optional<torch::Tensor> out);
+torch::Tensor gemm_uni_nn(torch::Tensor a, torch::Tensor w,
+                          c10::optional<torch::Tensor> out);
+torch::Tensor gemm_uni_tn(torch::Tensor dy, torch::Tensor x,
+                          c10::optional<torch::Tensor> out, int64_t splitr);
+bool gemm_uni_viable(int M, int N, int K);
 torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y);
 torch::Tensor smoke_add(torch::Tensor a, torch::Tensor b);
 
@@ -100,6 +109,22 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("a"), pybind11::arg("b"),
         pybind11::arg("out") = pybind11::none());
   m.def("transpose2d", &transpose2d);
+  m.def("transpose2d_into", &transpose2d_into,
+        "transpose a into out (out rows may be padded past a's row count)");
+  m.def("gemm_uni_nt", &gemm_uni_nt,
+        "deep-pipelined 256-tile NT fwd GEMM (+bias/ReLU)",
+        pybind11::arg("a"), pybind11::arg("w"),
+        pybind11::arg("bias") = c10::nullopt, pybind11::arg("epilogue") = 0,
+        pybind11::arg("out") = c10::nullopt);
+  m.def("gemm_uni_nn", &gemm_uni_nn,
+        "dX = dY @ W, W read red-major via tr16 (no weight transpose)",
+        pybind11::arg("a"), pybind11::arg("w"),
+        pybind11::arg("out") = c10::nullopt);
+  m.def("gemm_uni_tn", &gemm_uni_tn,
+        "dW = dY^T @ X, both operands tr16-read; optional split contraction",
+        pybind11::arg("dy"), pybind11::arg("x"),
+        pybind11::arg("out") = c10::nullopt, pybind11::arg("splitr") = 1);
+  m.def("gemm_uni_viable", &gemm_uni_viable);
   m.def("colsum", &colsum, pybind11::arg("a"),
         pybind11::arg("out") = pybind11::none());
   m.def("relu_bwd", &relu_bwd);

@@ -115,28 +115,27 @@ void ce_fwd_kernel(const short* __restrict__ logits,
 
 // dlogits[j] = dloss/batch * mask_row * (softmax_j - (1-eps)*onehot_j - eps/V)
 // dloss arrives as a device scalar so backward never synchronizes the host.
+// dlogits rows carry leading dim Vp (V rounded up to 256 — the GEMM tile
+// width, so TR staging of dY never reads past the allocation), with the
+// pad columns written as ZERO — the downstream dX / dW GEMMs treat the
+// contraction as Vp-long with no ragged-K handling (gemm_uni.hip).
 __global__ __launch_bounds__(256)
 void ce_bwd_kernel(const short* __restrict__ logits,
                    const long* __restrict__ targets,
                    const float* __restrict__ lse,
                    const float* __restrict__ dloss_dev,
                    short* __restrict__ dlogits,
-                   long R, int V, float inv_batch, float eps_ls) {
+                   long R, int V, int Vp, float inv_batch, float eps_ls) {
   const long row = blockIdx.x;
   if (row >= R) return;
   const float scale = dloss_dev[0] * inv_batch;
   const short* lrow = logits + row * V;
-  short* drow = dlogits + row * V;
+  short* drow = dlogits + row * Vp;
   const long tgt = targets[row];
   const int t = threadIdx.x;
-  if (tgt == 0) {  // padded position contributes no gradient
-    for (int c = t * 8; c < V; c += 256 * 8) {
-      if (c + 8 <= V) {
-        *(s16x8*)(drow + c) = (s16x8){0, 0, 0, 0, 0, 0, 0, 0};
-      } else {
-        for (int j = 0; c + j < V; ++j) drow[c + j] = 0;
-      }
-    }
+  if (tgt == 0) {  // padded position contributes no gradient (Vp % 8 == 0)
+    for (int c = t * 8; c < Vp; c += 256 * 8)
+      *(s16x8*)(drow + c) = (s16x8){0, 0, 0, 0, 0, 0, 0, 0};
     return;
   }
   const float l = lse[row];
@@ -160,6 +159,7 @@ void ce_bwd_kernel(const short* __restrict__ logits,
       }
     }
   }
+  for (int c = V + t; c < Vp; c += 256) drow[c] = 0;  // zero the pad cols
 }
 
 // one wave: sum the 256 banks into the loss scalar and re-zero them for
@@ -201,19 +201,23 @@ std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets,
   return {loss, lse};
 }
 
+// Returns (R, Vp) with Vp = V rounded up to 256 and zero pad columns;
+// callers slice [:, :V] for autograd and may use the full padded tensor
+// (contraction Vp) in the logits dX / dW GEMMs.
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets,
                      torch::Tensor lse, torch::Tensor dloss,
                      double batch_size, double label_smoothing) {
   const long R = logits.size(0);
   const int V = logits.size(1);
+  const int Vp = (V + 255) / 256 * 256;
   TORCH_CHECK(dloss.is_cuda() && dloss.dtype() == torch::kFloat32 &&
               dloss.numel() == 1, "dloss must be a device fp32 scalar");
-  auto dlogits = torch::empty_like(logits);
+  auto dlogits = torch::empty({R, Vp}, logits.options());
   auto stream = at::hip::getCurrentHIPStream();
   ce_bwd_kernel<<<R, 256, 0, stream>>>(
       (const short*)logits.data_ptr(), targets.data_ptr<long>(),
       lse.data_ptr<float>(), dloss.data_ptr<float>(),
-      (short*)dlogits.data_ptr(), R, V, (float)(1.0 / batch_size),
+      (short*)dlogits.data_ptr(), R, V, Vp, (float)(1.0 / batch_size),
       (float)label_smoothing);
   return dlogits;
 }

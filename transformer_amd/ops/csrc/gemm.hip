@@ -314,7 +314,7 @@ __global__ void gemm_finalize_kernel(const float* __restrict__ cw,
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256)
 void transpose2d_kernel(const short* __restrict__ in, short* __restrict__ out,
-                        int M, int N) {
+                        int M, int N, long ldo) {
   __shared__ short tile[64][64 + 8];  // +8 elems (16 B) row pad vs conflicts
   int tb = blockIdx.x, nbx = (N + 63) >> 6;
   int bm = (tb / nbx) << 6, bn = (tb % nbx) << 6;
@@ -336,7 +336,7 @@ void transpose2d_kernel(const short* __restrict__ in, short* __restrict__ out,
       s16x8 v;
 #pragma unroll
       for (int j = 0; j < 8; ++j) v[j] = tile[c8 + j][r];
-      *(s16x8*)(out + (long)(bn + r) * M + bm + c8) = v;
+      *(s16x8*)(out + (bn + r) * ldo + bm + c8) = v;
     }
   } else {
     for (int p = 0; p < 16; ++p) {
@@ -350,7 +350,7 @@ void transpose2d_kernel(const short* __restrict__ in, short* __restrict__ out,
       int idx = p * 256 + t;
       int r = idx >> 6, c = idx & 63;
       int gr = bn + r, gc = bm + c;
-      if (gr < N && gc < M) out[(long)gr * M + gc] = tile[c][r];
+      if (gr < N && gc < M) out[gr * ldo + gc] = tile[c][r];
     }
   }
 }
@@ -365,7 +365,7 @@ void transpose2d_kernel(const short* __restrict__ in, short* __restrict__ out,
 #define COLSUM_ROWS 128
 __global__ __launch_bounds__(256)
 void colsum_part_kernel(const short* __restrict__ dy, float* __restrict__ acc,
-                        int M, int N) {
+                        int M, int N, long lda) {
   int n = blockIdx.x * 256 + threadIdx.x;
   if (n >= N) return;
   long m0 = (long)blockIdx.y * COLSUM_ROWS;
@@ -374,12 +374,12 @@ void colsum_part_kernel(const short* __restrict__ dy, float* __restrict__ acc,
   float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
   long m = m0;
   for (; m + 4 <= m1; m += 4) {
-    s0 += bfbits2f(dy[m * N + n]);
-    s1 += bfbits2f(dy[(m + 1) * N + n]);
-    s2 += bfbits2f(dy[(m + 2) * N + n]);
-    s3 += bfbits2f(dy[(m + 3) * N + n]);
+    s0 += bfbits2f(dy[m * lda + n]);
+    s1 += bfbits2f(dy[(m + 1) * lda + n]);
+    s2 += bfbits2f(dy[(m + 2) * lda + n]);
+    s3 += bfbits2f(dy[(m + 3) * lda + n]);
   }
-  for (; m < m1; ++m) s0 += bfbits2f(dy[m * N + n]);
+  for (; m < m1; ++m) s0 += bfbits2f(dy[m * lda + n]);
   atomicAdd(&acc[n], (s0 + s1) + (s2 + s3));
 }
 
@@ -560,13 +560,31 @@ torch::Tensor transpose2d(torch::Tensor a) {
   int nb = cdiv(M, 64) * cdiv(N, 64);
   auto stream = at::hip::getCurrentHIPStream();
   transpose2d_kernel<<<nb, 256, 0, stream>>>(
-      (const short*)a.data_ptr(), (short*)out.data_ptr(), M, N);
+      (const short*)a.data_ptr(), (short*)out.data_ptr(), M, N, M);
   return out;
 }
 
-torch::Tensor colsum(torch::Tensor a, c10::optional<torch::Tensor> out_opt) {
+// Transpose into a caller buffer whose rows may be longer than M (pad
+// columns untouched) — used for the padded-transposed logits weight that
+// feeds the NT dX path when the vocab is not a multiple of 64.
+void transpose2d_into(torch::Tensor a, torch::Tensor out) {
   CHECK_BF16_2D(a);
   int M = a.size(0), N = a.size(1);
+  TORCH_CHECK(out.is_cuda() && out.dtype() == torch::kBFloat16 &&
+              out.dim() == 2 && out.stride(1) == 1 && out.size(0) == N &&
+              out.stride(0) >= M, "transpose2d_into: bad out");
+  int nb = cdiv(M, 64) * cdiv(N, 64);
+  auto stream = at::hip::getCurrentHIPStream();
+  transpose2d_kernel<<<nb, 256, 0, stream>>>(
+      (const short*)a.data_ptr(), (short*)out.data_ptr(), M, N,
+      (long)out.stride(0));
+}
+
+torch::Tensor colsum(torch::Tensor a, c10::optional<torch::Tensor> out_opt) {
+  TORCH_CHECK(a.is_cuda() && a.dtype() == torch::kBFloat16 && a.dim() == 2 &&
+              a.stride(1) == 1, "colsum: bad a");
+  int M = a.size(0), N = a.size(1);
+  const long lda = a.stride(0);
   static std::map<std::pair<int, int>, torch::Tensor> ws_cache;
   auto key = std::make_pair((int)a.get_device(), N);
   auto it = ws_cache.find(key);
@@ -585,7 +603,7 @@ torch::Tensor colsum(torch::Tensor a, c10::optional<torch::Tensor> out_opt) {
   auto stream = at::hip::getCurrentHIPStream();
   dim3 grid(cdiv(N, 256), cdiv(M, COLSUM_ROWS));
   colsum_part_kernel<<<grid, 256, 0, stream>>>(
-      (const short*)a.data_ptr(), acc.data_ptr<float>(), M, N);
+      (const short*)a.data_ptr(), acc.data_ptr<float>(), M, N, lda);
   cast_colsum_kernel<<<cdiv(N, 256), 256, 0, stream>>>(
       acc.data_ptr<float>(), (short*)out.data_ptr(), N);
   return out;

@@ -67,13 +67,43 @@ def _flat(p):
 # ---------------------------------------------------------------------------
 
 
-# Forward GEMM backend (measured, tools/head_gemm_bench.py on MI355X):
-# hipBLASLt's fused-bias/ReLU epilogues (F.linear / torch._addmm_activation)
-# beat the hand-written NT path at every model forward shape (qkv 515->617
-# TF, ffn2 733->1034, logits 611->716; the original fwd A/B had only
-# compared matmul + a separate bias pass, which loses).  TFMX_FWD_GEMM=hip
-# forces the hand-written kernel (A/B + fallback).
-_FWD_HIP = os.environ.get("TFMX_FWD_GEMM", "") == "hip"
+# GEMM backend (round 2): every training-shape GEMM runs hand-written —
+# the deep-pipelined gemm_uni kernels (counted-vmcnt schedule + tr16
+# transpose-read operand modes, ops/csrc/gemm_uni.hip) carry fwd, dX and
+# the wide dW; hipBLASLt remains only as the TFMX_FWD_GEMM=blaslt A/B arm.
+# TFMX_FWD_GEMM=hip forces the older 128/256-tile NT path.
+_FWD_BACKEND = os.environ.get("TFMX_FWD_GEMM", "uni")
+_FWD_HIP = _FWD_BACKEND == "hip"
+
+
+# Version counter for weight-derived caches (the padded-transposed logits
+# weight): bumped by NoamAdam.step so caches refresh once per step.
+_WEIGHT_VERSION = [0]
+
+
+def bump_weight_version():
+    _WEIGHT_VERSION[0] += 1
+
+
+_WTP_CACHE: dict = {}  # id(weight) -> [version, padded-transposed copy]
+
+
+def _wt_padded(E, w):
+    """W[N,K]^T into a (K, Np) buffer, Np = N rounded up to 64, pad columns
+    zero — the NT B-operand for the ragged-vocab dX GEMM.  Refreshed once
+    per optimizer step (bump_weight_version); under graph capture the
+    transpose is launched unconditionally so replays recompute it."""
+    key = id(w)
+    ent = _WTP_CACHE.get(key)
+    npad = (w.shape[0] + 255) // 256 * 256
+    if ent is None or ent[1].shape[1] != npad:
+        buf = torch.zeros(w.shape[1], npad, device=w.device, dtype=w.dtype)
+        ent = [None, buf]
+        _WTP_CACHE[key] = ent
+    if ent[0] != _WEIGHT_VERSION[0] or _GRAPH_SEED_T is not None:
+        E.transpose2d_into(w, ent[1])
+        ent[0] = _WEIGHT_VERSION[0]
+    return ent[1]
 
 
 _CAPTURE_HINT = False
@@ -89,33 +119,65 @@ def set_capture_hint(on: bool):
 
 
 def _fwd_gemm(E, x, w, b, activation):
-    # small-M (decode/serving) rows: the hand-written kernel's launch is
-    # cheaper than hipBLASLt's (eager B=1 decode measured 1.8 vs 2.4
-    # ms/token through F.linear); training shapes (M ~ 16k) go library.
-    if _FWD_HIP or b is None or (x.shape[0] <= 1024 and not _CAPTURE_HINT):
-        return E.gemm_nt(x, w, b if b is not None else torch.Tensor(),
-                         1 if activation == "relu" else 0)
-    if activation == "relu":
-        if hasattr(torch, "_addmm_activation"):
-            return torch._addmm_activation(b, x, w.t())
-        return torch.relu(torch.nn.functional.linear(x, w, b))
-    return torch.nn.functional.linear(x, w, b)
+    epi = 1 if activation == "relu" else 0
+    if _FWD_BACKEND == "blaslt" and b is not None and \
+            (x.shape[0] > 1024 or _CAPTURE_HINT):
+        # A/B arm: hipBLASLt fused bias/ReLU epilogues (round-1 default)
+        if activation == "relu":
+            if hasattr(torch, "_addmm_activation"):
+                return torch._addmm_activation(b, x, w.t())
+            return torch.relu(torch.nn.functional.linear(x, w, b))
+        return torch.nn.functional.linear(x, w, b)
+    if not _FWD_HIP and E.gemm_uni_viable(x.shape[0], w.shape[0],
+                                          x.shape[1]):
+        return E.gemm_uni_nt(x, w, b, epi)
+    # small-M (decode/serving) and non-viable shapes: 128/256-tile path,
+    # whose launch is cheaper at M<=1024 (1.8 vs 2.4 ms/token at B=1)
+    return E.gemm_nt(x, w, b if b is not None else torch.Tensor(), epi)
 
 def _dw_gemm(dy, x, out=None):
-    """dW[N,K] = dY^T @ X — plain GEMM; backend measured per shape on
-    MI355X (tools/gemm_bench.py): hipBLASLt TN wins at the wide logits
-    head (N>=8k: 2x), the hand-written transpose2d+NT path wins at the
-    d_model-sized heads."""
+    """dW[N,K] = dY^T @ X — plain GEMM, hand-written both ways: the wide
+    logits head goes to the deep-pipelined TRxTR gemm_uni_tn (both
+    operands tr16 transpose-read), the d_model-sized heads to the
+    split-contraction gemm_dw kernel (tools/gemm_bench.py)."""
     E = ext()
-    if dy.shape[1] >= 8192:
-        if out is not None:
-            return torch.matmul(dy.t(), x, out=out)
-        return torch.matmul(dy.t(), x)
+    npad = (dy.shape[1] + 255) // 256 * 256
+    if dy.shape[1] >= 8192 and dy.shape[0] % 64 == 0 \
+            and (dy.shape[1] % 256 == 0 or dy.stride(0) >= npad) \
+            and x.shape[1] % 128 == 0:
+        return E.gemm_uni_tn(dy, x, out)
     # gemm_dw writes DISJOINT per-slice fp32 partials (no atomics, no
     # zeroing — the fp32-atomic epilogue measured at the chip's atomic
     # rate); it sizes its own torch::empty workspace via the caching
     # allocator, so no Python-side workspace is needed.
-    return E.gemm_dw(dy, x, out, None, None)
+    return E.gemm_dw(dy.contiguous(), x, out, None, None)
+
+
+def _dx_gemm(E, dy, w):
+    """dX[M,K] = dY[M,N] @ W[N,K] — hand-written both ways: NTxTR (W read
+    red-major via tr16, no transpose) when N is 64-aligned; for the ragged
+    vocab head, dY arrives as ce_bwd's zero-padded row-strided view and is
+    contracted against the padded-transposed weight through the NT kernel."""
+    N = w.shape[0]
+    M = dy.shape[0]
+    if N % 64 == 0 and E.gemm_uni_viable(M, w.shape[1], N):
+        return E.gemm_uni_nn(dy, w)
+    npad = (N + 255) // 256 * 256
+    if npad != N and dy.stride(1) == 1 and dy.stride(0) == npad \
+            and dy.storage_offset() == 0 \
+            and E.gemm_uni_viable(M, w.shape[1], npad):
+        # ce_bwd wrote the pad columns as zeros; widen the view back to
+        # the padded buffer so the contraction is 64-aligned
+        full = dy.as_strided((M, npad), (npad, 1))
+        return E.gemm_uni_nt(full, _wt_padded(E, w), None, 0)
+    return torch.matmul(dy, w)
+
+
+def _dense2d(t):
+    """Row-strided dense (stride(1)==1) passes through — ce_bwd's padded
+    dlogits view must NOT be copied back to contiguous."""
+    return t if (t.stride(1) == 1 and t.stride(0) >= t.shape[1]) \
+        else t.contiguous()
 
 
 def _dw_db_gemm(dy, x, has_bias, w_out=None, b_out=None):
@@ -144,14 +206,11 @@ class _LinearFn(torch.autograd.Function):
     def backward(ctx, dy):
         E = ext()
         x, w, y = ctx.saved_tensors
-        dy = dy.contiguous()
         if ctx.activation == "relu":
-            dy = E.relu_bwd(dy, y)  # dz = dy * (y > 0)
-        # The backward GEMMs carry no epilogue, so they go to hipBLASLt
-        # (torch.matmul), which handles the NN/TN layouts natively — no
-        # physical transposes, and measured faster than both hand-written
-        # paths at every model shape (tools/gemm_bench.py on MI355X).
-        dx = torch.matmul(dy, w)        # dX[M,K] = dY[M,N] @ W[N,K]
+            dy = E.relu_bwd(dy.contiguous(), y)  # dz = dy * (y > 0)
+        else:
+            dy = _dense2d(dy)
+        dx = _dx_gemm(E, dy, w)         # dX[M,K] = dY[M,N] @ W[N,K]
         dw, db = _dw_db_gemm(dy, x, ctx.has_bias)
         return dx, dw, db, None
 
@@ -178,12 +237,11 @@ class _LinearFlatFn(torch.autograd.Function):
         E = ext()
         x, y = ctx.saved_tensors
         w, b = ctx.wb
-        dy = dy.contiguous()
         if ctx.activation == "relu":
-            dy = E.relu_bwd(dy, y)
-        # Plain GEMMs -> hipBLASLt, straight into the flat-grad view
-        # (see _LinearFn.backward for the measurement rationale).
-        dx = torch.matmul(dy, w)
+            dy = E.relu_bwd(dy.contiguous(), y)
+        else:
+            dy = _dense2d(dy)
+        dx = _dx_gemm(E, dy, w)
         _dw_db_gemm(dy, x, b is not None,
                     w_out=_flat(w).view(w.shape[0], -1),
                     b_out=_flat(b).view(-1) if b is not None else None)
@@ -612,9 +670,14 @@ class _CrossEntropyFn(torch.autograd.Function):
     def backward(ctx, dloss):
         E = ext()
         logits, targets, lse = ctx.saved_tensors
-        dlogits = E.ce_bwd(logits, targets, lse,
-                           dloss.to(torch.float32).reshape(1).contiguous(),
-                           ctx.batch_size, ctx.ls)
+        # ce_bwd returns (R, Vp) with Vp = V rounded up to 64 and ZERO pad
+        # columns; the [:, :V] view keeps the padded stride so the logits
+        # dX/dW GEMMs can contract over the 64-aligned Vp (gemm_uni.hip).
+        dfull = E.ce_bwd(logits, targets, lse,
+                         dloss.to(torch.float32).reshape(1).contiguous(),
+                         ctx.batch_size, ctx.ls)
+        V = logits.shape[1]
+        dlogits = dfull if dfull.shape[1] == V else dfull[:, :V]
         return dlogits, None, None, None
 
 

@@ -98,4 +98,6 @@ class CapturedTrainStep:
                                                          non_blocking=True)
         self.graph.replay()
         self.opt.step_count += 1
+        from ..ops import functional as _F
+        _F.bump_weight_version()  # replay updated weights in-graph
         return self.loss

@@ -156,6 +156,8 @@ class NoamAdam:
 
     @torch.no_grad()
     def step(self):
+        from ..ops import functional as _F
+        _F.bump_weight_version()  # invalidate weight-derived caches
         self.step_count += 1
         lr = self.schedule(self.step_count)
         b1, b2 = self.betas
