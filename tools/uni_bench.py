@@ -74,11 +74,40 @@ DW = [
 ]
 
 
+SCHED_NAMES = {0: "cnt+prio", 1: "cnt", 2: "r1-style", 3: "cnt+stat",
+               4: "cnt+Bq1"}
+
+
+def sched_ab(iters):
+    print("== NT schedule A/B (gemm_uni_nt_ab) ==")
+    dt = torch.bfloat16
+    for M, N, K, _, tag in [FWD[0], FWD[3], FWD[4], FWD[6]]:
+        a = torch.randn(M, K, device="cuda", dtype=dt)
+        w = torch.randn(N, K, device="cuda", dtype=dt) * 0.05
+        scheds = [0, 1, 2, 3, 4]
+        fns = [lambda s=s: E.gemm_uni_nt_ab(a, w, s) for s in scheds]
+        fns.append(lambda: E.gemm_nt(a, w, torch.Tensor(), 0))
+        r = bench_pair(fns, iters)
+        parts = " | ".join(f"{SCHED_NAMES[s]} {tf(M,N,K,r[i]):5.0f}TF"
+                           for i, s in enumerate(scheds))
+        print(f"{tag:18s} {parts} | r1-nt {tf(M,N,K,r[-1]):5.0f}TF")
+        # numerics sanity on one round
+        ref = (a.float() @ w.float().T)
+        for s in scheds:
+            c = E.gemm_uni_nt_ab(a, w, s).float()
+            err = (c - ref).abs().max() / ref.abs().max().clamp(min=1)
+            assert err < 0.03, (tag, s, err)
+
+
 def main():
     iters = int(sys.argv[1]) if len(sys.argv) > 1 else 30
     dev = "cuda"
     dt = torch.bfloat16
     torch.manual_seed(0)
+
+    if len(sys.argv) > 2 and sys.argv[2] == "sched":
+        sched_ab(iters)
+        return
 
     print("== forward: uni vs r1 gemm_nt vs hipBLASLt ==")
     for M, N, K, epi, tag in FWD:

@@ -31,6 +31,8 @@ torch::Tensor gemm_uni_nn(torch::Tensor a, torch::Tensor w,
 torch::Tensor gemm_uni_tn(torch::Tensor dy, torch::Tensor x,
                           c10::optional<torch::Tensor> out, int64_t splitr);
 bool gemm_uni_viable(int M, int N, int K);
+torch::Tensor gemm_uni_nt_ab(torch::Tensor a, torch::Tensor w,
+                             int64_t sched);
 torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y);
 torch::Tensor smoke_add(torch::Tensor a, torch::Tensor b);
 
@@ -125,6 +127,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("dy"), pybind11::arg("x"),
         pybind11::arg("out") = c10::nullopt, pybind11::arg("splitr") = 1);
   m.def("gemm_uni_viable", &gemm_uni_viable);
+  m.def("gemm_uni_nt_ab", &gemm_uni_nt_ab,
+        "schedule A/B variants of the uni NT kernel (bench only)");
   m.def("colsum", &colsum, pybind11::arg("a"),
         pybind11::arg("out") = pybind11::none());
   m.def("relu_bwd", &relu_bwd);

@@ -152,7 +152,15 @@ DEV_INLINE bf16x8u u_frag_tr(const short* lds, int ms, int colb) {
 // 1 ReLU.  SPLITR: >0 = red axis sliced across blockIdx.y, fp32 partials.
 // ---------------------------------------------------------------------------
 
-template <int EPILOGUE, int BN_, bool TRA, bool TRB, bool SPLITR = false>
+// SCHED variants (A/B; tools/uni_bench.py — the default is the measured
+// winner):
+//   0 counted waits + per-quad s_setprio
+//   1 counted waits, no setprio
+//   2 r1-style: all A staged q0, B q1, vmcnt(0) at end-q3, no setprio
+//   3 counted waits + static young-half priority (T5 static form)
+//   4 counted waits, no setprio, B issued at q1 (balanced bursts)
+template <int EPILOGUE, int BN_, bool TRA, bool TRB, bool SPLITR = false,
+          int SCHED = 1>
 __global__ __launch_bounds__(U_THREADS, 1)
 void gemm_uni_kernel(const short* __restrict__ A, const short* __restrict__ B,
                      const short* __restrict__ bias, short* __restrict__ C,
@@ -232,6 +240,9 @@ void gemm_uni_kernel(const short* __restrict__ A, const short* __restrict__ B,
   U_WAIT_VM(0);
   U_BARRIER();
 
+  if (SCHED == 3 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);  // static young-half priority
+
   bf16x8u afr[MF / 2][2], bfr[NF][2];
 
   auto lda_frag = [&](int i, int ks, int rh, const short* a_lds) {
@@ -246,7 +257,7 @@ void gemm_uni_kernel(const short* __restrict__ A, const short* __restrict__ B,
   };
 
 #define U_MFMA_QUAD(RH, CH)                                                \
-  __builtin_amdgcn_s_setprio(1);                                          \
+  if (SCHED == 0) __builtin_amdgcn_s_setprio(1);                          \
   _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                         \
     _Pragma("unroll") for (int i = 0; i < MF / 2; ++i)                     \
       _Pragma("unroll") for (int j = 0; j < NF / 2; ++j)                   \
@@ -254,7 +265,7 @@ void gemm_uni_kernel(const short* __restrict__ A, const short* __restrict__ B,
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(                       \
                 afr[i][ks], bfr[(CH) * (NF / 2) + j][ks],                  \
                 acc[(RH) * (MF / 2) + i][(CH) * (NF / 2) + j], 0, 0, 0);   \
-  __builtin_amdgcn_s_setprio(0);
+  if (SCHED == 0) __builtin_amdgcn_s_setprio(0);
 
   for (int t = 0; t < ntiles; ++t) {
     const short* a_lds = smem + (t & 1) * SLOT;
@@ -274,7 +285,8 @@ void gemm_uni_kernel(const short* __restrict__ A, const short* __restrict__ B,
       for (int ks = 0; ks < 2; ++ks) bfr[j][ks] = ldb_frag(j, ks, b_lds);
     if (do_pf) {
       stage_a(t + 1, 0, pa_lds);
-      stage_b(t + 1, pb_lds);
+      if (SCHED != 4) stage_b(t + 1, pb_lds);
+      if (SCHED == 2) stage_a(t + 1, 1, pa_lds);
     }
     U_BARRIER();
     U_WAIT_LGKM0();
@@ -286,17 +298,22 @@ void gemm_uni_kernel(const short* __restrict__ A, const short* __restrict__ B,
     for (int j = NF / 2; j < NF; ++j)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) bfr[j][ks] = ldb_frag(j, ks, b_lds);
-    if (do_pf) stage_a(t + 1, 1, pa_lds);
+    if (do_pf) {
+      if (SCHED == 4) stage_b(t + 1, pb_lds);
+      if (SCHED != 2) stage_a(t + 1, 1, pa_lds);
+    }
     U_BARRIER();
     U_WAIT_LGKM0();
     U_MFMA_QUAD(0, 1)
     // end-q1 wait: drain THIS tile's A-piece1 (issued q1 of t-1, 4 phases
     // ago, consumed by q2's reads one barrier from here); t+1's
     // A-piece0+B (2+NB) + A-piece1 (2) stay in flight.
-    if (do_pf) {
-      if (BN_ == 256) U_WAIT_VM(8); else U_WAIT_VM(6);
-    } else {
-      U_WAIT_VM(0);
+    if (SCHED != 2) {
+      if (do_pf) {
+        if (BN_ == 256) U_WAIT_VM(8); else U_WAIT_VM(6);
+      } else {
+        U_WAIT_VM(0);
+      }
     }
     U_BARRIER();
 
@@ -314,10 +331,10 @@ void gemm_uni_kernel(const short* __restrict__ A, const short* __restrict__ B,
     U_MFMA_QUAD(1, 1)
     // end-q3 wait: drain t+1's A-piece0+B (issued q0, 3 phases ago);
     // its A-piece1 (2 glds) stays in flight across the tile boundary.
-    if (do_pf) {
-      U_WAIT_VM(2);
-    } else {
+    if (SCHED == 2 || !do_pf) {
       U_WAIT_VM(0);
+    } else {
+      U_WAIT_VM(2);
     }
     U_BARRIER();
   }
@@ -463,6 +480,40 @@ static torch::Tensor gemm_uni_launch(
     long mn = (long)M * N;
     uni_reduce_kernel<<<(mn + 255) / 256, 256, 0, stream>>>(
         cwp, (short*)c.data_ptr(), mn, nslices);
+  }
+  return c;
+}
+
+// Bench-only schedule A/B: NT x NT, no epilogue, BN=256 grid assumed.
+torch::Tensor gemm_uni_nt_ab(torch::Tensor a, torch::Tensor w,
+                             int64_t sched) {
+  const int M = a.size(0), K = a.size(1), N = w.size(0);
+  auto c = torch::empty({M, N}, a.options());
+  const int BN = ((long)cdiv(M, 256) * cdiv(N, 256) >= 224) ? 256 : 128;
+  const int nbm = cdiv(M, 256), nbn = cdiv(N, BN);
+  const size_t smem = 2 * ((size_t)256 + BN) * U_BK * sizeof(short);
+  auto stream = at::hip::getCurrentHIPStream();
+  auto go = [&](auto bnc, auto sc) {
+    constexpr int BNv = decltype(bnc)::value;
+    constexpr int SC = decltype(sc)::value;
+    auto kfn = gemm_uni_kernel<0, BNv, false, false, false, SC>;
+    (void)hipFuncSetAttribute((const void*)kfn,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)smem);
+    kfn<<<nbm * nbn, U_THREADS, smem, stream>>>(
+        (const short*)a.data_ptr(), (const short*)w.data_ptr(), nullptr,
+        (short*)c.data_ptr(), nullptr, M, N, K, K, K, 0, nbm, nbn, K);
+  };
+  auto go2 = [&](auto sc) {
+    if (BN == 256) go(std::integral_constant<int, 256>{}, sc);
+    else go(std::integral_constant<int, 128>{}, sc);
+  };
+  switch ((int)sched) {
+    case 0: go2(std::integral_constant<int, 0>{}); break;
+    case 2: go2(std::integral_constant<int, 2>{}); break;
+    case 3: go2(std::integral_constant<int, 3>{}); break;
+    case 4: go2(std::integral_constant<int, 4>{}); break;
+    default: go2(std::integral_constant<int, 1>{}); break;
   }
   return c;
 }
