@@ -126,14 +126,17 @@ def main():
               f"r1 {r[1]*1000:7.1f}us {tf(M,N,K,r[1]):6.0f}TF | "
               f"blaslt {r[2]*1000:7.1f}us {tf(M,N,K,r[2]):6.0f}TF")
 
-    print("== dX: uni_nn (NTxTR) vs torch.matmul ==")
+    print("== dX: uni_nn (NTxTR) vs wt+gemm_nt vs torch.matmul ==")
     for M, N, K, tag in DX:
         dy = torch.randn(M, N, device=dev, dtype=dt) * 0.05
         w = torch.randn(N, K, device=dev, dtype=dt) * 0.05
+        wt = w.t().contiguous()
         r = bench_pair([lambda: E.gemm_uni_nn(dy, w),
+                        lambda: E.gemm_nt(dy, wt, torch.Tensor(), 0),
                         lambda: torch.matmul(dy, w)], iters)
         print(f"{tag:22s} uni {r[0]*1000:7.1f}us {tf(M,N,K,r[0]):6.0f}TF | "
-              f"blaslt {r[1]*1000:7.1f}us {tf(M,N,K,r[1]):6.0f}TF")
+              f"wt+nt {r[1]*1000:7.1f}us {tf(M,N,K,r[1]):6.0f}TF | "
+              f"blaslt {r[2]*1000:7.1f}us {tf(M,N,K,r[2]):6.0f}TF")
 
     print("== dW: uni_tn (TRxTR) vs gemm_dw vs torch.matmul ==")
     for Mt, N, K, tag in DW:

@@ -15,20 +15,15 @@
 // (reference contract rows K1/K7/K8/K12: Attention.py:46-50, point_ffn.py:
 // 5-6, Transformer.py:16 — their backward GEMMs.)
 //
-// Schedule (guide T3+T4+T5, vs gemm256.hip's per-K-tile vmcnt(0) drain):
-// 4 quadrant phases per K-tile (BK=64), per-wave glds prefetch of tile t+1
-// issued in CONSUMPTION order —
-//   q0: issue A-piece0 (2 glds) + all of B (NB glds)   [consumed at q0]
-//   q1: issue A-piece1 (2 glds)                        [consumed at q2]
-// and only counted waits, placed before the barrier whose consumers need
-// the data:
-//   end-q1: s_waitcnt vmcnt(4+NB)  (drains t's A-piece1; t+1's 4+NB fly)
-//   end-q3: s_waitcnt vmcnt(2)     (drains t+1's A-piece0+B; A-piece1 flies)
-// so every load has >=4 phases (>1300 cycles > HBM latency) before its
-// wait, and no wave ever drains the queue inside the loop.  s_setprio(1)
-// wraps each MFMA quadrant (T5: the 8-phase role-split is what makes it
-// pay).  A-piece0 = the glds covering the rows/cols the quadrant-0 MFMAs
-// read (see the band derivation in the stage helpers).
+// Schedule: 4 quadrant phases per K-tile (BK=64), glds prefetch of tile
+// t+1 issued in the first two phases, one vmcnt(0) per K-tile at the
+// boundary (the gemm256.hip schedule).  A full counted-vmcnt /
+// consumption-ordered variant and per-quad / static s_setprio forms were
+// built and MEASURED SLOWER on MI355X at every model shape
+// (profiles/r2 sched A/B: counted 557-930 TF vs drain 600-1003; setprio
+// -2..-4% — T5 is null-to-negative on this lockstep phase structure, and
+// the per-tile drain already gives loads >=2.5 phases to land); they are
+// kept as SCHED template variants for the bench tool only.
 //
 // Constraints (dispatched shapes all qualify; host falls back otherwise):
 // K % 64 == 0, and TR-mode operands must be allocated out to the tile
@@ -160,7 +155,7 @@ DEV_INLINE bf16x8u u_frag_tr(const short* lds, int ms, int colb) {
 //   3 counted waits + static young-half priority (T5 static form)
 //   4 counted waits, no setprio, B issued at q1 (balanced bursts)
 template <int EPILOGUE, int BN_, bool TRA, bool TRB, bool SPLITR = false,
-          int SCHED = 1>
+          int SCHED = 2>
 __global__ __launch_bounds__(U_THREADS, 1)
 void gemm_uni_kernel(const short* __restrict__ A, const short* __restrict__ B,
                      const short* __restrict__ bias, short* __restrict__ C,

@@ -67,12 +67,13 @@ def _flat(p):
 # ---------------------------------------------------------------------------
 
 
-# GEMM backend (round 2): every training-shape GEMM runs hand-written —
-# the deep-pipelined gemm_uni kernels (counted-vmcnt schedule + tr16
-# transpose-read operand modes, ops/csrc/gemm_uni.hip) carry fwd, dX and
-# the wide dW; hipBLASLt remains only as the TFMX_FWD_GEMM=blaslt A/B arm.
-# TFMX_FWD_GEMM=hip forces the older 128/256-tile NT path.
-_FWD_BACKEND = os.environ.get("TFMX_FWD_GEMM", "uni")
+# GEMM backend (round 2): every training-shape GEMM runs hand-written.
+# Forward goes to the gemm256/128 NT dispatch (measured best hand path:
+# 600-1040 TF; the uni counted-vmcnt schedule variants measured slower —
+# see gemm_uni.hip header); dX and the wide dW go to the gemm_uni TR-mode
+# kernels (tr16 transpose-read operands).  hipBLASLt remains only as the
+# TFMX_FWD_GEMM=blaslt A/B arm; TFMX_FWD_GEMM=uni forces uni NT forward.
+_FWD_BACKEND = os.environ.get("TFMX_FWD_GEMM", "hand")
 _FWD_HIP = _FWD_BACKEND == "hip"
 
 
@@ -128,11 +129,11 @@ def _fwd_gemm(E, x, w, b, activation):
                 return torch._addmm_activation(b, x, w.t())
             return torch.relu(torch.nn.functional.linear(x, w, b))
         return torch.nn.functional.linear(x, w, b)
-    if not _FWD_HIP and E.gemm_uni_viable(x.shape[0], w.shape[0],
-                                          x.shape[1]):
+    if _FWD_BACKEND == "uni" and E.gemm_uni_viable(x.shape[0], w.shape[0],
+                                                   x.shape[1]):
         return E.gemm_uni_nt(x, w, b, epi)
-    # small-M (decode/serving) and non-viable shapes: 128/256-tile path,
-    # whose launch is cheaper at M<=1024 (1.8 vs 2.4 ms/token at B=1)
+    # default: gemm256/128 dispatch (best-measured hand-written NT path;
+    # launch is cheaper than the library's at small M, decode/serving)
     return E.gemm_nt(x, w, b if b is not None else torch.Tensor(), epi)
 
 def _dw_gemm(dy, x, out=None):
@@ -153,24 +154,33 @@ def _dw_gemm(dy, x, out=None):
     return E.gemm_dw(dy.contiguous(), x, out, None, None)
 
 
+# dX backend: "wt" contracts against a per-step cached transposed weight
+# through the fast NT path; "uni" reads W red-major via tr16 (no
+# transpose); "blaslt" is the library A/B arm.  Default = measured winner.
+_DX_BACKEND = os.environ.get("TFMX_DX", "wt")
+
+
 def _dx_gemm(E, dy, w):
-    """dX[M,K] = dY[M,N] @ W[N,K] — hand-written both ways: NTxTR (W read
-    red-major via tr16, no transpose) when N is 64-aligned; for the ragged
-    vocab head, dY arrives as ce_bwd's zero-padded row-strided view and is
-    contracted against the padded-transposed weight through the NT kernel."""
+    """dX[M,K] = dY[M,N] @ W[N,K] — hand-written: either NT against the
+    cached W^T (refreshed once per optimizer step) or NTxTR with W read
+    red-major via tr16.  The ragged-vocab head's dY arrives as ce_bwd's
+    zero-padded row-strided view; the contraction widens to the 64-aligned
+    pad (zero columns) against the padded-transposed weight."""
     N = w.shape[0]
     M = dy.shape[0]
+    if _DX_BACKEND == "blaslt":
+        return torch.matmul(dy, w)
+    npad = (N + 255) // 256 * 256
+    padded = (npad != N and dy.stride(1) == 1 and dy.stride(0) == npad
+              and dy.storage_offset() == 0)
+    if padded and E.gemm_uni_viable(M, w.shape[1], npad):
+        full = dy.as_strided((M, npad), (npad, 1))
+        return E.gemm_nt(full, _wt_padded(E, w), torch.Tensor(), 0)
+    if _DX_BACKEND == "wt" and N % 64 == 0 and M >= 2048:
+        return E.gemm_nt(_dense2d(dy), _wt_padded(E, w), torch.Tensor(), 0)
     if N % 64 == 0 and E.gemm_uni_viable(M, w.shape[1], N):
         return E.gemm_uni_nn(dy, w)
-    npad = (N + 255) // 256 * 256
-    if npad != N and dy.stride(1) == 1 and dy.stride(0) == npad \
-            and dy.storage_offset() == 0 \
-            and E.gemm_uni_viable(M, w.shape[1], npad):
-        # ce_bwd wrote the pad columns as zeros; widen the view back to
-        # the padded buffer so the contraction is 64-aligned
-        full = dy.as_strided((M, npad), (npad, 1))
-        return E.gemm_uni_nt(full, _wt_padded(E, w), None, 0)
-    return torch.matmul(dy, w)
+    return torch.matmul(dy, w)  # small/odd shapes (not a training shape)
 
 
 def _dense2d(t):
