@@ -659,3 +659,66 @@ def test_dx_gemm_padded_logits_path():
     F.bump_weight_version()
     dx2 = F._dx_gemm(_ext(), dy, w)
     assert_close(dx2, ref * 2.0, 0.05, "padded dx after update")
+
+
+def test_colsum_fused_finalize_repeated():
+    """colsum's single-launch last-arriver finalize: repeated L1-warm
+    calls on the same cached workspace must stay exact (stale-L1 reads by
+    the finalizing block are the failure mode — guide G16 pitfall 3)."""
+    torch.manual_seed(9)
+    for N in (512, 1536, 2048):
+        for it in range(20):
+            M = 1024 + 128 * (it % 5)
+            a = torch.randn(M, N, device="cuda", dtype=torch.bfloat16)
+            got = _ext().colsum(a)
+            ref = a.float().sum(dim=0)
+            assert_close(got, ref, 0.02, f"colsum N={N} it={it}")
+
+
+def test_ln_gb_fused_finalize_repeated():
+    """ln_bwd's dgamma/dbeta single-launch finalize under repeated calls."""
+    torch.manual_seed(10)
+    D = 512
+    gamma = torch.randn(D, device="cuda", dtype=torch.bfloat16)
+    beta = torch.randn(D, device="cuda", dtype=torch.bfloat16)
+    for it in range(15):
+        R = 2048 + 256 * (it % 3)
+        x = torch.randn(R, D, device="cuda", dtype=torch.bfloat16)
+        res = torch.randn(R, D, device="cuda", dtype=torch.bfloat16)
+        y, s, mean, rstd = _ext().ln_fwd(x, res, gamma, beta, 1e-6, 0.0, 0,
+                                         None)
+        dy = torch.randn(R, D, device="cuda", dtype=torch.bfloat16)
+        dx, dgamma, dbeta = _ext().ln_bwd(dy, s, gamma, mean, rstd, None,
+                                          None, None, 0.0)
+        sf = s.float()
+        xh = (sf - sf.mean(-1, keepdim=True)) \
+            * torch.rsqrt(sf.var(-1, unbiased=False, keepdim=True) + 1e-6)
+        ref_g = (dy.float() * xh).sum(0)
+        ref_b = dy.float().sum(0)
+        assert_close(dgamma, ref_g, 0.03, f"dgamma it={it}")
+        assert_close(dbeta, ref_b, 0.03, f"dbeta it={it}")
+
+
+def test_transpose_batch_descriptor_table():
+    """Batched weight transpose: several weights in one launch, padded
+    destinations, values match per-weight transpose."""
+    torch.manual_seed(11)
+    shapes = [(512, 512), (1536, 512), (1000, 128), (2048, 512)]
+    rows = []
+    pairs = []
+    for (m, n) in shapes:
+        w = torch.randn(m, n, device="cuda", dtype=torch.bfloat16)
+        npad = (m + 255) // 256 * 256
+        buf = torch.zeros(n, npad, device="cuda", dtype=torch.bfloat16)
+        pairs.append((w, buf))
+        for bm in range(0, m, 64):
+            for bn in range(0, n, 64):
+                rows.append([w.data_ptr(), buf.data_ptr(), m, n,
+                             buf.stride(0), bm, bn])
+    desc = torch.tensor(rows, dtype=torch.int64).cuda()
+    _ext().transpose_batch(desc)
+    torch.cuda.synchronize()
+    for w, buf in pairs:
+        m = w.shape[0]
+        assert torch.equal(buf[:, :m], w.t().contiguous()), w.shape
+        assert (buf[:, m:].float() == 0).all()

@@ -22,6 +22,7 @@ torch::Tensor gemm_tn(torch::Tensor a, torch::Tensor b,
                       c10::optional<torch::Tensor> out);
 torch::Tensor transpose2d(torch::Tensor a);
 void transpose2d_into(torch::Tensor a, torch::Tensor out);
+void transpose_batch(torch::Tensor desc);
 torch::Tensor colsum(torch::Tensor a, c10::optional<torch::Tensor> out);
 torch::Tensor gemm_uni_nt(torch::Tensor a, torch::Tensor w,
                           c10::optional<torch::Tensor> bias, int64_t epilogue,
@@ -113,6 +114,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("transpose2d", &transpose2d);
   m.def("transpose2d_into", &transpose2d_into,
         "transpose a into out (out rows may be padded past a's row count)");
+  m.def("transpose_batch", &transpose_batch,
+        "batched 64x64-tile transposes from an int64 descriptor table");
   m.def("gemm_uni_nt", &gemm_uni_nt,
         "deep-pipelined 256-tile NT fwd GEMM (+bias/ReLU)",
         pybind11::arg("a"), pybind11::arg("w"),

@@ -55,3 +55,30 @@ DEV_INLINE float wave_max(float v) {
   } while (0)
 
 constexpr int cdiv(int a, int b) { return (a + b - 1) / b; }
+
+// Last-arriver gate (guide G16 counter form) for single-launch
+// partial-reduce + finalize kernels.  Call with ALL of the block's threads
+// AFTER the block's device-scope atomicAdds.  Returns true in exactly one
+// (the last-arriving) block, whose threads may then read every block's
+// atomics with plain loads: each wave drains its own adds (vmcnt) before
+// the one-lane relaxed agent ticket, and the winner takes a one-lane
+// agent acquire (drops this CU's stale L1 lines — the accumulator may
+// have been read by a PREVIOUS call's finalize on this CU) followed by a
+// barrier.  The caller re-zeroes *cnt (any store: gfx950 stores write
+// through to the coherent point; the next launch's atomics see it).
+DEV_INLINE bool last_arriver(unsigned* cnt, unsigned total) {
+  __shared__ unsigned la_flag;
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    unsigned old = __hip_atomic_fetch_add(cnt, 1u, __ATOMIC_RELAXED,
+                                          __HIP_MEMORY_SCOPE_AGENT);
+    la_flag = (old + 1 == total) ? 1u : 0u;
+  }
+  __syncthreads();
+  if (la_flag == 0) return false;
+  if (threadIdx.x == 0)
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  __syncthreads();
+  return true;
+}

@@ -312,12 +312,10 @@ __global__ void gemm_finalize_kernel(const float* __restrict__ cw,
 // transpose2d: [M,N] bf16 -> [N,M].  64x64 tiles, padded LDS, short4 IO.
 // (No longer on the linear-backward hot path — kept as a utility op.)
 // ---------------------------------------------------------------------------
-__global__ __launch_bounds__(256)
-void transpose2d_kernel(const short* __restrict__ in, short* __restrict__ out,
-                        int M, int N, long ldo) {
-  __shared__ short tile[64][64 + 8];  // +8 elems (16 B) row pad vs conflicts
-  int tb = blockIdx.x, nbx = (N + 63) >> 6;
-  int bm = (tb / nbx) << 6, bn = (tb % nbx) << 6;
+DEV_INLINE void transpose_tile(const short* __restrict__ in,
+                               short* __restrict__ out, int M, int N,
+                               long ldo, int bm, int bn,
+                               short tile[64][64 + 8]) {
   int t = threadIdx.x;
   const bool interior = (bm + 64 <= M) && (bn + 64 <= N);
   if (interior) {
@@ -355,6 +353,26 @@ void transpose2d_kernel(const short* __restrict__ in, short* __restrict__ out,
   }
 }
 
+__global__ __launch_bounds__(256)
+void transpose2d_kernel(const short* __restrict__ in, short* __restrict__ out,
+                        int M, int N, long ldo) {
+  __shared__ short tile[64][64 + 8];  // +8 elems (16 B) row pad vs conflicts
+  int tb = blockIdx.x, nbx = (N + 63) >> 6;
+  transpose_tile(in, out, M, N, ldo, (tb / nbx) << 6, (tb % nbx) << 6, tile);
+}
+
+// Batched transpose over a device descriptor table: ONE launch per step
+// refreshes every cached transposed weight (the per-weight transpose2d
+// launches were ~67/step = 0.35 ms of launch floor).  desc row (int64 x7):
+// [src_ptr, dst_ptr, M, N, ldo, bm, bn] — one row per 64x64 tile.
+__global__ __launch_bounds__(256)
+void transpose_batch_kernel(const long* __restrict__ desc) {
+  __shared__ short tile[64][64 + 8];
+  const long* d = desc + (long)blockIdx.x * 7;
+  transpose_tile((const short*)d[0], (short*)d[1], (int)d[2], (int)d[3],
+                 d[4], (int)d[5], (int)d[6], tile);
+}
+
 // colsum: db[N] = sum_m dy[M,N] (fp32 accumulate, bf16 out).
 // Two-stage: (col-chunk, row-chunk) blocks fill the chip, fp32 atomics
 // into a workspace, then cast.  The workspace is CACHED per (device, N)
@@ -363,32 +381,36 @@ void transpose2d_kernel(const short* __restrict__ in, short* __restrict__ out,
 // disjoint-partials scheme (no atomics) was tried and measured SLOWER:
 // its gy-deep serial reduction ran on a 2-8 block grid.
 #define COLSUM_ROWS 128
+// Single launch: partial atomics + last-arriver finalize (cast to bf16 and
+// re-zero the cached workspace) — the separate cast kernel was ~67
+// launch-floor dispatches per step.  acc holds N fp32 + 1 counter word.
 __global__ __launch_bounds__(256)
-void colsum_part_kernel(const short* __restrict__ dy, float* __restrict__ acc,
-                        int M, int N, long lda) {
+void colsum_kernel(const short* __restrict__ dy, float* __restrict__ acc,
+                   short* __restrict__ out, int M, int N, long lda) {
   int n = blockIdx.x * 256 + threadIdx.x;
-  if (n >= N) return;
-  long m0 = (long)blockIdx.y * COLSUM_ROWS;
-  long m1 = min((long)M, m0 + COLSUM_ROWS);
-  // 4 independent accumulators for memory-level parallelism
-  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
-  long m = m0;
-  for (; m + 4 <= m1; m += 4) {
-    s0 += bfbits2f(dy[m * lda + n]);
-    s1 += bfbits2f(dy[(m + 1) * lda + n]);
-    s2 += bfbits2f(dy[(m + 2) * lda + n]);
-    s3 += bfbits2f(dy[(m + 3) * lda + n]);
+  if (n < N) {
+    long m0 = (long)blockIdx.y * COLSUM_ROWS;
+    long m1 = min((long)M, m0 + COLSUM_ROWS);
+    // 4 independent accumulators for memory-level parallelism
+    float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+    long m = m0;
+    for (; m + 4 <= m1; m += 4) {
+      s0 += bfbits2f(dy[m * lda + n]);
+      s1 += bfbits2f(dy[(m + 1) * lda + n]);
+      s2 += bfbits2f(dy[(m + 2) * lda + n]);
+      s3 += bfbits2f(dy[(m + 3) * lda + n]);
+    }
+    for (; m < m1; ++m) s0 += bfbits2f(dy[m * lda + n]);
+    atomicAdd(&acc[n], (s0 + s1) + (s2 + s3));
   }
-  for (; m < m1; ++m) s0 += bfbits2f(dy[m * lda + n]);
-  atomicAdd(&acc[n], (s0 + s1) + (s2 + s3));
-}
-
-__global__ void cast_colsum_kernel(float* __restrict__ in,
-                                   short* __restrict__ out, int n) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < n) {
-    out[i] = f2bfbits(in[i]);
-    in[i] = 0.f;  // workspace stays zeroed for the next same-shape call
+  if (last_arriver((unsigned*)(acc + N), gridDim.x * gridDim.y)) {
+    for (int i = threadIdx.x; i < N; i += 256) {
+      out[i] = f2bfbits(acc[i]);
+      acc[i] = 0.f;  // workspace stays zeroed for the next call
+    }
+    if (threadIdx.x == 0)
+      __hip_atomic_store((unsigned*)(acc + N), 0u, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
   }
 }
 
@@ -580,6 +602,16 @@ void transpose2d_into(torch::Tensor a, torch::Tensor out) {
       (long)out.stride(0));
 }
 
+// One launch, whole registry: desc built by ops/functional._wt_padded.
+void transpose_batch(torch::Tensor desc) {
+  TORCH_CHECK(desc.is_cuda() && desc.dtype() == torch::kInt64 &&
+              desc.dim() == 2 && desc.size(1) == 7 && desc.is_contiguous(),
+              "transpose_batch: bad descriptor table");
+  auto stream = at::hip::getCurrentHIPStream();
+  transpose_batch_kernel<<<(unsigned)desc.size(0), 256, 0, stream>>>(
+      desc.data_ptr<long>());
+}
+
 torch::Tensor colsum(torch::Tensor a, c10::optional<torch::Tensor> out_opt) {
   TORCH_CHECK(a.is_cuda() && a.dtype() == torch::kBFloat16 && a.dim() == 2 &&
               a.stride(1) == 1, "colsum: bad a");
@@ -590,7 +622,7 @@ torch::Tensor colsum(torch::Tensor a, c10::optional<torch::Tensor> out_opt) {
   auto it = ws_cache.find(key);
   if (it == ws_cache.end())
     it = ws_cache.emplace(key, torch::zeros(
-        {N}, a.options().dtype(torch::kFloat32))).first;
+        {N + 1}, a.options().dtype(torch::kFloat32))).first;  // +counter
   torch::Tensor acc = it->second;
   torch::Tensor out;
   if (out_opt.has_value()) {
@@ -602,10 +634,9 @@ torch::Tensor colsum(torch::Tensor a, c10::optional<torch::Tensor> out_opt) {
   }
   auto stream = at::hip::getCurrentHIPStream();
   dim3 grid(cdiv(N, 256), cdiv(M, COLSUM_ROWS));
-  colsum_part_kernel<<<grid, 256, 0, stream>>>(
-      (const short*)a.data_ptr(), acc.data_ptr<float>(), M, N, lda);
-  cast_colsum_kernel<<<cdiv(N, 256), 256, 0, stream>>>(
-      acc.data_ptr<float>(), (short*)out.data_ptr(), N);
+  colsum_kernel<<<grid, 256, 0, stream>>>(
+      (const short*)a.data_ptr(), acc.data_ptr<float>(),
+      (short*)out.data_ptr(), M, N, lda);
   return out;
 }
 

@@ -200,11 +200,12 @@ __global__ __launch_bounds__(256)
 void ln_gb_kernel(const short* __restrict__ dy, const short* __restrict__ s,
                   const float* __restrict__ mean,
                   const float* __restrict__ rstd, float* __restrict__ acc_g,
-                  float* __restrict__ acc_b, int R, int D) {
+                  float* __restrict__ acc_b, short* __restrict__ dgamma,
+                  short* __restrict__ dbeta, int R, int D) {
   // 2 columns per thread (s16x2 loads) x 4-row unroll: the scalar
   // one-col-per-thread version was latency-bound at ~0.9 TB/s.
   const int c = (blockIdx.x * 64 + threadIdx.x) * 2;
-  if (c >= D) return;
+  if (c < D) {
   const long r0 = (long)blockIdx.y * LNGB_ROWS;
   const long r1 = min((long)R, r0 + LNGB_ROWS);
   float sg0 = 0.f, sg1 = 0.f, sb0 = 0.f, sb1 = 0.f;
@@ -251,20 +252,18 @@ void ln_gb_kernel(const short* __restrict__ dy, const short* __restrict__ s,
       atomicAdd(&acc_b[cc], sb);
     }
   }
-}
-
-// cast pass also re-zeroes the cached fp32 accumulators so the next
-// same-shape call skips the zero-fill launch.
-__global__ void ln_gb_cast_kernel(float* __restrict__ ag,
-                                  float* __restrict__ ab,
-                                  short* __restrict__ dgamma,
-                                  short* __restrict__ dbeta, int D) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < D) {
-    dgamma[i] = f2bfbits(ag[i]);
-    dbeta[i] = f2bfbits(ab[i]);
-    ag[i] = 0.f;
-    ab[i] = 0.f;
+  }
+  // single-launch finalize: cast to bf16 + re-zero (was ln_gb_cast_kernel)
+  if (last_arriver((unsigned*)(acc_b + D), gridDim.x * gridDim.y)) {
+    for (int i = threadIdx.x; i < D; i += 64) {
+      dgamma[i] = f2bfbits(acc_g[i]);
+      dbeta[i] = f2bfbits(acc_b[i]);
+      acc_g[i] = 0.f;
+      acc_b[i] = 0.f;
+    }
+    if (threadIdx.x == 0)
+      __hip_atomic_store((unsigned*)(acc_b + D), 0u, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
   }
 }
 
@@ -349,19 +348,17 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor s,
   auto wit = ws_cache.find(wkey);
   if (wit == ws_cache.end())
     wit = ws_cache.emplace(wkey, torch::zeros(
-        {2L * D}, dy.options().dtype(torch::kFloat32))).first;
+        {2L * D + 1}, dy.options().dtype(torch::kFloat32))).first;  // +cnt
   auto acc_g = wit->second.narrow(0, 0, D);
-  auto acc_b = wit->second.narrow(0, D, D);
+  auto acc_b = wit->second.narrow(0, D, D + 1);
   // 64-thread blocks: at d_model=512 a 256-thread block grid is only 128
   // workgroups — half the 256-CU chip idle.
   dim3 gbgrid(cdiv(cdiv(D, 2), 64), cdiv(R, LNGB_ROWS));
   ln_gb_kernel<<<gbgrid, 64, 0, stream>>>(
       (const short*)dy.data_ptr(), (const short*)s.data_ptr(),
       mean.data_ptr<float>(), rstd.data_ptr<float>(),
-      acc_g.data_ptr<float>(), acc_b.data_ptr<float>(), R, D);
-  ln_gb_cast_kernel<<<cdiv(D, 256), 256, 0, stream>>>(
       acc_g.data_ptr<float>(), acc_b.data_ptr<float>(),
-      (short*)dgamma.data_ptr(), (short*)dbeta.data_ptr(), D);
+      (short*)dgamma.data_ptr(), (short*)dbeta.data_ptr(), R, D);
   if (mask.has_value()) return {dx, dgamma, dbeta, dxm};
   return {dx, dgamma, dbeta};
 }

@@ -86,24 +86,45 @@ def bump_weight_version():
     _WEIGHT_VERSION[0] += 1
 
 
-_WTP_CACHE: dict = {}  # id(weight) -> [version, padded-transposed copy]
+_WTP_CACHE: dict = {}  # id(weight) -> [version, buf, weight]
+_WTP_TABLE = [None]    # cached transpose_batch descriptor table
+
+
+def _build_wtp_table(device):
+    rows = []
+    for ver, buf, w in _WTP_CACHE.values():
+        M, N = w.shape
+        ldo = buf.stride(0)
+        for bm in range(0, M, 64):
+            for bn in range(0, N, 64):
+                rows.append([w.data_ptr(), buf.data_ptr(), M, N, ldo,
+                             bm, bn])
+    return torch.tensor(rows, dtype=torch.int64).to(device)
 
 
 def _wt_padded(E, w):
-    """W[N,K]^T into a (K, Np) buffer, Np = N rounded up to 64, pad columns
-    zero — the NT B-operand for the ragged-vocab dX GEMM.  Refreshed once
-    per optimizer step (bump_weight_version); under graph capture the
-    transpose is launched unconditionally so replays recompute it."""
+    """W[N,K]^T into a (K, Np) buffer, Np = N rounded up to 256 (GEMM tile
+    width; pad columns zero) — the NT B-operand for dX.  All registered
+    buffers refresh together in ONE transpose_batch launch per optimizer
+    step (bump_weight_version) — a captured step records that launch, so
+    graph replays refresh too (graph.py bumps before capture)."""
     key = id(w)
     ent = _WTP_CACHE.get(key)
     npad = (w.shape[0] + 255) // 256 * 256
     if ent is None or ent[1].shape[1] != npad:
         buf = torch.zeros(w.shape[1], npad, device=w.device, dtype=w.dtype)
-        ent = [None, buf]
+        ent = [_WEIGHT_VERSION[0], buf, w]
         _WTP_CACHE[key] = ent
-    if ent[0] != _WEIGHT_VERSION[0] or _GRAPH_SEED_T is not None:
-        E.transpose2d_into(w, ent[1])
-        ent[0] = _WEIGHT_VERSION[0]
+        _WTP_TABLE[0] = None  # registry changed; rebuild lazily
+        E.transpose2d_into(w, buf)
+        return buf
+    if ent[0] != _WEIGHT_VERSION[0]:
+        if _WTP_TABLE[0] is None:
+            _WTP_TABLE[0] = _build_wtp_table(w.device)
+        E.transpose_batch(_WTP_TABLE[0])
+        v = _WEIGHT_VERSION[0]
+        for e in _WTP_CACHE.values():
+            e[0] = v
     return ent[1]
 
 
@@ -146,6 +167,21 @@ def _dw_gemm(dy, x, out=None):
     if dy.shape[1] >= 8192 and dy.shape[0] % 64 == 0 \
             and (dy.shape[1] % 256 == 0 or dy.stride(0) >= npad) \
             and x.shape[1] % 128 == 0:
+        N = dy.shape[1]
+        main = (N // 256) * 256
+        if main != N and main > 0:
+            if out is None:
+                out = torch.empty(N, x.shape[1], device=dy.device,
+                                  dtype=dy.dtype)
+            # Grid-quantization cliff: ceil(N/256) tiles puts the grid a
+            # couple of blocks past 256 (one per CU at this kernel's LDS),
+            # and the stragglers cost a whole second pass.  Compute the
+            # 256-aligned main span at exactly one block wave, then the
+            # ragged vocab tail (a few rows, full contraction) split-K so
+            # it also fills the chip for its instant of work.
+            E.gemm_uni_tn(dy[:, :main], x, out[:main], 1)
+            E.gemm_uni_tn(dy[:, main:], x, out[main:], 64)
+            return out
         return E.gemm_uni_tn(dy, x, out)
     # gemm_dw writes DISJOINT per-slice fp32 partials (no atomics, no
     # zeroing — the fp32-atomic epilogue measured at the chip's atomic

@@ -69,6 +69,10 @@ class CapturedTrainStep:
                 optimizer.master.to(optimizer.flat.flat_w.dtype))
             del snap
 
+            # make the weight-derived caches (transposed-weight buffers)
+            # stale so the capture RECORDS their batched refresh — replays
+            # then refresh them after each in-graph Adam step
+            ops.functional.bump_weight_version()
             self.graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(self.graph):
                 self.loss, self.logits, self.tar_real = run()
