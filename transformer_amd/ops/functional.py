@@ -102,6 +102,15 @@ def _build_wtp_table(device):
     return torch.tensor(rows, dtype=torch.int64).to(device)
 
 
+def prepare_weight_caches():
+    """Build the transpose_batch descriptor table eagerly (it involves a
+    host->device copy, which is illegal inside hipGraph capture) — called
+    by CapturedTrainStep between warmup and capture."""
+    if _WTP_CACHE and _WTP_TABLE[0] is None:
+        dev = next(iter(_WTP_CACHE.values()))[1].device
+        _WTP_TABLE[0] = _build_wtp_table(dev)
+
+
 def _wt_padded(E, w):
     """W[N,K]^T into a (K, Np) buffer, Np = N rounded up to 256 (GEMM tile
     width; pad columns zero) — the NT B-operand for dX.  All registered

@@ -71,7 +71,9 @@ class CapturedTrainStep:
 
             # make the weight-derived caches (transposed-weight buffers)
             # stale so the capture RECORDS their batched refresh — replays
-            # then refresh them after each in-graph Adam step
+            # then refresh them after each in-graph Adam step.  The
+            # descriptor table must exist BEFORE capture (H2D copy).
+            ops.functional.prepare_weight_caches()
             ops.functional.bump_weight_version()
             self.graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(self.graph):
