@@ -722,3 +722,33 @@ def test_transpose_batch_descriptor_table():
         m = w.shape[0]
         assert torch.equal(buf[:, :m], w.t().contiguous()), w.shape
         assert (buf[:, m:].float() == 0).all()
+
+
+def test_attn_defer_max_spike_forces_rescale():
+    """T13/rule-26: the defer-max branch is data-dependent — spike one K
+    row in a LATE tile so the running max jumps past THR=8 after several
+    deferred tiles; output must still match the fp32 reference."""
+    from transformer_amd.ops import reference as R
+    torch.manual_seed(12)
+    B, S, H, dh = 2, 512, 4, 64
+    q = torch.randn(B, S, H, dh, device="cuda", dtype=torch.bfloat16) * 0.5
+    k = torch.randn(B, S, H, dh, device="cuda", dtype=torch.bfloat16) * 0.5
+    v = torch.randn(B, S, H, dh, device="cuda", dtype=torch.bfloat16)
+    # spike key row 350 (tile 5 of 8) against everything: raw q.k ~ dh*4
+    k[:, 350] = 4.0
+    q[:, :, :, :8] = 2.0  # give q mass so the spike dominates post-scale
+    o, lse = _ext().attn_fwd(q, k, v, torch.Tensor(), False,
+                             1.0 / math.sqrt(dh), 1)
+    qt, kt, vt = (t.permute(0, 2, 1, 3).float() for t in (q, k, v))
+    ref, _ = R.scaled_dot_product_attention(qt, kt, vt, None,
+                                            return_weights=True)
+    ref = ref.permute(0, 2, 1, 3)
+    assert_close(o, ref, 0.04, "defer-max spike")
+    # and the no-spike path still matches (deferred tiles only)
+    k2 = torch.randn_like(k) * 0.3
+    o2, _ = _ext().attn_fwd(q, k2, v, torch.Tensor(), False,
+                            1.0 / math.sqrt(dh), 1)
+    kt2 = k2.permute(0, 2, 1, 3).float()
+    ref2, _ = R.scaled_dot_product_attention(qt, kt2, vt, None,
+                                             return_weights=True)
+    assert_close(o2, ref2.permute(0, 2, 1, 3), 0.04, "defer-max smooth")

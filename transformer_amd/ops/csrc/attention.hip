@@ -223,6 +223,17 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
       stage_kv<DH, false, true, FKVT>(Vp + (long)k0 * kv_rs, kv_rs, kc,
                                       nullptr, v_lds);
     __syncthreads();
+    // tile-uniform mask facts: most tiles (all of them when Sk%64==0 and
+    // nothing is padded; all-but-diagonal ones under causal) need NO
+    // per-element mask work — the kernel is issue-bound, and the mask
+    // loop is ~76 VALU per 64-key tile per row-fragment.
+    bool tile_pad_any = false;
+    if (pad) {
+      unsigned char pb =
+          (lane < FKVT && k0 + lane < Sk) ? pad[k0 + lane] : 0;
+      tile_pad_any = __any(pb != 0);
+    }
+    const bool tile_full = (kc == FKVT) && !tile_pad_any;
 
 #pragma unroll
     for (int rf = 0; rf < RF; ++rf) {
@@ -248,18 +259,32 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
         }
       }
       __builtin_amdgcn_s_setprio(0);
+      // clean: whole tile unmasked for this wave's rows (wave-uniform)
+      const bool clean =
+          tile_full && (!causal || k0 + FKVT - 1 <= qr0 + kg * 4);
+      if (clean) {
 #pragma unroll
-      for (int half = 0; half < NHALF; ++half) {
-        const int kcol = k0 + half * 16 + fr;   // C col = lane&15
-        const bool col_pad = (kcol >= Sk) || (pad && pad[min(kcol, Sk - 1)]);
+        for (int half = 0; half < NHALF; ++half)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int qrow = qr0 + kg * 4 + r;    // C row = (lane>>4)*4+r
-          float x = s2h[half][r] * scale;
-          if (col_pad) x += NEG_BIG;
-          if (causal && kcol > qrow) x += NEG_BIG;
-          p_raw[half][r] = x;
-          tile_pmax[r] = fmaxf(tile_pmax[r], x);
+          for (int r = 0; r < 4; ++r) {
+            float x = s2h[half][r] * scale;
+            p_raw[half][r] = x;
+            tile_pmax[r] = fmaxf(tile_pmax[r], x);
+          }
+      } else {
+#pragma unroll
+        for (int half = 0; half < NHALF; ++half) {
+          const int kcol = k0 + half * 16 + fr;   // C col = lane&15
+          const bool col_pad = (kcol >= Sk) || (pad && pad[min(kcol, Sk - 1)]);
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int qrow = qr0 + kg * 4 + r;    // C row = (lane>>4)*4+r
+            float x = s2h[half][r] * scale;
+            if (col_pad) x += NEG_BIG;
+            if (causal && kcol > qrow) x += NEG_BIG;
+            p_raw[half][r] = x;
+            tile_pmax[r] = fmaxf(tile_pmax[r], x);
+          }
         }
       }
       // row max across the 16 cols held by the 16-lane group
@@ -269,18 +294,31 @@ void attn_fwd_kernel(const short* __restrict__ Q, const short* __restrict__ K,
         for (int off = 1; off < 16; off <<= 1)
           tile_pmax[r] = fmaxf(tile_pmax[r], __shfl_xor(tile_pmax[r], off));
       }
-      // online rescale
+      // defer-max (guide T13, THR=8): when no row's max grew past m+THR,
+      // keep the old max (P bounded by e^THR, fp32 l/acc absorb it, the
+      // final O/l and LSE are unchanged) and skip the alpha exps + the
+      // O-wide rescale.  The decision covers this tile's P entirely
+      // (previous tile's PV is complete) — the T13 hazard order holds.
+      float need = 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        need = fmaxf(need, tile_pmax[r] - m_run[rf][r]);
       float alpha[4];
+      if (!__all(need <= 8.0f)) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float mn = fmaxf(m_run[rf][r], tile_pmax[r]);
-        alpha[r] = __expf(m_run[rf][r] - mn);
-        m_run[rf][r] = mn;
+        for (int r = 0; r < 4; ++r) {
+          float mn = fmaxf(m_run[rf][r], tile_pmax[r]);
+          alpha[r] = __expf(m_run[rf][r] - mn);
+          m_run[rf][r] = mn;
+        }
+#pragma unroll
+        for (int i = 0; i < D16; ++i)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) acc[rf][i][r] *= alpha[r];
+      } else {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) alpha[r] = 1.0f;
       }
-#pragma unroll
-      for (int i = 0; i < D16; ++i)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) acc[rf][i][r] *= alpha[r];
 
       // P = exp(S - m), park bf16 P in LDS (A-layout).  The row-sum for l
       // is NOT shuffle-reduced here: it comes out of the PV phase below as
@@ -477,15 +515,28 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
       const float lse_q = q_ok ? lse[min(qcol, Sq - 1)] : 0.f;
       const float d_q = q_ok ? dl[min(qcol, Sq - 1)] : 0.f;
       s16x4 pw;
+      // per-lane clean: this lane's 4 keys valid+unpadded, its q col in
+      // range, and (under causal) all its keys visible to that col
+      const bool clean2 = q_ok && !k_pad && (k0w + kg * 4 + 3 < Sk) &&
+                          (!causal || k0w + kg * 4 + 3 <= qcol);
+      if (clean2) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int key = k0w + kg * 4 + r;
-        float x = st[r] * scale;
-        bool masked = !q_ok || key >= Sk || (pad && key < Sk && pad[key]) ||
-                      (causal && key > qcol);
-        float p = masked ? 0.f : __expf(x - lse_q);
-        ds_reg[half][r] = p * (dpt[r] - d_q) * scale;
-        pw[r] = f2bfbits(p);
+        for (int r = 0; r < 4; ++r) {
+          float p = __expf(st[r] * scale - lse_q);
+          ds_reg[half][r] = p * (dpt[r] - d_q) * scale;
+          pw[r] = f2bfbits(p);
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int key = k0w + kg * 4 + r;
+          float x = st[r] * scale;
+          bool masked = !q_ok || key >= Sk || (pad && key < Sk && pad[key]) ||
+                        (causal && key > qcol);
+          float p = masked ? 0.f : __expf(x - lse_q);
+          ds_reg[half][r] = p * (dpt[r] - d_q) * scale;
+          pw[r] = f2bfbits(p);
+        }
       }
       scrT_write4(x_lds[wid], half * 16 + fr, kg * 4, pw);
     }
@@ -611,15 +662,28 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
       __builtin_amdgcn_s_setprio(0);
       const int kcol = k0 + half * 16 + fr;
       const bool col_pad = (kcol >= Sk) || (pad && pad[min(kcol, Sk - 1)]);
+      // per-lane clean: col valid+unpadded, all 4 q rows in range and
+      // (under causal) at/after this key col
+      const bool clean2 = !col_pad && (q0 + kg * 4 + 3 < Sq) &&
+                          (!causal || kcol <= q0 + kg * 4);
+      if (clean2) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int qrow = q0 + kg * 4 + r;
-        const bool row_ok = qrow < Sq;
-        float lse_q = row_ok ? lse[min(qrow, Sq - 1)] : 0.f;
-        float d_q = row_ok ? dl[min(qrow, Sq - 1)] : 0.f;
-        bool masked = !row_ok || col_pad || (causal && kcol > qrow);
-        float p = masked ? 0.f : __expf(s[r] * scale - lse_q);
-        dsw[r] = f2bfbits(p * (dp[r] - d_q) * scale);
+        for (int r = 0; r < 4; ++r) {
+          const int qrow = q0 + kg * 4 + r;
+          float p = __expf(s[r] * scale - lse[qrow]);
+          dsw[r] = f2bfbits(p * (dp[r] - dl[qrow]) * scale);
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int qrow = q0 + kg * 4 + r;
+          const bool row_ok = qrow < Sq;
+          float lse_q = row_ok ? lse[min(qrow, Sq - 1)] : 0.f;
+          float d_q = row_ok ? dl[min(qrow, Sq - 1)] : 0.f;
+          bool masked = !row_ok || col_pad || (causal && kcol > qrow);
+          float p = masked ? 0.f : __expf(s[r] * scale - lse_q);
+          dsw[r] = f2bfbits(p * (dp[r] - d_q) * scale);
+        }
       }
       scrT_write4(x_lds[wid], half * 16 + fr, kg * 4, dsw);
     }
