@@ -487,6 +487,14 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
   // causal: q-rows before this key-block are fully masked
   const int q_start = causal ? (kb * (WAVES * QW)) / KVT * KVT : 0;
 
+  // any pad among this wave's 16 OUTPUT keys (C rows k0w..k0w+15 — note
+  // k_pad above is the lane's A-fragment row k0w+fr, a different role)
+  bool wave_pad_any = false;
+  if (pad) {
+    const int kk = k0w + (lane & 15);
+    wave_pad_any = __any(kk < Sk && pad[min(kk, Sk - 1)]);
+  }
+
   for (int j0 = q_start; j0 < Sq; j0 += KVT) {
     const int jc = min(KVT, Sq - j0);
     stage_kv<DH, true, false>(Qp + (long)j0 * q_rs, q_rs, jc,
@@ -515,9 +523,11 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
       const float lse_q = q_ok ? lse[min(qcol, Sq - 1)] : 0.f;
       const float d_q = q_ok ? dl[min(qcol, Sq - 1)] : 0.f;
       s16x4 pw;
-      // per-lane clean: this lane's 4 keys valid+unpadded, its q col in
-      // range, and (under causal) all its keys visible to that col
-      const bool clean2 = q_ok && !k_pad && (k0w + kg * 4 + 3 < Sk) &&
+      // per-lane clean: this lane's 4 OUTPUT keys (k0w+kg*4+r) valid,
+      // no pad in the wave's keys, its q col in range, and (under
+      // causal) all its keys visible to that col
+      const bool clean2 = q_ok && !wave_pad_any &&
+                          (k0w + kg * 4 + 3 < Sk) &&
                           (!causal || k0w + kg * 4 + 3 <= qcol);
       if (clean2) {
 #pragma unroll
