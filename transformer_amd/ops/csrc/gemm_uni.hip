@@ -479,6 +479,158 @@ static torch::Tensor gemm_uni_launch(
   return c;
 }
 
+// ---------------------------------------------------------------------------
+// Ring-schedule NT GEMM (SCHED=5 in the A/B): ONE LDS slot, half-granular
+// staging — each phase stages exactly one piece of tile t+1 into the
+// region the previous phase's readers just vacated (piece death points:
+// A-piece0 after q0's barrier, B after q1's, A-piece1 after q2's), with
+// two counted s_waitcnt vmcnt(2) per K-tile (end-q1 covers A-piece1 of
+// the CURRENT tile, end-q3 covers A-piece0+B of the NEXT) — the faithful
+// reconstruction of the guide's 8-phase example's load pipeline.  Halves
+// the LDS of the double-buffered schedule.
+// ---------------------------------------------------------------------------
+
+template <int EPILOGUE, int BN_>
+__global__ __launch_bounds__(U_THREADS, 1)
+void gemm_ring_kernel(const short* __restrict__ A, const short* __restrict__ B,
+                      const short* __restrict__ bias, short* __restrict__ C,
+                      int M, int N, int K, long lda, long ldb, int has_bias,
+                      int nbm, int nbn) {
+  constexpr int BM_ = 256;
+  constexpr int MF = 8;
+  constexpr int NF = BN_ / 64;
+  constexpr int A_ELEMS = BM_ * U_BK;
+  extern __shared__ short smem[];
+  const int lane = threadIdx.x & 63;
+
+  int nwg = nbm * nbn;
+  int wg = blockIdx.x;
+  {
+    int q = nwg / 8, r = nwg % 8, x = wg % 8, o = wg / 8;
+    wg = (x < r ? x * (q + 1) : r * (q + 1) + (x - r) * q) + o;
+  }
+  const int bm0 = (wg / nbn) * BM_;
+  const int bn0 = (wg % nbn) * BN_;
+
+  const int wid = threadIdx.x >> 6;
+  const int wm = (wid >> 2) * (BM_ / 2);
+  const int wn = (wid & 3) * (BN_ / 4);
+  const int fr = lane & 15;
+  const int kg = lane >> 4;
+
+  f32x4 acc[MF][NF];
+#pragma unroll
+  for (int i = 0; i < MF; ++i)
+#pragma unroll
+    for (int j = 0; j < NF; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int ntiles = K / U_BK;
+  short* a_lds = smem;
+  short* b_lds = smem + A_ELEMS;
+
+  auto stage_a = [&](int kt, int piece) {
+    u_stage_nt<256>(A + (long)kt * U_BK, lda, bm0, M, piece, a_lds);
+  };
+  auto stage_b = [&](int kt) {
+    if (BN_ == 256) {
+      u_stage_nt<256>(B + (long)kt * U_BK, ldb, bn0, N, 0, b_lds);
+      u_stage_nt<256>(B + (long)kt * U_BK, ldb, bn0, N, 1, b_lds);
+    } else {
+      u_stage_nt<128>(B + (long)kt * U_BK, ldb, bn0, N, 0, b_lds);
+    }
+  };
+
+  stage_a(0, 0);
+  stage_a(0, 1);
+  stage_b(0);
+  U_WAIT_VM(0);
+  U_BARRIER();
+
+  bf16x8u afr[MF / 2][2], bfr[NF][2];
+
+#define R_MFMA_QUAD(RH, CH)                                                \
+  _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                         \
+    _Pragma("unroll") for (int i = 0; i < MF / 2; ++i)                     \
+      _Pragma("unroll") for (int j = 0; j < NF / 2; ++j)                   \
+        acc[(RH) * (MF / 2) + i][(CH) * (NF / 2) + j] =                    \
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(                       \
+                afr[i][ks], bfr[(CH) * (NF / 2) + j][ks],                  \
+                acc[(RH) * (MF / 2) + i][(CH) * (NF / 2) + j], 0, 0, 0);
+
+  for (int t = 0; t < ntiles; ++t) {
+    const bool do_pf = t + 1 < ntiles;
+
+    // q0: read A-piece0 + B-low fragments; MFMA (0,0)
+#pragma unroll
+    for (int i = 0; i < MF / 2; ++i)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+        afr[i][ks] = u_frag_nt(a_lds, wm + i * 16 + fr, ks * 64 + kg * 16);
+#pragma unroll
+    for (int j = 0; j < NF / 2; ++j)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+        bfr[j][ks] = u_frag_nt(b_lds, wn + j * 16 + fr, ks * 64 + kg * 16);
+    U_BARRIER();
+    U_WAIT_LGKM0();
+    R_MFMA_QUAD(0, 0)
+    U_BARRIER();
+    // q1: stage A-piece0(t+1) into the region q0 vacated; read B-high
+    if (do_pf) stage_a(t + 1, 0);
+#pragma unroll
+    for (int j = NF / 2; j < NF; ++j)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+        bfr[j][ks] = u_frag_nt(b_lds, wn + j * 16 + fr, ks * 64 + kg * 16);
+    U_BARRIER();
+    U_WAIT_LGKM0();
+    R_MFMA_QUAD(0, 1)
+    // end-q1: drain THIS tile's A-piece1 (staged q3 of t-1); t+1's
+    // A-piece0 stays in flight
+    if (do_pf) U_WAIT_VM(2); else U_WAIT_VM(0);
+    U_BARRIER();
+    // q2: stage B(t+1) over the dead B regions; read A-piece1
+    if (do_pf) stage_b(t + 1);
+#pragma unroll
+    for (int i = 0; i < MF / 2; ++i)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+        afr[i][ks] = u_frag_nt(a_lds, wm + (BM_ / 4) + i * 16 + fr,
+                               ks * 64 + kg * 16);
+    U_BARRIER();
+    U_WAIT_LGKM0();
+    R_MFMA_QUAD(1, 0)
+    U_BARRIER();
+    // q3: stage A-piece1(t+1); MFMA (1,1) from registers
+    if (do_pf) stage_a(t + 1, 1);
+    R_MFMA_QUAD(1, 1)
+    // end-q3: drain t+1's A-piece0+B (2-3 phases in flight); its
+    // A-piece1 rides across the boundary (covered by next end-q1)
+    if (do_pf) U_WAIT_VM(2); else U_WAIT_VM(0);
+    U_BARRIER();
+  }
+#undef R_MFMA_QUAD
+
+#pragma unroll
+  for (int i = 0; i < MF; ++i) {
+    const int grow_base = bm0 + wm + i * 16 + kg * 4;
+#pragma unroll
+    for (int j = 0; j < NF; ++j) {
+      const int gcol = bn0 + wn + j * 16 + fr;
+      if (gcol >= N) continue;
+      const float bv = (has_bias && bias) ? bfbits2f(bias[gcol]) : 0.0f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int grow = grow_base + r;
+        if (grow >= M) continue;
+        float v = acc[i][j][r] + bv;
+        if (EPILOGUE == 1) v = fmaxf(v, 0.0f);
+        C[(long)grow * N + gcol] = f2bfbits(v);
+      }
+    }
+  }
+}
+
 // Bench-only schedule A/B: NT x NT, no epilogue, BN=256 grid assumed.
 torch::Tensor gemm_uni_nt_ab(torch::Tensor a, torch::Tensor w,
                              int64_t sched) {
@@ -503,6 +655,22 @@ torch::Tensor gemm_uni_nt_ab(torch::Tensor a, torch::Tensor w,
     if (BN == 256) go(std::integral_constant<int, 256>{}, sc);
     else go(std::integral_constant<int, 128>{}, sc);
   };
+  if ((int)sched == 5) {
+    const size_t rsm = ((size_t)256 + BN) * U_BK * sizeof(short);
+    auto ring = [&](auto bnc) {
+      constexpr int BNv = decltype(bnc)::value;
+      auto kfn = gemm_ring_kernel<0, BNv>;
+      (void)hipFuncSetAttribute((const void*)kfn,
+                                hipFuncAttributeMaxDynamicSharedMemorySize,
+                                (int)rsm);
+      kfn<<<nbm * nbn, U_THREADS, rsm, stream>>>(
+          (const short*)a.data_ptr(), (const short*)w.data_ptr(), nullptr,
+          (short*)c.data_ptr(), M, N, K, K, K, 0, nbm, nbn);
+    };
+    if (BN == 256) ring(std::integral_constant<int, 256>{});
+    else ring(std::integral_constant<int, 128>{});
+    return c;
+  }
   switch ((int)sched) {
     case 0: go2(std::integral_constant<int, 0>{}); break;
     case 2: go2(std::integral_constant<int, 2>{}); break;
