@@ -40,6 +40,10 @@ def init_distributed(backend: str | None = None) -> tuple[int, int, int]:
     local = int(os.environ.get("LOCAL_RANK", rank))
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
+    # TFMX_BACKEND=gloo lets 2 ranks share ONE GPU (RCCL rejects
+    # duplicate devices) — used by tools/dp_parity.py to exercise the
+    # real bucket manager + GPU kernels on a single-GPU box.
+    backend = os.environ.get("TFMX_BACKEND", backend)
     if not dist.is_initialized():
         dist.init_process_group(backend=backend, rank=rank, world_size=world,
                                 timeout=datetime.timedelta(seconds=300))
@@ -63,6 +67,10 @@ class BucketedDataParallel:
         self.flat = flat
         self.pg = process_group
         self.world = dist.get_world_size(process_group) if dist.is_initialized() else 1
+        # TFMX_DDP_FORCE=1: launch the bucket collectives even at world=1
+        # (self all-reduce) so a single-GPU rocprof run shows the REAL
+        # RCCL launches from the enqueue callbacks overlapping backward.
+        self.active = self.world > 1 or os.environ.get("TFMX_DDP_FORCE") == "1"
         if bucket_mb is None:
             bucket_mb = float(os.environ.get("TFMX_BUCKET_MB", "25"))
         cap = int(bucket_mb * 2 ** 20 / flat.flat_g.element_size())
@@ -105,7 +113,7 @@ class BucketedDataParallel:
         self._hooks = []
 
     def _on_param_ready(self, p):
-        if self.world <= 1:
+        if not self.active:
             return
         bi = self._param_bucket.get(id(p))
         if bi is None:
@@ -135,7 +143,7 @@ class BucketedDataParallel:
         bi = self._param_bucket[id(p)]
 
         def hook(_param):
-            if self.world <= 1:
+            if not self.active:
                 return
             b = self.buckets[bi]
             b["pending"] -= 1
@@ -155,7 +163,7 @@ class BucketedDataParallel:
 
     def finalize(self):
         """Call after loss.backward(), before optimizer.step()."""
-        if self.world <= 1:
+        if not self.active:
             return
         for b in self.buckets:
             if b["work"] is None and b["pending"] > 0:

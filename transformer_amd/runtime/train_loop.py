@@ -294,9 +294,11 @@ class DistributedTrain(Train):
         kwargs.setdefault("use_flat", True)
         super().__init__(*args, **kwargs)
         if ddp is None:
+            import os
             import torch.distributed as dist
             if dist.is_available() and dist.is_initialized() \
-                    and dist.get_world_size() > 1:
+                    and (dist.get_world_size() > 1
+                         or os.environ.get("TFMX_DDP_FORCE") == "1"):
                 from ..parallel import BucketedDataParallel
                 ddp = BucketedDataParallel(self.optimizer.flat)
         self.ddp = ddp
