@@ -12,15 +12,22 @@ import sys
 def main():
     db = sys.argv[1]
     c = sqlite3.connect(db)
-    uuid = [r[0] for r in c.execute(
+    uuids = [r[0].split("rocpd_kernel_dispatch_")[1] for r in c.execute(
         "SELECT name FROM sqlite_master WHERE type='table' "
-        "AND name LIKE 'rocpd_kernel_dispatch%'")][0].split(
-            "rocpd_kernel_dispatch_")[1]
-    rows = list(c.execute(f"""
-        SELECT s.string, d.start, d.end
-        FROM rocpd_kernel_dispatch_{uuid} d
-        JOIN rocpd_info_kernel_symbol_{uuid} k ON d.kernel_id = k.id
-        JOIN rocpd_string_{uuid} s ON k.kernel_name_id = s.id"""))
+        "AND name LIKE 'rocpd_kernel_dispatch%'")]
+    rows = []
+    for uuid in uuids:
+        cols = [r[1] for r in c.execute(
+            f"PRAGMA table_info(rocpd_info_kernel_symbol_{uuid})")]
+        name_expr = ("s.string" if "kernel_name_id" in cols
+                     else "k.kernel_name")
+        join = (f"JOIN rocpd_string_{uuid} s ON k.kernel_name_id = s.id"
+                if "kernel_name_id" in cols else "")
+        rows += list(c.execute(f"""
+            SELECT {name_expr}, d.start, d.end
+            FROM rocpd_kernel_dispatch_{uuid} d
+            JOIN rocpd_info_kernel_symbol_{uuid} k ON d.kernel_id = k.id
+            {join}"""))
     comm = [(s, e) for n, s, e in rows if "ccl" in n.lower()]
     comp = [(s, e) for n, s, e in rows if "ccl" not in n.lower()]
     if not comm:

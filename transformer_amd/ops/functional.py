@@ -111,15 +111,17 @@ def prepare_weight_caches():
         _WTP_TABLE[0] = _build_wtp_table(dev)
 
 
-def _wt_padded(E, w):
-    """W[N,K]^T into a (K, Np) buffer, Np = N rounded up to 256 (GEMM tile
-    width; pad columns zero) — the NT B-operand for dX.  All registered
-    buffers refresh together in ONE transpose_batch launch per optimizer
-    step (bump_weight_version) — a captured step records that launch, so
-    graph replays refresh too (graph.py bumps before capture)."""
+def _wt_padded(E, w, gran=64):
+    """W[N,K]^T into a (K, Np) buffer, Np = N rounded up to `gran` (pad
+    columns zero) — the NT B-operand for dX.  gran=256 matches ce_bwd's
+    padded dlogits contraction (the ragged-vocab head); the generic path
+    uses 64 (the GEMM K-step) so the contraction equals dY's width.  All
+    registered buffers refresh together in ONE transpose_batch launch per
+    optimizer step (bump_weight_version) — a captured step records that
+    launch, so graph replays refresh too (graph.py bumps before capture)."""
     key = id(w)
     ent = _WTP_CACHE.get(key)
-    npad = (w.shape[0] + 255) // 256 * 256
+    npad = (w.shape[0] + gran - 1) // gran * gran
     if ent is None or ent[1].shape[1] != npad:
         buf = torch.zeros(w.shape[1], npad, device=w.device, dtype=w.dtype)
         ent = [_WEIGHT_VERSION[0], buf, w]
@@ -220,7 +222,7 @@ def _dx_gemm(E, dy, w):
               and dy.storage_offset() == 0)
     if padded and E.gemm_uni_viable(M, w.shape[1], npad):
         full = dy.as_strided((M, npad), (npad, 1))
-        return E.gemm_nt(full, _wt_padded(E, w), torch.Tensor(), 0)
+        return E.gemm_nt(full, _wt_padded(E, w, 256), torch.Tensor(), 0)
     if _DX_BACKEND == "wt" and N % 64 == 0 and M >= 2048:
         return E.gemm_nt(_dense2d(dy), _wt_padded(E, w), torch.Tensor(), 0)
     if N % 64 == 0 and E.gemm_uni_viable(M, w.shape[1], N):
