@@ -96,8 +96,8 @@ def main():
     if dev.type == "cuda":
         torch.cuda.synchronize()
 
-    wbytes = tr.optimizer.flat.flat_w.detach().cpu().numpy().tobytes()
-    whash = hashlib.sha256(wbytes).hexdigest()
+    wbits = tr.optimizer.flat.flat_w.detach().cpu().view(torch.int16)
+    whash = hashlib.sha256(wbits.numpy().tobytes()).hexdigest()
 
     if world > 1:
         # cross-rank: weights must be bit-identical after N synced steps
@@ -110,9 +110,13 @@ def main():
         dist.all_reduce(lt)  # sum of per-shard losses = global-batch loss
         losses = lt.tolist()
 
+    # launch-origin accounting: every bucket should go out from the
+    # backward-time readiness callbacks (overlap-by-construction), not
+    # the finalize() straggler fallback
+    stats = getattr(tr.ddp, "stats", None) if tr.ddp is not None else None
     out = {"world": world, "losses": losses, "weights_sha256": whash,
            "backend": dist.get_backend() if dist.is_initialized() else None,
-           "device": str(dev)}
+           "bucket_launches": stats, "device": str(dev)}
     if rank == 0:
         os.makedirs("gpurun_out", exist_ok=True)
         ref_path = "gpurun_out/dp_parity_ref.json"
@@ -138,6 +142,7 @@ def main():
                 "final_loss_world2": out["losses"][-1],
                 "final_rel_diff": drift,
                 "rank_weights_identical": True,
+                "bucket_launches": out["bucket_launches"],
                 "device": out["device"]}))
     if dist.is_initialized():
         dist.barrier()

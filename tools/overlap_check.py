@@ -28,8 +28,12 @@ def main():
             FROM rocpd_kernel_dispatch_{uuid} d
             JOIN rocpd_info_kernel_symbol_{uuid} k ON d.kernel_id = k.id
             {join}"""))
-    comm = [(s, e) for n, s, e in rows if "ccl" in n.lower()]
-    comp = [(s, e) for n, s, e in rows if "ccl" not in n.lower()]
+    def is_comm(n):
+        nl = n.lower()
+        return ("ncclDevKernel" in n) or ("rccl" in nl) or \
+               ("nccl" in nl and "rocclr" not in nl)
+    comm = [(s, e) for n, s, e in rows if is_comm(n)]
+    comp = [(s, e) for n, s, e in rows if not is_comm(n)]
     if not comm:
         print(json.dumps({"error": "no RCCL kernels in trace"}))
         return

@@ -101,6 +101,9 @@ class BucketedDataParallel:
         ]
         from ..ops import functional as _F
         _F.set_grad_ready_callback(flat.params, self._on_param_ready)
+        # launch-origin accounting (overlap evidence: callback launches
+        # happen DURING backward, finalize launches are stragglers)
+        self.stats = {"callback": 0, "finalize": 0}
         self._reset_step()
 
     def detach(self):
@@ -121,6 +124,7 @@ class BucketedDataParallel:
         b = self.buckets[bi]
         b["pending"] -= 1
         if b["pending"] == 0:
+            self.stats["callback"] += 1
             self._launch(b)
 
     def _push_bucket(self, plist):
@@ -148,6 +152,7 @@ class BucketedDataParallel:
             b = self.buckets[bi]
             b["pending"] -= 1
             if b["pending"] == 0:
+                self.stats["callback"] += 1
                 self._launch(b)
         return hook
 
@@ -168,6 +173,7 @@ class BucketedDataParallel:
         for b in self.buckets:
             if b["work"] is None and b["pending"] > 0:
                 # params that produced no grad this step (robustness)
+                self.stats["finalize"] += 1
                 self._launch(b)
         for b in self.buckets:
             if b["work"] is not None:
