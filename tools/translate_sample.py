@@ -31,8 +31,9 @@ def main():
     dev = torch.device("cuda" if torch.cuda.is_available() else "cpu")
     dt = torch.bfloat16 if dev.type == "cuda" else torch.float32
     _, _, src_tok, tgt_tok = load_dataset(
-        a.dataset_path, "src_vocab.txt", "tgt_vocab.txt", 64,
-        a.sequence_length, buffer_size=1000)
+        a.dataset_path, "src_vocab.txt", "tgt_vocab.txt",
+        sequence_length=a.sequence_length, batch_size=64,
+        buffer_size=1000)
     model = Transformer(
         num_layers=a.num_layers, d_model=a.d_model, num_heads=a.num_heads,
         dff=a.dff, input_vocab_size=src_tok.vocab_size + 2,
