@@ -131,3 +131,26 @@ def test_serve_endpoints(tmp_path, toy_corpus):
     body = r.json()
     assert isinstance(body["tokens"], list) and len(body["tokens"]) >= 1
     assert "text" in body
+
+
+def test_restore_tolerates_shape_mismatch(tmp_path, capsys):
+    """expect_partial semantics: a checkpoint from a differently-sized
+    model must not crash restore — mismatched entries are skipped with a
+    warning (reference train.py:163)."""
+    small = Transformer(num_layers=1, d_model=16, num_heads=2, dff=32,
+                        input_vocab_size=50, target_vocab_size=50, rate=0.0,
+                        max_position=32)
+    CheckpointManager(small, None, str(tmp_path), 2).save(1)
+    big = Transformer(num_layers=1, d_model=32, num_heads=2, dff=64,
+                      input_vocab_size=50, target_vocab_size=50, rate=0.0,
+                      max_position=32)
+    mgr = CheckpointManager(big, None, str(tmp_path), 2)
+    before = {k: v.clone() for k, v in big.state_dict().items()}
+    meta = mgr.restore()
+    assert meta is not None and meta["step"] == 1
+    out = capsys.readouterr().out
+    assert "shape_mismatch" in out
+    # mismatched params untouched
+    for k, v in big.state_dict().items():
+        if v.shape != small.state_dict().get(k, v).shape:
+            assert torch.equal(v, before[k])

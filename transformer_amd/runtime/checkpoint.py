@@ -75,10 +75,19 @@ class CheckpointManager:
         if path is None or not os.path.exists(path):
             return None
         blob = torch.load(path, map_location="cpu", weights_only=False)
-        missing, unexpected = self.model.load_state_dict(blob["model"], strict=False)
-        if missing or unexpected:
+        # expect_partial semantics (reference train.py:163): tolerate
+        # missing/unexpected keys AND shape-mismatched entries (strict=False
+        # alone still raises on size mismatch)
+        state = blob["model"]
+        current = self.model.state_dict()
+        skipped = [k for k, v in state.items()
+                   if k in current and current[k].shape != v.shape]
+        for k in skipped:
+            state.pop(k)
+        missing, unexpected = self.model.load_state_dict(state, strict=False)
+        if missing or unexpected or skipped:
             print(f"[checkpoint] partial restore: missing={missing} "
-                  f"unexpected={unexpected}")
+                  f"unexpected={unexpected} shape_mismatch={skipped}")
         if self.optimizer is not None and blob.get("optimizer") is not None:
             self.optimizer.load_state_dict(blob["optimizer"])
         return {"step": blob.get("step", 0), "epoch": blob.get("epoch")}
