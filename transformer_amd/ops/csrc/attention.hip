@@ -427,8 +427,10 @@ void attn_bwd_dot_kernel(const short* __restrict__ dO,
   Dl[((b * H) + h) * S + s] = acc;
 }
 
-// dK/dV kernel: one 4-wave block per 64-key tile; loops q-tiles of 32.
-template <int DH>
+// dK/dV kernel: one 4-wave block per 64-key tile; loops q-tiles of QT
+// (32, or 64 at long sequence — halves the per-tile staging/sync fixed
+// costs, which dominate the S=4096 backward).
+template <int DH, int QT = KVT>
 __global__ __launch_bounds__(256)
 void attn_bwd_kv_kernel(const short* __restrict__ Q,
                         const short* __restrict__ K,
@@ -442,9 +444,10 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
                         long q_rs, long kv_rs, long do_rs, long dkv_rs) {
   constexpr int D32 = DH / 32;
   constexpr int D16 = DH / 16;
-  __shared__ short q_lds[KVT * DH];    // q-tile natural [32][DH]
-  __shared__ short do_lds[KVT * DH];   // dO-tile natural
-  __shared__ short x_lds[WAVES][QW * KVT];  // per-wave P^T / dS^T scratch
+  constexpr int QH = QT / 16;          // 16-q halves per tile
+  __shared__ short q_lds[QT * DH];     // q-tile natural [QT][DH]
+  __shared__ short do_lds[QT * DH];    // dO-tile natural
+  __shared__ short x_lds[WAVES][QW * QT];  // per-wave P^T / dS^T scratch
 
   const int bh = blockIdx.x;
   const int kb = blockIdx.y;           // key-block of 64
@@ -485,7 +488,7 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
   }
 
   // causal: q-rows before this key-block are fully masked
-  const int q_start = causal ? (kb * (WAVES * QW)) / KVT * KVT : 0;
+  const int q_start = causal ? (kb * (WAVES * QW)) / QT * QT : 0;
 
   // any pad among this wave's 16 OUTPUT keys (C rows k0w..k0w+15 — note
   // k_pad above is the lane's A-fragment row k0w+fr, a different role)
@@ -495,18 +498,18 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
     wave_pad_any = __any(kk < Sk && pad[min(kk, Sk - 1)]);
   }
 
-  for (int j0 = q_start; j0 < Sq; j0 += KVT) {
-    const int jc = min(KVT, Sq - j0);
-    stage_kv<DH, true, false>(Qp + (long)j0 * q_rs, q_rs, jc,
-                              q_lds, nullptr);
-    stage_kv<DH, true, false>(dOp + (long)j0 * do_rs, do_rs, jc,
-                              do_lds, nullptr);
+  for (int j0 = q_start; j0 < Sq; j0 += QT) {
+    const int jc = min(QT, Sq - j0);
+    stage_kv<DH, true, false, QT>(Qp + (long)j0 * q_rs, q_rs, jc,
+                                  q_lds, nullptr);
+    stage_kv<DH, true, false, QT>(dOp + (long)j0 * do_rs, do_rs, jc,
+                                  do_lds, nullptr);
     __syncthreads();
 
-    // pass 1: both halves of P^T into x_lds; dS^T kept in registers
-    float ds_reg[2][4];
+    // pass 1: all halves of P^T into x_lds; dS^T kept in registers
+    float ds_reg[QH][4];
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
+    for (int half = 0; half < QH; ++half) {
       // S^T[key][q] = K·Q^T ; dP^T[key][q] = V·dO^T   (C row=key, col=q)
       f32x4 st = {0, 0, 0, 0}, dpt = {0, 0, 0, 0};
       __builtin_amdgcn_s_setprio(1);
@@ -551,18 +554,23 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
       scrT_write4(x_lds[wid], half * 16 + fr, kg * 4, pw);
     }
     __builtin_amdgcn_wave_barrier();
-    // dV += P^T · dO   (A = full P^T[key][q0..31], B = dO^T[d][q])
+    // dV += P^T · dO   (A = full P^T[key][q-chunk], B = dO^T[d][q])
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int i = 0; i < D16; ++i) {
-      bf16x8 pa = scrT_read8(x_lds[wid], kg * 8);
-      bf16x8 db = lds_read8_tr<DH * 2>(do_lds, kg * 8, i * 16);
-      acc_dv[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, db, acc_dv[i], 0, 0, 0);
+    for (int ks = 0; ks < QT / 32; ++ks) {
+      bf16x8 pa = scrT_read8(x_lds[wid], ks * 32 + kg * 8);
+#pragma unroll
+      for (int i = 0; i < D16; ++i) {
+        bf16x8 db = lds_read8_tr<DH * 2>(do_lds, ks * 32 + kg * 8, i * 16);
+        acc_dv[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, db,
+                                                            acc_dv[i],
+                                                            0, 0, 0);
+      }
     }
     __builtin_amdgcn_s_setprio(0);
-    // pass 2: both halves of dS^T, then dK += dS^T · Q
+    // pass 2: all halves of dS^T, then dK += dS^T · Q
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
+    for (int half = 0; half < QH; ++half) {
       s16x4 dw = {f2bfbits(ds_reg[half][0]), f2bfbits(ds_reg[half][1]),
                   f2bfbits(ds_reg[half][2]), f2bfbits(ds_reg[half][3])};
       scrT_write4(x_lds[wid], half * 16 + fr, kg * 4, dw);
@@ -570,10 +578,15 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
     __builtin_amdgcn_wave_barrier();
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int i = 0; i < D16; ++i) {
-      bf16x8 sa = scrT_read8(x_lds[wid], kg * 8);
-      bf16x8 qb2 = lds_read8_tr<DH * 2>(q_lds, kg * 8, i * 16);
-      acc_dk[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(sa, qb2, acc_dk[i], 0, 0, 0);
+    for (int ks = 0; ks < QT / 32; ++ks) {
+      bf16x8 sa = scrT_read8(x_lds[wid], ks * 32 + kg * 8);
+#pragma unroll
+      for (int i = 0; i < D16; ++i) {
+        bf16x8 qb2 = lds_read8_tr<DH * 2>(q_lds, ks * 32 + kg * 8, i * 16);
+        acc_dk[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(sa, qb2,
+                                                            acc_dk[i],
+                                                            0, 0, 0);
+      }
     }
     __builtin_amdgcn_s_setprio(0);
     __syncthreads();
@@ -595,8 +608,8 @@ void attn_bwd_kv_kernel(const short* __restrict__ Q,
   }
 }
 
-// dQ kernel: one 4-wave block per 64-q tile; loops kv-tiles of 32.
-template <int DH>
+// dQ kernel: one 4-wave block per 64-q tile; loops kv-tiles of QT.
+template <int DH, int QT = KVT>
 __global__ __launch_bounds__(256)
 void attn_bwd_q_kernel(const short* __restrict__ Q,
                        const short* __restrict__ K,
@@ -610,9 +623,10 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
                        long do_rs, long dq_rs) {
   constexpr int D32 = DH / 32;
   constexpr int D16 = DH / 16;
-  __shared__ short k_lds[KVT * DH];
-  __shared__ short v_lds[KVT * DH];
-  __shared__ short x_lds[WAVES][QW * KVT];
+  constexpr int QH = QT / 16;
+  __shared__ short k_lds[QT * DH];
+  __shared__ short v_lds[QT * DH];
+  __shared__ short x_lds[WAVES][QW * QT];
 
   const int bh = blockIdx.x;
   const int qb = blockIdx.y;
@@ -649,16 +663,16 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
 
   const int kend = causal ? min(Sk, qb * (WAVES * QW) + WAVES * QW) : Sk;
 
-  for (int k0 = 0; k0 < kend; k0 += KVT) {
-    const int kc = min(KVT, Sk - k0);
-    stage_kv<DH, true, false>(Kp + (long)k0 * kv_rs, kv_rs, kc,
-                              k_lds, nullptr);
-    stage_kv<DH, true, false>(Vp + (long)k0 * kv_rs, kv_rs, kc,
-                              v_lds, nullptr);
+  for (int k0 = 0; k0 < kend; k0 += QT) {
+    const int kc = min(QT, Sk - k0);
+    stage_kv<DH, true, false, QT>(Kp + (long)k0 * kv_rs, kv_rs, kc,
+                                  k_lds, nullptr);
+    stage_kv<DH, true, false, QT>(Vp + (long)k0 * kv_rs, kv_rs, kc,
+                                  v_lds, nullptr);
     __syncthreads();
 
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
+    for (int half = 0; half < QH; ++half) {
       f32x4 s = {0, 0, 0, 0}, dp = {0, 0, 0, 0};
       s16x4 dsw;
       __builtin_amdgcn_s_setprio(1);
@@ -701,10 +715,15 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
     // dQ += dS · K   (A = dS[q][key] from LDS, B = K^T[d][key])
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int i = 0; i < D16; ++i) {
-      bf16x8 sa = scrT_read8(x_lds[wid], kg * 8);
-      bf16x8 kb2 = lds_read8_tr<DH * 2>(k_lds, kg * 8, i * 16);
-      acc_dq[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(sa, kb2, acc_dq[i], 0, 0, 0);
+    for (int ks = 0; ks < QT / 32; ++ks) {
+      bf16x8 sa = scrT_read8(x_lds[wid], ks * 32 + kg * 8);
+#pragma unroll
+      for (int i = 0; i < D16; ++i) {
+        bf16x8 kb2 = lds_read8_tr<DH * 2>(k_lds, ks * 32 + kg * 8, i * 16);
+        acc_dq[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(sa, kb2,
+                                                            acc_dq[i],
+                                                            0, 0, 0);
+      }
     }
     __builtin_amdgcn_s_setprio(0);
     __syncthreads();
@@ -862,17 +881,32 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   dim3 grid_q(B * H, cdiv(Sq, WAVES * QW));
   const long q_rs = q.stride(1), kv_rs = k.stride(1);
   const long do_rs = (long)H * DH;
+  // QT=64 at long sequence halves the per-tile staging/sync fixed costs
+  // (the S=4096 backward's dominant overhead); QT=32 keeps short-seq
+  // tail efficiency.  TFMX_ATTN_QT forces either for A/B.
+  static const int qt_force = [] {
+    const char* e = getenv("TFMX_ATTN_QT");
+    return e ? atoi(e) : 0;
+  }();
+  const bool qt64 = qt_force ? (qt_force == 64)
+                             : (Sq >= 1024 && Sk >= 1024);
   DISPATCH_DH(DH, {
-    attn_bwd_kv_kernel<DHC><<<grid_kv, 256, 0, stream>>>(
-        (const short*)q.data_ptr(), (const short*)k.data_ptr(),
-        (const short*)v.data_ptr(), (const short*)dout.data_ptr(),
-        lse.data_ptr<float>(), dl.data_ptr<float>(), pad, dk_p, dv_p, B, H,
-        Sq, Sk, causal ? 1 : 0, (float)scale, q_rs, kv_rs, do_rs, dkv_rs);
-    attn_bwd_q_kernel<DHC><<<grid_q, 256, 0, stream>>>(
-        (const short*)q.data_ptr(), (const short*)k.data_ptr(),
-        (const short*)v.data_ptr(), (const short*)dout.data_ptr(),
-        lse.data_ptr<float>(), dl.data_ptr<float>(), pad, dq_p, B, H, Sq, Sk,
-        causal ? 1 : 0, (float)scale, q_rs, kv_rs, do_rs, dq_rs);
+    auto run = [&](auto qtc) {
+      constexpr int QTC = decltype(qtc)::value;
+      attn_bwd_kv_kernel<DHC, QTC><<<grid_kv, 256, 0, stream>>>(
+          (const short*)q.data_ptr(), (const short*)k.data_ptr(),
+          (const short*)v.data_ptr(), (const short*)dout.data_ptr(),
+          lse.data_ptr<float>(), dl.data_ptr<float>(), pad, dk_p, dv_p, B,
+          H, Sq, Sk, causal ? 1 : 0, (float)scale, q_rs, kv_rs, do_rs,
+          dkv_rs);
+      attn_bwd_q_kernel<DHC, QTC><<<grid_q, 256, 0, stream>>>(
+          (const short*)q.data_ptr(), (const short*)k.data_ptr(),
+          (const short*)v.data_ptr(), (const short*)dout.data_ptr(),
+          lse.data_ptr<float>(), dl.data_ptr<float>(), pad, dq_p, B, H, Sq,
+          Sk, causal ? 1 : 0, (float)scale, q_rs, kv_rs, do_rs, dq_rs);
+    };
+    if (qt64) run(std::integral_constant<int, 64>{});
+    else run(std::integral_constant<int, 32>{});
   });
   return outs;
 }
