@@ -661,6 +661,18 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
 #pragma unroll
   for (int i = 0; i < D16; ++i) acc_dq[i] = {0, 0, 0, 0};
 
+  // this lane's 4 q-rows are FIXED for the whole kv loop — hoist their
+  // lse/D values (they were re-loaded per tile x half: 2 scalar loads x
+  // 4 rows x 2 halves x Sk/QT tiles)
+  float lse_r[4], dl_r[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = q0 + kg * 4 + r;
+    const bool ok = qrow < Sq;
+    lse_r[r] = ok ? lse[min(qrow, Sq - 1)] : 0.f;
+    dl_r[r] = ok ? dl[min(qrow, Sq - 1)] : 0.f;
+  }
+
   const int kend = causal ? min(Sk, qb * (WAVES * QW) + WAVES * QW) : Sk;
 
   for (int k0 = 0; k0 < kend; k0 += QT) {
@@ -693,20 +705,16 @@ void attn_bwd_q_kernel(const short* __restrict__ Q,
       if (clean2) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          const int qrow = q0 + kg * 4 + r;
-          float p = __expf(s[r] * scale - lse[qrow]);
-          dsw[r] = f2bfbits(p * (dp[r] - dl[qrow]) * scale);
+          float p = __expf(s[r] * scale - lse_r[r]);
+          dsw[r] = f2bfbits(p * (dp[r] - dl_r[r]) * scale);
         }
       } else {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int qrow = q0 + kg * 4 + r;
-          const bool row_ok = qrow < Sq;
-          float lse_q = row_ok ? lse[min(qrow, Sq - 1)] : 0.f;
-          float d_q = row_ok ? dl[min(qrow, Sq - 1)] : 0.f;
-          bool masked = !row_ok || col_pad || (causal && kcol > qrow);
-          float p = masked ? 0.f : __expf(s[r] * scale - lse_q);
-          dsw[r] = f2bfbits(p * (dp[r] - d_q) * scale);
+          bool masked = (qrow >= Sq) || col_pad || (causal && kcol > qrow);
+          float p = masked ? 0.f : __expf(s[r] * scale - lse_r[r]);
+          dsw[r] = f2bfbits(p * (dp[r] - dl_r[r]) * scale);
         }
       }
       scrT_write4(x_lds[wid], half * 16 + fr, kg * 4, dsw);
