@@ -237,11 +237,20 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
 // Tile choice: widest instance whose grid still fills the 256-CU chip
 // (narrower tiles re-read operands more, so wider wins when it fits —
 // measured: BN=128 wins FFN2-class shapes +8% but loses on the logits dW
-// where the wide grid already fills).
+// where the wide grid already fills), EXCEPT when the wide grid lands
+// badly off a multiple of 256 blocks: at 1 block/CU a 1.5-wave grid
+// (e.g. QKV fwd: 64x6 = 384) idles a quarter of the chip for half the
+// kernel — the narrower tile's full waves win despite the re-reads.
 static void gemm256_tile(int M, int N, int& BM, int& BN) {
   const long g22 = (long)cdiv(M, 256) * cdiv(N, 256);
   const long g21 = (long)cdiv(M, 256) * cdiv(N, 128);
-  if (g22 >= 224) { BM = 256; BN = 256; }
+  auto util = [](long nwg) {
+    return nwg < 1 ? 0.0 : (double)nwg / ((nwg + 255) / 256 * 256);
+  };
+  if (g22 >= 224) {
+    BM = 256;
+    BN = (util(g21) > util(g22) + 0.1) ? 128 : 256;
+  }
   else if (g21 >= 224) { BM = 256; BN = 128; }
   else { BM = 128; BN = 128; }
 }

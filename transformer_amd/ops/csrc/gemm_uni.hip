@@ -401,8 +401,15 @@ static torch::Tensor gemm_uni_launch(
   TORCH_CHECK(K % U_BK == 0 && K >= 2 * U_BK, "gemm_uni: K must be n*64");
   TORCH_CHECK(splitr <= 1 || (epilogue == 0 && !has_bias),
               "gemm_uni: split-contraction excludes bias/activation");
-  // widest tile whose grid still fills the 256-CU chip
-  const int BN = ((long)cdiv(M, 256) * cdiv(N, 256) >= 224) ? 256 : 128;
+  // widest tile whose grid still fills the 256-CU chip; prefer the
+  // narrow tile when the wide grid is badly off a 256-block wave (the
+  // 1-block/CU quantization cliff — see gemm256_tile)
+  const long g22 = (long)cdiv(M, 256) * cdiv(N, 256);
+  const long g21 = (long)cdiv(M, 256) * cdiv(N, 128);
+  auto util = [](long nwg) {
+    return nwg < 1 ? 0.0 : (double)nwg / ((nwg + 255) / 256 * 256);
+  };
+  const int BN = (g22 >= 224 && !(util(g21) > util(g22) + 0.1)) ? 256 : 128;
   const int nbm = cdiv(M, 256), nbn = cdiv(N, BN);
   // TR staging reads the full tile span of the out axis — the allocation
   // must cover it (ragged extents arrive padded: ce_bwd / _wt_padded).
