@@ -82,7 +82,7 @@ DEV_INLINE bf16x8g frag256(const short* lds, int row, int kb /*bytes*/) {
 // BM_/BN_ in {256, 128}: per-wave output (BM_/2) x (BN_/4); narrower
 // instances keep the chip full on small-M/N shapes with the same
 // pipelined phase structure.
-template <int EPILOGUE, int BM_, int BN_>
+template <int EPILOGUE, int BM_, int BN_, int ORDER = 0, int XCDMAP = 1>
 __global__ __launch_bounds__(G_THREADS, 1)
 void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
                     const short* __restrict__ bias, short* __restrict__ C,
@@ -96,15 +96,18 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
   extern __shared__ short smem[];
   const int lane = threadIdx.x & 63;
 
-  // XCD-aware bijective workgroup remap (8 XCDs, private L2s).
+  // XCD-aware bijective workgroup remap (8 XCDs, private L2s) + tile
+  // walk order (ORDER 0: m-major — consecutive wgs share the A panel;
+  // 1: n-major — share the B panel.  At L3-resident huge-N shapes the
+  // walk order decides which operand streams from HBM).
   int nwg = nbm * nbn;
   int wg = blockIdx.x;
-  {
+  if (XCDMAP) {
     int q = nwg / 8, r = nwg % 8, x = wg % 8, o = wg / 8;
     wg = (x < r ? x * (q + 1) : r * (q + 1) + (x - r) * q) + o;
   }
-  const int bm0 = (wg / nbn) * BM_;
-  const int bn0 = (wg % nbn) * BN_;
+  const int bm0 = (ORDER ? (wg % nbm) : (wg / nbn)) * BM_;
+  const int bn0 = (ORDER ? (wg / nbm) : (wg % nbn)) * BN_;
 
   const int wid = threadIdx.x >> 6;
   const int wm = (wid >> 2) * (BM_ / 2);  // wave rows
@@ -293,21 +296,43 @@ torch::Tensor gemm256_nt(torch::Tensor a, torch::Tensor w, torch::Tensor bias,
   const size_t smem = 2 * ((size_t)BMv + BNv) * G_BK * sizeof(short);
   auto stream = at::hip::getCurrentHIPStream();
   static bool attr_set[2][3] = {};
+  // tile-walk order / XCD-remap variants (TFMX_G256_ORDER=nm,
+  // TFMX_G256_XCD=0 for the A/B prober; huge-N shapes pick n-major by
+  // the measured rule below)
+  static const int env_order = [] {
+    const char* e = getenv("TFMX_G256_ORDER");
+    return e ? (e[0] == 'n' ? 1 : 0) : -1;
+  }();
+  static const int env_xcd = [] {
+    const char* e = getenv("TFMX_G256_XCD");
+    return e ? atoi(e) : -1;
+  }();
+  const int order = env_order >= 0 ? env_order : 0;
+  const int xcd = env_xcd >= 0 ? env_xcd : 1;
   auto launch = [&](auto epi, auto bmc, auto bnc) {
     constexpr int E = decltype(epi)::value;
     constexpr int BM_ = decltype(bmc)::value;
     constexpr int BN_ = decltype(bnc)::value;
     constexpr int ti = BM_ == 128 ? 2 : (BN_ == 128 ? 1 : 0);
-    if (!attr_set[E][ti]) {
-      (void)hipFuncSetAttribute((const void*)gemm256_kernel<E, BM_, BN_>,
-                                hipFuncAttributeMaxDynamicSharedMemorySize,
-                                (int)smem);
-      attr_set[E][ti] = true;
-    }
-    gemm256_kernel<E, BM_, BN_><<<nbm * nbn, G_THREADS, smem, stream>>>(
-        (const short*)a.data_ptr(), (const short*)w.data_ptr(),
-        has_bias ? (const short*)bias.data_ptr() : nullptr,
-        (short*)c.data_ptr(), M, N, K, K, K, has_bias, nbm, nbn);
+    auto go = [&](auto oc, auto xc) {
+      constexpr int OV = decltype(oc)::value;
+      constexpr int XV = decltype(xc)::value;
+      if (!attr_set[E][ti]) {
+        (void)hipFuncSetAttribute(
+            (const void*)gemm256_kernel<E, BM_, BN_, OV, XV>,
+            hipFuncAttributeMaxDynamicSharedMemorySize, (int)smem);
+        attr_set[E][ti] = true;
+      }
+      gemm256_kernel<E, BM_, BN_, OV, XV>
+          <<<nbm * nbn, G_THREADS, smem, stream>>>(
+              (const short*)a.data_ptr(), (const short*)w.data_ptr(),
+              has_bias ? (const short*)bias.data_ptr() : nullptr,
+              (short*)c.data_ptr(), M, N, K, K, K, has_bias, nbm, nbn);
+    };
+    using I0 = std::integral_constant<int, 0>;
+    using I1 = std::integral_constant<int, 1>;
+    if (order) { if (xcd) go(I1{}, I1{}); else go(I1{}, I0{}); }
+    else       { if (xcd) go(I0{}, I1{}); else go(I0{}, I0{}); }
   };
   using E0 = std::integral_constant<int, 0>;
   using E1 = std::integral_constant<int, 1>;
