@@ -254,8 +254,11 @@ class Transformer(nn.Module):
         inp, tar = inputs
         # mask construction (C5) folds into the attention kernels' predicate
         # logic: only per-token pad flags cross the boundary.
-        src_pad = (inp == 0)
-        tgt_pad = (tar == 0)
+        # Converted to uint8 ONCE here — the per-layer kernel wrappers'
+        # .to(uint8).contiguous() then no-op instead of copying per layer
+        # (was 18 copy kernels/step).
+        src_pad = (inp == 0).to(torch.uint8)
+        tgt_pad = (tar == 0).to(torch.uint8)
         enc_output = self.encoder(inp, src_pad, training)
         dec_output, attention_weights = self.decoder(
             tar, enc_output, tgt_pad, src_pad, training, return_weights)
@@ -304,7 +307,7 @@ class DecodeCache:
 @torch.no_grad()
 def encode(model, inp, training=False):
     """Run the encoder once; returns (enc_output, src_pad)."""
-    src_pad = (inp == 0)
+    src_pad = (inp == 0).to(torch.uint8)
     return model.encoder(inp, src_pad, training), src_pad
 
 
