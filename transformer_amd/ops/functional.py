@@ -88,6 +88,10 @@ def bump_weight_version():
 
 _WTP_CACHE: dict = {}  # id(weight) -> [version, buf, weight]
 _WTP_TABLE = [None]    # cached transpose_batch descriptor table
+# A captured hipGraph holds the table ADDRESS it recorded; if the registry
+# grows later and the table is rebuilt, the old tensor must stay alive or
+# replays would read freed memory.
+_WTP_TABLES_KEEP: list = []
 
 
 def _build_wtp_table(device):
@@ -99,7 +103,9 @@ def _build_wtp_table(device):
             for bn in range(0, N, 64):
                 rows.append([w.data_ptr(), buf.data_ptr(), M, N, ldo,
                              bm, bn])
-    return torch.tensor(rows, dtype=torch.int64).to(device)
+    t = torch.tensor(rows, dtype=torch.int64).to(device)
+    _WTP_TABLES_KEEP.append(t)
+    return t
 
 
 def prepare_weight_caches():
