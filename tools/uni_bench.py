@@ -71,6 +71,9 @@ DW = [
     (16384, 512, 2048, "FFN2 dW"),
     (16320, 32768, 512, "logits dW (aligned probe)"),
     (16384, 3072, 1024, "big QKV dW"),
+    (16384, 1024, 1024, "big O dW"),
+    (16384, 4096, 1024, "big FFN1 dW"),
+    (16384, 1024, 4096, "big FFN2 dW"),
 ]
 
 

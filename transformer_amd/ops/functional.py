@@ -1,16 +1,17 @@
 """Autograd-integrated op layer with device dispatch.
 
 GPU (CUDA/ROCm) tensors run via `torch.autograd.Function` wrappers
-(SURVEY.md §2.3): the fused/custom ops (attention fwd+bwd, dW-transpose
-GEMM, residual-LN(+dropout), CE, Adam, embedding+PE, dropout, argmax/
-accuracy) are hand-written CDNA4 HIP kernels; PLAIN GEMMs go to the
-backend that measured fastest per shape (hipBLASLt fused bias/ReLU
-epilogues for training-shape forwards and for dX / the wide logits dW;
-the hand-written NT/dW kernels for small-M forwards and the d_model dW
-shapes — see docs/PERF.md "GEMM backends").  CPU tensors run the fp32
-reference math in reference.py through plain differentiable torch ops.
-The GPU path never falls back silently: if the extension is missing it
-raises (see ops/__init__.py).
+(SURVEY.md §2.3), and — since round 2 — EVERY training-shape op runs a
+hand-written CDNA4 HIP kernel, GEMMs included: forwards on the
+gemm256/gemm128 NT dispatch, dX against a per-step cached transposed
+weight (ONE batched-transpose launch/step), the d_model dW shapes on the
+split-contraction tr16 kernel and the wide logits dW on the TRxTR
+gemm_uni kernel.  Every dispatch choice is measured per shape
+(docs/PERF.md round 2); hipBLASLt survives only as the
+TFMX_FWD_GEMM=blaslt / TFMX_DX=blaslt A/B arms.  CPU tensors run the
+fp32 reference math in reference.py through plain differentiable torch
+ops.  The GPU path never falls back silently: if the extension is
+missing it raises (see ops/__init__.py).
 """
 
 from __future__ import annotations
