@@ -752,3 +752,17 @@ def test_attn_defer_max_spike_forces_rescale():
     ref2, _ = R.scaled_dot_product_attention(qt, kt2, vt, None,
                                              return_weights=True)
     assert_close(o2, ref2.permute(0, 2, 1, 3), 0.04, "defer-max smooth")
+
+
+def test_relu_bwd_db_fused():
+    """dz = dy*(y>0) with fused bias grad vs separate reference, under
+    repeated L1-warm calls on the cached workspace."""
+    torch.manual_seed(13)
+    for it in range(10):
+        M, N = 2048 + 256 * (it % 3), 2048
+        dy = torch.randn(M, N, device="cuda", dtype=torch.bfloat16)
+        y = torch.randn(M, N, device="cuda", dtype=torch.bfloat16)
+        dz, db = _ext().relu_bwd_db(dy, y)
+        ref_dz = dy.float() * (y.float() > 0)
+        assert_close(dz, ref_dz, 1e-3, f"relu_bwd_db dz it={it}")
+        assert_close(db, ref_dz.sum(0), 0.02, f"relu_bwd_db db it={it}")

@@ -35,6 +35,8 @@ bool gemm_uni_viable(int M, int N, int K);
 torch::Tensor gemm_uni_nt_ab(torch::Tensor a, torch::Tensor w,
                              int64_t sched);
 torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y);
+std::vector<torch::Tensor> relu_bwd_db(torch::Tensor dy, torch::Tensor y,
+                                       c10::optional<torch::Tensor> db_out);
 torch::Tensor smoke_add(torch::Tensor a, torch::Tensor b);
 
 std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor res,
@@ -135,6 +137,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum", &colsum, pybind11::arg("a"),
         pybind11::arg("out") = pybind11::none());
   m.def("relu_bwd", &relu_bwd);
+  m.def("relu_bwd_db", &relu_bwd_db,
+        "dz = dy*(y>0) with the bias grad db = colsum(dz) fused",
+        pybind11::arg("dy"), pybind11::arg("y"),
+        pybind11::arg("db") = pybind11::none());
   m.def("smoke_add", &smoke_add);
   m.def("ln_fwd", &ln_fwd, pybind11::arg("x"), pybind11::arg("res"),
         pybind11::arg("gamma"), pybind11::arg("beta"), pybind11::arg("eps"),

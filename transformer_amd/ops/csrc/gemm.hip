@@ -414,6 +414,49 @@ void colsum_kernel(const short* __restrict__ dy, float* __restrict__ acc,
   }
 }
 
+// relu_bwd fused with the bias grad: dz = dy * (y > 0), db = colsum(dz).
+// Column-parallel layout (thread = column, row loop — the same coalesced
+// pattern as colsum_kernel) so the separate colsum pass's full re-read of
+// dz (64 MB per FFN1 backward) disappears; finalize via last-arriver.
+#define RELU_ROWS 128
+__global__ __launch_bounds__(256)
+void relu_bwd_db_kernel(const short* __restrict__ dy,
+                        const short* __restrict__ y,
+                        short* __restrict__ dz, float* __restrict__ acc,
+                        short* __restrict__ db, int M, int N) {
+  int n = blockIdx.x * 256 + threadIdx.x;
+  if (n < N) {
+    long m0 = (long)blockIdx.y * RELU_ROWS;
+    long m1 = min((long)M, m0 + RELU_ROWS);
+    float s0 = 0.f, s1 = 0.f;
+    long m = m0;
+    for (; m + 2 <= m1; m += 2) {
+      short a0 = (bfbits2f(y[m * N + n]) > 0.f) ? dy[m * N + n] : (short)0;
+      short a1 = (bfbits2f(y[(m + 1) * N + n]) > 0.f) ? dy[(m + 1) * N + n]
+                                                      : (short)0;
+      dz[m * N + n] = a0;
+      dz[(m + 1) * N + n] = a1;
+      s0 += bfbits2f(a0);
+      s1 += bfbits2f(a1);
+    }
+    for (; m < m1; ++m) {
+      short a = (bfbits2f(y[m * N + n]) > 0.f) ? dy[m * N + n] : (short)0;
+      dz[m * N + n] = a;
+      s0 += bfbits2f(a);
+    }
+    atomicAdd(&acc[n], s0 + s1);
+  }
+  if (last_arriver((unsigned*)(acc + N), gridDim.x * gridDim.y)) {
+    for (int i = threadIdx.x; i < N; i += 256) {
+      db[i] = f2bfbits(acc[i]);
+      acc[i] = 0.f;
+    }
+    if (threadIdx.x == 0)
+      __hip_atomic_store((unsigned*)(acc + N), 0u, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+  }
+}
+
 // relu_bwd: dz = dy * (y > 0)
 __global__ void relu_bwd_kernel(const short* __restrict__ dy,
                                 const short* __restrict__ y,
@@ -638,6 +681,38 @@ torch::Tensor colsum(torch::Tensor a, c10::optional<torch::Tensor> out_opt) {
       (const short*)a.data_ptr(), acc.data_ptr<float>(),
       (short*)out.data_ptr(), M, N, lda);
   return out;
+}
+
+// Fused variant: returns (dz, db).  Reuses colsum's cached fp32
+// workspace (N floats + counter) keyed per (device, N).
+std::vector<torch::Tensor> relu_bwd_db(torch::Tensor dy, torch::Tensor y,
+                                       c10::optional<torch::Tensor> db_out) {
+  TORCH_CHECK(dy.is_cuda() && dy.dtype() == torch::kBFloat16 &&
+              dy.dim() == 2 && dy.is_contiguous() && y.is_contiguous() &&
+              y.sizes() == dy.sizes(), "relu_bwd_db: bad inputs");
+  const int M = dy.size(0), N = dy.size(1);
+  auto dz = torch::empty_like(dy);
+  torch::Tensor db;
+  if (db_out.has_value()) {
+    db = *db_out;
+    TORCH_CHECK(db.is_cuda() && db.dtype() == torch::kBFloat16 &&
+                db.is_contiguous() && db.numel() == N);
+  } else {
+    db = torch::empty({N}, dy.options());
+  }
+  static std::map<std::pair<int, int>, torch::Tensor> ws_cache;
+  auto key = std::make_pair((int)dy.get_device(), N);
+  auto it = ws_cache.find(key);
+  if (it == ws_cache.end())
+    it = ws_cache.emplace(key, torch::zeros(
+        {N + 1}, dy.options().dtype(torch::kFloat32))).first;
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 grid(cdiv(N, 256), cdiv(M, RELU_ROWS));
+  relu_bwd_db_kernel<<<grid, 256, 0, stream>>>(
+      (const short*)dy.data_ptr(), (const short*)y.data_ptr(),
+      (short*)dz.data_ptr(), it->second.data_ptr<float>(),
+      (short*)db.data_ptr(), M, N);
+  return {dz, db};
 }
 
 torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor y) {

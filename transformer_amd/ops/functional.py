@@ -270,12 +270,21 @@ class _LinearFn(torch.autograd.Function):
     def backward(ctx, dy):
         E = ext()
         x, w, y = ctx.saved_tensors
+        db = None
         if ctx.activation == "relu":
-            dy = E.relu_bwd(dy.contiguous(), y)  # dz = dy * (y > 0)
+            if ctx.has_bias:
+                # dz and its column sum in one pass (the separate colsum
+                # re-read of dz was a full extra HBM sweep per FFN1 bwd)
+                dy, db = E.relu_bwd_db(dy.contiguous(), y)
+            else:
+                dy = E.relu_bwd(dy.contiguous(), y)  # dz = dy * (y > 0)
         else:
             dy = _dense2d(dy)
         dx = _dx_gemm(E, dy, w)         # dX[M,K] = dY[M,N] @ W[N,K]
-        dw, db = _dw_db_gemm(dy, x, ctx.has_bias)
+        if db is None:
+            dw, db = _dw_db_gemm(dy, x, ctx.has_bias)
+        else:
+            dw = _dw_gemm(dy, x)
         return dx, dw, db, None
 
 
@@ -301,14 +310,23 @@ class _LinearFlatFn(torch.autograd.Function):
         E = ext()
         x, y = ctx.saved_tensors
         w, b = ctx.wb
+        done_db = False
         if ctx.activation == "relu":
-            dy = E.relu_bwd(dy.contiguous(), y)
+            if b is not None:
+                dy, _ = E.relu_bwd_db(dy.contiguous(), y,
+                                      _flat(b).view(-1))
+                done_db = True
+            else:
+                dy = E.relu_bwd(dy.contiguous(), y)
         else:
             dy = _dense2d(dy)
         dx = _dx_gemm(E, dy, w)
-        _dw_db_gemm(dy, x, b is not None,
-                    w_out=_flat(w).view(w.shape[0], -1),
-                    b_out=_flat(b).view(-1) if b is not None else None)
+        if done_db:
+            _dw_gemm(dy, x, out=_flat(w).view(w.shape[0], -1))
+        else:
+            _dw_db_gemm(dy, x, b is not None,
+                        w_out=_flat(w).view(w.shape[0], -1),
+                        b_out=_flat(b).view(-1) if b is not None else None)
         _grad_ready(w, b)
         return dx, None, None
 
