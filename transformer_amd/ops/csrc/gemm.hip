@@ -418,33 +418,46 @@ void colsum_kernel(const short* __restrict__ dy, float* __restrict__ acc,
 // Column-parallel layout (thread = column, row loop — the same coalesced
 // pattern as colsum_kernel) so the separate colsum pass's full re-read of
 // dz (64 MB per FFN1 backward) disappears; finalize via last-arriver.
-#define RELU_ROWS 128
+#define RELU_ROWS 64
 __global__ __launch_bounds__(256)
 void relu_bwd_db_kernel(const short* __restrict__ dy,
                         const short* __restrict__ y,
                         short* __restrict__ dz, float* __restrict__ acc,
                         short* __restrict__ db, int M, int N) {
-  int n = blockIdx.x * 256 + threadIdx.x;
-  if (n < N) {
+  // 8 adjacent columns per thread with s16x8 loads/stores (scalar
+  // column loads measured 2.3x off roofline), 8 fp32 accumulators,
+  // 8 atomics per thread at the end.
+  const int n8 = (blockIdx.x * 256 + threadIdx.x) * 8;
+  if (n8 + 8 <= N) {
     long m0 = (long)blockIdx.y * RELU_ROWS;
     long m1 = min((long)M, m0 + RELU_ROWS);
-    float s0 = 0.f, s1 = 0.f;
-    long m = m0;
-    for (; m + 2 <= m1; m += 2) {
-      short a0 = (bfbits2f(y[m * N + n]) > 0.f) ? dy[m * N + n] : (short)0;
-      short a1 = (bfbits2f(y[(m + 1) * N + n]) > 0.f) ? dy[(m + 1) * N + n]
+    float sc[8] = {0.f};
+    for (long m = m0; m < m1; ++m) {
+      s16x8 dv = *(const s16x8*)(dy + m * N + n8);
+      s16x8 yv = *(const s16x8*)(y + m * N + n8);
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        o[j] = (bfbits2f(yv[j]) > 0.f) ? dv[j] : (short)0;
+        sc[j] += bfbits2f(o[j]);
+      }
+      *(s16x8*)(dz + m * N + n8) = o;
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) atomicAdd(&acc[n8 + j], sc[j]);
+  } else if (n8 < N) {
+    long m0 = (long)blockIdx.y * RELU_ROWS;
+    long m1 = min((long)M, m0 + RELU_ROWS);
+    for (int j = 0; n8 + j < N; ++j) {
+      float sc = 0.f;
+      for (long m = m0; m < m1; ++m) {
+        short a = (bfbits2f(y[m * N + n8 + j]) > 0.f) ? dy[m * N + n8 + j]
                                                       : (short)0;
-      dz[m * N + n] = a0;
-      dz[(m + 1) * N + n] = a1;
-      s0 += bfbits2f(a0);
-      s1 += bfbits2f(a1);
+        dz[m * N + n8 + j] = a;
+        sc += bfbits2f(a);
+      }
+      atomicAdd(&acc[n8 + j], sc);
     }
-    for (; m < m1; ++m) {
-      short a = (bfbits2f(y[m * N + n]) > 0.f) ? dy[m * N + n] : (short)0;
-      dz[m * N + n] = a;
-      s0 += bfbits2f(a);
-    }
-    atomicAdd(&acc[n], s0 + s1);
   }
   if (last_arriver((unsigned*)(acc + N), gridDim.x * gridDim.y)) {
     for (int i = threadIdx.x; i < N; i += 256) {
@@ -707,7 +720,7 @@ std::vector<torch::Tensor> relu_bwd_db(torch::Tensor dy, torch::Tensor y,
     it = ws_cache.emplace(key, torch::zeros(
         {N + 1}, dy.options().dtype(torch::kFloat32))).first;
   auto stream = at::hip::getCurrentHIPStream();
-  dim3 grid(cdiv(N, 256), cdiv(M, RELU_ROWS));
+  dim3 grid(cdiv(N, 256 * 8), cdiv(M, RELU_ROWS));
   relu_bwd_db_kernel<<<grid, 256, 0, stream>>>(
       (const short*)dy.data_ptr(), (const short*)y.data_ptr(),
       (short*)dz.data_ptr(), it->second.data_ptr<float>(),
