@@ -281,3 +281,35 @@ def test_distributed_train_entry_point_subprocess(tmp_path, toy_corpus):
         capture_output=True, text=True, timeout=540)
     assert out.returncode == 0, (out.stderr[-3000:], out.stdout[-1000:])
     assert os.path.isdir(tmp_path / "ckpt"), "no rank-0 checkpoint directory"
+
+
+def test_ddp_force_mode_self_collectives(monkeypatch):
+    """TFMX_DDP_FORCE=1 runs the bucket collectives even at world=1 (the
+    single-GPU overlap-evidence mode): buckets must launch from the
+    readiness callbacks and finalize must complete."""
+    import os
+    import torch
+    import torch.distributed as dist
+    from transformer_amd.models import Transformer
+    from transformer_amd.runtime.optimizer import FlatParams
+    from transformer_amd.parallel import BucketedDataParallel
+
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29511")
+        dist.init_process_group("gloo", rank=0, world_size=1)
+    monkeypatch.setenv("TFMX_DDP_FORCE", "1")
+    m = Transformer(num_layers=1, d_model=16, num_heads=2, dff=32,
+                    input_vocab_size=30, target_vocab_size=30, rate=0.0,
+                    max_position=16)
+    flat = FlatParams(m)
+    ddp = BucketedDataParallel(flat, bucket_mb=0.01)
+    assert ddp.active
+    inp = torch.randint(1, 30, (2, 6))
+    tar = torch.randint(1, 30, (2, 6))
+    logits, _ = m((inp, tar), training=True)
+    logits.sum().backward()
+    ddp.finalize()
+    assert ddp.stats["callback"] + ddp.stats["finalize"] == len(ddp.buckets)
+    assert ddp.stats["callback"] > 0
+    ddp.detach()
