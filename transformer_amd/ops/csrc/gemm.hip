@@ -418,7 +418,7 @@ void colsum_kernel(const short* __restrict__ dy, float* __restrict__ acc,
 // Column-parallel layout (thread = column, row loop — the same coalesced
 // pattern as colsum_kernel) so the separate colsum pass's full re-read of
 // dz (64 MB per FFN1 backward) disappears; finalize via last-arriver.
-#define RELU_ROWS 16
+#define RELU_ROWS 64
 __global__ __launch_bounds__(256)
 void relu_bwd_db_kernel(const short* __restrict__ dy,
                         const short* __restrict__ y,

@@ -270,21 +270,17 @@ class _LinearFn(torch.autograd.Function):
     def backward(ctx, dy):
         E = ext()
         x, w, y = ctx.saved_tensors
-        db = None
         if ctx.activation == "relu":
-            if ctx.has_bias:
-                # dz and its column sum in one pass (the separate colsum
-                # re-read of dz was a full extra HBM sweep per FFN1 bwd)
-                dy, db = E.relu_bwd_db(dy.contiguous(), y)
-            else:
-                dy = E.relu_bwd(dy.contiguous(), y)  # dz = dy * (y > 0)
+            # (A fused relu+colsum kernel — relu_bwd_db — was built and
+            # measured SLOWER in context at every grid shape: the column-
+            # parallel layout + per-column atomics lose to the flat
+            # 16k-block stream + separate colsum, the same result as
+            # round 1's column-reduction experiments.  docs/PERF.md.)
+            dy = E.relu_bwd(dy.contiguous(), y)  # dz = dy * (y > 0)
         else:
             dy = _dense2d(dy)
         dx = _dx_gemm(E, dy, w)         # dX[M,K] = dY[M,N] @ W[N,K]
-        if db is None:
-            dw, db = _dw_db_gemm(dy, x, ctx.has_bias)
-        else:
-            dw = _dw_gemm(dy, x)
+        dw, db = _dw_db_gemm(dy, x, ctx.has_bias)
         return dx, dw, db, None
 
 
@@ -310,23 +306,14 @@ class _LinearFlatFn(torch.autograd.Function):
         E = ext()
         x, y = ctx.saved_tensors
         w, b = ctx.wb
-        done_db = False
         if ctx.activation == "relu":
-            if b is not None:
-                dy, _ = E.relu_bwd_db(dy.contiguous(), y,
-                                      _flat(b).view(-1))
-                done_db = True
-            else:
-                dy = E.relu_bwd(dy.contiguous(), y)
+            dy = E.relu_bwd(dy.contiguous(), y)
         else:
             dy = _dense2d(dy)
         dx = _dx_gemm(E, dy, w)
-        if done_db:
-            _dw_gemm(dy, x, out=_flat(w).view(w.shape[0], -1))
-        else:
-            _dw_db_gemm(dy, x, b is not None,
-                        w_out=_flat(w).view(w.shape[0], -1),
-                        b_out=_flat(b).view(-1) if b is not None else None)
+        _dw_db_gemm(dy, x, b is not None,
+                    w_out=_flat(w).view(w.shape[0], -1),
+                    b_out=_flat(b).view(-1) if b is not None else None)
         _grad_ready(w, b)
         return dx, None, None
 
