@@ -195,39 +195,52 @@ void ln_bwd_kernel(const short* __restrict__ dy, const short* __restrict__ s,
 // column reduction: dgamma[c] = sum_r dy[r][c]*xhat[r][c]; dbeta[c] = sum dy.
 // Thread t of block b owns column b*256+t; every row read is coalesced
 // across the block's 256 consecutive columns.
-#define LNGB_ROWS 64
+#define LNGB_ROWS 128
 __global__ __launch_bounds__(256)
 void ln_gb_kernel(const short* __restrict__ dy, const short* __restrict__ s,
                   const float* __restrict__ mean,
                   const float* __restrict__ rstd, float* __restrict__ acc_g,
                   float* __restrict__ acc_b, short* __restrict__ dgamma,
                   short* __restrict__ dbeta, int R, int D) {
-  // 8 columns per thread (s16x8 loads — the s16x2 version read at half
-  // the coalescing width, 2x off the HBM bound).
-  const int c = (blockIdx.x * 64 + threadIdx.x) * 8;
+  // 2 columns per thread (s16x2 loads) x 4-row unroll: the scalar
+  // one-col-per-thread version was latency-bound at ~0.9 TB/s.
+  const int c = (blockIdx.x * 64 + threadIdx.x) * 2;
   if (c < D) {
   const long r0 = (long)blockIdx.y * LNGB_ROWS;
   const long r1 = min((long)R, r0 + LNGB_ROWS);
-  if (c + 8 <= D) {
-    float sg[8] = {0.f}, sb[8] = {0.f};
-    for (long r = r0; r < r1; ++r) {
-      s16x8 dv = *(const s16x8*)(dy + r * D + c);
-      s16x8 sv = *(const s16x8*)(s + r * D + c);
-      float mu = mean[r], rs = rstd[r];
+  float sg0 = 0.f, sg1 = 0.f, sb0 = 0.f, sb1 = 0.f;
+  const bool pair = (c + 1 < D) && (D % 2 == 0);
+  if (pair) {
+    long r = r0;
+    for (; r + 4 <= r1; r += 4) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float d = bfbits2f(dv[j]);
-        sg[j] += d * (bfbits2f(sv[j]) - mu) * rs;
-        sb[j] += d;
+      for (int u = 0; u < 4; ++u) {
+        s16x2 dv = *(const s16x2*)(dy + (r + u) * D + c);
+        s16x2 sv = *(const s16x2*)(s + (r + u) * D + c);
+        float mu = mean[r + u], rs = rstd[r + u];
+        float d0 = bfbits2f(dv[0]), d1 = bfbits2f(dv[1]);
+        sg0 += d0 * (bfbits2f(sv[0]) - mu) * rs;
+        sg1 += d1 * (bfbits2f(sv[1]) - mu) * rs;
+        sb0 += d0;
+        sb1 += d1;
       }
     }
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      atomicAdd(&acc_g[c + j], sg[j]);
-      atomicAdd(&acc_b[c + j], sb[j]);
+    for (; r < r1; ++r) {
+      s16x2 dv = *(const s16x2*)(dy + r * D + c);
+      s16x2 sv = *(const s16x2*)(s + r * D + c);
+      float mu = mean[r], rs = rstd[r];
+      float d0 = bfbits2f(dv[0]), d1 = bfbits2f(dv[1]);
+      sg0 += d0 * (bfbits2f(sv[0]) - mu) * rs;
+      sg1 += d1 * (bfbits2f(sv[1]) - mu) * rs;
+      sb0 += d0;
+      sb1 += d1;
     }
+    atomicAdd(&acc_g[c], sg0);
+    atomicAdd(&acc_b[c], sb0);
+    atomicAdd(&acc_g[c + 1], sg1);
+    atomicAdd(&acc_b[c + 1], sb1);
   } else {
-    for (int cc = c; cc < D; ++cc) {
+    for (int cc = c; cc < min(c + 2, D); ++cc) {
       float sg = 0.f, sb = 0.f;
       for (long r = r0; r < r1; ++r) {
         float dyv = bfbits2f(dy[r * D + cc]);
@@ -338,9 +351,9 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor s,
         {2L * D + 1}, dy.options().dtype(torch::kFloat32))).first;  // +cnt
   auto acc_g = wit->second.narrow(0, 0, D);
   auto acc_b = wit->second.narrow(0, D, D + 1);
-  // 64-thread blocks, 8 cols/thread; row blocks sized so the grid keeps
-  // several blocks per CU in flight (streaming kernel).
-  dim3 gbgrid(cdiv(cdiv(D, 8), 64), cdiv(R, LNGB_ROWS));
+  // 64-thread blocks: at d_model=512 a 256-thread block grid is only 128
+  // workgroups — half the 256-CU chip idle.
+  dim3 gbgrid(cdiv(cdiv(D, 2), 64), cdiv(R, LNGB_ROWS));
   ln_gb_kernel<<<gbgrid, 64, 0, stream>>>(
       (const short*)dy.data_ptr(), (const short*)s.data_ptr(),
       mean.data_ptr<float>(), rstd.data_ptr<float>(),
