@@ -766,3 +766,22 @@ def test_relu_bwd_db_fused():
         ref_dz = dy.float() * (y.float() > 0)
         assert_close(dz, ref_dz, 1e-3, f"relu_bwd_db dz it={it}")
         assert_close(db, ref_dz.sum(0), 0.02, f"relu_bwd_db db it={it}")
+
+
+def test_dx_gemm_wt_path_nonpadded():
+    """The default dX backend (cached transposed weight through the NT
+    kernel) vs the fp32 reference at a 64-aligned training shape, across
+    a weight update (cache refresh)."""
+    from transformer_amd.ops import functional as F
+    torch.manual_seed(14)
+    M, N, K = 4096, 1536, 512
+    dy = torch.randn(M, N, device="cuda", dtype=torch.bfloat16) * 0.05
+    w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16) * 0.05
+    F.bump_weight_version()
+    dx = F._dx_gemm(_ext(), dy, w)
+    assert_close(dx, dy.float() @ w.float(), 0.03, "wt dx")
+    with torch.no_grad():
+        w.mul_(0.5)
+    F.bump_weight_version()
+    dx2 = F._dx_gemm(_ext(), dy, w)
+    assert_close(dx2, dy.float() @ w.float(), 0.03, "wt dx refreshed")

@@ -306,3 +306,49 @@ def test_recapture_is_state_neutral():
     assert opt.step_count == snap[4]
     step_t, _ = opt.graph_state()
     assert int(step_t.item()) == opt.step_count
+
+
+def test_graph_replay_survives_wt_registry_growth():
+    """A captured step records the transpose_batch descriptor-table
+    ADDRESS; creating another model afterwards rebuilds the table — the
+    old one must stay alive and replays must keep refreshing correctly."""
+    import torch
+    from transformer_amd.models import Transformer
+    from transformer_amd.runtime import NoamAdam
+    from transformer_amd.runtime.graph import CapturedTrainStep
+    from transformer_amd import ops
+
+    torch.manual_seed(15)
+    m1 = Transformer(num_layers=1, d_model=64, num_heads=2, dff=128,
+                     input_vocab_size=90, target_vocab_size=90,
+                     rate=0.0, max_position=32).cuda().bfloat16()
+    opt1 = NoamAdam(m1, 64, warmup_steps=50, use_flat=True)
+    B = 4
+    lf = lambda real, pred: ops.masked_cross_entropy(pred, real, B, 0.0)
+    cap = CapturedTrainStep(m1, opt1, lf, (B, 8), (B, 8),
+                            torch.device("cuda"))
+    src = torch.randint(1, 80, (B, 8), device="cuda")
+    tar = torch.randint(1, 80, (B, 8), device="cuda")
+    l0 = cap(src, tar).item()
+    # second model grows the registry and rebuilds the table
+    m2 = Transformer(num_layers=1, d_model=64, num_heads=2, dff=128,
+                     input_vocab_size=90, target_vocab_size=90,
+                     rate=0.0, max_position=32).cuda().bfloat16()
+    opt2 = NoamAdam(m2, 64, warmup_steps=50, use_flat=True)
+    logits, _ = m2((src, tar[:, :-1].contiguous()), training=True)
+    loss2 = lf(tar[:, 1:].contiguous(), logits)
+    opt2.zero_grad()
+    loss2.backward()
+    opt2.step()
+    torch.cuda.synchronize()
+    # replays still train m1 (the captured transpose_batch reads the OLD
+    # table, which must be kept alive)
+    losses = [cap(src, tar).item() for _ in range(6)]
+    torch.cuda.synchronize()
+    assert losses[-1] < l0, (l0, losses)
+    # and m1's cached W^T actually tracks its weights (dX correctness):
+    # one more eager step on m1 must also reduce the loss
+    ops.functional.bump_weight_version()
+    logits1, _ = m1((src, tar[:, :-1].contiguous()), training=True)
+    l_eager = lf(tar[:, 1:].contiguous(), logits1).item()
+    assert l_eager < l0
