@@ -57,7 +57,12 @@ def main():
     assert world == args.gpus or world == 1, \
         f"launched world {world} != --gpus {args.gpus}"
     use_cuda = torch.cuda.is_available()
-    device = torch.device(f"cuda:{local_rank}") if use_cuda else torch.device("cpu")
+    # clamp to the visible device count so the world-2-on-one-GPU gloo
+    # smoke (tools/dp_parity.py setup) runs; on a real node each rank
+    # keeps its own GPU
+    dev_idx = min(local_rank, max(torch.cuda.device_count() - 1, 0)) \
+        if use_cuda else 0
+    device = torch.device(f"cuda:{dev_idx}") if use_cuda else torch.device("cpu")
     dtype = torch.bfloat16 if use_cuda else torch.float32
     if use_cuda:
         ops.ext()  # fail loudly if the HIP extension is missing
