@@ -269,6 +269,7 @@ void gemm_uni_kernel(const short* __restrict__ A, const short* __restrict__ B,
     short* pb_lds = pa_lds + A_ELEMS;
     const bool do_pf = t + 1 < ntiles;
 
+    constexpr bool LOCKSTEP = (SCHED < 6);
     // ---- q0: quadrant (0,0); issue t+1's A-piece0 + all B -------------
 #pragma unroll
     for (int i = 0; i < MF / 2; ++i)
@@ -281,12 +282,12 @@ void gemm_uni_kernel(const short* __restrict__ A, const short* __restrict__ B,
     if (do_pf) {
       stage_a(t + 1, 0, pa_lds);
       if (SCHED != 4) stage_b(t + 1, pb_lds);
-      if (SCHED == 2) stage_a(t + 1, 1, pa_lds);
+      if (SCHED == 2 || SCHED >= 6) stage_a(t + 1, 1, pa_lds);
     }
-    U_BARRIER();
+    if (LOCKSTEP) U_BARRIER();
     U_WAIT_LGKM0();
     U_MFMA_QUAD(0, 0)
-    U_BARRIER();
+    if (LOCKSTEP) U_BARRIER();
 
     // ---- q1: quadrant (0,1); issue t+1's A-piece1 ---------------------
 #pragma unroll
@@ -295,38 +296,38 @@ void gemm_uni_kernel(const short* __restrict__ A, const short* __restrict__ B,
       for (int ks = 0; ks < 2; ++ks) bfr[j][ks] = ldb_frag(j, ks, b_lds);
     if (do_pf) {
       if (SCHED == 4) stage_b(t + 1, pb_lds);
-      if (SCHED != 2) stage_a(t + 1, 1, pa_lds);
+      if (SCHED != 2 && SCHED < 6) stage_a(t + 1, 1, pa_lds);
     }
-    U_BARRIER();
+    if (LOCKSTEP) U_BARRIER();
     U_WAIT_LGKM0();
     U_MFMA_QUAD(0, 1)
     // end-q1 wait: drain THIS tile's A-piece1 (issued q1 of t-1, 4 phases
     // ago, consumed by q2's reads one barrier from here); t+1's
     // A-piece0+B (2+NB) + A-piece1 (2) stay in flight.
-    if (SCHED != 2) {
+    if (SCHED != 2 && SCHED < 6) {
       if (do_pf) {
         if (BN_ == 256) U_WAIT_VM(8); else U_WAIT_VM(6);
       } else {
         U_WAIT_VM(0);
       }
     }
-    U_BARRIER();
+    if (LOCKSTEP || SCHED == 7) U_BARRIER();
 
     // ---- q2: quadrant (1,0) -------------------------------------------
 #pragma unroll
     for (int i = 0; i < MF / 2; ++i)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) afr[i][ks] = lda_frag(i, ks, 1, a_lds);
-    U_BARRIER();
+    if (LOCKSTEP) U_BARRIER();
     U_WAIT_LGKM0();
     U_MFMA_QUAD(1, 0)
-    U_BARRIER();
+    if (LOCKSTEP) U_BARRIER();
 
     // ---- q3: quadrant (1,1), operands already in registers ------------
     U_MFMA_QUAD(1, 1)
     // end-q3 wait: drain t+1's A-piece0+B (issued q0, 3 phases ago);
     // its A-piece1 (2 glds) stays in flight across the tile boundary.
-    if (SCHED == 2 || !do_pf) {
+    if (SCHED == 2 || SCHED >= 6 || !do_pf) {
       U_WAIT_VM(0);
     } else {
       U_WAIT_VM(2);
@@ -683,6 +684,8 @@ torch::Tensor gemm_uni_nt_ab(torch::Tensor a, torch::Tensor w,
     case 2: go2(std::integral_constant<int, 2>{}); break;
     case 3: go2(std::integral_constant<int, 3>{}); break;
     case 4: go2(std::integral_constant<int, 4>{}); break;
+    case 6: go2(std::integral_constant<int, 6>{}); break;
+    case 7: go2(std::integral_constant<int, 7>{}); break;
     default: go2(std::integral_constant<int, 1>{}); break;
   }
   return c;
