@@ -10,7 +10,13 @@
 // fragments from LDS into registers (12/4/8/0 ds_read_b128 — A-halves and
 // B-quadrant-halves are REUSED across phases from registers), prefetches
 // half-tiles of K-tile t+1 into the other slot with global_load_lds, then
-// raw s_barrier + s_waitcnt lgkmcnt(0) + 16 MFMAs + s_barrier:
+// s_waitcnt lgkmcnt(0) + 16 MFMAs.  NO intra-tile barriers (round 2,
+// measured +5-6% per shape): within a tile every wave reads the SAME
+// slot (synced at the boundary) and staging targets the other slot, so
+// the per-phase barriers only enforced lockstep — dropping them lets
+// waves drift and overlap each other's staging/MFMA phases (the 35%
+// wave-parked time in profiles/r02_pmc_summary.md was these barriers).
+// One vmcnt(0) + s_barrier per K-tile at the boundary remains:
 //   q0: read A-half0(8) + B-ch0(4); stage A0',A1'  ; mfma quadrant (0,0)
 //   q1: read B-ch1(4)             ; stage B0',B1'  ; mfma quadrant (0,1)
 //   q2: read A-half1(8)           ;                ; mfma quadrant (1,0)
@@ -173,10 +179,8 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
       stage256<BM_>(A + pk0, lda, bm0, M, 0, pa_lds);
       stage256<BM_>(A + pk0, lda, bm0, M, 1, pa_lds);
     }
-    G_BARRIER();
     G_WAIT_LGKM0();
     G_MFMA_QUAD(0, 0)
-    G_BARRIER();
 
     // ---- q1: quadrant (0, 1) ------------------------------------------
 #pragma unroll
@@ -188,10 +192,8 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
       stage256<BN_>(B + pk0, ldb, bn0, N, 0, pb_lds);
       stage256<BN_>(B + pk0, ldb, bn0, N, 1, pb_lds);
     }
-    G_BARRIER();
     G_WAIT_LGKM0();
     G_MFMA_QUAD(0, 1)
-    G_BARRIER();
 
     // ---- q2: quadrant (1, 0) ------------------------------------------
 #pragma unroll
@@ -200,10 +202,8 @@ void gemm256_kernel(const short* __restrict__ A, const short* __restrict__ B,
       for (int ks = 0; ks < 2; ++ks)
         afr[i][ks] = frag256(a_lds, wm + (BM_ / 4) + i * 16 + fr,
                              ks * 64 + kg * 16);
-    G_BARRIER();
     G_WAIT_LGKM0();
     G_MFMA_QUAD(1, 0)
-    G_BARRIER();
 
     // ---- q3: quadrant (1, 1), all operands already in registers -------
     G_MFMA_QUAD(1, 1)

@@ -155,7 +155,7 @@ DEV_INLINE bf16x8u u_frag_tr(const short* lds, int ms, int colb) {
 //   3 counted waits + static young-half priority (T5 static form)
 //   4 counted waits, no setprio, B issued at q1 (balanced bursts)
 template <int EPILOGUE, int BN_, bool TRA, bool TRB, bool SPLITR = false,
-          int SCHED = 2>
+          int SCHED = 6>
 __global__ __launch_bounds__(U_THREADS, 1)
 void gemm_uni_kernel(const short* __restrict__ A, const short* __restrict__ B,
                      const short* __restrict__ bias, short* __restrict__ C,
