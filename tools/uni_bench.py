@@ -78,7 +78,7 @@ DW = [
 
 
 SCHED_NAMES = {0: "cnt+prio", 1: "cnt", 2: "r1-style", 3: "cnt+stat",
-               4: "cnt+Bq1", 5: "ring", 6: "nobar", 7: "2bar"}
+               4: "cnt+Bq1", 5: "ring", 6: "nobar", 7: "2bar", 8: "ring128x2"}
 
 
 def sched_ab(iters):
@@ -87,7 +87,7 @@ def sched_ab(iters):
     for M, N, K, _, tag in [FWD[0], FWD[3], FWD[4], FWD[6]]:
         a = torch.randn(M, K, device="cuda", dtype=dt)
         w = torch.randn(N, K, device="cuda", dtype=dt) * 0.05
-        scheds = [0, 2, 5, 6, 7]
+        scheds = [2, 5, 6, 8]
         fns = [lambda s=s: E.gemm_uni_nt_ab(a, w, s) for s in scheds]
         fns.append(lambda: E.gemm_nt(a, w, torch.Tensor(), 0))
         r = bench_pair(fns, iters)

@@ -498,8 +498,8 @@ static torch::Tensor gemm_uni_launch(
 // the LDS of the double-buffered schedule.
 // ---------------------------------------------------------------------------
 
-template <int EPILOGUE, int BN_>
-__global__ __launch_bounds__(U_THREADS, 1)
+template <int EPILOGUE, int BN_, int WPS = 1>
+__global__ __launch_bounds__(U_THREADS, WPS)
 void gemm_ring_kernel(const short* __restrict__ A, const short* __restrict__ B,
                       const short* __restrict__ bias, short* __restrict__ C,
                       int M, int N, int K, long lda, long ldb, int has_bias,
@@ -677,6 +677,22 @@ torch::Tensor gemm_uni_nt_ab(torch::Tensor a, torch::Tensor w,
     };
     if (BN == 256) ring(std::integral_constant<int, 256>{});
     else ring(std::integral_constant<int, 128>{});
+    return c;
+  }
+  if ((int)sched == 8) {
+    // ring @ BN128 with a 4-waves/SIMD VGPR cap: 48 KiB LDS + <=128
+    // VGPR admit TWO co-resident blocks per CU, overlapping prologue
+    // drains and boundary stalls across blocks (the K=512 1-block/CU
+    // shapes idle the CU during their prologue)
+    const int nbn8 = cdiv(N, 128);
+    const size_t rsm = ((size_t)256 + 128) * U_BK * sizeof(short);
+    auto kfn = gemm_ring_kernel<0, 128, 4>;
+    (void)hipFuncSetAttribute((const void*)kfn,
+                              hipFuncAttributeMaxDynamicSharedMemorySize,
+                              (int)rsm);
+    kfn<<<nbm * nbn8, U_THREADS, rsm, stream>>>(
+        (const short*)a.data_ptr(), (const short*)w.data_ptr(), nullptr,
+        (short*)c.data_ptr(), M, N, K, K, K, 0, nbm, nbn8);
     return c;
   }
   switch ((int)sched) {
