@@ -42,7 +42,9 @@ def main(epochs, enable_function, buffer_size, batch_size, sequence_length,
     rank, local_rank, world_size = init_distributed()
     torch.manual_seed(seed)  # same init on every rank (then X1 broadcast)
     if device is None and torch.cuda.is_available():
-        device = f"cuda:{local_rank}"
+        # clamp to visible devices (world-2-on-one-GPU gloo smoke; on a
+        # real node each rank keeps its own GPU)
+        device = f"cuda:{min(local_rank, torch.cuda.device_count() - 1)}"
     device, dtype = pick_device_dtype(device, dtype)
 
     # global-batch divisibility guard (reference distributed_train.py:154-158)
