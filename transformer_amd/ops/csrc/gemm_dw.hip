@@ -143,18 +143,22 @@ typedef __bf16 bf16x4d __attribute__((ext_vector_type(4)));
 // 4-VGPR operand (seen in the compiled loop; ~24% of the MFMA issue
 // time).  Same LDS traffic, twice the (half-size) MFMA instructions.
 template <bool SPLIT, bool DB, bool NOBAR = true, bool X16 = false,
-          int BM = DW_BM>
+          int BM = DW_BM, int RING = 2>
 __global__ __launch_bounds__(DW_THREADS)
 void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
                     short* __restrict__ C, float* __restrict__ CW,
                     float* __restrict__ DBW, int M,
                     int N, int K, long m_per_slice, int nbk) {
-  // Two-slot ring (PMC: this kernel fetches each operand from HBM once —
-  // L2 absorbs all tile re-reads — so it is pipeline-stall bound, not
-  // traffic bound; the old single-buffer __syncthreads() drain exposed
-  // the full staging latency every 64-token block).
-  __shared__ short a_img[2][BM * DW_BN];  // dY block
-  __shared__ short b_img[2][BM * DW_BK];  // X block
+  // RING-slot staging ring (PMC: this kernel fetches each operand from
+  // HBM once — L2 absorbs all tile re-reads — so it is pipeline-stall
+  // bound, not traffic bound).  RING=2: prefetch block t+1, drain fully
+  // at the boundary — but a 64-token block is only ~550 MFMA cycles,
+  // under the ~900-cycle HBM latency, so the boundary drain stalls
+  // (SQ_WAIT_ANY ~50%, profiles/r02_pmc_summary.md).  RING=3 stages
+  // t+2 as well and waits the boundary with a COUNTED vmcnt (t+2's
+  // loads ride across), trading LDS 64->96 KiB (2 -> 1 block/CU).
+  __shared__ short a_img[RING][BM * DW_BN];  // dY block
+  __shared__ short b_img[RING][BM * DW_BK];  // X block
 
   // XCD-aware bijective remap: consecutive logical tiles (same n-block,
   // varying k) land on the SAME XCD, so a dY slice is read into one XCD's
@@ -194,18 +198,30 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
   const int db_c4 = (threadIdx.x & 31) << 2;
   float db_acc[4] = {0.f, 0.f, 0.f, 0.f};
 
-  // Prologue: stage block 0 into slot 0, drain once.
-  if (m_lo < m_hi) {
-    if (m_lo + BM <= m_hi && a_full)
-      dw_stage_glds<128, BM>(dY, N, m_lo, bn0, a_img[0]);
+  // staging helper: block starting at token row `mb` into slot `sl`
+  auto stage_blk = [&](long mb, int sl) {
+    if (mb >= m_hi) return;
+    if (mb + BM <= m_hi && a_full)
+      dw_stage_glds<128, BM>(dY, N, mb, bn0, a_img[sl]);
     else
-      dw_stage<128, BM>(dY, N, m_lo, m_hi, bn0, N, a_img[0]);
-    if (m_lo + BM <= m_hi && b_full)
-      dw_stage_glds<128, BM>(X, K, m_lo, bk0, b_img[0]);
+      dw_stage<128, BM>(dY, N, mb, m_hi, bn0, N, a_img[sl]);
+    if (mb + BM <= m_hi && b_full)
+      dw_stage_glds<128, BM>(X, K, mb, bk0, b_img[sl]);
     else
-      dw_stage<128, BM>(X, K, m_lo, m_hi, bk0, K, b_img[0]);
+      dw_stage<128, BM>(X, K, mb, m_hi, bk0, K, b_img[sl]);
+  };
+
+  // Prologue: stage block 0 (and, at RING=3, block 1), drain block 0.
+  // The counted wait (vmcnt(8): block 1's 8 glds stay in flight) is only
+  // valid when block 1 was glds-staged (full rows + full cols) — the
+  // guarded register path issues no glds to count.
+  stage_blk(m_lo, 0);
+  if (RING == 3) stage_blk(m_lo + BM, 1 % RING);
+  if (RING == 3 && a_full && b_full && m_lo + 2 * BM <= m_hi) {
+    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   }
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
   for (long m0 = m_lo; m0 < m_hi; m0 += BM) {
@@ -226,8 +242,8 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
       }
     }
     const int cur = (int)((m0 - m_lo) / BM);
-    const short* a_lds = a_img[cur & 1];
-    const short* b_lds = b_img[cur & 1];
+    const short* a_lds = a_img[cur % RING];
+    const short* b_lds = b_img[cur % RING];
     // Prologue staged block 0; here prefetch block cur+1 into the other
     // slot (its previous tenant was consumed last block) — all glds
     // issued in phase 0, the rest of the block to land before the
@@ -260,16 +276,9 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
           bf_[j] = dw_frag<BM>(b_lds, ko, wk + j * 16);
       }
       if (ph == 0 && pf) {
-        short* pa = a_img[(cur + 1) & 1];
-        short* pb = b_img[(cur + 1) & 1];
-        if (mN + BM <= m_hi && a_full)
-          dw_stage_glds<128, BM>(dY, N, mN, bn0, pa);
-        else
-          dw_stage<128, BM>(dY, N, mN, m_hi, bn0, N, pa);
-        if (mN + BM <= m_hi && b_full)
-          dw_stage_glds<128, BM>(X, K, mN, bk0, pb);
-        else
-          dw_stage<128, BM>(X, K, mN, m_hi, bk0, K, pb);
+        // stage the NEXT un-staged block: t+1 at RING=2, t+2 at RING=3
+        const long mS = (RING == 3) ? mN + BM : mN;
+        if (mS < m_hi) stage_blk(mS, (int)((cur + RING - 1) % RING));
       }
       // NOBAR: no intra-block barriers — all phases read the SAME LDS
       // slot (synced at the block boundary), fragments are wave-private
@@ -297,8 +306,16 @@ void gemm_dw_kernel(const short* __restrict__ dY, const short* __restrict__ X,
           }
         }
     }
-    // boundary: prefetched block cur+1 must have landed
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    // boundary: block t+1's stages must have landed.  RING=3 leaves
+    // block t+2's 8 glds in flight across the barrier (vmcnt(8)) — valid
+    // only when t+2 really was glds-staged (m0+3*BM fits and the tile is
+    // full); otherwise drain (the guarded path's loads are
+    // compiler-waited before its ds_writes, as at RING=2).
+    if (RING == 3 && a_full && b_full && m0 + 3 * BM <= m_hi) {
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
     __builtin_amdgcn_s_barrier();
   }
 
@@ -418,6 +435,10 @@ torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x,
     const char* e = getenv("TFMX_DW_BAR");
     return e && atoi(e) != 0;
   }();
+  static const bool ring3 = [] {
+    const char* e = getenv("TFMX_DW_RING");
+    return e && atoi(e) == 3;
+  }();
   static const bool use_x16 = [] {
     const char* e = getenv("TFMX_DW_X16");
     return e && atoi(e) != 0;
@@ -426,11 +447,13 @@ torch::Tensor gemm_dw(torch::Tensor dy, torch::Tensor x,
     constexpr bool SP = decltype(split)::value, DBV = decltype(dbc)::value;
     auto kfn = use_bm32
         ? gemm_dw_kernel<SP, DBV, true, false, 32>
-        : (use_x16
-               ? (use_bar ? gemm_dw_kernel<SP, DBV, false, true>
-                          : gemm_dw_kernel<SP, DBV, true, true>)
-               : (use_bar ? gemm_dw_kernel<SP, DBV, false, false>
-                          : gemm_dw_kernel<SP, DBV, true, false>));
+        : (ring3
+               ? gemm_dw_kernel<SP, DBV, true, false, DW_BM, 3>
+               : (use_x16
+                      ? (use_bar ? gemm_dw_kernel<SP, DBV, false, true>
+                                 : gemm_dw_kernel<SP, DBV, true, true>)
+                      : (use_bar ? gemm_dw_kernel<SP, DBV, false, false>
+                                 : gemm_dw_kernel<SP, DBV, true, false>)));
     kfn<<<grid, DW_THREADS, 0, stream>>>(
             (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
             (short*)c.data_ptr(), cwp, dbw_p, (int)M, N, K, m_per_slice,
