@@ -1,3 +1,8 @@
+"""gemm256 vs the 128-tile dispatch at K=512 256-block shapes
+(the round-2 viability-rule relax: +4-24% measured — gemm256.hip).
+
+    python tools/probe128.py   (GPU box)
+"""
 import sys, torch
 sys.path.insert(0, ".")
 from transformer_amd.ops import ext
