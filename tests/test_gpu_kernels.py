@@ -785,3 +785,50 @@ def test_dx_gemm_wt_path_nonpadded():
     F.bump_weight_version()
     dx2 = F._dx_gemm(_ext(), dy, w)
     assert_close(dx2, dy.float() @ w.float(), 0.03, "wt dx refreshed")
+
+
+def test_ce_fused_matches_reference_and_seed_scaling():
+    """ce_fused (loss + grad in one kernel): loss and grad vs the fp32
+    reference; ce_scale no-ops at seed 1.0 and scales exactly otherwise."""
+    from transformer_amd.ops import reference as Rf
+    torch.manual_seed(16)
+    R, V = 96, 32770
+    vp = (V + 255) // 256 * 256
+    logits = (torch.randn(R, V, device="cuda") * 2).bfloat16()
+    tgt = torch.randint(1, V, (R,), device="cuda")
+    tgt[::5] = 0
+    loss, dfull = _ext().ce_fused(logits, tgt, 8.0, 0.1)
+    assert dfull.shape == (R, vp)
+    assert (dfull[:, V:].float() == 0).all()
+    lf = logits.detach().float().requires_grad_(True)
+    ref_loss = Rf.masked_cross_entropy(lf.view(1, R, V), tgt.view(1, R),
+                                       8, 0.1)
+    ref_loss.backward()
+    assert abs(loss.item() - ref_loss.item()) / max(abs(ref_loss.item()),
+                                                    1) < 0.01
+    assert_close(dfull[:, :V], lf.grad, 0.05, "ce_fused grad")
+    # seed 1.0: no-op
+    before = dfull.clone()
+    _ext().ce_scale(dfull, torch.ones(1, device="cuda"))
+    assert torch.equal(dfull, before)
+    # seed 2.5: exact scale
+    _ext().ce_scale(dfull, torch.full((1,), 2.5, device="cuda"))
+    assert_close(dfull[:, :V], lf.grad * 2.5, 0.06, "ce_scale 2.5")
+
+
+def test_ce_autograd_path_nonunit_seed():
+    """masked_cross_entropy through autograd with a NON-unit upstream
+    gradient (loss * 3).backward() must scale the logits grad by 3."""
+    from transformer_amd import ops
+    torch.manual_seed(17)
+    B, T, V = 2, 24, 1000
+    logits = torch.randn(B, T, V, device="cuda",
+                         dtype=torch.bfloat16).requires_grad_(True)
+    tgt = torch.randint(1, V, (B, T), device="cuda")
+    loss = ops.masked_cross_entropy(logits, tgt, B, 0.1)
+    (loss * 3.0).backward()
+    g3 = logits.grad.clone()
+    logits.grad = None
+    loss2 = ops.masked_cross_entropy(logits, tgt, B, 0.1)
+    loss2.backward()
+    assert_close(g3, logits.grad.float() * 3.0, 0.03, "seed-3 scaling")

@@ -76,6 +76,11 @@ std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets,
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets,
                      torch::Tensor lse, torch::Tensor dloss,
                      double batch_size, double label_smoothing);
+std::vector<torch::Tensor> ce_fused(torch::Tensor logits,
+                                    torch::Tensor targets,
+                                    double batch_size,
+                                    double label_smoothing);
+void ce_scale(torch::Tensor dlogits, torch::Tensor dloss);
 
 void adam_fused_dev(torch::Tensor master, torch::Tensor m, torch::Tensor v,
                     torch::Tensor grad, torch::Tensor param,
@@ -168,6 +173,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_bwd", &dropout_bwd);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
+  m.def("ce_fused", &ce_fused,
+        "loss + gradient in one kernel (the grad sweep re-reads L2-hot "
+        "rows); seed applied later by ce_scale (no-op at 1.0)");
+  m.def("ce_scale", &ce_scale);
   m.def("adam_fused", &adam_fused);
   m.def("adam_fused_dev", &adam_fused_dev,
         "graph-capturable Adam: device step tensor + in-kernel Noam lr");

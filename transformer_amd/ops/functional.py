@@ -726,27 +726,32 @@ def dropout(x, p: float, training: bool):
 # ---------------------------------------------------------------------------
 
 class _CrossEntropyFn(torch.autograd.Function):
+    """Loss AND gradient in ONE kernel at forward time (the gradient
+    sweep re-reads rows that are still L2-hot, saving ce_bwd's separate
+    full logits pass); the autograd seed is applied in backward by a
+    device-side kernel that NO-OPS when the seed is 1.0 — the value
+    loss.backward() always feeds — so numerics are exact for any seed
+    without a host sync.  Memory is neutral: dfull (R, Vp) is saved
+    instead of (logits, lse)."""
+
     @staticmethod
     def forward(ctx, logits, targets, batch_size, label_smoothing):
         E = ext()
-        loss, lse = E.ce_fwd(logits, targets, float(batch_size), label_smoothing)
-        ctx.batch_size = float(batch_size)
-        ctx.ls = label_smoothing
-        ctx.save_for_backward(logits, targets, lse)
+        loss, dfull = E.ce_fused(logits, targets, float(batch_size),
+                                 label_smoothing)
+        ctx.V = logits.shape[1]
+        ctx.save_for_backward(dfull)
         return loss
 
     @staticmethod
     def backward(ctx, dloss):
         E = ext()
-        logits, targets, lse = ctx.saved_tensors
-        # ce_bwd returns (R, Vp) with Vp = V rounded up to 64 and ZERO pad
-        # columns; the [:, :V] view keeps the padded stride so the logits
-        # dX/dW GEMMs can contract over the 64-aligned Vp (gemm_uni.hip).
-        dfull = E.ce_bwd(logits, targets, lse,
-                         dloss.to(torch.float32).reshape(1).contiguous(),
-                         ctx.batch_size, ctx.ls)
-        V = logits.shape[1]
-        dlogits = dfull if dfull.shape[1] == V else dfull[:, :V]
+        (dfull,) = ctx.saved_tensors
+        E.ce_scale(dfull, dloss.to(torch.float32).reshape(1).contiguous())
+        # dfull is (R, Vp), Vp = roundup(V, 256), zero pad columns; the
+        # [:, :V] view keeps the padded stride so the logits dX/dW GEMMs
+        # contract over the aligned Vp (gemm_uni.hip)
+        dlogits = dfull if dfull.shape[1] == ctx.V else dfull[:, :ctx.V]
         return dlogits, None, None, None
 
 
