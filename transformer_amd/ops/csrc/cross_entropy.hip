@@ -274,16 +274,21 @@ void ce_fused_kernel(const short* __restrict__ logits,
 // training case — loss.backward() seeds 1.0), else scale in place.
 __global__ void ce_scale_kernel(short* __restrict__ d, long n,
                                 const float* __restrict__ dloss) {
+  // grid-stride with a CAPPED grid: the no-op case (seed 1.0) must cost
+  // one early-exit sweep of ~2k blocks, not the dispatch of 262k
   const float sc = dloss[0];
   if (sc == 1.0f) return;
-  long i = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8;
-  if (i + 8 <= n) {
-    s16x8 v = *(s16x8*)(d + i);
+  const long stride = (long)gridDim.x * blockDim.x * 8;
+  for (long i = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8; i < n;
+       i += stride) {
+    if (i + 8 <= n) {
+      s16x8 v = *(s16x8*)(d + i);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) v[j] = f2bfbits(bfbits2f(v[j]) * sc);
-    *(s16x8*)(d + i) = v;
-  } else {
-    for (long j = i; j < n; ++j) d[j] = f2bfbits(bfbits2f(d[j]) * sc);
+      for (int j = 0; j < 8; ++j) v[j] = f2bfbits(bfbits2f(v[j]) * sc);
+      *(s16x8*)(d + i) = v;
+    } else {
+      for (long j = i; j < n; ++j) d[j] = f2bfbits(bfbits2f(d[j]) * sc);
+    }
   }
 }
 
@@ -366,7 +371,8 @@ void ce_scale(torch::Tensor dlogits, torch::Tensor dloss) {
               dloss.numel() == 1);
   long n = dlogits.numel();
   auto stream = at::hip::getCurrentHIPStream();
-  ce_scale_kernel<<<((n + 7) / 8 + 255) / 256, 256, 0, stream>>>(
+  long blocks = std::min<long>(((n + 7) / 8 + 255) / 256, 2048);
+  ce_scale_kernel<<<(unsigned)blocks, 256, 0, stream>>>(
       (short*)dlogits.data_ptr(), n, dloss.data_ptr<float>());
 }
 
