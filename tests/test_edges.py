@@ -147,3 +147,24 @@ def test_length_filter_boundary(tmp_path, toy_corpus):
     assert 0 < n_lo < n_hi  # the cap really filters, without emptying
     for src, tgt in train_lo:
         assert src.shape[1] <= 8 and tgt.shape[1] <= 8
+
+
+def test_tokenizer_build_deterministic():
+    """Two builds from the same corpus give identical vocab + ids (DP
+    rank-0-builds-and-persists relies on rebuilds being reproducible)."""
+    from transformer_amd.data.tokenizer import SubwordTokenizer
+    corpus = ["the cat sat", "the dog sat", "ein hund sass"] * 20
+    a = SubwordTokenizer.build_from_corpus(corpus, target_vocab_size=200)
+    b = SubwordTokenizer.build_from_corpus(corpus, target_vocab_size=200)
+    assert a.subwords == b.subwords
+    assert a.encode("the cat sass") == b.encode("the cat sass")
+
+
+def test_noam_schedule_step_zero_guard():
+    """step 0 is clamped to 1 (the reference's schedule is undefined at 0);
+    warmup ramp is linear and the peak sits at warmup_steps."""
+    from transformer_amd.runtime.schedule import NoamSchedule
+    s = NoamSchedule(512, warmup_steps=4000)
+    assert s(0) == s(1) > 0
+    assert abs(s(2000) - 2 * s(1000)) < 1e-12  # linear ramp
+    assert s(4000) >= s(3999) and s(4000) >= s(4001)  # peak at warmup
